@@ -1,0 +1,247 @@
+/* mz_gpu.h — C ABI of the MI355X-native incremental join/reduce engine.
+ *
+ * This is the drop-in boundary for Materialize's compute hot path
+ * (SURVEY.md §8b). Each entry point states the reference interface it
+ * replaces (file:line under /root/reference). A Rust host would bind these
+ * over plain `extern "C"` FFI; no torch types appear
+ * here — plain pointers and sizes only.
+ *
+ * Ownership: the caller owns all host/device memory passed in via
+ * descriptors until the call returns. The library owns arrangements,
+ * operators and returned out-batches; out-batches are freed with
+ * mz_gpu_out_release. All calls on one context must be serialized by the
+ * caller (one driver thread per GPU, mirroring one timely worker per core —
+ * src/compute/src/server.rs:327-377).
+ *
+ * Errors: non-zero int return; mz_gpu_last_error(ctx) gives a message.
+ */
+#ifndef MZ_GPU_H
+#define MZ_GPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct mz_gpu_ctx mz_gpu_ctx;
+typedef struct mz_gpu_arr mz_gpu_arr;   /* an arrangement (trace/spine) */
+typedef struct mz_gpu_join mz_gpu_join; /* a linear-join operator (mz_join_core state) */
+typedef struct mz_gpu_red mz_gpu_red;   /* an accumulable-reduce operator */
+
+/* ---------------------------------------------------------------- schema */
+
+/* Arrangement schema: fixed-width keys (1..2 u64 words) and vals (bytes).
+ * Blueprint: RowRowSpine columnar layout, src/row-spine/src/lib.rs:56-135. */
+typedef struct {
+  uint32_t key_words;  /* 1 or 2; keys compared as i64-tuple ascending */
+  uint32_t val_bytes;  /* 0..64; vals compared as lexicographic bytes   */
+} mz_gpu_schema;
+
+/* A set of updates ((key, val), time, diff), SoA columns.
+ * `on_device` = 1 when the pointers are HIP device pointers on the ctx's
+ * device; 0 for host memory (the library stages them in).
+ * This is both the delta-batch input format and the sealed-batch push
+ * format; sealed batches must additionally be sorted by (key, val, time)
+ * and consolidated (duplicate (key,val,time) diffs summed, zeros dropped) —
+ * use mz_gpu_consolidate to produce that form. */
+typedef struct {
+  const uint64_t *keys;   /* [key_words * n] key words, key-major          */
+  const uint8_t  *vals;   /* [val_bytes * n], may be NULL if val_bytes==0  */
+  const uint64_t *times;  /* [n] */
+  const int64_t  *diffs;  /* [n] */
+  uint64_t n;
+  uint64_t lower, upper;  /* batch time bounds [lower, upper)              */
+  int32_t  on_device;
+} mz_gpu_updates;
+
+/* ----------------------------------------------------------- closures
+ * JoinClosure (src/compute-types/src/plan/join.rs:60-86) restated as a
+ * filter + field-map spec evaluated inside the probe kernel. Fields are
+ * copied from the probe inputs into the output (key,val) row. */
+enum {
+  MZ_SRC_KEY = 0,      /* the (shared) join key words                  */
+  MZ_SRC_VAL_STREAM = 1, /* the delta/stream side's val bytes          */
+  MZ_SRC_VAL_LOOKUP = 2, /* the arrangement side's val bytes           */
+  MZ_SRC_COMPUTE = 3   /* computed field; `off` = compute id           */
+};
+enum { MZ_CMP_LT = 0, MZ_CMP_LE, MZ_CMP_GT, MZ_CMP_GE, MZ_CMP_EQ, MZ_CMP_NE };
+enum {
+  /* revenue = extendedprice_cents * (10000 - discount_bp) as i64 1e-4
+   * units — Q3/Q5/Q10's `l_extendedprice * (1 - l_discount)` with TPC-H
+   * fixed-scale decimals mapped to exact integers. Operand
+   * offsets are given by arg0/arg1 on the out field. */
+  MZ_COMPUTE_REVENUE = 0
+};
+
+typedef struct {
+  uint8_t  src;     /* MZ_SRC_* (filters: KEY/VAL_STREAM/VAL_LOOKUP)   */
+  uint16_t off;     /* byte offset into src                            */
+  uint8_t  width;   /* 4 or 8 (signed little-endian integer)           */
+  uint8_t  cmp;     /* MZ_CMP_*                                        */
+  int64_t  imm;     /* literal operand                                 */
+} mz_gpu_filter;
+
+typedef struct {
+  uint8_t  src;     /* MZ_SRC_*                                        */
+  uint16_t off;     /* byte offset (or compute id for MZ_SRC_COMPUTE)  */
+  uint8_t  width;   /* bytes copied / produced                         */
+  uint16_t arg0, arg1; /* compute operand byte offsets (both in VAL_*) */
+  uint8_t  arg0_src, arg1_src;
+} mz_gpu_field;
+
+#define MZ_GPU_MAX_FILTERS 4
+#define MZ_GPU_MAX_FIELDS  8
+
+typedef struct {
+  uint32_t n_filters;
+  mz_gpu_filter filters[MZ_GPU_MAX_FILTERS];
+  uint32_t n_key_fields;             /* output key (next stage / reduce) */
+  mz_gpu_field key_fields[MZ_GPU_MAX_FIELDS];
+  uint32_t n_val_fields;             /* output val                        */
+  mz_gpu_field val_fields[MZ_GPU_MAX_FIELDS];
+  mz_gpu_schema out;                 /* shape of the output rows          */
+} mz_gpu_closure;
+
+/* ------------------------------------------------------------- reduce
+ * AccumulablePlan (src/compute-types/src/plan/reduce.rs:233) restated.
+ * Accum semantics follow src/compute/src/render/reduce.rs:1611-2270:
+ * SimpleNumber = wrapping i128 + non_nulls; Float = 24-frac-bit fixed-point
+ * wrapping i128 + inf/nan/non_null counts; COUNT = non_nulls only. */
+enum {
+  MZ_AGG_COUNT = 0,
+  MZ_AGG_SUM_I64,    /* also exact-decimal sums (i64 cents etc.)       */
+  MZ_AGG_SUM_F64     /* fixed-point accumulation, reduce.rs:1641-1697  */
+};
+
+typedef struct {
+  uint8_t  func;     /* MZ_AGG_*                                       */
+  uint16_t off;      /* byte offset of the datum in the input val      */
+  uint8_t  width;    /* 4 or 8                                         */
+  uint8_t  is_float; /* datum is f64 (for SUM_F64)                     */
+  uint8_t  nullable; /* datum has a null indicator byte at off+width   */
+} mz_gpu_aggregate;
+
+#define MZ_GPU_MAX_AGGS 4
+
+typedef struct {
+  uint32_t n_aggs;
+  mz_gpu_aggregate aggs[MZ_GPU_MAX_AGGS];
+  mz_gpu_schema in;   /* input (key,val) schema                        */
+  mz_gpu_schema out;  /* output: key + finalized aggregate row         */
+} mz_gpu_reduce_spec;
+
+/* --------------------------------------------------------------- context */
+
+typedef struct {
+  uint64_t hbm_pool_bytes;   /* 0 = default                            */
+  uint32_t device_index;
+} mz_gpu_cfg;
+
+mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg);
+void        mz_gpu_fini(mz_gpu_ctx *ctx);
+const char *mz_gpu_last_error(mz_gpu_ctx *ctx);
+/* Synchronize the context's stream (all prior calls complete). */
+int         mz_gpu_sync(mz_gpu_ctx *ctx);
+
+/* ----------------------------------------------------------- arrangement
+ * Replaces mz_arrange_core + Spine maintenance
+ * (src/compute/src/extensions/arrange.rs:69-114,
+ *  src/compute/src/arrangement/manager.rs:54). */
+mz_gpu_arr *mz_gpu_arr_create(mz_gpu_ctx *ctx, const mz_gpu_schema *schema);
+void        mz_gpu_arr_drop(mz_gpu_ctx *ctx, mz_gpu_arr *arr);
+/* Push a sealed, sorted, consolidated batch (see mz_gpu_updates docs). */
+int  mz_gpu_arr_push_batch(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                           const mz_gpu_updates *batch);
+/* Advance the logical compaction frontier (times advance to it on merge) —
+ * cf. set_logical_compaction, mz_join_core.rs:461. */
+int  mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                                       uint64_t frontier);
+/* Perform up to `fuel` rows of spine merge work — cf. manager.rs:54. */
+int  mz_gpu_arr_maintain(mz_gpu_ctx *ctx, mz_gpu_arr *arr, uint64_t fuel);
+/* Introspection (arrangement-size logging, extensions/arrange.rs:249). */
+int  mz_gpu_arr_stats(mz_gpu_ctx *ctx, mz_gpu_arr *arr, uint64_t *n_batches,
+                      uint64_t *n_updates, uint64_t *hbm_bytes);
+
+/* ------------------------------------------------------------- out batch */
+typedef struct {
+  uint64_t *keys;  uint8_t *vals;  uint64_t *times;  int64_t *diffs;
+  uint64_t n;
+  int32_t on_device;      /* 1: device pointers (default)              */
+  mz_gpu_schema schema;
+} mz_gpu_out;
+
+void mz_gpu_out_release(mz_gpu_ctx *ctx, mz_gpu_out *out);
+/* Copy an out-batch's columns to caller host buffers (sized n). */
+int  mz_gpu_out_to_host(mz_gpu_ctx *ctx, const mz_gpu_out *out,
+                        uint64_t *keys, uint8_t *vals, uint64_t *times,
+                        int64_t *diffs);
+
+/* Sort by (key,val,time), consolidate diffs, drop zeros — DD
+ * consolidate_updates as used at mz_join_core.rs:604. */
+int  mz_gpu_consolidate(mz_gpu_ctx *ctx, const mz_gpu_schema *schema,
+                        const mz_gpu_updates *in, mz_gpu_out **out);
+
+/* ------------------------------------------------------------ linear join
+ * Replaces mz_join_core (mz_join_core.rs:57-496). The operator tracks the
+ * acknowledged frontier of each side implicitly: a push probes the opposing
+ * arrangement AS OF the call. Push concurrent batches side-1-first to
+ * reproduce the reference's drain order (exactly-once). */
+mz_gpu_join *mz_gpu_join_create(mz_gpu_ctx *ctx, mz_gpu_arr *arr1,
+                                mz_gpu_arr *arr2, const mz_gpu_closure *cl);
+void mz_gpu_join_drop(mz_gpu_ctx *ctx, mz_gpu_join *op);
+/* Join `delta` (side = 1 or 2, already consolidated) against the opposing
+ * arrangement; output consolidated. NOTE: push the delta to its own
+ * arrangement via mz_gpu_arr_push_batch BEFORE or AFTER this call per the
+ * reference discipline: arr_push(arr1,b1); join_push(1,b1);
+ * arr_push(arr2,b2); join_push(2,b2). */
+int  mz_gpu_join_push(mz_gpu_ctx *ctx, mz_gpu_join *op, int side,
+                      const mz_gpu_updates *delta, mz_gpu_out **out);
+
+/* ------------------------------------------------------------- half join
+ * Replaces half_join2 (delta_join.rs:500,544): probe `delta`'s updates
+ * (whose `times` are the promoted data-times) against `lookup`; a trace
+ * update at t' matches a stream update at t iff t' <= t (le=1; source
+ * relation precedes lookup relation) or t' < t (le=0) —
+ * delta_join.rs:362,372. Output time = t. Output consolidated.
+ * `stream_val_bytes` = the delta updates' val stride (the stream row was
+ * re-keyed/thinned by the previous stage and need not match `lookup`'s). */
+int  mz_gpu_halfjoin(mz_gpu_ctx *ctx, mz_gpu_arr *lookup,
+                     const mz_gpu_updates *delta, uint32_t stream_val_bytes,
+                     int le, const mz_gpu_closure *cl, mz_gpu_out **out);
+
+/* --------------------------------------------------------------- reduce
+ * Replaces build_accumulable + mz_reduce_abelian (reduce.rs:1357-1581,
+ * extensions/reduce.rs:131). The operator owns the resident accumulator
+ * table AND the output arrangement; each push returns output corrections
+ * (new minus old finalized rows, diffs ±1) per changed key. Input updates
+ * are (key, input-val) rows; the datum→accumulator move (explode_one,
+ * reduce.rs:1409-1431) happens inside. Multi-timestamp batches are
+ * processed in time order. */
+mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *ctx,
+                                 const mz_gpu_reduce_spec *spec);
+void mz_gpu_reduce_drop(mz_gpu_ctx *ctx, mz_gpu_red *op);
+int  mz_gpu_reduce_push(mz_gpu_ctx *ctx, mz_gpu_red *op,
+                        const mz_gpu_updates *delta, mz_gpu_out **out);
+
+/* ------------------------------------------------------------ exchange
+ * Replaces the Exchange pact routing (linear_join.rs:390,
+ * extensions/arrange.rs:134): shard = splitmix64(key words) % nshards
+ * (hash substitution per DESIGN.md §2.2). Writes each update into its
+ * shard's contiguous region of the caller-provided output columns (device
+ * or host, matching `in->on_device`) and fills counts[nshards]. The
+ * collective itself is the caller's (RCCL all-to-all-v over xGMI). */
+int  mz_gpu_partition(mz_gpu_ctx *ctx, const mz_gpu_schema *schema,
+                      const mz_gpu_updates *in, uint32_t nshards,
+                      uint64_t *out_keys, uint8_t *out_vals,
+                      uint64_t *out_times, int64_t *out_diffs,
+                      uint64_t *counts);
+
+/* The routing hash itself (host helper; device code uses the same). */
+uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* MZ_GPU_H */

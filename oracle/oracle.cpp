@@ -1,0 +1,910 @@
+// oracle/oracle.cpp — CPU restatement of Materialize's incremental
+// join/reduce hot path. *** TEST INFRASTRUCTURE ONLY ***
+//
+// This library is the parity oracle: only tests/, __graft_entry__.smoke()
+// and bench.py's cpu_baseline leg may load it. The product path
+// (materialize_amd/) never links or imports it.
+//
+// Restates, line-for-line in behavior (citations are file:line under
+// /root/reference):
+//  - linear join:  src/compute/src/render/join/mz_join_core.rs
+//      per-key merge scan (:644-663), EditList load+consolidation
+//      (:857-878), simple cross-product strategy (:755-767, chosen when
+//      either side has <10 edits, :743), linear time scan (:770-834),
+//      output consolidation (:604).
+//  - delta join:   src/compute/src/render/join/delta_join.rs
+//      half-join probe with le/lt time tie-break by relation order
+//      (:356-399); output time = the stream tuple's promoted data-time.
+//  - reduce:       src/compute/src/render/reduce.rs
+//      accumulable reduce (:1357-1581), Accum semigroup with wrapping
+//      i128 adds (:2102-2203), multiply by diff (:2205-2266), fixed-point
+//      float sums with 24 fractional bits (:1641-1697), finalize_accum
+//      (:1840-1997); corrections = new minus old output per changed key
+//      (reduce_abelian contract, src/compute/src/extensions/reduce.rs:131).
+//  - consolidation: DD consolidate_updates as used at mz_join_core.rs:604 —
+//      sort by (data, time), sum diffs (wrapping i64), drop zeros; ordering
+//      matches ColInternalMerger::merge
+//      (src/timely-util/src/columnation.rs:653-713).
+//  - arrangement:  sorted immutable columnar batches + merge with logical
+//      compaction (times advance to the frontier so add/retract pairs
+//      cancel) — RowRowSpine blueprint, src/row-spine/src/lib.rs:56-135.
+//
+// Diff = Overflowing<i64> wraps in release builds
+// (src/ore/src/overflowing.rs:24-31): all diff arithmetic here is wrapping.
+//
+// Parity pinning: tests/golden/ fixtures extracted from the reference's own
+// test/sqllogictest/{joins,aggregates}.slt literal results (see
+// tests/golden/README.md). The DD third-party boundary (batch/cursor
+// internals) is unpinned in-repo and covered by property
+// tests instead.
+
+#include "../include/mz_gpu.h"
+
+#include <algorithm>
+#include <cstdint>
+#include <cstring>
+#include <cstdio>
+#include <cmath>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+using u8 = uint8_t;
+using u32 = uint32_t;
+using u64 = uint64_t;
+using i64 = int64_t;
+using i128 = __int128;
+using u128 = unsigned __int128;
+
+namespace {
+
+// ------------------------------------------------------------------ rows
+
+struct Schema {
+  u32 kw;  // key words
+  u32 vb;  // val bytes
+};
+
+// SoA update columns (mirrors mz_gpu_updates host layout).
+struct Cols {
+  std::vector<u64> keys;  // kw * n
+  std::vector<u8> vals;   // vb * n
+  std::vector<u64> times;
+  std::vector<i64> diffs;
+  size_t size() const { return times.size(); }
+  void push(const u64 *k, u32 kw, const u8 *v, u32 vb, u64 t, i64 d) {
+    keys.insert(keys.end(), k, k + kw);
+    if (vb) vals.insert(vals.end(), v, v + vb);
+    times.push_back(t);
+    diffs.push_back(d);
+  }
+};
+
+// Key order: i64-tuple ascending. Val order: lexicographic
+// bytes (Row byte-lex order analog, row-spine batch ordering).
+inline int cmp_key(const u64 *a, const u64 *b, u32 kw) {
+  for (u32 i = 0; i < kw; i++) {
+    i64 x = (i64)a[i], y = (i64)b[i];
+    if (x < y) return -1;
+    if (x > y) return 1;
+  }
+  return 0;
+}
+inline int cmp_val(const u8 *a, const u8 *b, u32 vb) {
+  return vb ? std::memcmp(a, b, vb) : 0;
+}
+
+// consolidate_updates semantics: sort by (key, val, time), sum diffs
+// (wrapping), drop zeros. Cf. mz_join_core.rs:604;
+// timely-util/src/columnation.rs:686-694 (equal keys sum via plus_equals,
+// zeros dropped).
+void consolidate(const Schema &s, Cols &c) {
+  size_t n = c.size();
+  if (n == 0) return;
+  std::vector<u32> idx(n);
+  for (size_t i = 0; i < n; i++) idx[i] = (u32)i;
+  const u64 *K = c.keys.data();
+  const u8 *V = c.vals.data();
+  const u64 *T = c.times.data();
+  u32 kw = s.kw, vb = s.vb;
+  std::sort(idx.begin(), idx.end(), [&](u32 a, u32 b) {
+    int ck = cmp_key(K + (size_t)a * kw, K + (size_t)b * kw, kw);
+    if (ck) return ck < 0;
+    int cv = cmp_val(V + (size_t)a * vb, V + (size_t)b * vb, vb);
+    if (cv) return cv < 0;
+    if (T[a] != T[b]) return T[a] < T[b];
+    return a < b;  // stable
+  });
+  Cols out;
+  out.keys.reserve(c.keys.size());
+  out.vals.reserve(c.vals.size());
+  out.times.reserve(n);
+  out.diffs.reserve(n);
+  size_t i = 0;
+  while (i < n) {
+    size_t j = i;
+    i64 d = 0;
+    while (j < n &&
+           cmp_key(K + (size_t)idx[i] * kw, K + (size_t)idx[j] * kw, kw) == 0 &&
+           cmp_val(V + (size_t)idx[i] * vb, V + (size_t)idx[j] * vb, vb) == 0 &&
+           T[idx[i]] == T[idx[j]]) {
+      d = (i64)((u64)d + (u64)c.diffs[idx[j]]);  // wrapping add
+      j++;
+    }
+    if (d != 0)
+      out.push(K + (size_t)idx[i] * kw, kw, V + (size_t)idx[i] * vb, vb,
+               T[idx[i]], d);
+    i = j;
+  }
+  c = std::move(out);
+}
+
+// ------------------------------------------------------------ arrangement
+
+// A sealed batch: updates sorted by (key,val,time), consolidated.
+struct Batch {
+  Cols cols;
+  u64 lower, upper;
+  // key index: start offset of each distinct key run (plus end sentinel)
+  std::vector<size_t> key_starts;
+  void build_index(const Schema &s) {
+    key_starts.clear();
+    size_t n = cols.size();
+    const u64 *K = cols.keys.data();
+    for (size_t i = 0; i < n; i++)
+      if (i == 0 || cmp_key(K + (i - 1) * s.kw, K + i * s.kw, s.kw) != 0)
+        key_starts.push_back(i);
+    key_starts.push_back(n);
+  }
+  // binary search for key; returns [lo,hi) update range or empty
+  std::pair<size_t, size_t> seek(const Schema &s, const u64 *key) const {
+    if (key_starts.size() <= 1) return {0, 0};
+    size_t lo = 0, hi = key_starts.size() - 1;  // distinct-key count
+    const u64 *K = cols.keys.data();
+    while (lo < hi) {
+      size_t mid = (lo + hi) / 2;
+      if (cmp_key(K + key_starts[mid] * s.kw, key, s.kw) < 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    if (lo == key_starts.size() - 1) return {0, 0};
+    if (cmp_key(K + key_starts[lo] * s.kw, key, s.kw) != 0) return {0, 0};
+    return {key_starts[lo], key_starts[lo + 1]};
+  }
+};
+
+struct Arr {
+  Schema schema;
+  std::vector<std::unique_ptr<Batch>> batches;
+  u64 logical_compaction = 0;  // times advance to this on merge
+  u64 upper = 0;               // acknowledged frontier = max pushed upper
+
+  // Merge all batches into one, advancing times to the logical compaction
+  // frontier (add/retract pairs then cancel) — Spine merge + logical
+  // compaction semantics (mz_join_core.rs:458-465 contract; manager.rs:54).
+  void merge_all() {
+    Cols all;
+    u64 lo = UINT64_MAX, hi = 0;
+    for (auto &b : batches) {
+      lo = std::min(lo, b->lower);
+      hi = std::max(hi, b->upper);
+      size_t n = b->cols.size();
+      for (size_t i = 0; i < n; i++) {
+        u64 t = std::max(b->cols.times[i], logical_compaction);
+        all.push(b->cols.keys.data() + i * schema.kw, schema.kw,
+                 b->cols.vals.data() + i * schema.vb, schema.vb, t,
+                 b->cols.diffs[i]);
+      }
+    }
+    consolidate(schema, all);
+    batches.clear();
+    auto nb = std::make_unique<Batch>();
+    nb->cols = std::move(all);
+    nb->lower = lo == UINT64_MAX ? 0 : lo;
+    nb->upper = hi;
+    nb->build_index(schema);
+    batches.push_back(std::move(nb));
+  }
+};
+
+// ------------------------------------------------------------- closures
+
+// JoinClosure restatement (plan/join.rs:60-86): filters then field map.
+// For the linear join, VAL_STREAM = input 1's val and VAL_LOOKUP = input 2's
+// val regardless of which side the delta arrived on (result_fn(key,v1,v2),
+// mz_join_core.rs:69). For the half join, VAL_STREAM = the stream tuple's
+// val, VAL_LOOKUP = the arrangement's val (delta_join.rs:511-528).
+inline i64 read_int(const u8 *p, u8 width) {
+  if (width == 4) {
+    int32_t v;
+    std::memcpy(&v, p, 4);
+    return v;
+  }
+  i64 v;
+  std::memcpy(&v, p, 8);
+  return v;
+}
+
+inline const u8 *cl_src(const u64 *key, const u8 *v1, const u8 *v2, u8 src) {
+  switch (src) {
+    case MZ_SRC_KEY: return (const u8 *)key;
+    case MZ_SRC_VAL_STREAM: return v1;
+    default: return v2;
+  }
+}
+
+bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
+                   const u8 *v2, u64 *out_key, u8 *out_val) {
+  for (u32 i = 0; i < cl->n_filters; i++) {
+    const auto &f = cl->filters[i];
+    i64 x = read_int(cl_src(key, v1, v2, f.src) + f.off, f.width);
+    bool ok;
+    switch (f.cmp) {
+      case MZ_CMP_LT: ok = x < f.imm; break;
+      case MZ_CMP_LE: ok = x <= f.imm; break;
+      case MZ_CMP_GT: ok = x > f.imm; break;
+      case MZ_CMP_GE: ok = x >= f.imm; break;
+      case MZ_CMP_EQ: ok = x == f.imm; break;
+      default: ok = x != f.imm; break;
+    }
+    if (!ok) return false;
+  }
+  u8 *outs[2] = {(u8 *)out_key, out_val};
+  for (int which = 0; which < 2; which++) {
+    u32 nf = which ? cl->n_val_fields : cl->n_key_fields;
+    const mz_gpu_field *fs = which ? cl->val_fields : cl->key_fields;
+    u8 *dst = outs[which];
+    for (u32 i = 0; i < nf; i++) {
+      const auto &f = fs[i];
+      if (f.src == MZ_SRC_COMPUTE) {
+        // MZ_COMPUTE_REVENUE: extendedprice_cents * (10000 - discount_bp),
+        // exact i64 1e-4 currency units.
+        i64 ep = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+        i64 disc = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+        i64 rev = ep * (10000 - disc);
+        std::memcpy(dst, &rev, 8);
+        dst += 8;
+      } else {
+        std::memcpy(dst, cl_src(key, v1, v2, f.src) + f.off, f.width);
+        dst += f.width;
+      }
+    }
+  }
+  return true;
+}
+
+// ------------------------------------------------------ linear join (oracle)
+
+// EditList (mz_join_core.rs:841-891): per distinct val, consolidated
+// (time,diff) edits with times advanced by join(meet).
+struct EditList {
+  // values[i] = (val bytes offset into vals, end index into edits)
+  std::vector<std::pair<size_t, size_t>> values;
+  std::vector<u8> vals;  // vb-strided storage of distinct vals
+  std::vector<std::pair<u64, i64>> edits;
+  size_t len() const { return edits.size(); }
+};
+
+// Load a key's edits from one or more batches (the trace side presents a
+// merged cursor over batches, CursorList; vals iterate in sorted order with
+// per-val times from every batch — mz_join_core.rs:857-878).
+void editlist_load(const Schema &s, EditList &el,
+                   const std::vector<std::pair<const Batch *, std::pair<size_t, size_t>>> &ranges,
+                   u64 meet) {
+  el.values.clear();
+  el.vals.clear();
+  el.edits.clear();
+  u32 vb = s.vb;
+  // k-way merge of per-batch val runs in val order
+  struct Cur {
+    const Batch *b;
+    size_t i, end;
+  };
+  std::vector<Cur> cur;
+  for (auto &r : ranges)
+    if (r.second.first < r.second.second)
+      cur.push_back({r.first, r.second.first, r.second.second});
+  size_t edit_idx = 0;
+  while (!cur.empty()) {
+    // find smallest current val
+    const u8 *minv = nullptr;
+    for (auto &c : cur) {
+      const u8 *v = c.b->cols.vals.data() + c.i * vb;
+      if (!minv || cmp_val(v, minv, vb) < 0) minv = v;
+    }
+    u8 minv_copy[64];
+    if (vb) std::memcpy(minv_copy, minv, vb);
+    // collect all edits for this val across batches; times join with meet
+    for (auto &c : cur) {
+      while (c.i < c.end &&
+             cmp_val(c.b->cols.vals.data() + c.i * vb, minv_copy, vb) == 0) {
+        u64 t = std::max(c.b->cols.times[c.i], meet);  // time.join_assign(meet)
+        el.edits.emplace_back(t, c.b->cols.diffs[c.i]);
+        c.i++;
+      }
+    }
+    cur.erase(std::remove_if(cur.begin(), cur.end(),
+                             [](const Cur &c) { return c.i >= c.end; }),
+              cur.end());
+    // consolidate_from(edits, edit_idx) — mz_join_core.rs:869
+    std::sort(el.edits.begin() + edit_idx, el.edits.end());
+    size_t w = edit_idx;
+    for (size_t r = edit_idx; r < el.edits.size();) {
+      size_t q = r;
+      i64 d = 0;
+      while (q < el.edits.size() && el.edits[q].first == el.edits[r].first) {
+        d = (i64)((u64)d + (u64)el.edits[q].second);
+        q++;
+      }
+      if (d != 0) el.edits[w++] = {el.edits[r].first, d};
+      r = q;
+    }
+    el.edits.resize(w);
+    if (el.edits.size() > edit_idx) {
+      size_t off = el.vals.size();
+      el.vals.insert(el.vals.end(), minv_copy, minv_copy + vb);
+      edit_idx = el.edits.size();
+      el.values.emplace_back(off, edit_idx);
+    }
+  }
+}
+
+struct JoinOp {
+  Arr *arr1, *arr2;
+  mz_gpu_closure cl;
+};
+
+// ValueHistory for the linear time scan (mz_join_core.rs:893-975).
+struct ValueHistory {
+  EditList *edits;
+  // future entries: (time, meet, value_idx, diff), sorted DESCENDING;
+  // popped from the back (ascending time).
+  std::vector<std::tuple<u64, u64, size_t, i64>> future;
+  // past entries: (value_idx, time, diff)
+  std::vector<std::tuple<size_t, u64, i64>> past;
+
+  void replay() {  // :940-959
+    future.clear();
+    past.clear();
+    for (size_t idx = 0; idx < edits->values.size(); idx++) {
+      size_t start = idx == 0 ? 0 : edits->values[idx - 1].second;
+      size_t end = edits->values[idx].second;
+      for (size_t e = start; e < end; e++)
+        future.emplace_back(edits->edits[e].first, edits->edits[e].first, idx,
+                            edits->edits[e].second);
+    }
+    std::sort(future.begin(), future.end(),
+              [](const auto &x, const auto &y) { return y < x; });
+    for (size_t i = 1; i < future.size(); i++)
+      std::get<1>(future[i]) =
+          std::min(std::get<1>(future[i]), std::get<1>(future[i - 1]));  // meet
+  }
+  bool empty() const { return future.empty(); }
+  void step() {  // :962-966
+    auto [t, m, v, r] = future.back();
+    future.pop_back();
+    past.emplace_back(v, t, r);
+  }
+  void advance_past_by(u64 meet) {  // :969-974
+    for (auto &p : past) std::get<1>(p) = std::max(std::get<1>(p), meet);
+    // consolidate_updates on (value_idx, time, diff)
+    std::sort(past.begin(), past.end());
+    size_t w = 0;
+    for (size_t r = 0; r < past.size();) {
+      size_t q = r;
+      i64 d = 0;
+      while (q < past.size() &&
+             std::get<0>(past[q]) == std::get<0>(past[r]) &&
+             std::get<1>(past[q]) == std::get<1>(past[r])) {
+        d = (i64)((u64)d + (u64)std::get<2>(past[q]));
+        q++;
+      }
+      if (d != 0) past[w++] = {std::get<0>(past[r]), std::get<1>(past[r]), d};
+      r = q;
+    }
+    past.resize(w);
+  }
+};
+
+// Produce matches for one key (mz_join_core.rs:727-834). `swap` = true when
+// the delta came from input 2 (history1 is then the trace side) — result_fn
+// argument order is always (key, val-of-input1, val-of-input2).
+void join_key(const mz_gpu_closure *cl, const u64 *key, EditList &h1,
+              EditList &h2, Cols &out, const Schema &os) {
+  std::vector<u64> okey(os.kw);
+  std::vector<u8> oval(os.vb ? os.vb : 1);
+  auto emit = [&](const u8 *v1, const u8 *v2, u64 t, i64 r) {
+    if (closure_apply(cl, key, v1, v2, okey.data(), oval.data()))
+      out.push(okey.data(), os.kw, oval.data(), os.vb, t, r);
+  };
+  if (h1.len() < 10 || h2.len() < 10) {
+    // simple strategy: full cross product (:755-767)
+    for (size_t i1 = 0; i1 < h1.values.size(); i1++) {
+      size_t s1 = i1 == 0 ? 0 : h1.values[i1 - 1].second;
+      for (size_t e1 = s1; e1 < h1.values[i1].second; e1++) {
+        for (size_t i2 = 0; i2 < h2.values.size(); i2++) {
+          size_t s2 = i2 == 0 ? 0 : h2.values[i2 - 1].second;
+          for (size_t e2 = s2; e2 < h2.values[i2].second; e2++) {
+            u64 t = std::max(h1.edits[e1].first, h2.edits[e2].first);
+            i64 r = (i64)((u64)h1.edits[e1].second * (u64)h2.edits[e2].second);
+            emit(h1.vals.data() + h1.values[i1].first,
+                 h2.vals.data() + h2.values[i2].first, t, r);
+          }
+        }
+      }
+    }
+  } else {
+    // linear time scan (:770-834)
+    ValueHistory vh1{&h1}, vh2{&h2};
+    vh1.replay();
+    vh2.replay();
+    auto work1 = [&]() {
+      auto [t1, meet, v1, r1] = vh1.future.back();
+      vh2.advance_past_by(meet);
+      for (auto &[v2, t2, r2] : vh2.past)
+        emit(h1.vals.data() + h1.values[v1].first,
+             h2.vals.data() + h2.values[v2].first, std::max(t1, t2),
+             (i64)((u64)r1 * (u64)r2));
+      vh1.step();
+    };
+    auto work2 = [&]() {
+      auto [t2, meet, v2, r2] = vh2.future.back();
+      vh1.advance_past_by(meet);
+      for (auto &[v1, t1, r1] : vh1.past)
+        emit(h1.vals.data() + h1.values[v1].first,
+             h2.vals.data() + h2.values[v2].first, std::max(t1, t2),
+             (i64)((u64)r1 * (u64)r2));
+      vh2.step();
+    };
+    while (!vh1.empty() && !vh2.empty()) {
+      if (std::get<0>(vh1.future.back()) < std::get<0>(vh2.future.back()))
+        work1();
+      else
+        work2();
+    }
+    while (!vh1.empty()) work1();
+    while (!vh2.empty()) work2();
+  }
+}
+
+// ------------------------------------------------------------ reduce
+
+// Accum restatement (reduce.rs:1611-2270). The benchmark path carries
+// COUNT / SUM_I64 / SUM_F64; each slot keeps the full Float bookkeeping.
+struct Accum {
+  i128 accum = 0;  // wrapping
+  i64 non_nulls = 0, pos_infs = 0, neg_infs = 0, nans = 0;
+  bool is_zero() const {
+    return accum == 0 && non_nulls == 0 && pos_infs == 0 && neg_infs == 0 &&
+           nans == 0;
+  }
+  void plus(const Accum &o) {  // :2102-2149 (wrapping adds)
+    accum = (i128)((u128)accum + (u128)o.accum);
+    non_nulls = (i64)((u64)non_nulls + (u64)o.non_nulls);
+    pos_infs = (i64)((u64)pos_infs + (u64)o.pos_infs);
+    neg_infs = (i64)((u64)neg_infs + (u64)o.neg_infs);
+    nans = (i64)((u64)nans + (u64)o.nans);
+  }
+  void mul(i64 f) {  // :2205-2236
+    accum = (i128)((u128)accum * (u128)(i128)f);
+    non_nulls = (i64)((u64)non_nulls * (u64)f);
+    pos_infs = (i64)((u64)pos_infs * (u64)f);
+    neg_infs = (i64)((u64)neg_infs * (u64)f);
+    nans = (i64)((u64)nans * (u64)f);
+  }
+};
+
+// float_to_fixed_point (reduce.rs:1663-1697): trunc(n * 2^24) mod 2^128,
+// wrapping (group homomorphism), via integer_decode.
+i128 float_to_fixed_point(double n) {
+  u64 bits;
+  std::memcpy(&bits, &n, 8);
+  // Rust Float::integer_decode: mantissa incl. implicit bit, exponent-1075
+  u64 mantissa = bits & ((1ULL << 52) - 1);
+  int exp_bits = (int)((bits >> 52) & 0x7ff);
+  int exponent;
+  if (exp_bits == 0) {
+    exponent = -1074;  // subnormal: no implicit bit
+  } else {
+    mantissa |= 1ULL << 52;
+    exponent = exp_bits - 1075;
+  }
+  int sign = (bits >> 63) ? -1 : 1;
+  long exp = (long)exponent + 24;
+  u128 significand = (u128)mantissa;
+  u128 magnitude;
+  if (exp >= 0) {
+    magnitude = exp < 128 ? (significand << exp) : (u128)0;
+  } else {
+    long sh = -exp;
+    magnitude = sh < 128 ? (significand >> sh) : (u128)0;
+  }
+  i128 m = (i128)magnitude;
+  return sign < 0 ? (i128)(~(u128)m + 1) : m;
+}
+
+// datum_to_accumulator (reduce.rs:1699-1838) for the three funcs.
+Accum datum_to_accum(const mz_gpu_aggregate &a, const u8 *val) {
+  Accum r;
+  bool null = a.nullable && val[a.off + a.width] != 0;
+  switch (a.func) {
+    case MZ_AGG_COUNT:
+      r.non_nulls = null ? 0 : 1;
+      break;
+    case MZ_AGG_SUM_I64:
+      if (!null) {
+        r.accum = (i128)read_int(val + a.off, a.width);
+        r.non_nulls = 1;
+      }
+      break;
+    case MZ_AGG_SUM_F64:
+      if (!null) {
+        double n;
+        if (a.width == 4) {
+          float f;
+          std::memcpy(&f, val + a.off, 4);
+          n = (double)f;
+        } else {
+          std::memcpy(&n, val + a.off, 8);
+        }
+        r.nans = std::isnan(n) ? 1 : 0;
+        r.pos_infs = (n == HUGE_VAL) ? 1 : 0;
+        r.neg_infs = (n == -HUGE_VAL) ? 1 : 0;
+        r.non_nulls = 1;
+        if (!r.nans && !r.pos_infs && !r.neg_infs)
+          r.accum = float_to_fixed_point(n);
+      }
+      break;
+  }
+  return r;
+}
+
+// finalize_accum (reduce.rs:1840-1997). Output slot layout:
+// 24 bytes per aggregate: { u8 null; u8 pad[7]; 16-byte value }.
+//   COUNT    -> i64 non_nulls in low 8 bytes
+//   SUM_I64  -> i128 little-endian (matches Datum::from(i128), :1882)
+//   SUM_F64  -> f64 bits in low 8 bytes (fixed-point decode, :1952)
+void finalize_accum(const mz_gpu_aggregate &a, const Accum &ac, i64 total,
+                    u8 *slot) {
+  std::memset(slot, 0, 24);
+  // reduce.rs:1844: total positive, accum zero, func != COUNT => NULL
+  if (total > 0 && ac.is_zero() && a.func != MZ_AGG_COUNT) {
+    slot[0] = 1;
+    return;
+  }
+  switch (a.func) {
+    case MZ_AGG_COUNT: {
+      i64 c = ac.non_nulls;
+      std::memcpy(slot + 8, &c, 8);
+      break;
+    }
+    case MZ_AGG_SUM_I64: {
+      i128 v = ac.accum;
+      std::memcpy(slot + 8, &v, 16);
+      break;
+    }
+    case MZ_AGG_SUM_F64: {
+      double v;
+      if (ac.nans > 0 || (ac.pos_infs > 0 && ac.neg_infs > 0))
+        v = NAN;
+      else if (ac.pos_infs > 0)
+        v = HUGE_VAL;
+      else if (ac.neg_infs > 0)
+        v = -HUGE_VAL;
+      else
+        v = (double)ac.accum / 16777216.0;  // / 2^24, :1952
+      std::memcpy(slot + 8, &v, 8);
+      break;
+    }
+  }
+}
+
+struct AccumRow {
+  Accum a[MZ_GPU_MAX_AGGS];
+  i64 total = 0;
+  bool exists() const {
+    if (total != 0) return true;
+    for (auto &x : a)
+      if (!x.is_zero()) return true;
+    return false;
+  }
+};
+
+struct RedOp {
+  mz_gpu_reduce_spec spec;
+  // deterministic iteration: ordered map on key words
+  std::map<std::vector<u64>, AccumRow> state;
+};
+
+// ------------------------------------------------------------- context
+
+struct Out {
+  mz_gpu_out pub_;
+  Cols cols;
+  Schema schema;
+};
+
+}  // namespace
+
+struct orc_ctx {
+  std::string err;
+  std::vector<std::unique_ptr<Arr>> arrs;
+  std::vector<std::unique_ptr<JoinOp>> joins;
+  std::vector<std::unique_ptr<RedOp>> reds;
+};
+
+extern "C" {
+
+orc_ctx *orc_init() { return new orc_ctx(); }
+void orc_fini(orc_ctx *c) { delete c; }
+const char *orc_last_error(orc_ctx *c) { return c->err.c_str(); }
+
+Arr *orc_arr_create(orc_ctx *c, const mz_gpu_schema *s) {
+  auto a = std::make_unique<Arr>();
+  a->schema = {s->key_words, s->val_bytes};
+  Arr *p = a.get();
+  c->arrs.push_back(std::move(a));
+  return p;
+}
+
+static void cols_from_updates(const mz_gpu_updates *u, const Schema &s,
+                              Cols &c) {
+  c.keys.assign(u->keys, u->keys + u->n * s.kw);
+  if (s.vb)
+    c.vals.assign(u->vals, u->vals + u->n * s.vb);
+  c.times.assign(u->times, u->times + u->n);
+  c.diffs.assign(u->diffs, u->diffs + u->n);
+}
+
+int orc_arr_push_batch(orc_ctx *c, Arr *a, const mz_gpu_updates *u) {
+  auto b = std::make_unique<Batch>();
+  cols_from_updates(u, a->schema, b->cols);
+  b->lower = u->lower;
+  b->upper = u->upper;
+  b->build_index(a->schema);
+  a->batches.push_back(std::move(b));
+  a->upper = std::max(a->upper, u->upper);
+  return 0;
+}
+
+int orc_arr_set_logical_compaction(orc_ctx *c, Arr *a, u64 f) {
+  a->logical_compaction = f;
+  return 0;
+}
+
+int orc_arr_maintain(orc_ctx *c, Arr *a, u64 fuel) {
+  (void)fuel;  // oracle merges eagerly (scheduling policy)
+  a->merge_all();
+  return 0;
+}
+
+int orc_arr_stats(orc_ctx *c, Arr *a, u64 *nb, u64 *nu, u64 *bytes) {
+  *nb = a->batches.size();
+  u64 n = 0, by = 0;
+  for (auto &b : a->batches) {
+    n += b->cols.size();
+    by += b->cols.keys.size() * 8 + b->cols.vals.size() +
+          b->cols.size() * 16;
+  }
+  *nu = n;
+  *bytes = by;
+  return 0;
+}
+
+JoinOp *orc_join_create(orc_ctx *c, Arr *a1, Arr *a2,
+                        const mz_gpu_closure *cl) {
+  auto j = std::make_unique<JoinOp>();
+  j->arr1 = a1;
+  j->arr2 = a2;
+  j->cl = *cl;
+  JoinOp *p = j.get();
+  c->joins.push_back(std::move(j));
+  return p;
+}
+
+static Out *make_out(Cols &&c, const Schema &s) {
+  Out *o = new Out();
+  o->cols = std::move(c);
+  o->schema = s;
+  o->pub_.keys = o->cols.keys.data();
+  o->pub_.vals = o->cols.vals.data();
+  o->pub_.times = o->cols.times.data();
+  o->pub_.diffs = o->cols.diffs.data();
+  o->pub_.n = o->cols.size();
+  o->pub_.on_device = 0;
+  o->pub_.schema = {s.kw, s.vb};
+  return o;
+}
+
+// mz_join_core's per-batch work: join `delta` (side 1 or 2) against the
+// opposing arrangement's batches as of now (= cursor_through(acknowledged),
+// mz_join_core.rs:237-368). meet = delta batch's capability time (lower).
+int orc_join_push(orc_ctx *c, JoinOp *op, int side, const mz_gpu_updates *u,
+                  mz_gpu_out **out) {
+  Arr *own = side == 1 ? op->arr1 : op->arr2;
+  Arr *opp = side == 1 ? op->arr2 : op->arr1;
+  const Schema &ds = own->schema;  // delta rows use their own side's schema
+  const Schema &ts = opp->schema;  // trace rows use the opposing schema
+  Batch delta;
+  cols_from_updates(u, ds, delta.cols);
+  delta.lower = u->lower;
+  delta.upper = u->upper;
+  delta.build_index(ds);
+  u64 meet = u->lower;
+  Schema os = {op->cl.out.key_words, op->cl.out.val_bytes};
+  Cols result;
+  EditList el_delta, el_trace;
+  // merge scan over delta's distinct keys (start_work, :644-663): the delta
+  // cursor drives; trace cursors seek.
+  size_t ndk = delta.key_starts.size() ? delta.key_starts.size() - 1 : 0;
+  for (size_t ki = 0; ki < ndk; ki++) {
+    const u64 *key = delta.cols.keys.data() + delta.key_starts[ki] * ds.kw;
+    std::vector<std::pair<const Batch *, std::pair<size_t, size_t>>> tr;
+    bool any = false;
+    for (auto &b : opp->batches) {
+      auto r = b->seek(ts, key);
+      if (r.first < r.second) {
+        tr.push_back({b.get(), r});
+        any = true;
+      }
+    }
+    if (!any) continue;
+    editlist_load(ds, el_delta,
+                  {{&delta, {delta.key_starts[ki], delta.key_starts[ki + 1]}}},
+                  meet);
+    editlist_load(ts, el_trace, tr, meet);
+    if (el_delta.len() == 0 || el_trace.len() == 0) continue;
+    if (side == 1)
+      join_key(&op->cl, key, el_delta, el_trace, result, os);
+    else
+      join_key(&op->cl, key, el_trace, el_delta, result, os);
+  }
+  consolidate(os, result);
+  *out = &make_out(std::move(result), os)->pub_;
+  return 0;
+}
+
+// half_join probe (delta_join.rs:500-583 / half_join2 contract): for each
+// stream update (key, val, t) probe `lookup`; trace updates at t' match iff
+// le ? t' <= t : t' < t; emit (closure(key, stream_val, lookup_val), t,
+// d_stream * d_lookup). Output consolidated. `stream_vb` = the stream
+// updates' val stride (may differ from the lookup arrangement's).
+int orc_halfjoin(orc_ctx *c, Arr *lookup, const mz_gpu_updates *u,
+                 u32 stream_vb, int le, const mz_gpu_closure *cl,
+                 mz_gpu_out **out) {
+  const Schema &ls = lookup->schema;
+  Schema os = {cl->out.key_words, cl->out.val_bytes};
+  Cols result;
+  std::vector<u64> okey(os.kw);
+  std::vector<u8> oval(os.vb ? os.vb : 1);
+  for (u64 i = 0; i < u->n; i++) {
+    const u64 *key = u->keys + i * ls.kw;
+    const u8 *sval = stream_vb ? u->vals + i * stream_vb : nullptr;
+    u64 t = u->times[i];
+    i64 d1 = u->diffs[i];
+    for (auto &b : lookup->batches) {
+      auto r = b->seek(ls, key);
+      for (size_t j = r.first; j < r.second; j++) {
+        u64 t2 = b->cols.times[j];
+        if (le ? (t2 <= t) : (t2 < t)) {
+          const u8 *lval = ls.vb ? b->cols.vals.data() + j * ls.vb : nullptr;
+          if (closure_apply(cl, key, sval, lval, okey.data(), oval.data())) {
+            i64 d = (i64)((u64)d1 * (u64)b->cols.diffs[j]);
+            result.push(okey.data(), os.kw, oval.data(), os.vb, t, d);
+          }
+        }
+      }
+    }
+  }
+  consolidate(os, result);
+  *out = &make_out(std::move(result), os)->pub_;
+  return 0;
+}
+
+RedOp *orc_reduce_create(orc_ctx *c, const mz_gpu_reduce_spec *spec) {
+  auto r = std::make_unique<RedOp>();
+  r->spec = *spec;
+  RedOp *p = r.get();
+  c->reds.push_back(std::move(r));
+  return p;
+}
+
+// build_accumulable push: move datums into accumulators scaled by diff
+// (explode_one, reduce.rs:1409-1431 + Multiply :2205), merge into resident
+// state per key (Semigroup :2102), emit corrections new-minus-old per
+// changed key at each timestamp in order (reduce_abelian contract).
+int orc_reduce_push(orc_ctx *c, RedOp *op, const mz_gpu_updates *u,
+                    mz_gpu_out **out) {
+  const Schema in = {op->spec.in.key_words, op->spec.in.val_bytes};
+  Schema os = {op->spec.out.key_words, op->spec.out.val_bytes};
+  u32 na = op->spec.n_aggs;
+  Cols result;
+  // group updates by time (ascending), then process per time
+  std::vector<u64> order(u->n);
+  for (u64 i = 0; i < u->n; i++) order[i] = i;
+  std::stable_sort(order.begin(), order.end(),
+                   [&](u64 a, u64 b) { return u->times[a] < u->times[b]; });
+  std::vector<u8> oldrow(os.vb), newrow(os.vb);
+  size_t p = 0;
+  while (p < order.size()) {
+    u64 t = u->times[order[p]];
+    // apply all updates at time t, tracking changed keys
+    std::map<std::vector<u64>, AccumRow> olds;
+    while (p < order.size() && u->times[order[p]] == t) {
+      u64 i = order[p++];
+      std::vector<u64> key(u->keys + i * in.kw, u->keys + (i + 1) * in.kw);
+      auto it = op->state.find(key);
+      if (olds.find(key) == olds.end())
+        olds[key] = it != op->state.end() ? it->second : AccumRow();
+      AccumRow &st = op->state[key];
+      const u8 *val = in.vb ? u->vals + i * in.vb : nullptr;
+      i64 d = u->diffs[i];
+      for (u32 a = 0; a < na; a++) {
+        Accum ac = datum_to_accum(op->spec.aggs[a], val);
+        ac.mul(d);
+        st.a[a].plus(ac);
+      }
+      st.total = (i64)((u64)st.total + (u64)d);
+    }
+    // emit corrections per changed key
+    for (auto &[key, old] : olds) {
+      AccumRow &nw = op->state[key];
+      bool oe = old.exists(), ne = nw.exists();
+      if (oe) {
+        for (u32 a = 0; a < na; a++)
+          finalize_accum(op->spec.aggs[a], old.a[a], old.total,
+                         oldrow.data() + 24 * a);
+      }
+      if (ne) {
+        for (u32 a = 0; a < na; a++)
+          finalize_accum(op->spec.aggs[a], nw.a[a], nw.total,
+                         newrow.data() + 24 * a);
+      }
+      if (oe && ne && std::memcmp(oldrow.data(), newrow.data(), os.vb) == 0)
+        continue;
+      if (oe) result.push(key.data(), os.kw, oldrow.data(), os.vb, t, -1);
+      if (ne) result.push(key.data(), os.kw, newrow.data(), os.vb, t, 1);
+      if (!ne) op->state.erase(key);
+    }
+  }
+  consolidate(os, result);
+  *out = &make_out(std::move(result), os)->pub_;
+  return 0;
+}
+
+int orc_consolidate(orc_ctx *c, const mz_gpu_schema *sc,
+                    const mz_gpu_updates *u, mz_gpu_out **out) {
+  Schema s = {sc->key_words, sc->val_bytes};
+  Cols cols;
+  cols_from_updates(u, s, cols);
+  consolidate(s, cols);
+  Out *o = make_out(std::move(cols), s);
+  o->pub_.schema = *sc;
+  *out = &o->pub_;
+  return 0;
+}
+
+void orc_out_release(orc_ctx *c, mz_gpu_out *o) {
+  // Out::pub_ is the first member; the public pointer IS the Out pointer.
+  delete reinterpret_cast<Out *>(o);
+}
+
+// Routing hash: splitmix64 over the key words (substitute for fixed-seed
+// ahash, used identically by oracle and GPU — DESIGN.md §2.2).
+u64 orc_route_hash(const u64 *kw, u32 n) {
+  u64 h = 0x9E3779B97F4A7C15ULL;
+  for (u32 i = 0; i < n; i++) {
+    u64 x = kw[i] + h;
+    x ^= x >> 30;
+    x *= 0xBF58476D1CE4E5B9ULL;
+    x ^= x >> 27;
+    x *= 0x94D049BB133111EBULL;
+    x ^= x >> 31;
+    h = x;
+  }
+  return h;
+}
+
+}  // extern "C"

@@ -1,0 +1,139 @@
+"""oracle/pyoracle.py — ctypes wrapper over liboracle.so.
+
+*** TEST INFRASTRUCTURE ONLY *** — only tests/, __graft_entry__.smoke() and
+bench.py's cpu_baseline leg may import this module. The product package
+(materialize_amd/) never does.
+"""
+import ctypes as C
+import os
+import subprocess
+
+from materialize_amd._abi import (  # type defs of the shared boundary
+    Closure, OutBatch, ReduceSpec, Schema, Updates, out_to_numpy,
+)
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB = None
+
+
+def _load():
+    global _LIB
+    if _LIB is not None:
+        return _LIB
+    path = os.path.join(_HERE, "liboracle.so")
+    if not os.path.exists(path):
+        subprocess.run(["make", "-C", _HERE], check=True,
+                       capture_output=True)
+    lib = C.CDLL(path)
+    lib.orc_init.restype = C.c_void_p
+    lib.orc_fini.argtypes = [C.c_void_p]
+    lib.orc_last_error.restype = C.c_char_p
+    lib.orc_last_error.argtypes = [C.c_void_p]
+    lib.orc_arr_create.restype = C.c_void_p
+    lib.orc_arr_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
+    lib.orc_arr_push_batch.argtypes = [C.c_void_p, C.c_void_p,
+                                       C.POINTER(Updates)]
+    lib.orc_arr_set_logical_compaction.argtypes = [C.c_void_p, C.c_void_p,
+                                                   C.c_uint64]
+    lib.orc_arr_maintain.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
+    lib.orc_arr_stats.argtypes = [C.c_void_p, C.c_void_p] + \
+        [C.POINTER(C.c_uint64)] * 3
+    lib.orc_join_create.restype = C.c_void_p
+    lib.orc_join_create.argtypes = [C.c_void_p, C.c_void_p, C.c_void_p,
+                                    C.POINTER(Closure)]
+    lib.orc_join_push.argtypes = [C.c_void_p, C.c_void_p, C.c_int,
+                                  C.POINTER(Updates),
+                                  C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_halfjoin.argtypes = [C.c_void_p, C.c_void_p, C.POINTER(Updates),
+                                 C.c_uint32, C.c_int, C.POINTER(Closure),
+                                 C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_reduce_create.restype = C.c_void_p
+    lib.orc_reduce_create.argtypes = [C.c_void_p, C.POINTER(ReduceSpec)]
+    lib.orc_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
+                                    C.POINTER(Updates),
+                                    C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
+                                    C.POINTER(Updates),
+                                    C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_out_release.argtypes = [C.c_void_p, C.POINTER(OutBatch)]
+    lib.orc_route_hash.restype = C.c_uint64
+    lib.orc_route_hash.argtypes = [C.POINTER(C.c_uint64), C.c_uint32]
+    _LIB = lib
+    return lib
+
+
+class OracleCtx:
+    def __init__(self):
+        self.lib = _load()
+        self.ctx = self.lib.orc_init()
+
+    def close(self):
+        if self.ctx:
+            self.lib.orc_fini(self.ctx)
+            self.ctx = None
+
+    def __del__(self):
+        self.close()
+
+    def _take(self, outp):
+        res = out_to_numpy(outp.contents)
+        self.lib.orc_out_release(self.ctx, outp)
+        return res
+
+    def arr_create(self, sch):
+        return self.lib.orc_arr_create(self.ctx, C.byref(sch))
+
+    def arr_push(self, arr, upd):
+        rc = self.lib.orc_arr_push_batch(self.ctx, arr, C.byref(upd))
+        assert rc == 0
+    def arr_set_logical_compaction(self, arr, frontier):
+        self.lib.orc_arr_set_logical_compaction(self.ctx, arr, frontier)
+
+    def arr_maintain(self, arr, fuel=0):
+        self.lib.orc_arr_maintain(self.ctx, arr, fuel)
+
+    def arr_stats(self, arr):
+        nb, nu, by = C.c_uint64(), C.c_uint64(), C.c_uint64()
+        self.lib.orc_arr_stats(self.ctx, arr, C.byref(nb), C.byref(nu),
+                               C.byref(by))
+        return nb.value, nu.value, by.value
+
+    def join_create(self, a1, a2, cl):
+        return self.lib.orc_join_create(self.ctx, a1, a2, C.byref(cl))
+
+    def join_push(self, op, side, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_join_push(self.ctx, op, side, C.byref(upd),
+                                    C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
+
+    def halfjoin(self, lookup, upd, stream_vb, le, cl):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_halfjoin(self.ctx, lookup, C.byref(upd), stream_vb,
+                                   1 if le else 0, C.byref(cl),
+                                   C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
+
+    def reduce_create(self, spec):
+        return self.lib.orc_reduce_create(self.ctx, C.byref(spec))
+
+    def reduce_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_reduce_push(self.ctx, op, C.byref(upd),
+                                      C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
+
+    def consolidate(self, sch, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_consolidate(self.ctx, C.byref(sch), C.byref(upd),
+                                      C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
+
+    def route_hash(self, words):
+        arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
+                                          for w in words])
+        return self.lib.orc_route_hash(arr, len(words))
