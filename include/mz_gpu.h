@@ -72,7 +72,9 @@ enum {
    * units — Q3/Q5/Q10's `l_extendedprice * (1 - l_discount)` with TPC-H
    * fixed-scale decimals mapped to exact integers. Operand
    * offsets are given by arg0/arg1 on the out field. */
-  MZ_COMPUTE_REVENUE = 0
+  MZ_COMPUTE_REVENUE = 0,
+  /* 8 zero bytes (re-key to a constant, e.g. cross-join stages). */
+  MZ_COMPUTE_CONST0 = 1
 };
 
 typedef struct {

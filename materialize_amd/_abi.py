@@ -16,6 +16,7 @@ MZ_SRC_COMPUTE = 3
 
 MZ_CMP_LT, MZ_CMP_LE, MZ_CMP_GT, MZ_CMP_GE, MZ_CMP_EQ, MZ_CMP_NE = range(6)
 MZ_COMPUTE_REVENUE = 0
+MZ_COMPUTE_CONST0 = 1
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 
 MZ_GPU_MAX_FILTERS = 4

@@ -259,12 +259,15 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
     for (u32 i = 0; i < nf; i++) {
       const auto &f = fs[i];
       if (f.src == MZ_SRC_COMPUTE) {
-        // MZ_COMPUTE_REVENUE: extendedprice_cents * (10000 - discount_bp),
-        // exact i64 1e-4 currency units.
-        i64 ep = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
-        i64 disc = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
-        i64 rev = ep * (10000 - disc);
-        std::memcpy(dst, &rev, 8);
+        i64 v = 0;
+        if (f.off == MZ_COMPUTE_REVENUE) {
+          // extendedprice_cents * (10000 - discount_bp), exact i64 1e-4
+          // currency units.
+          i64 ep = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 disc = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          v = ep * (10000 - disc);
+        }  // MZ_COMPUTE_CONST0 leaves v = 0
+        std::memcpy(dst, &v, 8);
         dst += 8;
       } else {
         std::memcpy(dst, cl_src(key, v1, v2, f.src) + f.off, f.width);
