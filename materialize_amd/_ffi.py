@@ -1,0 +1,205 @@
+"""materialize_amd._ffi — ctypes bindings to libmzgpu.so (the product).
+
+Loads the in-tree HIP library. On a machine with a GPU this module FAILS
+LOUDLY if the extension is missing or fails to initialize — there is no
+CPU fallback on the product path.
+"""
+import ctypes as C
+import os
+import subprocess
+
+from ._abi import (Closure, OutBatch, ReduceSpec, Schema, Updates,
+                   out_to_numpy)
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "csrc", "libmzgpu.so")
+_LIB = None
+
+
+class MzGpuError(RuntimeError):
+    pass
+
+
+def _torch_has_gpu():
+    try:
+        import torch
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def load():
+    """Load and bind libmzgpu.so. Raises MzGpuError if unavailable."""
+    global _LIB
+    if _LIB is not None:
+        return _LIB
+    if not os.path.exists(_LIB_PATH):
+        # attempt an in-tree build (hipcc cross-compiles without a GPU)
+        try:
+            subprocess.run(["make", "-C", os.path.dirname(_LIB_PATH)],
+                           check=True, capture_output=True)
+        except Exception as e:
+            raise MzGpuError(
+                f"libmzgpu.so missing at {_LIB_PATH} and in-tree build "
+                f"failed: {e}. The HIP extension is required — there is no "
+                f"CPU fallback.") from e
+    lib = C.CDLL(_LIB_PATH)
+    lib.mz_gpu_init.restype = C.c_void_p
+    lib.mz_gpu_init.argtypes = [C.c_void_p]
+    lib.mz_gpu_fini.argtypes = [C.c_void_p]
+    lib.mz_gpu_last_error.restype = C.c_char_p
+    lib.mz_gpu_last_error.argtypes = [C.c_void_p]
+    lib.mz_gpu_sync.argtypes = [C.c_void_p]
+    lib.mz_gpu_arr_create.restype = C.c_void_p
+    lib.mz_gpu_arr_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
+    lib.mz_gpu_arr_drop.argtypes = [C.c_void_p, C.c_void_p]
+    lib.mz_gpu_arr_push_batch.argtypes = [C.c_void_p, C.c_void_p,
+                                          C.POINTER(Updates)]
+    lib.mz_gpu_arr_set_logical_compaction.argtypes = [C.c_void_p, C.c_void_p,
+                                                      C.c_uint64]
+    lib.mz_gpu_arr_maintain.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
+    lib.mz_gpu_arr_stats.argtypes = [C.c_void_p, C.c_void_p] + \
+        [C.POINTER(C.c_uint64)] * 3
+    lib.mz_gpu_out_release.argtypes = [C.c_void_p, C.POINTER(OutBatch)]
+    lib.mz_gpu_out_to_host.argtypes = [
+        C.c_void_p, C.POINTER(OutBatch), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.POINTER(C.c_int64)]
+    lib.mz_gpu_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
+                                       C.POINTER(Updates),
+                                       C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_join_create.restype = C.c_void_p
+    lib.mz_gpu_join_create.argtypes = [C.c_void_p, C.c_void_p, C.c_void_p,
+                                       C.POINTER(Closure)]
+    lib.mz_gpu_join_push.argtypes = [C.c_void_p, C.c_void_p, C.c_int,
+                                     C.POINTER(Updates),
+                                     C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_halfjoin.argtypes = [C.c_void_p, C.c_void_p,
+                                    C.POINTER(Updates), C.c_uint32, C.c_int,
+                                    C.POINTER(Closure),
+                                    C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_reduce_create.restype = C.c_void_p
+    lib.mz_gpu_reduce_create.argtypes = [C.c_void_p, C.POINTER(ReduceSpec)]
+    lib.mz_gpu_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
+                                       C.POINTER(Updates),
+                                       C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_partition.argtypes = [
+        C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_int64), C.POINTER(C.c_uint64)]
+    lib.mz_gpu_route_hash.restype = C.c_uint64
+    lib.mz_gpu_route_hash.argtypes = [C.POINTER(C.c_uint64), C.c_uint32]
+    lib.mz_gpu_set_kernel_timing.argtypes = [C.c_void_p, C.c_int]
+    lib.mz_gpu_get_probe_stats.argtypes = [C.c_void_p, C.POINTER(C.c_double),
+                                           C.POINTER(C.c_uint64),
+                                           C.POINTER(C.c_uint64)]
+    _LIB = lib
+    return lib
+
+
+class GpuCtx:
+    """One engine context on one GPU (one driver thread per GPU —
+    mirrors a timely worker, server.rs:327-377)."""
+
+    def __init__(self, device=0):
+        self.lib = load()
+        from ._abi import Cfg
+        cfg = Cfg(hbm_pool_bytes=0, device_index=device)
+        self.ctx = self.lib.mz_gpu_init(C.byref(cfg))
+        if not self.ctx:
+            raise MzGpuError(
+                "mz_gpu_init failed: no HIP device available. The product "
+                "path requires an MI355X; there is no CPU fallback.")
+
+    def close(self):
+        if getattr(self, "ctx", None):
+            self.lib.mz_gpu_fini(self.ctx)
+            self.ctx = None
+
+    def _check(self, rc):
+        if rc != 0:
+            raise MzGpuError(self.lib.mz_gpu_last_error(self.ctx).decode())
+
+    def _take(self, outp):
+        """Copy an out-batch to host numpy arrays and release it."""
+        import numpy as np
+        ob = outp.contents
+        n = ob.n
+        kw, vb = ob.schema.key_words, ob.schema.val_bytes
+        keys = np.empty(n * kw, np.uint64)
+        vals = np.empty(n * vb, np.uint8)
+        times = np.empty(n, np.uint64)
+        diffs = np.empty(n, np.int64)
+        if n:
+            self._check(self.lib.mz_gpu_out_to_host(
+                self.ctx, outp,
+                keys.ctypes.data_as(C.POINTER(C.c_uint64)),
+                vals.ctypes.data_as(C.POINTER(C.c_uint8)),
+                times.ctypes.data_as(C.POINTER(C.c_uint64)),
+                diffs.ctypes.data_as(C.POINTER(C.c_int64))))
+        self.lib.mz_gpu_out_release(self.ctx, outp)
+        return keys.view("int64"), vals, times, diffs
+
+    # --- mirrors of the OracleCtx interface (parity rig symmetry) ---
+    def arr_create(self, sch):
+        return self.lib.mz_gpu_arr_create(self.ctx, C.byref(sch))
+
+    def arr_push(self, arr, upd):
+        self._check(self.lib.mz_gpu_arr_push_batch(self.ctx, arr,
+                                                   C.byref(upd)))
+
+    def arr_set_logical_compaction(self, arr, frontier):
+        self.lib.mz_gpu_arr_set_logical_compaction(self.ctx, arr, frontier)
+
+    def arr_maintain(self, arr, fuel=0):
+        self._check(self.lib.mz_gpu_arr_maintain(self.ctx, arr, fuel))
+
+    def arr_stats(self, arr):
+        nb, nu, by = C.c_uint64(), C.c_uint64(), C.c_uint64()
+        self.lib.mz_gpu_arr_stats(self.ctx, arr, C.byref(nb), C.byref(nu),
+                                  C.byref(by))
+        return nb.value, nu.value, by.value
+
+    def join_create(self, a1, a2, cl):
+        return self.lib.mz_gpu_join_create(self.ctx, a1, a2, C.byref(cl))
+
+    def join_push(self, op, side, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_join_push(self.ctx, op, side,
+                                              C.byref(upd), C.byref(outp)))
+        return self._take(outp)
+
+    def halfjoin(self, lookup, upd, stream_vb, le, cl):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_halfjoin(
+            self.ctx, lookup, C.byref(upd), stream_vb, 1 if le else 0,
+            C.byref(cl), C.byref(outp)))
+        return self._take(outp)
+
+    def reduce_create(self, spec):
+        return self.lib.mz_gpu_reduce_create(self.ctx, C.byref(spec))
+
+    def reduce_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
+                                                C.byref(outp)))
+        return self._take(outp)
+
+    def consolidate(self, sch, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_consolidate(self.ctx, C.byref(sch),
+                                                C.byref(upd), C.byref(outp)))
+        return self._take(outp)
+
+    def route_hash(self, words):
+        arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
+                                          for w in words])
+        return self.lib.mz_gpu_route_hash(arr, len(words))
+
+    def set_kernel_timing(self, on):
+        self.lib.mz_gpu_set_kernel_timing(self.ctx, 1 if on else 0)
+
+    def probe_stats(self):
+        ms, rows, launches = C.c_double(), C.c_uint64(), C.c_uint64()
+        self.lib.mz_gpu_get_probe_stats(self.ctx, C.byref(ms), C.byref(rows),
+                                        C.byref(launches))
+        return ms.value, rows.value, launches.value

@@ -1,0 +1,1583 @@
+// mzgpu.hip — MI355X-native incremental join/reduce engine (PRODUCT).
+//
+// Implements the C ABI of include/mz_gpu.h with hand-written HIP kernels
+// for gfx950. All hot work is HBM-bound integer/byte processing (hash
+// probes, sort-based consolidation, segmented accumulation) — no MFMA.
+// Sort/scan primitives come from rocPRIM (AMD-native); the probe,
+// consolidate-emit, batch-build, hash and reduce kernels are hand-written.
+//
+// Replaces (reference file:line under /root/reference — algorithms
+// restated, not translated):
+//   mz_join_core            src/compute/src/render/join/mz_join_core.rs:57-496
+//   half_join stages        src/compute/src/render/join/delta_join.rs:338-583
+//   build_accumulable       src/compute/src/render/reduce.rs:1357-1581,1611-2270
+//   arrangement/spine       src/compute/src/extensions/arrange.rs:69-114,
+//                           src/row-spine/src/lib.rs:56-135
+//   Exchange routing        src/compute/src/render/join/linear_join.rs:390
+//
+// The cross-product ("simple") join strategy is applied at update
+// granularity; after consolidation its results are identical to the
+// reference's EditList/linear-scan pipeline (bilinearity — DESIGN.md §5),
+// which the oracle (which implements both strategies) verifies.
+
+#include "common.h"
+#include "../../include/mz_gpu.h"
+
+#include <algorithm>
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include <rocprim/rocprim.hpp>
+
+// ------------------------------------------------------------------ utils
+
+static constexpr int BLK = 256;
+static inline u32 ngrid(u64 n) {
+  u64 g = (n + BLK - 1) / BLK;
+  return (u32)std::min<u64>(g, 8 * 256);  // cap; kernels grid-stride
+}
+#define GRID_STRIDE(i, n)                                     \
+  for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < (n); \
+       i += (u64)gridDim.x * blockDim.x)
+
+__device__ __forceinline__ i64 wadd(i64 a, i64 b) {
+  return (i64)((u64)a + (u64)b);
+}
+__device__ __forceinline__ i64 wmul(i64 a, i64 b) {
+  return (i64)((u64)a * (u64)b);
+}
+
+// ------------------------------------------------- elementary kernels
+
+__global__ void k_iota(u32 *p, u64 n) { GRID_STRIDE(i, n) p[i] = (u32)i; }
+
+__global__ void k_gather_u64(const u64 *src, const u32 *perm, u64 *dst,
+                             u64 n) {
+  GRID_STRIDE(i, n) dst[i] = src[perm[i]];
+}
+__global__ void k_gather_i64(const i64 *src, const u32 *perm, i64 *dst,
+                             u64 n) {
+  GRID_STRIDE(i, n) dst[i] = src[perm[i]];
+}
+__global__ void k_gather_keyrows(const u64 *src, u32 kw, const u32 *perm,
+                                 u64 *dst, u64 n) {
+  GRID_STRIDE(i, n) {
+    for (u32 w = 0; w < kw; w++) dst[i * kw + w] = src[(u64)perm[i] * kw + w];
+  }
+}
+__global__ void k_gather_valrows(const u8 *src, u32 vb, const u32 *perm,
+                                 u8 *dst, u64 n) {
+  GRID_STRIDE(i, n) {
+    const u8 *s = src + (u64)perm[i] * vb;
+    u8 *d = dst + i * vb;
+    for (u32 b = 0; b < vb; b++) d[b] = s[b];
+  }
+}
+
+// sort key for one key word: flip sign bit -> unsigned order == i64 order
+__global__ void k_sortkey_key(const u64 *keys, u32 kw, u32 word,
+                              const u32 *perm, u64 *out, u64 n) {
+  GRID_STRIDE(i, n)
+  out[i] = keys[(u64)perm[i] * kw + word] ^ 0x8000000000000000ULL;
+}
+// sort key for one 8-byte val group: big-endian load == byte-lex order
+__global__ void k_sortkey_val(const u8 *vals, u32 vb, u32 word,
+                              const u32 *perm, u64 *out, u64 n) {
+  GRID_STRIDE(i, n) {
+    const u8 *v = vals + (u64)perm[i] * vb + word * 8;
+    u32 rem = vb - word * 8;
+    u64 x = 0;
+    for (u32 b = 0; b < 8 && b < rem; b++) x |= (u64)v[b] << (56 - 8 * b);
+    out[i] = x;
+  }
+}
+__global__ void k_sortkey_u64(const u64 *src, const u32 *perm, u64 *out,
+                              u64 n) {
+  GRID_STRIDE(i, n) out[i] = src[perm[i]];
+}
+
+// head flags over the permuted (key,val,time) order
+__global__ void k_head_flags(const u64 *keys, u32 kw, const u8 *vals, u32 vb,
+                             const u64 *times, const u32 *perm, u32 *flags,
+                             u64 n, int with_time) {
+  GRID_STRIDE(i, n) {
+    if (i == 0) {
+      flags[0] = 1;
+      continue;
+    }
+    u64 a = perm[i], b = perm[i - 1];
+    bool neq = false;
+    for (u32 w = 0; w < kw; w++) neq |= keys[a * kw + w] != keys[b * kw + w];
+    for (u32 c = 0; c < vb && !neq; c++) neq |= vals[a * vb + c] != vals[b * vb + c];
+    if (with_time && !neq) neq = times[a] != times[b];
+    flags[i] = neq ? 1u : 0u;
+  }
+}
+
+__global__ void k_group_starts(const u32 *flags, const u32 *gid, u32 *starts,
+                               u64 n) {
+  GRID_STRIDE(i, n) if (flags[i]) starts[gid[i] - 1] = (u32)i;
+}
+
+// per-group wrapping diff sums via prefix differences; emit nonzero flags
+__global__ void k_group_sums(const u32 *starts, u64 G, u64 n,
+                             const u64 *diff_prefix /* inclusive, permuted */,
+                             i64 *gsum, u32 *nz) {
+  GRID_STRIDE(g, G) {
+    u64 lo = starts[g];
+    u64 hi = (g + 1 < G) ? starts[g + 1] : n;
+    u64 s = diff_prefix[hi - 1] - (lo ? diff_prefix[lo - 1] : 0);
+    gsum[g] = (i64)s;
+    nz[g] = s != 0;
+  }
+}
+
+__global__ void k_emit_consolidated(const u64 *keys, u32 kw, const u8 *vals,
+                                    u32 vb, const u64 *times, const u32 *perm,
+                                    const u32 *starts, const i64 *gsum,
+                                    const u32 *nz, const u32 *nzpos, u64 G,
+                                    u64 *okeys, u8 *ovals, u64 *otimes,
+                                    i64 *odiffs) {
+  GRID_STRIDE(g, G) {
+    if (!nz[g]) continue;
+    u64 o = nzpos[g];  // exclusive scan of nz
+    u64 r = perm[starts[g]];
+    for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = keys[r * kw + w];
+    for (u32 c = 0; c < vb; c++) ovals[o * vb + c] = vals[r * vb + c];
+    otimes[o] = times[r];
+    odiffs[o] = gsum[g];
+  }
+}
+
+__global__ void k_advance_times(u64 *times, u64 n, u64 frontier) {
+  GRID_STRIDE(i, n) times[i] = times[i] < frontier ? frontier : times[i];
+}
+
+// ----------------------------------------------- batch structure build
+
+// flags over SEALED (already sorted) arrays
+__global__ void k_change_flags(const u64 *keys, u32 kw, const u8 *vals,
+                               u32 vb, u32 *kc, u32 *vc, u64 n) {
+  GRID_STRIDE(i, n) {
+    if (i == 0) {
+      kc[0] = 1;
+      vc[0] = 1;
+      continue;
+    }
+    bool kneq = false;
+    for (u32 w = 0; w < kw; w++)
+      kneq |= keys[i * kw + w] != keys[(i - 1) * kw + w];
+    bool vneq = kneq;
+    for (u32 c = 0; c < vb && !vneq; c++)
+      vneq |= vals[i * vb + c] != vals[(i - 1) * vb + c];
+    kc[i] = kneq ? 1u : 0u;
+    vc[i] = vneq ? 1u : 0u;
+  }
+}
+
+__global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
+                                    u32 vb, const u32 *kc, const u32 *vc,
+                                    const u32 *kid /*inclusive*/,
+                                    const u32 *vid /*inclusive*/, u64 n,
+                                    u64 n_keys, u64 n_vals, u64 *bkeys,
+                                    u32 *kv_off, u8 *bvals, u32 *vu_off,
+                                    u32 *val_key, u32 *upd_val) {
+  GRID_STRIDE(i, n) {
+    u32 k = kid[i] - 1, v = vid[i] - 1;
+    upd_val[i] = v;
+    if (kc[i]) {
+      for (u32 w = 0; w < kw; w++) bkeys[(u64)k * kw + w] = keys[i * kw + w];
+      kv_off[k] = v;
+      if (k == 0) kv_off[n_keys] = (u32)n_vals;
+    }
+    if (vc[i]) {
+      for (u32 c = 0; c < vb; c++) bvals[(u64)v * vb + c] = vals[i * vb + c];
+      vu_off[v] = (u32)i;
+      val_key[v] = k;
+      if (v == 0) vu_off[n_vals] = (u32)n;
+    }
+  }
+}
+
+__global__ void k_hash_init(u64 *hash, u64 slots, u32 kw) {
+  GRID_STRIDE(i, slots) hash[i * (kw + 1) + kw] = ~0ull;
+}
+
+__global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
+                             u64 n_keys) {
+  GRID_STRIDE(i, n_keys) {
+    u64 h = route_hash(keys + i * kw, kw) & (slots - 1);
+    for (;;) {
+      u64 *slot = hash + h * (kw + 1);
+      unsigned long long expected = ~0ull;
+      unsigned long long got = atomicCAS(
+          (unsigned long long *)(slot + kw), expected, (unsigned long long)i);
+      if (got == ~0ull) {
+        for (u32 w = 0; w < kw; w++) slot[w] = keys[i * kw + w];
+        break;
+      }
+      h = (h + 1) & (slots - 1);
+    }
+  }
+}
+
+__device__ __forceinline__ int hash_lookup(const u64 *hash, u64 slots,
+                                           const u64 *key, u32 kw) {
+  if (slots == 0) return -1;
+  u64 h = route_hash(key, kw) & (slots - 1);
+  for (;;) {
+    const u64 *slot = hash + h * (kw + 1);
+    u64 iw = slot[kw];
+    if (iw == ~0ull) return -1;
+    bool eq = true;
+    for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
+    if (eq) return (int)(u32)iw;
+    h = (h + 1) & (slots - 1);
+  }
+}
+
+// ------------------------------------------------------- closure (device)
+
+__device__ __forceinline__ i64 d_read_int(const u8 *p, u8 width) {
+  if (width == 4) {
+    int32_t v;
+    memcpy(&v, p, 4);
+    return v;
+  }
+  i64 v;
+  memcpy(&v, p, 8);
+  return v;
+}
+
+__device__ __forceinline__ const u8 *d_cl_src(const u64 *key, const u8 *v1,
+                                              const u8 *v2, u8 src) {
+  switch (src) {
+    case MZ_SRC_KEY: return (const u8 *)key;
+    case MZ_SRC_VAL_STREAM: return v1;
+    default: return v2;
+  }
+}
+
+// mirrors oracle closure_apply; v1 = input-1/stream val, v2 = input-2/lookup
+__device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
+                                const u8 *v1, const u8 *v2, u64 *okey,
+                                u8 *oval) {
+  for (u32 i = 0; i < cl->n_filters; i++) {
+    const mz_gpu_filter f = cl->filters[i];
+    i64 x = d_read_int(d_cl_src(key, v1, v2, f.src) + f.off, f.width);
+    bool ok;
+    switch (f.cmp) {
+      case MZ_CMP_LT: ok = x < f.imm; break;
+      case MZ_CMP_LE: ok = x <= f.imm; break;
+      case MZ_CMP_GT: ok = x > f.imm; break;
+      case MZ_CMP_GE: ok = x >= f.imm; break;
+      case MZ_CMP_EQ: ok = x == f.imm; break;
+      default: ok = x != f.imm; break;
+    }
+    if (!ok) return false;
+  }
+  if (!okey) return true;
+  for (int which = 0; which < 2; which++) {
+    u32 nf = which ? cl->n_val_fields : cl->n_key_fields;
+    const mz_gpu_field *fs = which ? cl->val_fields : cl->key_fields;
+    u8 *dst = which ? oval : (u8 *)okey;
+    for (u32 i = 0; i < nf; i++) {
+      const mz_gpu_field f = fs[i];
+      if (f.src == MZ_SRC_COMPUTE) {
+        i64 v = 0;
+        if (f.off == MZ_COMPUTE_REVENUE) {
+          i64 ep = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 disc = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          v = ep * (10000 - disc);
+        }
+        memcpy(dst, &v, 8);
+        dst += 8;
+      } else {
+        const u8 *s = d_cl_src(key, v1, v2, f.src) + f.off;
+        for (u32 b = 0; b < f.width; b++) dst[b] = s[b];
+        dst += f.width;
+      }
+    }
+  }
+  return true;
+}
+
+// --------------------------------------------------------------- probe
+
+// A probe target list (snapshot of an arrangement's batches).
+struct BatchList {
+  int n;
+  DevBatch b[12];
+};
+
+enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
+
+// Phase 1: count emitted pairs per delta update.
+// delta: n updates (keys/vals/times/diffs columns); swap: delta is input 2
+// of a linear join (closure arg order is (key, v1=input1, v2=input2)).
+__global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                              const u64 *dtimes, u64 n, u32 kw, u32 lvb,
+                              BatchList bl, int mode, int swap,
+                              const mz_gpu_closure cl, u32 *count) {
+  GRID_STRIDE(i, n) {
+    const u64 *key = dkeys + i * kw;
+    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+    u64 t = dtimes[i];
+    u32 c = 0;
+    for (int bi = 0; bi < bl.n; bi++) {
+      const DevBatch &b = bl.b[bi];
+      int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
+      if (ki < 0) continue;
+      for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
+        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+        const u8 *v1 = swap ? lv : dv;
+        const u8 *v2 = swap ? dv : lv;
+        if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
+        u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+        if (mode == PM_JOIN) {
+          c += hi - lo;
+        } else {
+          for (u32 u = lo; u < hi; u++) {
+            u64 t2 = b.times[u];
+            c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+          }
+        }
+      }
+    }
+    count[i] = c;
+  }
+}
+
+// Phase 2: emit pairs at offsets (deterministic: offset is a function of
+// the input index).
+__global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                             const u64 *dtimes, const i64 *ddiffs, u64 n,
+                             u32 kw, u32 lvb, BatchList bl, int mode,
+                             int swap, const mz_gpu_closure cl,
+                             const u32 *offs /*exclusive*/, u64 *okeys,
+                             u8 *ovals, u64 *otimes, i64 *odiffs) {
+  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
+  GRID_STRIDE(i, n) {
+    const u64 *key = dkeys + i * kw;
+    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+    u64 t = dtimes[i];
+    i64 d1 = ddiffs[i];
+    u64 o = offs[i];
+    for (int bi = 0; bi < bl.n; bi++) {
+      const DevBatch &b = bl.b[bi];
+      int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
+      if (ki < 0) continue;
+      for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
+        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+        const u8 *v1 = swap ? lv : dv;
+        const u8 *v2 = swap ? dv : lv;
+        u64 okey[MAX_KW];
+        u8 oval[MAX_VB];
+        if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
+        for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
+          u64 t2 = b.times[u];
+          u64 tout;
+          if (mode == PM_JOIN) {
+            tout = t2 > t ? t2 : t;
+          } else {
+            if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
+            tout = t;
+          }
+          for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
+          for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oval[c];
+          otimes[o] = tout;
+          odiffs[o] = wmul(d1, b.diffs[u]);
+          o++;
+        }
+      }
+    }
+  }
+}
+
+// expansion: flatten a DevBatch back to per-update (key,val,time,diff)
+__global__ void k_expand_batch(DevBatch b, u32 kw, u32 vb, u64 frontier,
+                               u64 *okeys, u8 *ovals, u64 *otimes,
+                               i64 *odiffs, u64 base) {
+  GRID_STRIDE(i, b.n_upds) {
+    u32 v = b.upd_val[i];
+    u32 k = b.val_key[v];
+    u64 o = base + i;
+    for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = b.keys[(u64)k * kw + w];
+    for (u32 c = 0; c < vb; c++) ovals[o * vb + c] = b.vals[(u64)v * vb + c];
+    u64 t = b.times[i];
+    otimes[o] = t < frontier ? frontier : t;
+    odiffs[o] = b.diffs[i];
+  }
+}
+
+// ------------------------------------------------------------ partition
+
+__global__ void k_shard_of(const u64 *keys, u32 kw, u64 n, u32 nshards,
+                           u32 *shard) {
+  GRID_STRIDE(i, n) shard[i] = (u32)(route_hash(keys + i * kw, kw) % nshards);
+}
+
+// ------------------------------------------------------------ reduce
+
+// Accumulator per aggregate (Accum restatement, reduce.rs:1611-2270):
+// 48 bytes: { u128 accum; u64 non_nulls, pos_infs, neg_infs, nans }.
+struct Acc5 {
+  u128 accum;
+  u64 nn, pi, ni, nan;
+};
+
+__device__ __forceinline__ i128 d_float_to_fixed_point(double dv) {
+  // reduce.rs:1663-1697 (wrapping trunc(n * 2^24) mod 2^128)
+  u64 bits = __double_as_longlong(dv);
+  u64 mantissa = bits & ((1ULL << 52) - 1);
+  int exp_bits = (int)((bits >> 52) & 0x7ff);
+  int exponent;
+  if (exp_bits == 0) {
+    exponent = -1074;
+  } else {
+    mantissa |= 1ULL << 52;
+    exponent = exp_bits - 1075;
+  }
+  long e = (long)exponent + 24;
+  u128 significand = (u128)mantissa;
+  u128 magnitude;
+  if (e >= 0)
+    magnitude = e < 128 ? (significand << e) : (u128)0;
+  else
+    magnitude = (-e) < 128 ? (significand >> (-e)) : (u128)0;
+  i128 m = (i128)magnitude;
+  return (bits >> 63) ? (i128)(~(u128)m + 1) : m;
+}
+
+// exact round-to-nearest-even i128 -> double (matches host sitofp / Rust
+// `as f64`, used by finalize SUM_F64, reduce.rs:1952)
+__host__ __device__ inline double i128_to_double(i128 v) {
+  if (v == 0) return 0.0;
+  bool neg = v < 0;
+  u128 m = neg ? (u128)0 - (u128)v : (u128)v;
+  u64 hi = (u64)(m >> 64), lo = (u64)m;
+  int bits;
+#ifdef __HIP_DEVICE_COMPILE__
+  bits = hi ? 128 - __clzll(hi) : 64 - __clzll(lo);
+#else
+  bits = hi ? 128 - __builtin_clzll(hi) : 64 - __builtin_clzll(lo);
+#endif
+  double d;
+  if (bits <= 53) {
+    d = (double)lo;
+  } else {
+    int shift = bits - 54;
+    u64 top = (u64)(m >> shift);  // 54 bits
+    bool sticky = (m & (((u128)1 << shift) - 1)) != 0;
+    u64 mant = top >> 1;
+    bool rnd = top & 1;
+    if (rnd && (sticky || (mant & 1))) mant++;
+    d = ldexp((double)mant, shift + 1);
+  }
+  return neg ? -d : d;
+}
+
+// datum -> Acc5 (datum_to_accumulator, reduce.rs:1699-1838)
+__device__ __forceinline__ Acc5 d_datum_to_acc(const mz_gpu_aggregate a,
+                                               const u8 *val) {
+  Acc5 r{0, 0, 0, 0, 0};
+  bool null = a.nullable && val[a.off + a.width] != 0;
+  if (a.func == MZ_AGG_COUNT) {
+    r.nn = null ? 0 : 1;
+  } else if (a.func == MZ_AGG_SUM_I64) {
+    if (!null) {
+      r.accum = (u128)(i128)d_read_int(val + a.off, a.width);
+      r.nn = 1;
+    }
+  } else {  // SUM_F64
+    if (!null) {
+      double n;
+      if (a.width == 4) {
+        float f;
+        memcpy(&f, val + a.off, 4);
+        n = (double)f;
+      } else {
+        memcpy(&n, val + a.off, 8);
+      }
+      r.nan = isnan(n) ? 1 : 0;
+      r.pi = (n == HUGE_VAL) ? 1 : 0;
+      r.ni = (n == -HUGE_VAL) ? 1 : 0;
+      r.nn = 1;
+      if (!r.nan && !r.pi && !r.ni) r.accum = (u128)d_float_to_fixed_point(n);
+    }
+  }
+  return r;
+}
+
+// finalize one aggregate into its 24-byte output slot
+// (finalize_accum, reduce.rs:1840-1997; slot layout DESIGN.md §2.1)
+__device__ __forceinline__ void d_finalize(const mz_gpu_aggregate a,
+                                           const Acc5 acc, i64 total,
+                                           u8 *slot) {
+  for (int i = 0; i < 24; i++) slot[i] = 0;
+  bool zero = acc.accum == 0 && acc.nn == 0 && acc.pi == 0 && acc.ni == 0 &&
+              acc.nan == 0;
+  if (total > 0 && zero && a.func != MZ_AGG_COUNT) {
+    slot[0] = 1;
+    return;
+  }
+  if (a.func == MZ_AGG_COUNT) {
+    i64 c = (i64)acc.nn;
+    memcpy(slot + 8, &c, 8);
+  } else if (a.func == MZ_AGG_SUM_I64) {
+    u128 v = acc.accum;
+    memcpy(slot + 8, &v, 16);
+  } else {
+    double v;
+    if (acc.nan > 0 || (acc.pi > 0 && acc.ni > 0))
+      v = __longlong_as_double(0x7FF8000000000000LL);  // NaN
+    else if (acc.pi > 0)
+      v = HUGE_VAL;
+    else if (acc.ni > 0)
+      v = -HUGE_VAL;
+    else
+      v = i128_to_double((i128)acc.accum) / 16777216.0;
+    memcpy(slot + 8, &v, 8);
+  }
+}
+
+// Reduce state: open-addressing hash over persistent AccumRows.
+//   slot words: [key kw][idx]
+//   state row (u64 words): [key kw][total][na * 6 words of Acc5]
+//
+// Inserts are coherence-safe by construction: within one push, distinct
+// keys are grouped so no two threads ever insert the same key; lookups and
+// inserts run in SEPARATE kernel launches (kernel-boundary coherence —
+// per-XCD L2s are not coherent within a launch, MI355X_MICROARCH §XCD),
+// and the insert kernel's losers only CAS the idx word, never read another
+// insert's key words. State row index = base + miss order (deterministic).
+struct RedState {
+  u64 *hash;
+  u64 slots;
+  u64 *rows;     // stride words
+  u64 capacity;
+  u32 stride_w;  // kw + 1 + 6*na
+};
+
+// Phase A: lookup each key group in the (fully published) table.
+__global__ void k_red_lookup(const u64 *keys, u32 kw, const u32 *gstart,
+                             u64 G, RedState st, u32 *found, u32 *miss) {
+  GRID_STRIDE(g, G) {
+    const u64 *key = keys + (u64)gstart[g] * kw;
+    int idx = hash_lookup(st.hash, st.slots, key, kw);
+    found[g] = idx < 0 ? ~0u : (u32)idx;
+    miss[g] = idx < 0 ? 1u : 0u;
+  }
+}
+
+// Phase B: insert missing keys; row idx = base + rank among misses.
+__global__ void k_red_insert(const u64 *keys, u32 kw, const u32 *gstart,
+                             u64 G, const u32 *miss, const u32 *misspos,
+                             u64 base, RedState st) {
+  GRID_STRIDE(g, G) {
+    if (!miss[g]) continue;
+    const u64 *key = keys + (u64)gstart[g] * kw;
+    u64 idx = base + misspos[g];
+    u64 *row = st.rows + idx * st.stride_w;
+    for (u32 w = 0; w < kw; w++) row[w] = key[w];
+    for (u32 w = kw; w < st.stride_w; w++) row[w] = 0;
+    u64 h = route_hash(key, kw) & (st.slots - 1);
+    for (;;) {
+      u64 *slot = st.hash + h * (kw + 1);
+      unsigned long long prev = atomicCAS((unsigned long long *)(slot + kw),
+                                          ~0ull, (unsigned long long)idx);
+      if (prev == ~0ull) {
+        for (u32 w = 0; w < kw; w++) slot[w] = key[w];
+        break;
+      }
+      h = (h + 1) & (st.slots - 1);
+    }
+  }
+}
+
+// Phase C: one thread per distinct key in a time slice [lo,hi) of updates
+// sorted by (time, key). gstart: start row of each key group; emits
+// corrections (new minus old finalized rows).
+__global__ void k_reduce_apply(const u64 *keys, const u8 *vals, u32 kw,
+                               u32 vb, const i64 *diffs, const u32 *gstart,
+                               u64 G, u64 hi_row, u64 t, RedState st,
+                               const u32 *found, const u32 *miss,
+                               const u32 *misspos, u64 base,
+                               mz_gpu_reduce_spec spec, u64 *okeys, u8 *ovals,
+                               u64 *otimes, i64 *odiffs,
+                               unsigned long long *ocount) {
+  u32 na = spec.n_aggs;
+  u32 ovb = spec.out.val_bytes;
+  GRID_STRIDE(g, G) {
+    u64 lo = gstart[g];
+    u64 end = (g + 1 < G) ? gstart[g + 1] : hi_row;
+    const u64 *key = keys + lo * kw;
+    u64 idx = miss[g] ? base + misspos[g] : found[g];
+    u64 *row = st.rows + idx * st.stride_w;
+    i64 *total_p = (i64 *)(row + kw);
+    Acc5 *accs = (Acc5 *)(row + kw + 1);
+    // snapshot old
+    i64 old_total = *total_p;
+    Acc5 old_a[MZ_GPU_MAX_AGGS];
+    for (u32 a = 0; a < na; a++) old_a[a] = accs[a];
+    // apply updates (explode_one * diff + semigroup merge; wrapping)
+    for (u64 r = lo; r < end; r++) {
+      i64 d = diffs[r];
+      const u8 *v = vals + r * vb;
+      for (u32 a = 0; a < na; a++) {
+        Acc5 c = d_datum_to_acc(spec.aggs[a], v);
+        accs[a].accum += c.accum * (u128)(i128)d;
+        accs[a].nn += (u64)c.nn * (u64)d;
+        accs[a].pi += (u64)c.pi * (u64)d;
+        accs[a].ni += (u64)c.ni * (u64)d;
+        accs[a].nan += (u64)c.nan * (u64)d;
+      }
+      *total_p = wadd(*total_p, d);
+    }
+    i64 new_total = *total_p;
+    // exists = any nonzero component (reduce_abelian: keys with nonempty
+    // input accumulation produce one output row)
+    auto exists = [&](const Acc5 *as, i64 tot) {
+      if (tot != 0) return true;
+      for (u32 a = 0; a < na; a++)
+        if (as[a].accum != 0 || as[a].nn || as[a].pi || as[a].ni || as[a].nan)
+          return true;
+      return false;
+    };
+    bool oe = exists(old_a, old_total), ne = exists(accs, new_total);
+    u8 oldrow[MZ_GPU_MAX_AGGS * 24], newrow[MZ_GPU_MAX_AGGS * 24];
+    if (oe)
+      for (u32 a = 0; a < na; a++)
+        d_finalize(spec.aggs[a], old_a[a], old_total, oldrow + 24 * a);
+    if (ne)
+      for (u32 a = 0; a < na; a++)
+        d_finalize(spec.aggs[a], accs[a], new_total, newrow + 24 * a);
+    if (oe && ne) {
+      bool same = true;
+      for (u32 c = 0; c < ovb; c++) same &= oldrow[c] == newrow[c];
+      if (same) continue;
+    }
+    if (oe) {
+      u64 o = atomicAdd(ocount, 1ull);
+      for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = key[w];
+      for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oldrow[c];
+      otimes[o] = t;
+      odiffs[o] = -1;
+    }
+    if (ne) {
+      u64 o = atomicAdd(ocount, 1ull);
+      for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = key[w];
+      for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = newrow[c];
+      otimes[o] = t;
+      odiffs[o] = 1;
+    }
+  }
+}
+
+// time-change flags over sorted times
+__global__ void k_time_flags(const u64 *times, const u32 *perm, u32 *flags,
+                             u64 n) {
+  GRID_STRIDE(i, n)
+  flags[i] = (i == 0 || times[perm[i]] != times[perm[i - 1]]) ? 1u : 0u;
+}
+__global__ void k_key_flags_sorted(const u64 *keys, u32 kw, const u64 *times,
+                                   u32 *flags, u64 n) {
+  GRID_STRIDE(i, n) {
+    if (i == 0) {
+      flags[0] = 1;
+      continue;
+    }
+    bool neq = times[i] != times[i - 1];
+    for (u32 w = 0; w < kw && !neq; w++)
+      neq |= keys[i * kw + w] != keys[(i - 1) * kw + w];
+    flags[i] = neq ? 1u : 0u;
+  }
+}
+
+// ================================================================== host
+
+namespace {
+
+struct Ctx;
+
+struct Alloc {
+  void *p = nullptr;
+  size_t bytes = 0;
+};
+
+struct Scratch {
+  // bump arena of device memory, grown on demand, reset per call
+  void *base = nullptr;
+  size_t cap = 0, used = 0;
+  void reset() { used = 0; }
+  void *get(size_t bytes) {
+    bytes = (bytes + 255) & ~size_t(255);
+    if (used + bytes > cap) {
+      size_t ncap = std::max<size_t>(2 * cap, used + bytes + (64u << 20));
+      void *nb;
+      HIP_CHECK(hipMalloc(&nb, ncap));
+      // old allocations in this call remain valid until free; we leak-free
+      // the previous arena only when no allocations are outstanding (reset
+      // happens at call start, so this is safe at call boundaries). To stay
+      // safe mid-call we keep old arenas alive until ctx teardown.
+      if (base) retired.push_back(base);
+      base = nb;
+      cap = ncap;
+      // re-bump: prior in-call allocations still point into the retired
+      // arena; only new requests come from the new one.
+      used = 0;
+      if (used + bytes > cap) abort();
+    }
+    void *p = (char *)base + used;
+    used += bytes;
+    return p;
+  }
+  std::vector<void *> retired;
+};
+
+}  // namespace
+
+struct mz_gpu_arr {
+  DevSchema schema;
+  std::vector<DevBatch> batches;
+  u64 logical_compaction = 0;
+  u64 upper = 0;
+  Ctx *ctx = nullptr;
+};
+
+struct mz_gpu_join {
+  mz_gpu_arr *arr1, *arr2;
+  mz_gpu_closure cl;
+};
+
+struct mz_gpu_red {
+  mz_gpu_reduce_spec spec;
+  RedState st;
+  u64 capacity;
+  u64 n_rows = 0;  // host-tracked row count (insert order is deterministic)
+};
+
+namespace {
+
+struct Ctx {
+  hipStream_t stream = nullptr;
+  std::string err;
+  Scratch scratch;
+  std::vector<mz_gpu_arr *> arrs;
+  std::vector<mz_gpu_join *> joins;
+  std::vector<mz_gpu_red *> reds;
+  // probe-kernel timing (for bench roofline): accumulated ns and bytes
+  double probe_ms = 0;
+  u64 probe_rows = 0, probe_launches = 0;
+  hipEvent_t ev_a = nullptr, ev_b = nullptr;
+  int time_kernels = 0;
+};
+
+void *dmalloc(size_t bytes) {
+  void *p = nullptr;
+  if (bytes == 0) bytes = 16;
+  HIP_CHECK(hipMalloc(&p, bytes));
+  return p;
+}
+
+template <typename T>
+T *dnew(u64 n) {
+  return (T *)dmalloc(n * sizeof(T));
+}
+
+// composite stable sort: returns perm ordering rows by (key, val, time) —
+// or (time, key) when for_reduce (primary time).
+// Sort passes use rocprim radix_sort_pairs (AMD-native primitive).
+void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
+                  const u64 *times, u64 n, u32 *perm, bool time_major) {
+  auto &S = c->scratch;
+  u64 *skey = (u64 *)S.get(n * 8);
+  u64 *skey_out = (u64 *)S.get(n * 8);
+  u32 *perm_out = (u32 *)S.get(n * 4);
+  void *tmp = nullptr;
+  size_t tmp_bytes = 0;
+  hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, c->stream, perm,
+                     n);
+  // determine passes (LSD): least-significant first
+  struct Pass {
+    int kind;  // 0 = time, 1 = val word, 2 = key word
+    u32 word;
+  };
+  std::vector<Pass> passes;
+  if (time_major) {
+    // (time, key): key words LSD..MSD, then time last
+    for (int w = (int)kw - 1; w >= 0; w--) passes.push_back({2, (u32)w});
+    passes.push_back({0, 0});
+  } else {
+    // (key, val, time): time first, then val words MSD->... LSD order:
+    // least significant = time, then val last word .. first, then key
+    passes.push_back({0, 0});
+    u32 vwords = (vb + 7) / 8;
+    for (int w = (int)vwords - 1; w >= 0; w--) passes.push_back({1, (u32)w});
+    for (int w = (int)kw - 1; w >= 0; w--) passes.push_back({2, (u32)w});
+  }
+  for (auto &p : passes) {
+    if (p.kind == 0)
+      hipLaunchKernelGGL(k_sortkey_u64, dim3(ngrid(n)), dim3(BLK), 0,
+                         c->stream, times, perm, skey, n);
+    else if (p.kind == 1)
+      hipLaunchKernelGGL(k_sortkey_val, dim3(ngrid(n)), dim3(BLK), 0,
+                         c->stream, vals, vb, p.word, perm, skey, n);
+    else
+      hipLaunchKernelGGL(k_sortkey_key, dim3(ngrid(n)), dim3(BLK), 0,
+                         c->stream, keys, kw, p.word, perm, skey, n);
+    size_t need = 0;
+    (void)rocprim::radix_sort_pairs(nullptr, need, skey, skey_out, perm, perm_out,
+                              (unsigned)n, 0, 64, c->stream);
+    if (need > tmp_bytes) {
+      tmp = S.get(need);
+      tmp_bytes = need;
+    }
+    (void)rocprim::radix_sort_pairs(tmp, tmp_bytes, skey, skey_out, perm, perm_out,
+                              (unsigned)n, 0, 64, c->stream);
+    std::swap(perm, perm_out);
+  }
+  // ensure result is in `perm` (the caller's buffer): passes count parity
+  if (passes.size() % 2 == 1) {
+    // result currently in the buffer that started as perm_out
+    HIP_CHECK(hipMemcpyAsync(perm_out, perm, n * 4, hipMemcpyDeviceToDevice,
+                             c->stream));
+    std::swap(perm, perm_out);
+  }
+}
+
+u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
+  // returns total; out = exclusive prefix (out has n+1 slots).
+  // The n+1-sized scan would read in[n]; pad a copy to stay defined.
+  auto &S = c->scratch;
+  u32 *pad = (u32 *)S.get((n + 1) * 4);
+  HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
+                           c->stream));
+  HIP_CHECK(hipMemsetAsync(pad + n, 0, 4, c->stream));
+  size_t need = 0;
+  (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
+                          rocprim::plus<u32>(), c->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::exclusive_scan(tmp, need, pad, out, 0u, n + 1,
+                          rocprim::plus<u32>(), c->stream);
+  u32 total;
+  HIP_CHECK(hipMemcpyAsync(&total, out + n, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return total;
+}
+
+void inclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
+  auto &S = c->scratch;
+  size_t need = 0;
+  (void)rocprim::inclusive_scan(nullptr, need, in, out, n, rocprim::plus<u32>(),
+                          c->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::inclusive_scan(tmp, need, in, out, n, rocprim::plus<u32>(),
+                          c->stream);
+}
+
+void inclusive_scan_u64(Ctx *c, const u64 *in, u64 *out, u64 n) {
+  auto &S = c->scratch;
+  size_t need = 0;
+  (void)rocprim::inclusive_scan(nullptr, need, in, out, n, rocprim::plus<u64>(),
+                          c->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::inclusive_scan(tmp, need, in, out, n, rocprim::plus<u64>(),
+                          c->stream);
+}
+
+// Stage updates onto the device (if host) and return device pointers.
+struct DevUpdates {
+  const u64 *keys;
+  const u8 *vals;
+  const u64 *times;
+  const i64 *diffs;
+  u64 n;
+};
+
+DevUpdates stage_updates(Ctx *c, const mz_gpu_updates *u, u32 kw, u32 vb) {
+  DevUpdates d;
+  d.n = u->n;
+  if (u->on_device) {
+    d.keys = u->keys;
+    d.vals = u->vals;
+    d.times = u->times;
+    d.diffs = u->diffs;
+    return d;
+  }
+  auto &S = c->scratch;
+  u64 *k = (u64 *)S.get(u->n * kw * 8);
+  u8 *v = vb ? (u8 *)S.get(u->n * vb) : nullptr;
+  u64 *t = (u64 *)S.get(u->n * 8);
+  i64 *df = (i64 *)S.get(u->n * 8);
+  HIP_CHECK(hipMemcpyAsync(k, u->keys, u->n * kw * 8, hipMemcpyHostToDevice,
+                           c->stream));
+  if (vb)
+    HIP_CHECK(hipMemcpyAsync(v, u->vals, u->n * vb, hipMemcpyHostToDevice,
+                             c->stream));
+  HIP_CHECK(hipMemcpyAsync(t, u->times, u->n * 8, hipMemcpyHostToDevice,
+                           c->stream));
+  HIP_CHECK(hipMemcpyAsync(df, u->diffs, u->n * 8, hipMemcpyHostToDevice,
+                           c->stream));
+  d.keys = k;
+  d.vals = v;
+  d.times = t;
+  d.diffs = df;
+  return d;
+}
+
+struct OutOwned {
+  mz_gpu_out pub_;
+  // owned device arrays
+  u64 *keys;
+  u8 *vals;
+  u64 *times;
+  i64 *diffs;
+};
+
+mz_gpu_out *make_out(u64 *k, u8 *v, u64 *t, i64 *d, u64 n, u32 kw, u32 vb) {
+  OutOwned *o = new OutOwned();
+  o->keys = k;
+  o->vals = v;
+  o->times = t;
+  o->diffs = d;
+  o->pub_.keys = k;
+  o->pub_.vals = v;
+  o->pub_.times = t;
+  o->pub_.diffs = d;
+  o->pub_.n = n;
+  o->pub_.on_device = 1;
+  o->pub_.schema.key_words = kw;
+  o->pub_.schema.val_bytes = vb;
+  return &o->pub_;
+}
+
+// Core consolidation: sort + group + sum + compact. Returns owned device
+// arrays (exact-size). Input must be device-resident.
+void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
+                     u8 **ovals, u64 **otimes, i64 **odiffs, u64 *out_n) {
+  auto &S = c->scratch;
+  u64 n = in.n;
+  if (n == 0) {
+    *okeys = dnew<u64>(1);
+    *ovals = (u8 *)dmalloc(1);
+    *otimes = dnew<u64>(1);
+    *odiffs = dnew<i64>(1);
+    *out_n = 0;
+    return;
+  }
+  u32 *perm = (u32 *)S.get(n * 4);
+  sort_updates(c, in.keys, kw, in.vals, vb, in.times, n, perm, false);
+  u32 *flags = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_head_flags, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     in.keys, kw, in.vals, vb, in.times, perm, flags, n, 1);
+  u32 *gid = (u32 *)S.get(n * 4);
+  inclusive_scan_u32(c, flags, gid, n);
+  u32 G;
+  HIP_CHECK(hipMemcpyAsync(&G, gid + n - 1, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  u32 *starts = (u32 *)S.get((u64)G * 4);
+  hipLaunchKernelGGL(k_group_starts, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     flags, gid, starts, n);
+  // wrapping inclusive prefix of permuted diffs
+  u64 *pdiff = (u64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     (const u64 *)in.diffs, perm, pdiff, n);
+  u64 *pref = (u64 *)S.get(n * 8);
+  inclusive_scan_u64(c, pdiff, pref, n);
+  i64 *gsum = (i64 *)S.get((u64)G * 8);
+  u32 *nz = (u32 *)S.get((u64)G * 4 + 4);
+  hipLaunchKernelGGL(k_group_sums, dim3(ngrid(G)), dim3(BLK), 0, c->stream,
+                     starts, (u64)G, n, pref, gsum, nz);
+  u32 *nzpos = (u32 *)S.get(((u64)G + 1) * 4);
+  u64 M = exclusive_scan_u32(c, nz, nzpos, G);
+  *okeys = dnew<u64>(std::max<u64>(M, 1) * kw);
+  *ovals = (u8 *)dmalloc(std::max<u64>(M, 1) * vb);
+  *otimes = dnew<u64>(std::max<u64>(M, 1));
+  *odiffs = dnew<i64>(std::max<u64>(M, 1));
+  if (M)
+    hipLaunchKernelGGL(k_emit_consolidated, dim3(ngrid(G)), dim3(BLK), 0,
+                       c->stream, in.keys, kw, in.vals, vb, in.times, perm,
+                       starts, gsum, nz, nzpos, (u64)G, *okeys, *ovals,
+                       *otimes, *odiffs);
+  *out_n = M;
+}
+
+void free_batch(DevBatch &b) {
+  for (void *p : {(void *)b.keys, (void *)b.kv_off, (void *)b.vals,
+                  (void *)b.vu_off, (void *)b.val_key, (void *)b.times,
+                  (void *)b.diffs, (void *)b.upd_val, (void *)b.hash})
+    if (p) HIP_CHECK(hipFree(p));
+  b = DevBatch();
+}
+
+// Build a sealed DevBatch from sealed (sorted+consolidated) device arrays.
+// Takes ownership of the arrays (they become the batch's times/diffs after
+// compaction into structure arrays; keys/vals are re-packed).
+DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
+                     i64 *diffs, u64 n, u64 lower, u64 upper) {
+  auto &S = c->scratch;
+  DevBatch b;
+  b.lower = lower;
+  b.upper = upper;
+  b.n_upds = n;
+  if (n == 0) {
+    b.keys = keys;
+    b.vals = vals;
+    b.times = times;
+    b.diffs = diffs;
+    b.kv_off = dnew<u32>(1);
+    b.vu_off = dnew<u32>(1);
+    b.val_key = dnew<u32>(1);
+    b.upd_val = dnew<u32>(1);
+    HIP_CHECK(hipMemsetAsync(b.kv_off, 0, 4, c->stream));
+    HIP_CHECK(hipMemsetAsync(b.vu_off, 0, 4, c->stream));
+    return b;
+  }
+  u32 *kc = (u32 *)S.get(n * 4);
+  u32 *vc = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_change_flags, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     keys, kw, vals, vb, kc, vc, n);
+  u32 *kid = (u32 *)S.get(n * 4);
+  u32 *vid = (u32 *)S.get(n * 4);
+  inclusive_scan_u32(c, kc, kid, n);
+  inclusive_scan_u32(c, vc, vid, n);
+  u32 nk, nv;
+  HIP_CHECK(hipMemcpyAsync(&nk, kid + n - 1, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipMemcpyAsync(&nv, vid + n - 1, 4, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  b.n_keys = nk;
+  b.n_vals = nv;
+  b.keys = dnew<u64>((u64)nk * kw);
+  b.kv_off = dnew<u32>((u64)nk + 1);
+  b.vals = (u8 *)dmalloc((u64)nv * vb);
+  b.vu_off = dnew<u32>((u64)nv + 1);
+  b.val_key = dnew<u32>(nv);
+  b.upd_val = dnew<u32>(n);
+  hipLaunchKernelGGL(k_scatter_structure, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, keys, kw, vals, vb, kc, vc, kid, vid, n,
+                     (u64)nk, (u64)nv, b.keys, b.kv_off, b.vals, b.vu_off,
+                     b.val_key, b.upd_val);
+  b.times = times;
+  b.diffs = diffs;
+  // hash index: pow2 >= 2*nk
+  u64 slots = 16;
+  while (slots < 2 * (u64)nk) slots <<= 1;
+  b.hash_slots = slots;
+  b.hash = dnew<u64>(slots * (kw + 1));
+  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0, c->stream,
+                     b.hash, slots, kw);
+  hipLaunchKernelGGL(k_hash_build, dim3(ngrid(nk)), dim3(BLK), 0, c->stream,
+                     b.hash, slots, b.keys, kw, (u64)nk);
+  // keys/vals flat arrays were re-packed into structure; free originals
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  HIP_CHECK(hipFree(keys));
+  HIP_CHECK(hipFree(vals));
+  return b;
+}
+
+// Merge an arrangement's batches [from, to) into one (logical compaction
+// applied). Policy is the host's; semantics = concat + advance + consolidate.
+void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
+  if (to - from <= 1) return;
+  auto &S = c->scratch;
+  S.reset();
+  u32 kw = a->schema.kw, vb = a->schema.vb;
+  u64 total = 0, lo = UINT64_MAX, hi = 0;
+  for (size_t i = from; i < to; i++) {
+    total += a->batches[i].n_upds;
+    lo = std::min(lo, a->batches[i].lower);
+    hi = std::max(hi, a->batches[i].upper);
+  }
+  u64 *keys = (u64 *)S.get(total * kw * 8);
+  u8 *vals = (u8 *)S.get(std::max<u64>(total * vb, 1));
+  u64 *times = (u64 *)S.get(total * 8);
+  i64 *diffs = (i64 *)S.get(total * 8);
+  u64 base = 0;
+  for (size_t i = from; i < to; i++) {
+    DevBatch &b = a->batches[i];
+    if (b.n_upds)
+      hipLaunchKernelGGL(k_expand_batch, dim3(ngrid(b.n_upds)), dim3(BLK), 0,
+                         c->stream, b, kw, vb, a->logical_compaction, keys,
+                         vals, times, diffs, base);
+    base += b.n_upds;
+  }
+  DevUpdates in{keys, vals, times, diffs, total};
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 M;
+  consolidate_dev(c, kw, vb, in, &ok, &ov, &ot, &od, &M);
+  DevBatch merged = build_batch(c, kw, vb, ok, ov, ot, od, M,
+                                lo == UINT64_MAX ? 0 : lo, hi);
+  for (size_t i = from; i < to; i++) free_batch(a->batches[i]);
+  a->batches.erase(a->batches.begin() + from, a->batches.begin() + to);
+  a->batches.insert(a->batches.begin() + from, merged);
+}
+
+}  // namespace
+
+// ================================================================ C ABI
+
+struct mz_gpu_ctx {
+  Ctx impl;
+};
+
+extern "C" {
+
+mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
+  int ndev = 0;
+  if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) return nullptr;
+  if (cfg) HIP_CHECK(hipSetDevice((int)cfg->device_index));
+  mz_gpu_ctx *c = new mz_gpu_ctx();
+  HIP_CHECK(hipStreamCreate(&c->impl.stream));
+  HIP_CHECK(hipEventCreate(&c->impl.ev_a));
+  HIP_CHECK(hipEventCreate(&c->impl.ev_b));
+  return c;
+}
+
+void mz_gpu_fini(mz_gpu_ctx *c) {
+  if (!c) return;
+  hipStreamSynchronize(c->impl.stream);
+  delete c;
+}
+
+const char *mz_gpu_last_error(mz_gpu_ctx *c) { return c->impl.err.c_str(); }
+
+int mz_gpu_sync(mz_gpu_ctx *c) {
+  HIP_CHECK(hipStreamSynchronize(c->impl.stream));
+  return 0;
+}
+
+mz_gpu_arr *mz_gpu_arr_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
+  mz_gpu_arr *a = new mz_gpu_arr();
+  a->schema = {s->key_words, s->val_bytes};
+  c->impl.arrs.push_back(a);
+  return a;
+}
+
+void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
+  for (auto &b : a->batches) free_batch(b);
+  a->batches.clear();
+}
+
+int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
+                          const mz_gpu_updates *u) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  u32 kw = a->schema.kw, vb = a->schema.vb;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  // copy into owned arrays (batch owns its storage)
+  u64 *keys = dnew<u64>(std::max<u64>(d.n, 1) * kw);
+  u8 *vals = (u8 *)dmalloc(std::max<u64>(d.n * vb, 1));
+  u64 *times = dnew<u64>(std::max<u64>(d.n, 1));
+  i64 *diffs = dnew<i64>(std::max<u64>(d.n, 1));
+  if (d.n) {
+    HIP_CHECK(hipMemcpyAsync(keys, d.keys, d.n * kw * 8,
+                             hipMemcpyDeviceToDevice, ctx->stream));
+    if (vb)
+      HIP_CHECK(hipMemcpyAsync(vals, d.vals, d.n * vb,
+                               hipMemcpyDeviceToDevice, ctx->stream));
+    HIP_CHECK(hipMemcpyAsync(times, d.times, d.n * 8,
+                             hipMemcpyDeviceToDevice, ctx->stream));
+    HIP_CHECK(hipMemcpyAsync(diffs, d.diffs, d.n * 8,
+                             hipMemcpyDeviceToDevice, ctx->stream));
+  }
+  DevBatch b =
+      build_batch(ctx, kw, vb, keys, vals, times, diffs, d.n, u->lower,
+                  u->upper);
+  a->batches.push_back(b);
+  a->upper = std::max(a->upper, u->upper);
+  // geometric maintenance: keep batch count bounded (amortized merging —
+  // scheduling policy per DESIGN.md §2.4; semantics = Spine merges)
+  while (a->batches.size() > 8) {
+    // merge the two smallest adjacent tail batches
+    size_t nb = a->batches.size();
+    merge_range(ctx, a, nb - 2, nb);
+  }
+  return 0;
+}
+
+int mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *c, mz_gpu_arr *a,
+                                      uint64_t f) {
+  a->logical_compaction = f;
+  return 0;
+}
+
+int mz_gpu_arr_maintain(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t fuel) {
+  (void)fuel;
+  merge_range(&c->impl, a, 0, a->batches.size());
+  return 0;
+}
+
+int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
+                     uint64_t *n_updates, uint64_t *hbm_bytes) {
+  *n_batches = a->batches.size();
+  u64 n = 0, by = 0;
+  u32 kw = a->schema.kw, vb = a->schema.vb;
+  for (auto &b : a->batches) {
+    n += b.n_upds;
+    by += b.n_keys * kw * 8 + b.n_vals * vb + b.n_upds * 16 +
+          b.hash_slots * (kw + 1) * 8 + (b.n_keys + b.n_vals) * 4 +
+          b.n_upds * 4 + b.n_vals * 4;
+  }
+  *n_updates = n;
+  *hbm_bytes = by;
+  return 0;
+}
+
+void mz_gpu_out_release(mz_gpu_ctx *c, mz_gpu_out *o) {
+  OutOwned *oo = reinterpret_cast<OutOwned *>(o);
+  for (void *p : {(void *)oo->keys, (void *)oo->vals, (void *)oo->times,
+                  (void *)oo->diffs})
+    if (p) HIP_CHECK(hipFree(p));
+  delete oo;
+}
+
+int mz_gpu_out_to_host(mz_gpu_ctx *c, const mz_gpu_out *o, uint64_t *keys,
+                       uint8_t *vals, uint64_t *times, int64_t *diffs) {
+  Ctx *ctx = &c->impl;
+  u64 n = o->n;
+  if (n == 0) return 0;
+  u32 kw = o->schema.key_words, vb = o->schema.val_bytes;
+  HIP_CHECK(hipMemcpyAsync(keys, o->keys, n * kw * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  if (vb)
+    HIP_CHECK(hipMemcpyAsync(vals, o->vals, n * vb, hipMemcpyDeviceToHost,
+                             ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(times, o->times, n * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(diffs, o->diffs, n * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return 0;
+}
+
+int mz_gpu_consolidate(mz_gpu_ctx *c, const mz_gpu_schema *s,
+                       const mz_gpu_updates *u, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  u32 kw = s->key_words, vb = s->val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 M;
+  consolidate_dev(ctx, kw, vb, d, &ok, &ov, &ot, &od, &M);
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  *out = make_out(ok, ov, ot, od, M, kw, vb);
+  return 0;
+}
+
+mz_gpu_join *mz_gpu_join_create(mz_gpu_ctx *c, mz_gpu_arr *a1, mz_gpu_arr *a2,
+                                const mz_gpu_closure *cl) {
+  mz_gpu_join *j = new mz_gpu_join();
+  j->arr1 = a1;
+  j->arr2 = a2;
+  j->cl = *cl;
+  c->impl.joins.push_back(j);
+  return j;
+}
+
+void mz_gpu_join_drop(mz_gpu_ctx *c, mz_gpu_join *j) { (void)c; (void)j; }
+
+// shared probe path for linear join and half join
+static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
+                      u32 stream_vb, int mode, int swap,
+                      const mz_gpu_closure *cl, mz_gpu_out **out) {
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = lookup->schema.kw, lvb = lookup->schema.vb;
+  u32 okw = cl->out.key_words, ovb = cl->out.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, stream_vb);
+  BatchList bl;
+  bl.n = 0;
+  for (auto &b : lookup->batches) {
+    if (b.n_upds == 0) continue;
+    if (bl.n >= 12) {
+      // too many batches: merge first (bounded by push-time policy)
+      merge_range(ctx, lookup, 0, lookup->batches.size());
+      return probe_impl(ctx, lookup, u, stream_vb, mode, swap, cl, out);
+    }
+    bl.b[bl.n++] = b;
+  }
+  u64 n = d.n;
+  if (n == 0 || bl.n == 0) {
+    *out = make_out(dnew<u64>(1), (u8 *)dmalloc(1), dnew<u64>(1),
+                    dnew<i64>(1), 0, okw, ovb);
+    return 0;
+  }
+  u32 *count = (u32 *)S.get((n + 1) * 4);
+  if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+  hipLaunchKernelGGL(k_probe_count, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.keys, d.vals, stream_vb, d.times, n, kw, lvb, bl, mode,
+                     swap, *cl, count);
+  u32 *offs = (u32 *)S.get((n + 1) * 4);
+  u64 M = exclusive_scan_u32(ctx, count, offs, n);
+  u64 *pk = dnew<u64>(std::max<u64>(M, 1) * okw);
+  u8 *pv = (u8 *)dmalloc(std::max<u64>(M * ovb, 1));
+  u64 *pt = dnew<u64>(std::max<u64>(M, 1));
+  i64 *pd = dnew<i64>(std::max<u64>(M, 1));
+  if (M)
+    hipLaunchKernelGGL(k_probe_emit, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, offs, pk, pv,
+                       pt, pd);
+  if (ctx->time_kernels) {
+    HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_a, ctx->ev_b));
+    ctx->probe_ms += ms;
+    ctx->probe_rows += n;
+    ctx->probe_launches += 2;
+  }
+  // consolidate the emitted pairs
+  DevUpdates pin{pk, pv, pt, pd, M};
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 Mc;
+  consolidate_dev(ctx, okw, ovb, pin, &ok, &ov, &ot, &od, &Mc);
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
+    HIP_CHECK(hipFree(p));
+  *out = make_out(ok, ov, ot, od, Mc, okw, ovb);
+  return 0;
+}
+
+int mz_gpu_join_push(mz_gpu_ctx *c, mz_gpu_join *op, int side,
+                     const mz_gpu_updates *delta, mz_gpu_out **out) {
+  mz_gpu_arr *own = side == 1 ? op->arr1 : op->arr2;
+  mz_gpu_arr *opp = side == 1 ? op->arr2 : op->arr1;
+  // swap=1 when the delta is input 2 (closure args are (v1, v2) by input
+  // number — mz_join_core.rs:69)
+  return probe_impl(&c->impl, opp, delta, own->schema.vb, PM_JOIN,
+                    side == 2 ? 1 : 0, &op->cl, out);
+}
+
+int mz_gpu_halfjoin(mz_gpu_ctx *c, mz_gpu_arr *lookup,
+                    const mz_gpu_updates *delta, uint32_t stream_val_bytes,
+                    int le, const mz_gpu_closure *cl, mz_gpu_out **out) {
+  return probe_impl(&c->impl, lookup, delta, stream_val_bytes,
+                    le ? PM_HALF_LE : PM_HALF_LT, 0, cl, out);
+}
+
+mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
+                                 const mz_gpu_reduce_spec *spec) {
+  mz_gpu_red *r = new mz_gpu_red();
+  r->spec = *spec;
+  u32 kw = spec->in.key_words;
+  u64 cap = 1ull << 22;  // 4M keys default
+  r->capacity = cap;
+  u64 slots = 2 * cap;
+  r->st.hash = dnew<u64>(slots * (kw + 1));
+  r->st.slots = slots;
+  r->st.stride_w = kw + 1 + 6 * spec->n_aggs;
+  r->st.rows = dnew<u64>(cap * r->st.stride_w);
+  r->st.capacity = cap;
+  Ctx *ctx = &c->impl;
+  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
+                     ctx->stream, r->st.hash, slots, kw);
+  c->impl.reds.push_back(r);
+  return r;
+}
+
+void mz_gpu_reduce_drop(mz_gpu_ctx *c, mz_gpu_red *r) { (void)c; (void)r; }
+
+int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
+                       const mz_gpu_updates *u, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
+  u32 okw = op->spec.out.key_words, ovb = op->spec.out.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 n = d.n;
+  if (n == 0) {
+    *out = make_out(dnew<u64>(1), (u8 *)dmalloc(1), dnew<u64>(1),
+                    dnew<i64>(1), 0, okw, ovb);
+    return 0;
+  }
+  // sort by (time, key)
+  u32 *perm = (u32 *)S.get(n * 4);
+  sort_updates(ctx, d.keys, kw, d.vals, vb, d.times, n, perm, true);
+  // materialize sorted columns
+  u64 *sk = (u64 *)S.get(n * kw * 8);
+  u8 *sv = (u8 *)S.get(std::max<u64>(n * vb, 1));
+  u64 *stm = (u64 *)S.get(n * 8);
+  i64 *sd = (i64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, d.keys, kw, perm, sk, n);
+  if (vb)
+    hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.vals, vb, perm, sv, n);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.times, perm, stm, n);
+  hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.diffs, perm, sd, n);
+  // time slice boundaries (host; few distinct times per batch)
+  std::vector<u64> htimes(n);
+  HIP_CHECK(hipMemcpyAsync(htimes.data(), stm, n * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  std::vector<std::pair<u64, u64>> slices;  // (start, end)
+  for (u64 i = 0; i < n;) {
+    u64 j = i;
+    while (j < n && htimes[j] == htimes[i]) j++;
+    slices.push_back({i, j});
+    i = j;
+  }
+  // output buffer: capacity 2 * n corrections max (each input row can
+  // change at most one key per slice; 2 rows per changed key per slice)
+  u64 cap_out = 2 * n + 16;
+  u64 *pk = dnew<u64>(cap_out * okw);
+  u8 *pv = (u8 *)dmalloc(cap_out * ovb);
+  u64 *pt = dnew<u64>(cap_out);
+  i64 *pd = dnew<i64>(cap_out);
+  unsigned long long *ocount = (unsigned long long *)S.get(8);
+  HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+  u32 *flags = (u32 *)S.get(n * 4);
+  u32 *gid = (u32 *)S.get(n * 4);
+  for (auto [lo, hi] : slices) {
+    u64 m = hi - lo;
+    // key-group starts within the slice
+    hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, stm + lo, flags, m);
+    inclusive_scan_u32(ctx, flags, gid, m);
+    u32 G;
+    HIP_CHECK(hipMemcpyAsync(&G, gid + m - 1, 4, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    u32 *starts = (u32 *)S.get((u64)G * 4);
+    hipLaunchKernelGGL(k_group_starts, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, flags, gid, starts, m);
+    // Phase A: lookup; Phase B: insert misses (separate launches for
+    // coherence — see RedState docs); Phase C: apply + emit.
+    u32 *found = (u32 *)S.get((u64)G * 4);
+    u32 *miss = (u32 *)S.get((u64)G * 4);
+    hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, starts, (u64)G, op->st,
+                       found, miss);
+    u32 *misspos = (u32 *)S.get(((u64)G + 1) * 4);
+    u64 Mn = exclusive_scan_u32(ctx, miss, misspos, G);
+    if (op->n_rows + Mn > op->capacity) {
+      ctx->err = "reduce state capacity exceeded";
+      return -1;
+    }
+    if (Mn)
+      hipLaunchKernelGGL(k_red_insert, dim3(ngrid(G)), dim3(BLK), 0,
+                         ctx->stream, sk + lo * kw, kw, starts, (u64)G, miss,
+                         misspos, op->n_rows, op->st);
+    hipLaunchKernelGGL(k_reduce_apply, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, sv + lo * vb, kw, vb,
+                       sd + lo, starts, (u64)G, m, htimes[lo], op->st, found,
+                       miss, misspos, op->n_rows, op->spec, pk, pv, pt, pd,
+                       ocount);
+    op->n_rows += Mn;
+  }
+  unsigned long long M;
+  HIP_CHECK(hipMemcpyAsync(&M, ocount, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  // consolidate corrections (deterministic order)
+  DevUpdates pin{pk, pv, pt, pd, M};
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 Mc;
+  consolidate_dev(ctx, okw, ovb, pin, &ok, &ov, &ot, &od, &Mc);
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
+    HIP_CHECK(hipFree(p));
+  *out = make_out(ok, ov, ot, od, Mc, okw, ovb);
+  return 0;
+}
+
+int mz_gpu_partition(mz_gpu_ctx *c, const mz_gpu_schema *s,
+                     const mz_gpu_updates *u, uint32_t nshards,
+                     uint64_t *out_keys, uint8_t *out_vals,
+                     uint64_t *out_times, int64_t *out_diffs,
+                     uint64_t *counts) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = s->key_words, vb = s->val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 n = d.n;
+  for (u32 i = 0; i < nshards; i++) counts[i] = 0;
+  if (n == 0) return 0;
+  u32 *shard = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_shard_of, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.keys, kw, n, nshards, shard);
+  // stable order by shard: radix sort (shard, perm)
+  u32 *perm = (u32 *)S.get(n * 4);
+  u32 *perm_out = (u32 *)S.get(n * 4);
+  u32 *shard_out = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream, perm,
+                     n);
+  size_t need = 0;
+  (void)rocprim::radix_sort_pairs(nullptr, need, shard, shard_out, perm, perm_out,
+                            (unsigned)n, 0, 32, ctx->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::radix_sort_pairs(tmp, need, shard, shard_out, perm, perm_out,
+                            (unsigned)n, 0, 32, ctx->stream);
+  // gather into caller buffers (device or host staging)
+  bool dev_out = u->on_device;
+  u64 *gk = dev_out ? out_keys : (u64 *)S.get(n * kw * 8);
+  u8 *gv = vb ? (dev_out ? out_vals : (u8 *)S.get(n * vb)) : nullptr;
+  u64 *gt = dev_out ? out_times : (u64 *)S.get(n * 8);
+  i64 *gd = dev_out ? out_diffs : (i64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, d.keys, kw, perm_out, gk, n);
+  if (vb)
+    hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.vals, vb, perm_out, gv, n);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.times, perm_out, gt, n);
+  hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.diffs, perm_out, gd, n);
+  // shard counts on host
+  std::vector<u32> hshard(n);
+  HIP_CHECK(hipMemcpyAsync(hshard.data(), shard_out, n * 4,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  for (u64 i = 0; i < n; i++) counts[hshard[i]]++;
+  if (!dev_out) {
+    HIP_CHECK(hipMemcpy(out_keys, gk, n * kw * 8, hipMemcpyDeviceToHost));
+    if (vb) HIP_CHECK(hipMemcpy(out_vals, gv, n * vb, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out_times, gt, n * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out_diffs, gd, n * 8, hipMemcpyDeviceToHost));
+  }
+  return 0;
+}
+
+uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words) {
+  return route_hash(key_words, n_words);
+}
+
+// ---- bench instrumentation (not part of the drop-in surface)
+void mz_gpu_set_kernel_timing(mz_gpu_ctx *c, int on) {
+  c->impl.time_kernels = on;
+}
+void mz_gpu_get_probe_stats(mz_gpu_ctx *c, double *ms, uint64_t *rows,
+                            uint64_t *launches) {
+  *ms = c->impl.probe_ms;
+  *rows = c->impl.probe_rows;
+  *launches = c->impl.probe_launches;
+  c->impl.probe_ms = 0;
+  c->impl.probe_rows = 0;
+  c->impl.probe_launches = 0;
+}
+
+}  // extern "C"
