@@ -195,6 +195,50 @@ class GpuCtx:
                                           for w in words])
         return self.lib.mz_gpu_route_hash(arr, len(words))
 
+    # --- device-resident variants (outputs stay on the GPU; used by the
+    # render layer to chain stages without host round trips) ---
+    def _dev_out(self, outp):
+        from .render import DevOut
+        return DevOut(self, outp)
+
+    def _take_copy(self, outp):
+        """Copy an OutBatch to host WITHOUT releasing it."""
+        import numpy as np
+        ob = outp.contents
+        n = ob.n
+        kw, vb = ob.schema.key_words, ob.schema.val_bytes
+        keys = np.empty(n * kw, np.uint64)
+        vals = np.empty(n * vb, np.uint8)
+        times = np.empty(n, np.uint64)
+        diffs = np.empty(n, np.int64)
+        if n:
+            self._check(self.lib.mz_gpu_out_to_host(
+                self.ctx, outp,
+                keys.ctypes.data_as(C.POINTER(C.c_uint64)),
+                vals.ctypes.data_as(C.POINTER(C.c_uint8)),
+                times.ctypes.data_as(C.POINTER(C.c_uint64)),
+                diffs.ctypes.data_as(C.POINTER(C.c_int64))))
+        return keys.view("int64"), vals, times, diffs
+
+    def consolidate_dev(self, sch, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_consolidate(self.ctx, C.byref(sch),
+                                                C.byref(upd), C.byref(outp)))
+        return self._dev_out(outp)
+
+    def halfjoin_dev(self, lookup, upd, stream_vb, le, cl):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_halfjoin(
+            self.ctx, lookup, C.byref(upd), stream_vb, 1 if le else 0,
+            C.byref(cl), C.byref(outp)))
+        return self._dev_out(outp)
+
+    def reduce_push_dev(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
+                                                C.byref(outp)))
+        return self._dev_out(outp)
+
     def set_kernel_timing(self, on):
         self.lib.mz_gpu_set_kernel_timing(self.ctx, 1 if on else 0)
 

@@ -1566,6 +1566,39 @@ uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words) {
   return route_hash(key_words, n_words);
 }
 
+// ---- numeric debug probes (test support; not part of the drop-in surface)
+__global__ void k_dbg_f2fp(const double *in, u128 *out, u64 n) {
+  GRID_STRIDE(i, n) out[i] = (u128)d_float_to_fixed_point(in[i]);
+}
+__global__ void k_dbg_i128d(const u128 *in, double *out, u64 n) {
+  GRID_STRIDE(i, n) out[i] = i128_to_double((i128)in[i]) / 16777216.0;
+}
+// xs[n] doubles -> fp_out[2n] u64 (lo,hi of fixed-point encode);
+// fp_in[2n] -> dec_out[n] doubles (decode path incl. the /2^24).
+void mz_gpu_debug_float_paths(mz_gpu_ctx *c, const double *xs, uint64_t n,
+                              uint64_t *fp_out, const uint64_t *fp_in,
+                              double *dec_out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  double *dx = (double *)S.get(n * 8);
+  u128 *dfp = (u128 *)S.get(n * 16);
+  double *dd = (double *)S.get(n * 8);
+  HIP_CHECK(hipMemcpyAsync(dx, xs, n * 8, hipMemcpyHostToDevice,
+                           ctx->stream));
+  hipLaunchKernelGGL(k_dbg_f2fp, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     dx, dfp, n);
+  HIP_CHECK(hipMemcpyAsync(fp_out, dfp, n * 16, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(dfp, fp_in, n * 16, hipMemcpyHostToDevice,
+                           ctx->stream));
+  hipLaunchKernelGGL(k_dbg_i128d, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     dfp, dd, n);
+  HIP_CHECK(hipMemcpyAsync(dec_out, dd, n * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+}
+
 // ---- bench instrumentation (not part of the drop-in surface)
 void mz_gpu_set_kernel_timing(mz_gpu_ctx *c, int on) {
   c->impl.time_kernels = on;

@@ -1,0 +1,156 @@
+"""materialize_amd.render — mirror of the reference's rendering surface.
+
+Names and argument meanings follow src/compute/src/render/join/
+delta_join.rs (render_delta_join, DeltaPathPlan/DeltaStagePlan —
+src/compute-types/src/plan/join/delta_join.rs:38-78) and
+src/compute/src/render/reduce.rs (render_reduce). The plan types here are
+the drop-in surface: a hand-written plan (see workloads.py) plays the role
+of the optimizer-produced `Plan::Join { plan: DeltaJoinPlan }`.
+
+The engine context (`materialize_amd._ffi.GpuCtx`) executes stages on the
+GPU via the C ABI; intermediate stage outputs stay device-resident.
+"""
+import ctypes as C
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+import numpy as np
+
+from . import _abi as abi
+
+
+class DevOut:
+    """An engine out-batch kept where the engine produced it (device for
+    GpuCtx, host for OracleCtx), usable as the next stage's input."""
+
+    def __init__(self, ctx, outp, host_cols=None):
+        self.ctx = ctx
+        self.outp = outp  # POINTER(OutBatch) or None (host)
+        self.host_cols = host_cols
+
+    @property
+    def n(self):
+        if self.outp is not None:
+            return int(self.outp.contents.n)
+        return len(self.host_cols[2])
+
+    @property
+    def schema(self):
+        if self.outp is not None:
+            s = self.outp.contents.schema
+            return s.key_words, s.val_bytes
+        return self._host_schema
+
+    def updates(self, lower, upper):
+        if self.outp is not None:
+            ob = self.outp.contents
+            u = abi.Updates()
+            u.keys = ob.keys
+            u.vals = ob.vals
+            u.times = ob.times
+            u.diffs = ob.diffs
+            u.n = ob.n
+            u.lower = lower
+            u.upper = upper
+            u.on_device = 1
+            u._ref = self
+            return u
+        k, v, t, d = self.host_cols
+        return abi.make_updates(k, v, t, d, lower, upper)
+
+    def to_host(self):
+        if self.outp is not None:
+            return self.ctx._take_copy(self.outp)
+        return self.host_cols
+
+    def release(self):
+        if self.outp is not None:
+            self.ctx.lib.mz_gpu_out_release(self.ctx.ctx, self.outp)
+            self.outp = None
+
+
+@dataclass
+class DeltaStagePlan:
+    """One lookup stage of a delta path (delta_join.rs DeltaStagePlan:70).
+
+    `lookup_relation` names the arrangement probed; `le` is the time
+    tie-break (source_relation < lookup_relation -> le, else lt —
+    delta_join.rs:356-399); `closure` is the stage's JoinClosure."""
+    lookup_relation: str
+    le: bool
+    closure: abi.Closure
+    stream_val_bytes: int
+
+
+@dataclass
+class DeltaPathPlan:
+    """One update path (delta_join.rs DeltaPathPlan:48): reacts to
+    `source_relation`'s updates; `initial_prep` is the key preparation of
+    the source stream (the initial_closure analog)."""
+    source_relation: str
+    initial_prep: Callable  # (keys, vals, diffs) -> (keys, vals, diffs)
+    stages: list
+
+
+@dataclass
+class DeltaJoinPlan:
+    paths: list
+
+
+class DeltaJoinOp:
+    """render_delta_join (delta_join.rs:51-251): one operator per path,
+    outputs concatenated by the caller."""
+
+    def __init__(self, ctx, arrangements, plan: DeltaJoinPlan):
+        self.ctx = ctx
+        self.arrangements = arrangements  # name -> arr handle
+        self.plan = plan
+
+    def push_path(self, path: DeltaPathPlan, keys, vals, diffs, t):
+        """Run one path for a batch of source updates at time t.
+        Returns a DevOut of (key2words, revenue/val, time, diff) updates,
+        or None when the path produces nothing."""
+        ctx = self.ctx
+        keys, vals, diffs = path.initial_prep(keys, vals, diffs)
+        n = len(keys)
+        if n == 0:
+            return None
+        times = np.full(n, t, np.uint64)
+        u = abi.make_updates(keys, vals, times, diffs, t, t + 1)
+        cur = None
+        for i, st in enumerate(path.stages):
+            arr = self.arrangements[st.lookup_relation]
+            out = ctx.halfjoin_dev(arr, u, st.stream_val_bytes, st.le,
+                                   st.closure)
+            if cur is not None:
+                cur.release()
+            cur = out
+            if cur.n == 0:
+                cur.release()
+                return None
+            u = cur.updates(t, t + 1)
+        return cur
+
+
+@dataclass
+class ReducePlan:
+    """render_reduce / AccumulablePlan surface (reduce.rs:71,
+    plan/reduce.rs:233)."""
+    spec: abi.ReduceSpec
+
+
+class ReduceOp:
+    def __init__(self, ctx, plan: ReducePlan):
+        self.ctx = ctx
+        self.op = ctx.reduce_create(plan.spec)
+
+    def push(self, updates) -> DevOut:
+        return self.ctx.reduce_push_dev(self.op, updates)
+
+
+def render_delta_join(ctx, arrangements, plan) -> DeltaJoinOp:
+    return DeltaJoinOp(ctx, arrangements, plan)
+
+
+def render_reduce(ctx, plan) -> ReduceOp:
+    return ReduceOp(ctx, plan)

@@ -133,6 +133,25 @@ class OracleCtx:
         assert rc == 0
         return self._take(outp)
 
+    # --- DevOut-style variants mirroring GpuCtx (host-resident here) ---
+    def _host_out(self, res, kw, vb):
+        from materialize_amd.render import DevOut
+        o = DevOut(self, None, host_cols=res)
+        o._host_schema = (kw, vb)
+        return o
+
+    def consolidate_dev(self, sch, upd):
+        res = self.consolidate(sch, upd)
+        return self._host_out(res, sch.key_words, sch.val_bytes)
+
+    def halfjoin_dev(self, lookup, upd, stream_vb, le, cl):
+        res = self.halfjoin(lookup, upd, stream_vb, le, cl)
+        return self._host_out(res, cl.out.key_words, cl.out.val_bytes)
+
+    def reduce_push_dev(self, op, upd):
+        res = self.reduce_push(op, upd)
+        return self._host_out(res, 0, 0)
+
     def route_hash(self, words):
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
