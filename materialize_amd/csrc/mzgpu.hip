@@ -531,11 +531,13 @@ __device__ __forceinline__ void d_finalize(const mz_gpu_aggregate a,
     memcpy(slot + 8, &v, 16);
   } else {
     double v;
-    if (acc.nan > 0 || (acc.pi > 0 && acc.ni > 0))
+    // counts are SIGNED (Diff::is_positive, reduce.rs:1920-1927): a
+    // wrapped-negative inf/nan count is NOT positive
+    if ((i64)acc.nan > 0 || ((i64)acc.pi > 0 && (i64)acc.ni > 0))
       v = __longlong_as_double(0x7FF8000000000000LL);  // NaN
-    else if (acc.pi > 0)
+    else if ((i64)acc.pi > 0)
       v = HUGE_VAL;
-    else if (acc.ni > 0)
+    else if ((i64)acc.ni > 0)
       v = -HUGE_VAL;
     else
       v = i128_to_double((i128)acc.accum) / 16777216.0;

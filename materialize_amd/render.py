@@ -99,24 +99,48 @@ class DeltaJoinPlan:
 
 class DeltaJoinOp:
     """render_delta_join (delta_join.rs:51-251): one operator per path,
-    outputs concatenated by the caller."""
+    outputs concatenated by the caller. With an exchange (world > 1), the
+    stream is re-distributed by the next stage's key hash between stages —
+    the Exchange pacts of linear_join.rs:390 / delta_join.rs:442-464, one
+    all-to-all-v per stage per batch."""
 
-    def __init__(self, ctx, arrangements, plan: DeltaJoinPlan):
+    def __init__(self, ctx, arrangements, plan: DeltaJoinPlan,
+                 exchange=None):
         self.ctx = ctx
         self.arrangements = arrangements  # name -> arr handle
         self.plan = plan
+        self.exchange = exchange  # None or dist.TorchExchange
 
-    def push_path(self, path: DeltaPathPlan, keys, vals, diffs, t):
+    def _exchange_out(self, cur, t, kw, vb):
+        """Re-distribute a stage output by its (new) key hash. COLLECTIVE:
+        every rank calls this for every stage, with however few local rows
+        it has — a rank-dependent skip would desynchronize the
+        all-to-all."""
+        keys, vals, times, diffs = cur.to_host()
+        cur.release()
+        keys, vals, times, diffs = self.exchange.exchange(
+            keys, vals, times, diffs, kw, vb)
+        return abi.make_updates(keys, vals, times, diffs, t, t + 1)
+
+    def push_path(self, path: DeltaPathPlan, keys, vals, diffs, t,
+                  final_exchange=False):
         """Run one path for a batch of source updates at time t.
-        Returns a DevOut of (key2words, revenue/val, time, diff) updates,
-        or None when the path produces nothing."""
+        Returns a DevOut (or, post-exchange, a host-resident one), or None
+        when nothing is produced. Source updates must already be sharded by
+        the first stage's key (the caller filters them). When an exchange
+        is attached, EVERY rank runs every stage and every exchange, even
+        with zero local rows (collectives must match across ranks)."""
         ctx = self.ctx
         keys, vals, diffs = path.initial_prep(keys, vals, diffs)
         n = len(keys)
-        if n == 0:
+        collective = self.exchange is not None
+        if n == 0 and not collective:
             return None
         times = np.full(n, t, np.uint64)
-        u = abi.make_updates(keys, vals, times, diffs, t, t + 1)
+        u = abi.make_updates(np.ascontiguousarray(keys, np.int64),
+                             vals, times,
+                             np.ascontiguousarray(diffs, np.int64),
+                             t, t + 1)
         cur = None
         for i, st in enumerate(path.stages):
             arr = self.arrangements[st.lookup_relation]
@@ -125,11 +149,55 @@ class DeltaJoinOp:
             if cur is not None:
                 cur.release()
             cur = out
-            if cur.n == 0:
-                cur.release()
-                return None
-            u = cur.updates(t, t + 1)
+            more = i + 1 < len(path.stages)
+            if collective and (more or final_exchange):
+                okw = st.closure.out.key_words
+                ovb = st.closure.out.val_bytes
+                u = self._exchange_out(cur, t, okw, ovb)
+                cur = HostOut(u, okw, ovb)
+            elif more:
+                u = cur.updates(t, t + 1)
+        if cur.n == 0:
+            cur.release()
+            return None
         return cur
+
+
+class HostOut:
+    """A host-resident stage result (post-exchange) presenting the DevOut
+    interface."""
+
+    def __init__(self, upd, kw, vb):
+        self._upd = upd
+        self._schema = (kw, vb)
+
+    @property
+    def n(self):
+        return int(self._upd.n)
+
+    @property
+    def schema(self):
+        return self._schema
+
+    def updates(self, lower, upper):
+        self._upd.lower = lower
+        self._upd.upper = upper
+        return self._upd
+
+    def to_host(self):
+        import numpy as np
+        u = self._upd
+        kw, vb = self._schema
+        n = int(u.n)
+        keys = np.ctypeslib.as_array(u.keys, shape=(n * kw,)).view(np.int64)
+        vals = (np.ctypeslib.as_array(u.vals, shape=(n * vb,))
+                if vb else np.empty(0, np.uint8))
+        times = np.ctypeslib.as_array(u.times, shape=(n,))
+        diffs = np.ctypeslib.as_array(u.diffs, shape=(n,))
+        return keys.copy(), vals.copy(), times.copy(), diffs.copy()
+
+    def release(self):
+        self._upd = None
 
 
 @dataclass
@@ -148,8 +216,8 @@ class ReduceOp:
         return self.ctx.reduce_push_dev(self.op, updates)
 
 
-def render_delta_join(ctx, arrangements, plan) -> DeltaJoinOp:
-    return DeltaJoinOp(ctx, arrangements, plan)
+def render_delta_join(ctx, arrangements, plan, exchange=None) -> DeltaJoinOp:
+    return DeltaJoinOp(ctx, arrangements, plan, exchange=exchange)
 
 
 def render_reduce(ctx, plan) -> ReduceOp:

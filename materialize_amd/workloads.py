@@ -207,3 +207,100 @@ class Q3Dataflow:
     def stats(self):
         return {name: self.ctx.arr_stats(arr)
                 for name, arr in self.arrs.items()}
+
+
+class ShardedQ3Dataflow(Q3Dataflow):
+    """Q3 sharded across ranks: every arrangement is hash-partitioned by
+    its key (shard = route_hash(key) % world); stage outputs
+    are re-distributed by the next key's hash via the exchange (RCCL
+    all-to-all over xGMI at N GPUs; gloo in CPU tests). Every rank
+    generates the same deterministic churn and keeps its own shard."""
+
+    def __init__(self, ctx, exchange):
+        from .dist import shard_of
+        self.shard_of = shard_of
+        self.exchange = exchange
+        super().__init__(ctx)
+        self.join = render_delta_join(ctx, self.arrs, self.plan,
+                                      exchange=exchange)
+
+    def _filter_shard(self, keys, vals, diffs):
+        W, r = self.exchange.world, self.exchange.rank
+        if W == 1:
+            return keys, vals, diffs
+        m = self.shard_of(np.ascontiguousarray(keys, np.int64), 1, W) == r
+        vals = vals.reshape(len(keys), -1)
+        return keys[m], vals[m], diffs[m]
+
+    def _seal_push(self, name, keys, vals, diffs, t):
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.uint8).reshape(len(keys), -1)
+        diffs = np.ascontiguousarray(diffs, np.int64)
+        keys, vals, diffs = self._filter_shard(keys, vals, diffs)
+        if len(keys) == 0:
+            return
+        super()._seal_push(name, keys, vals, diffs, t)
+
+    def _push_sharded_path(self, name, keys, vals, diffs, t):
+        """Filter source updates to this rank's shard of the first stage's
+        key, then run the path with inter-stage exchanges."""
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.uint8).reshape(len(keys), -1)
+        diffs = np.ascontiguousarray(diffs, np.int64)
+        keys, vals, diffs = self._filter_shard(keys, vals, diffs)
+        return self.join.push_path(self.paths[name], keys, vals, diffs, t,
+                                   final_exchange=True)
+
+    def load(self, gen):
+        ones = lambda n: np.ones(n, np.int64)
+        ck, cm = gen.customer_updates()
+        self._seal_push("customer", ck, cm.view(np.uint8), ones(len(ck)), 0)
+        oidx = np.arange(gen.n_orders)
+        self._seal_push("orders_by_orderkey", gen.o_orderkey,
+                        gen.orders_vals(oidx), ones(gen.n_orders), 0)
+        self._seal_push("orders_by_custkey", gen.o_custkey,
+                        gen.orders_bycust_vals(oidx), ones(gen.n_orders), 0)
+        lk, lv = gen.lineitem_updates()
+        self._seal_push("lineitem", lk, lv, ones(len(lk)), 0)
+        out = self._push_sharded_path("customer", ck,
+                                      cm.view(np.uint8).reshape(-1, 8),
+                                      ones(len(ck)), 0)
+        n_corr = 0
+        if out is not None:
+            corr = self.reduce.push(out.updates(0, 1))
+            n_corr = corr.n
+            corr.release()
+            out.release()
+        return n_corr
+
+    def step(self, churn, t):
+        l_keys, l_vals, l_diffs = churn["lineitem"]
+        o_keys, o_vals, o_diffs = churn["orders"]
+        oc_keys, oc_vals, oc_diffs = churn["orders_by_cust"]
+        rows = len(l_keys) + len(o_keys)  # global input rows (same on all
+        # ranks; bench divides by wall time once, using the global count)
+        self._seal_push("lineitem", l_keys, l_vals, l_diffs, t)
+        self._seal_push("orders_by_orderkey", o_keys, o_vals, o_diffs, t)
+        self._seal_push("orders_by_custkey", oc_keys, oc_vals, oc_diffs, t)
+        outs = []
+        o = self._push_sharded_path("orders", oc_keys, oc_vals, oc_diffs, t)
+        if o is not None:
+            outs.append(o)
+        o = self._push_sharded_path("lineitem", l_keys, l_vals, l_diffs, t)
+        if o is not None:
+            outs.append(o)
+        if not outs:
+            return rows, None
+        if len(outs) == 1:
+            corr = self.reduce.push(outs[0].updates(t, t + 1))
+        else:
+            cols = [out.to_host() for out in outs]
+            keys = np.concatenate([c[0] for c in cols])
+            vals = np.concatenate([c[1] for c in cols])
+            times = np.concatenate([c[2] for c in cols])
+            diffs = np.concatenate([c[3] for c in cols])
+            u = abi.make_updates(keys, vals, times, diffs, t, t + 1)
+            corr = self.reduce.push(u)
+        for out in outs:
+            out.release()
+        return rows, corr

@@ -1,0 +1,91 @@
+"""Sharded Q3 over torch.distributed gloo (world_size 2, CPU).
+
+Covers the multi-GPU path's logic without GPUs: hash-sharded
+arrangements, inter-stage all-to-all-v exchanges, sharded reduce. The
+union of the two ranks' maintained results must equal the unsharded
+oracle run exactly.
+"""
+import multiprocessing as mp
+import os
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _final_state(df, gen, steps, batch_rows):
+    """Run load + churn steps, returning the maintained {key: revenue}."""
+    state = {}
+    holder = []
+    orig = df.reduce.push
+
+    def capture(u):
+        o = orig(u)
+        holder.append(o.to_host())
+        return o
+
+    df.reduce.push = capture
+    df.load(gen)
+    for t in range(1, steps + 1):
+        df.step(gen.churn(batch_rows), t)
+    for keys, vals, times, diffs in holder:
+        n = len(times)
+        vals = vals.reshape(n, 24) if n else vals
+        for i in range(n):
+            k = (int(keys[2 * i]), int(np.uint64(keys[2 * i + 1])))
+            lo = int(vals[i][8:16].view(np.uint64)[0])
+            hi = int(vals[i][16:24].view(np.int64)[0])
+            v = hi * 2**64 + lo
+            if int(diffs[i]) == 1:
+                state[k] = v
+            else:
+                assert state.pop(k) == v
+    return state
+
+
+def _worker(rank, world, port, ret):
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from materialize_amd.dist import TorchExchange
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import ShardedQ3Dataflow
+    from pyoracle import OracleCtx
+    ctx = OracleCtx()
+    df = ShardedQ3Dataflow(ctx, TorchExchange("cpu"))
+    gen = TpchGen(sf=0.002, seed=7)
+    state = _final_state(df, gen, steps=3, batch_rows=300)
+    ret[rank] = state
+    dist.destroy_process_group()
+
+
+def test_sharded_q3_gloo_world2():
+    from materialize_amd.workloads import Q3Dataflow
+    from materialize_amd.tpch import TpchGen
+    from pyoracle import OracleCtx
+    # unsharded reference run
+    df = Q3Dataflow(OracleCtx())
+    gen = TpchGen(sf=0.002, seed=7)
+    want = _final_state(df, gen, steps=3, batch_rows=300)
+    # sharded run, world 2
+    ctx = mp.get_context("spawn")
+    mgr = ctx.Manager()
+    ret = mgr.dict()
+    port = 29511
+    ps = [ctx.Process(target=_worker, args=(r, 2, port, ret))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    s0, s1 = ret[0], ret[1]
+    assert not (set(s0) & set(s1)), "shards overlap"
+    merged = {**s0, **s1}
+    assert merged == want
+    assert len(want) > 0
