@@ -1,0 +1,259 @@
+#!/usr/bin/env python3
+"""bench.py — TPC-H Q3 incremental maintenance on MI355X.
+
+Metric (BASELINE.json): input update rows/sec maintained (TPC-H Q3 delta
+join). A step = one churn batch (retract/insert order + lineitem rows,
+tpch.rs:204-241 shape) maintained end-to-end: arrangement pushes, the
+orders + lineitem delta paths (2 probe stages each) and the SUM reduce.
+
+N=1 workload = BASELINE config 2: TPC-H SF1 Q3 under 100k-row churn
+batches on one GPU. N>1 = config 3 (weak scaling): SF 1.25*N sharded by
+key hash across N GPUs with all-to-all-v exchanges over RCCL/xGMI,
+batches of 100k*N global rows.
+
+Inputs are pre-generated and pre-staged into HBM before the timed region
+(churn columns as torch CUDA tensors); the PCIe-inclusive rate is noted
+in DESIGN.md §7, never reported as `value`.
+
+Contract: W untimed warmup steps, then exactly K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX over ranks; rank 0
+prints ONE JSON line.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK = 8.0e12  # B/s, MI355X spec (MI355X_MICROARCH.md)
+METRIC = "input update rows/sec maintained (TPC-H Q3 delta join)"
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--sf", type=float, default=0.0, help="0 = auto by N")
+    p.add_argument("--batch-rows", type=int, default=0, help="0 = auto")
+    p.add_argument("--no-cpu-baseline", action="store_true")
+    p.add_argument("--seed", type=int, default=42)
+    return p.parse_args()
+
+
+def stage_churn(churn, t, device):
+    """Pre-stage one churn batch's columns into HBM as torch tensors and
+    build device Updates descriptors (times == t)."""
+    import torch
+    from materialize_amd import _abi as abi
+    out = {}
+    for name, (keys, vals, diffs) in churn.items():
+        tgt = {"orders_by_cust": "orders_by_custkey",
+               "orders": "orders_by_orderkey",
+               "lineitem": "lineitem"}[name]
+        n = len(keys)
+        kt = torch.from_numpy(np.ascontiguousarray(keys, np.int64)
+                              ).to(device)
+        vt = torch.from_numpy(np.ascontiguousarray(vals, np.uint8).reshape(-1)
+                              ).to(device)
+        tt = torch.full((n,), t, dtype=torch.int64, device=device)
+        dt = torch.from_numpy(np.ascontiguousarray(diffs, np.int64)
+                              ).to(device)
+        out[tgt] = abi.make_updates_from_torch(kt, vt, tt, dt, t, t + 1)
+    return out
+
+
+def filter_shard(churn, world, rank):
+    """Pre-filter churn columns to this rank's shard (untimed; the
+    generator is deterministic and identical on every rank)."""
+    from materialize_amd.dist import shard_of
+    if world == 1:
+        return churn
+    out = {}
+    for name, (keys, vals, diffs) in churn.items():
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.uint8).reshape(len(keys), -1)
+        m = shard_of(keys, 1, world) == rank
+        out[name] = (keys[m], vals[m], diffs[m])
+    return out
+
+
+def run_cpu_baseline(seed):
+    """Oracle (the CPU restatement, kind 'port') on a bounded sample of the
+    same workload: SF 0.05 Q3, 5k-row churn batches, single thread."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import Q3Dataflow
+    from pyoracle import OracleCtx
+    sf, batch, steps = 0.05, 5000, 6
+    gen = TpchGen(sf=sf, seed=seed)
+    df = Q3Dataflow(OracleCtx())
+    df.load(gen)
+    churns = [gen.churn(batch) for _ in range(steps + 1)]
+    rows, _ = df.step(churns[0], 1)  # warmup
+    t0 = time.perf_counter()
+    total = 0
+    for i in range(1, steps + 1):
+        r, corr = df.step(churns[i], i + 1)
+        if corr is not None:
+            corr.release()
+        total += r
+    dt = time.perf_counter() - t0
+    return {
+        "value": total / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": (f"oracle C++ restatement, TPC-H SF{sf} Q3, {steps} churn "
+                   f"batches of ~{batch} rows, single thread "
+                   f"({dt:.1f}s of CPU work)"),
+    }
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    N = world if world > 1 else args.gpus
+    if args.gpus > 1 and world == 1:
+        print("ERROR: launch N>1 via torch.distributed.run", file=sys.stderr)
+        sys.exit(2)
+
+    import torch
+    from materialize_amd import _abi as abi  # noqa: F401
+    from materialize_amd._ffi import GpuCtx
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import Q3Dataflow, ShardedQ3Dataflow
+
+    sf = args.sf or (1.0 if N == 1 else 1.25 * N)
+    batch_rows = args.batch_rows or 100_000 * N
+    device = f"cuda:{local_rank}"
+    torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("nccl")
+        from materialize_amd.dist import TorchExchange
+        ex = TorchExchange(device)
+    ctx = GpuCtx(device=local_rank)
+
+    gen = TpchGen(sf=sf, seed=args.seed)
+    if world > 1:
+        df = ShardedQ3Dataflow(ctx, ex)
+    else:
+        df = Q3Dataflow(ctx)
+    df.load(gen)
+    df.maintain()
+    ctx.lib.mz_gpu_sync(ctx.ctx)
+
+    # pre-generate + pre-filter + pre-stage all churn batches (untimed)
+    K, W = args.steps, args.warmup
+    staged = []
+    rows_per_step = []
+    for i in range(W + K):
+        churn = gen.churn(batch_rows)
+        rows_per_step.append(len(churn["lineitem"][0]) +
+                             len(churn["orders"][0]))
+        churn = filter_shard(churn, world, rank)
+        staged.append(stage_churn(churn, i + 1, device))
+
+    for i in range(W):
+        corr = df.step_dev(staged[i], i + 1)
+        if corr is not None:
+            corr.release()
+    ctx.lib.mz_gpu_sync(ctx.ctx)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+
+    ctx.set_kernel_timing(1)
+    t0 = time.perf_counter()
+    for i in range(W, W + K):
+        corr = df.step_dev(staged[i], i + 1)
+        if corr is not None:
+            corr.release()
+    ctx.lib.mz_gpu_sync(ctx.ctx)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    ctx.set_kernel_timing(0)
+
+    if dist:
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.cpu())
+
+    probe_ms, probe_rows, probe_launches = ctx.probe_stats()
+    import ctypes as C
+    pairs, batches, alg_bytes = C.c_uint64(), C.c_uint64(), C.c_uint64()
+    ctx.lib.mz_gpu_get_probe_stats2.argtypes = [C.c_void_p] + \
+        [C.POINTER(C.c_uint64)] * 3
+    ctx.lib.mz_gpu_get_probe_stats2(ctx.ctx, C.byref(pairs),
+                                    C.byref(batches), C.byref(alg_bytes))
+
+    total_rows = sum(rows_per_step[W:])
+    value = total_rows / elapsed
+    achieved = (alg_bytes.value / (probe_ms / 1e3)) if probe_ms > 0 else 0.0
+    roofline = {
+        "bound": "hbm",
+        "achieved": achieved,
+        "peak": HBM_PEAK,
+        "unit": "GB/s",
+        "frac": achieved / HBM_PEAK,
+        "traffic": None,
+    }
+    # normalize achieved/peak into GB/s for the printed unit
+    roofline["achieved"] = achieved / 1e9
+    roofline["peak"] = HBM_PEAK / 1e9
+
+    cpu_baseline = None
+    if rank == 0 and N == 1 and not args.no_cpu_baseline:
+        cpu_baseline = run_cpu_baseline(args.seed)
+
+    if rank == 0:
+        line = {
+            "metric": METRIC,
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": N,
+            "steps": K,
+            "warmup": W,
+            "ms_per_step": elapsed / K * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": (f"tpch_q3_sf{sf:g}_churn{batch_rows}" +
+                             ("" if N == 1 else f"_shards{N}")),
+                "sf": sf,
+                "batch_rows": batch_rows,
+                "parallelism": f"shards{N}",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+            "probe_kernel": {
+                "ms_total": probe_ms,
+                "delta_rows": probe_rows,
+                "emitted_pairs": pairs.value,
+                "launch_pairs": probe_launches // 2,
+                "alg_bytes": alg_bytes.value,
+            },
+        }
+        print(json.dumps(line))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

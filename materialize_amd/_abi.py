@@ -150,6 +150,25 @@ def make_updates(keys, vals, times, diffs, lower, upper, on_device=0):
     return u
 
 
+def make_updates_from_torch(keys_t, vals_t, times_t, diffs_t, lower, upper):
+    """Updates descriptor over torch CUDA tensors (on_device=1). Tensors:
+    keys int64 [n*kw], vals uint8 [n*vb] or None, times int64 (bit-pattern
+    u64) [n], diffs int64 [n]. Keeps tensor refs alive on the struct."""
+    u = Updates()
+    n = times_t.numel()
+    u.keys = C.cast(keys_t.data_ptr(), C.POINTER(C.c_uint64))
+    u.vals = (C.cast(vals_t.data_ptr(), C.POINTER(C.c_uint8))
+              if vals_t is not None and vals_t.numel() else None)
+    u.times = C.cast(times_t.data_ptr(), C.POINTER(C.c_uint64))
+    u.diffs = C.cast(diffs_t.data_ptr(), C.POINTER(C.c_int64))
+    u.n = n
+    u.lower = lower
+    u.upper = upper
+    u.on_device = 1
+    u._refs = (keys_t, vals_t, times_t, diffs_t)
+    return u
+
+
 def out_to_numpy(out, copy=True):
     """Read a host OutBatch into numpy arrays (keys, vals, times, diffs)."""
     n = out.n

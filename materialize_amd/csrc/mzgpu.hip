@@ -773,6 +773,7 @@ struct Ctx {
   // probe-kernel timing (for bench roofline): accumulated ns and bytes
   double probe_ms = 0;
   u64 probe_rows = 0, probe_launches = 0;
+  u64 probe_pairs = 0, probe_batches = 0, probe_alg_bytes = 0;
   hipEvent_t ev_a = nullptr, ev_b = nullptr;
   int time_kernels = 0;
 };
@@ -1341,6 +1342,15 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     ctx->probe_ms += ms;
     ctx->probe_rows += n;
     ctx->probe_launches += 2;
+    ctx->probe_pairs += M;
+    ctx->probe_batches += (u64)bl.n;
+    // Algorithmic bytes of the probe pair (SURVEY §8d model): delta tuple
+    // + one hash line per probed batch + matched val+upd read + output
+    // write. Counted ONCE for the count+emit pair — the 2-phase re-read is
+    // implementation overhead and must show as a lower roofline fraction.
+    ctx->probe_alg_bytes +=
+        n * (8ull * kw + stream_vb + 16) + n * 128ull * (u64)bl.n +
+        M * (lvb + 16ull) + M * (8ull * okw + ovb + 16);
   }
   // consolidate the emitted pairs
   DevUpdates pin{pk, pv, pt, pd, M};
@@ -1613,6 +1623,15 @@ void mz_gpu_get_probe_stats(mz_gpu_ctx *c, double *ms, uint64_t *rows,
   c->impl.probe_ms = 0;
   c->impl.probe_rows = 0;
   c->impl.probe_launches = 0;
+}
+void mz_gpu_get_probe_stats2(mz_gpu_ctx *c, uint64_t *pairs,
+                             uint64_t *batches, uint64_t *alg_bytes) {
+  *pairs = c->impl.probe_pairs;
+  *batches = c->impl.probe_batches;
+  *alg_bytes = c->impl.probe_alg_bytes;
+  c->impl.probe_pairs = 0;
+  c->impl.probe_batches = 0;
+  c->impl.probe_alg_bytes = 0;
 }
 
 }  // extern "C"

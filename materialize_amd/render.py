@@ -130,17 +130,25 @@ class DeltaJoinOp:
         the first stage's key (the caller filters them). When an exchange
         is attached, EVERY rank runs every stage and every exchange, even
         with zero local rows (collectives must match across ranks)."""
-        ctx = self.ctx
         keys, vals, diffs = path.initial_prep(keys, vals, diffs)
         n = len(keys)
-        collective = self.exchange is not None
-        if n == 0 and not collective:
+        if n == 0 and self.exchange is None:
             return None
         times = np.full(n, t, np.uint64)
         u = abi.make_updates(np.ascontiguousarray(keys, np.int64),
                              vals, times,
                              np.ascontiguousarray(diffs, np.int64),
                              t, t + 1)
+        return self.push_path_updates(path, u, t,
+                                      final_exchange=final_exchange)
+
+    def push_path_updates(self, path: DeltaPathPlan, u, t,
+                          final_exchange=False):
+        """As push_path, but the source stream arrives as a ready Updates
+        descriptor (host or device — the bench pre-stages churn batches
+        into HBM and passes device columns)."""
+        ctx = self.ctx
+        collective = self.exchange is not None
         cur = None
         for i, st in enumerate(path.stages):
             arr = self.arrangements[st.lookup_relation]

@@ -122,14 +122,16 @@ class TpchGen:
         # expected rows per churned order: 2 (order) + 2 * avg lines (~4)
         k = max(1, int(target_rows / 10))
         idx = rng.choice(self.n_orders, k, replace=False)
+        # flat positions of the chosen orders' lineitems (vectorized gather)
+        counts = self.l_count[idx].astype(np.int64)
+        starts = self.l_offs[idx]
+        cum = np.concatenate([[0], np.cumsum(counts)])
+        pos = np.repeat(starts - cum[:-1], counts) + np.arange(cum[-1])
         # --- retractions of current state
-        r_l_keys, r_l_vals = [], []
-        for i in idx:
-            lo, hi = self.l_offs[i], self.l_offs[i + 1]
-            r_l_keys.append(self.l_orderkey[lo:hi])
-            r_l_vals.append(self.lineitem_vals(
-                self.l_extendedprice[lo:hi], self.l_discount[lo:hi],
-                self.l_shipdate[lo:hi]))
+        r_l_keys = self.l_orderkey[pos].copy()
+        r_l_vals = self.lineitem_vals(self.l_extendedprice[pos].copy(),
+                                      self.l_discount[pos].copy(),
+                                      self.l_shipdate[pos].copy())
         o_retract_vals = self.orders_vals(idx)
         o_retract_bycust_vals = self.orders_bycust_vals(idx)
         o_retract_bycust_keys = self.o_custkey[idx].copy()
@@ -139,23 +141,19 @@ class TpchGen:
         ck = np.minimum(ck, max(self.n_customer, 1))
         self.o_custkey[idx] = ck
         self.o_orderdate[idx] = rng.integers(1, 2252, k).astype(np.int32)
-        # regenerate each order's lineitems in place (same count per order —
-        # keeps the flat arrays stable; the reference redraws 1..7, a shape
+        # regenerate the lineitems in place (same count per order — keeps
+        # the flat arrays stable; the reference redraws 1..7, a shape
         # detail that does not change the maintained row rate)
-        n_l, n_keys, n_vals = [], [], []
-        for i in idx:
-            lo, hi = self.l_offs[i], self.l_offs[i + 1]
-            m = hi - lo
-            oi = np.full(m, i)
-            ep, disc, sd = self._lineitem_cols(oi, rng)
-            self.l_extendedprice[lo:hi] = ep
-            self.l_discount[lo:hi] = disc
-            self.l_shipdate[lo:hi] = sd
-            n_keys.append(self.l_orderkey[lo:hi])
-            n_vals.append(self.lineitem_vals(ep, disc, sd))
-        l_keys = np.concatenate(r_l_keys + n_keys)
-        l_vals = np.concatenate(r_l_vals + n_vals)
-        nr = sum(len(x) for x in r_l_keys)
+        order_idx_rep = np.repeat(idx, counts)
+        ep, disc, sd = self._lineitem_cols(order_idx_rep, rng)
+        self.l_extendedprice[pos] = ep
+        self.l_discount[pos] = disc
+        self.l_shipdate[pos] = sd
+        n_keys = self.l_orderkey[pos].copy()
+        n_vals = self.lineitem_vals(ep, disc, sd)
+        l_keys = np.concatenate([r_l_keys, n_keys])
+        l_vals = np.concatenate([r_l_vals, n_vals])
+        nr = len(r_l_keys)
         l_diffs = np.concatenate([-np.ones(nr, np.int64),
                                   np.ones(len(l_keys) - nr, np.int64)])
         o_keys = np.concatenate([self.o_orderkey[idx], self.o_orderkey[idx]])

@@ -200,6 +200,45 @@ class Q3Dataflow:
             out.release()
         return rows, corr
 
+    _final_exchange = False
+
+    def step_dev(self, upd, t):
+        """Bench path: one churn step whose update columns are ALREADY
+        staged (device tensors at N GPUs — inputs resident in HBM when the
+        timed region starts). `upd` maps the three updated relations to
+        Updates descriptors with times == t. Returns the corrections
+        DevOut (or None)."""
+        ctx = self.ctx
+        for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
+            u = upd[name]
+            kw, vb = self.SCHEMAS[name]
+            sealed = ctx.consolidate_dev(abi.schema(kw, vb), u)
+            ctx.arr_push(self.arrs[name], sealed.updates(t, t + 1))
+            sealed.release()
+        outs = []
+        for rel, src in (("orders", "orders_by_custkey"),
+                         ("lineitem", "lineitem")):
+            o = self.join.push_path_updates(
+                self.paths[rel], upd[src], t,
+                final_exchange=self._final_exchange)
+            if o is not None:
+                outs.append(o)
+        if not outs:
+            return None
+        if len(outs) == 1:
+            corr = self.reduce.push(outs[0].updates(t, t + 1))
+        else:
+            cols = [out.to_host() for out in outs]
+            keys = np.concatenate([c[0] for c in cols])
+            vals = np.concatenate([c[1] for c in cols])
+            times = np.concatenate([c[2] for c in cols])
+            diffs = np.concatenate([c[3] for c in cols])
+            u = abi.make_updates(keys, vals, times, diffs, t, t + 1)
+            corr = self.reduce.push(u)
+        for out in outs:
+            out.release()
+        return corr
+
     def maintain(self):
         for arr in self.arrs.values():
             self.ctx.arr_maintain(arr)
@@ -215,6 +254,8 @@ class ShardedQ3Dataflow(Q3Dataflow):
     are re-distributed by the next key's hash via the exchange (RCCL
     all-to-all over xGMI at N GPUs; gloo in CPU tests). Every rank
     generates the same deterministic churn and keeps its own shard."""
+
+    _final_exchange = True
 
     def __init__(self, ctx, exchange):
         from .dist import shard_of
