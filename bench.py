@@ -85,17 +85,21 @@ def filter_shard(churn, world, rank):
 
 def run_cpu_baseline(seed):
     """Oracle (the CPU restatement, kind 'port') on a bounded sample of the
-    same workload: SF 0.05 Q3, 5k-row churn batches, single thread."""
+    SAME workload shape as the GPU run: TPC-H SF1 Q3, 100k-row churn
+    batches, single thread. The sample is a handful of batches (~10-30s of
+    CPU work including per-batch arrangement maintenance)."""
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     from materialize_amd.tpch import TpchGen
     from materialize_amd.workloads import Q3Dataflow
     from pyoracle import OracleCtx
-    sf, batch, steps = 0.05, 5000, 6
+    sf, batch, steps = 1.0, 100_000, 4
     gen = TpchGen(sf=sf, seed=seed)
     df = Q3Dataflow(OracleCtx())
     df.load(gen)
     churns = [gen.churn(batch) for _ in range(steps + 1)]
-    rows, _ = df.step(churns[0], 1)  # warmup
+    rows, corr = df.step(churns[0], 1)  # warmup
+    if corr is not None:
+        corr.release()
     t0 = time.perf_counter()
     total = 0
     for i in range(1, steps + 1):
@@ -109,9 +113,9 @@ def run_cpu_baseline(seed):
         "unit": "rows/s",
         "cores": 1,
         "kind": "port",
-        "sample": (f"oracle C++ restatement, TPC-H SF{sf} Q3, {steps} churn "
-                   f"batches of ~{batch} rows, single thread "
-                   f"({dt:.1f}s of CPU work)"),
+        "sample": (f"oracle C++ restatement, TPC-H SF{sf:g} Q3, {steps} "
+                   f"churn batches of ~{batch} rows, single thread "
+                   f"({dt:.1f}s of CPU work in the timed sample)"),
     }
 
 

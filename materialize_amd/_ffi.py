@@ -77,6 +77,7 @@ def load():
                                     C.POINTER(Updates), C.c_uint32, C.c_int,
                                     C.POINTER(Closure),
                                     C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_halfjoin_raw.argtypes = lib.mz_gpu_halfjoin.argtypes
     lib.mz_gpu_reduce_create.restype = C.c_void_p
     lib.mz_gpu_reduce_create.argtypes = [C.c_void_p, C.POINTER(ReduceSpec)]
     lib.mz_gpu_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
@@ -227,8 +228,10 @@ class GpuCtx:
         return self._dev_out(outp)
 
     def halfjoin_dev(self, lookup, upd, stream_vb, le, cl):
+        # raw (unconsolidated) output: the render layer's consumers —
+        # the next probe stage and the reduce — consolidate themselves
         outp = C.POINTER(OutBatch)()
-        self._check(self.lib.mz_gpu_halfjoin(
+        self._check(self.lib.mz_gpu_halfjoin_raw(
             self.ctx, lookup, C.byref(upd), stream_vb, 1 if le else 0,
             C.byref(cl), C.byref(outp)))
         return self._dev_out(outp)

@@ -78,24 +78,71 @@ __global__ void k_gather_valrows(const u8 *src, u32 vb, const u32 *perm,
 
 // sort key for one key word: flip sign bit -> unsigned order == i64 order
 __global__ void k_sortkey_key(const u64 *keys, u32 kw, u32 word,
-                              const u32 *perm, u64 *out, u64 n) {
+                              const u32 *perm, u64 *out, u64 n, u64 sub) {
   GRID_STRIDE(i, n)
-  out[i] = keys[(u64)perm[i] * kw + word] ^ 0x8000000000000000ULL;
+  out[i] = (keys[(u64)perm[i] * kw + word] ^ 0x8000000000000000ULL) - sub;
 }
-// sort key for one 8-byte val group: big-endian load == byte-lex order
+// sort key for one 8-byte val group: zero-padded little-endian u64 word
+// (the engine's canonical val order — matches oracle cmp_val)
+__device__ __forceinline__ u64 le_val_word(const u8 *v, u32 rem) {
+  u64 x = 0;
+  if (rem >= 8) {
+    memcpy(&x, v, 8);
+  } else {
+    for (u32 b = 0; b < rem; b++) x |= (u64)v[b] << (8 * b);
+  }
+  return x;
+}
+
 __global__ void k_sortkey_val(const u8 *vals, u32 vb, u32 word,
-                              const u32 *perm, u64 *out, u64 n) {
+                              const u32 *perm, u64 *out, u64 n, u64 sub) {
   GRID_STRIDE(i, n) {
     const u8 *v = vals + (u64)perm[i] * vb + word * 8;
-    u32 rem = vb - word * 8;
-    u64 x = 0;
-    for (u32 b = 0; b < 8 && b < rem; b++) x |= (u64)v[b] << (56 - 8 * b);
-    out[i] = x;
+    out[i] = le_val_word(v, vb - word * 8) - sub;
+  }
+}
+
+// per-pass min/max of all sort-key columns in one sweep (block LDS reduce
+// + one atomic per block per pass) — lets the radix passes subtract the
+// min and sort only the live bits.
+#define MAX_PASSES 12
+__global__ void k_pass_minmax(const u64 *keys, u32 kw, const u8 *vals,
+                              u32 vb, const u64 *times, u64 n, u32 vwords,
+                              int with_time, u64 *mins, u64 *maxs) {
+  __shared__ u64 smin[MAX_PASSES], smax[MAX_PASSES];
+  u32 np = (with_time ? 1 : 0) + vwords + kw;
+  if (threadIdx.x < np) {
+    smin[threadIdx.x] = ~0ull;
+    smax[threadIdx.x] = 0;
+  }
+  __syncthreads();
+  GRID_STRIDE(i, n) {
+    u32 p = 0;
+    if (with_time) {
+      atomicMin((unsigned long long *)&smin[p], times[i]);
+      atomicMax((unsigned long long *)&smax[p], times[i]);
+      p++;
+    }
+    for (u32 w = 0; w < vwords; w++, p++) {
+      u64 x = le_val_word(vals + i * vb + w * 8, vb - w * 8);
+      atomicMin((unsigned long long *)&smin[p], x);
+      atomicMax((unsigned long long *)&smax[p], x);
+    }
+    for (u32 w = 0; w < kw; w++, p++) {
+      u64 x = keys[i * kw + w] ^ 0x8000000000000000ULL;
+      atomicMin((unsigned long long *)&smin[p], x);
+      atomicMax((unsigned long long *)&smax[p], x);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < np) {
+    atomicMin((unsigned long long *)&mins[threadIdx.x], smin[threadIdx.x]);
+    atomicMax((unsigned long long *)&maxs[threadIdx.x], smax[threadIdx.x]);
   }
 }
 __global__ void k_sortkey_u64(const u64 *src, const u32 *perm, u64 *out,
-                              u64 n) {
-  GRID_STRIDE(i, n) out[i] = src[perm[i]];
+                              u64 n, u64 sub) {
+  GRID_STRIDE(i, n) out[i] = src[perm[i]] - sub;
 }
 
 // head flags over the permuted (key,val,time) order
@@ -803,48 +850,79 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
   size_t tmp_bytes = 0;
   hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, c->stream, perm,
                      n);
-  // determine passes (LSD): least-significant first
+  // Column min/max in one sweep: constant columns need no pass at all and
+  // the rest subtract the min and radix-sort only the live bits (end_bit)
+  // -- most benchmark columns are narrow (dates, keys, small decimals).
+  u32 vwords = time_major ? 0 : (vb + 7) / 8;
+  u64 *dminmax = (u64 *)S.get(2 * MAX_PASSES * 8);
+  u64 *dmin = dminmax, *dmax = dminmax + MAX_PASSES;
+  HIP_CHECK(hipMemsetAsync(dmin, 0xFF, MAX_PASSES * 8, c->stream));
+  HIP_CHECK(hipMemsetAsync(dmax, 0, MAX_PASSES * 8, c->stream));
+  if (n)
+    hipLaunchKernelGGL(k_pass_minmax, dim3(ngrid(n)), dim3(BLK), 0,
+                       c->stream, keys, kw, vals, vb, times, n, vwords, 1,
+                       dmin, dmax);
+  u64 hmin[MAX_PASSES], hmax[MAX_PASSES];
+  HIP_CHECK(hipMemcpyAsync(hmin, dmin, MAX_PASSES * 8,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipMemcpyAsync(hmax, dmax, MAX_PASSES * 8,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  // pass slot layout from k_pass_minmax: [time][val words][key words]
   struct Pass {
     int kind;  // 0 = time, 1 = val word, 2 = key word
     u32 word;
+    u64 sub;
+    int bits;
   };
-  std::vector<Pass> passes;
+  auto slot_of = [&](int kind, u32 word) -> u32 {
+    if (kind == 0) return 0;
+    if (kind == 1) return 1 + word;
+    return 1 + vwords + word;
+  };
+  // LSD order: least-significant column first
+  std::vector<std::pair<int, u32>> order;
   if (time_major) {
-    // (time, key): key words LSD..MSD, then time last
-    for (int w = (int)kw - 1; w >= 0; w--) passes.push_back({2, (u32)w});
-    passes.push_back({0, 0});
+    for (int w = (int)kw - 1; w >= 0; w--) order.push_back({2, (u32)w});
+    order.push_back({0, 0});
   } else {
-    // (key, val, time): time first, then val words MSD->... LSD order:
-    // least significant = time, then val last word .. first, then key
-    passes.push_back({0, 0});
-    u32 vwords = (vb + 7) / 8;
-    for (int w = (int)vwords - 1; w >= 0; w--) passes.push_back({1, (u32)w});
-    for (int w = (int)kw - 1; w >= 0; w--) passes.push_back({2, (u32)w});
+    order.push_back({0, 0});
+    for (int w = (int)vwords - 1; w >= 0; w--) order.push_back({1, (u32)w});
+    for (int w = (int)kw - 1; w >= 0; w--) order.push_back({2, (u32)w});
+  }
+  std::vector<Pass> passes;
+  for (auto &[kind, word] : order) {
+    u32 s = slot_of(kind, word);
+    if (n == 0 || hmax[s] <= hmin[s]) continue;  // constant column
+    u64 range = hmax[s] - hmin[s];
+    passes.push_back({kind, word, hmin[s],
+                      64 - (int)__builtin_clzll(range)});
   }
   for (auto &p : passes) {
     if (p.kind == 0)
       hipLaunchKernelGGL(k_sortkey_u64, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, times, perm, skey, n);
+                         c->stream, times, perm, skey, n, p.sub);
     else if (p.kind == 1)
       hipLaunchKernelGGL(k_sortkey_val, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, vals, vb, p.word, perm, skey, n);
+                         c->stream, vals, vb, p.word, perm, skey, n, p.sub);
     else
       hipLaunchKernelGGL(k_sortkey_key, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, keys, kw, p.word, perm, skey, n);
+                         c->stream, keys, kw, p.word, perm, skey, n, p.sub);
     size_t need = 0;
-    (void)rocprim::radix_sort_pairs(nullptr, need, skey, skey_out, perm, perm_out,
-                              (unsigned)n, 0, 64, c->stream);
+    (void)rocprim::radix_sort_pairs(nullptr, need, skey, skey_out, perm,
+                                    perm_out, (unsigned)n, 0, p.bits,
+                                    c->stream);
     if (need > tmp_bytes) {
       tmp = S.get(need);
       tmp_bytes = need;
     }
-    (void)rocprim::radix_sort_pairs(tmp, tmp_bytes, skey, skey_out, perm, perm_out,
-                              (unsigned)n, 0, 64, c->stream);
+    (void)rocprim::radix_sort_pairs(tmp, tmp_bytes, skey, skey_out, perm,
+                                    perm_out, (unsigned)n, 0, p.bits,
+                                    c->stream);
     std::swap(perm, perm_out);
   }
-  // ensure result is in `perm` (the caller's buffer): passes count parity
+  // ensure the result lands in the caller's buffer
   if (passes.size() % 2 == 1) {
-    // result currently in the buffer that started as perm_out
     HIP_CHECK(hipMemcpyAsync(perm_out, perm, n * 4, hipMemcpyDeviceToDevice,
                              c->stream));
     std::swap(perm, perm_out);
@@ -1198,13 +1276,20 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                   u->upper);
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
-  // geometric maintenance: keep batch count bounded (amortized merging —
-  // scheduling policy per DESIGN.md §2.4; semantics = Spine merges)
-  while (a->batches.size() > 8) {
-    // merge the two smallest adjacent tail batches
+  // Geometric spine maintenance (amortized merging — scheduling policy per
+  // DESIGN.md §2.4; semantics = DD Spine merges with logical compaction):
+  // keep batch sizes decreasing by >=2x tail-to-head; merging the tail
+  // whenever the invariant breaks costs O(log) amortized merge work per
+  // update and keeps the probe fan-out at ~log(arrangement/batch).
+  while (a->batches.size() >= 2) {
     size_t nb = a->batches.size();
-    merge_range(ctx, a, nb - 2, nb);
+    if (a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds)
+      merge_range(ctx, a, nb - 2, nb);
+    else
+      break;
   }
+  while (a->batches.size() > 10)  // hard cap (probe BatchList capacity)
+    merge_range(ctx, a, 0, a->batches.size());
   return 0;
 }
 
@@ -1292,10 +1377,13 @@ mz_gpu_join *mz_gpu_join_create(mz_gpu_ctx *c, mz_gpu_arr *a1, mz_gpu_arr *a2,
 
 void mz_gpu_join_drop(mz_gpu_ctx *c, mz_gpu_join *j) { (void)c; (void)j; }
 
-// shared probe path for linear join and half join
+// shared probe path for linear join and half join; consolidate_out=0
+// skips the output consolidation (legal when the consumer consolidates —
+// the reduce does — and an engine-internal optimization)
 static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                       u32 stream_vb, int mode, int swap,
-                      const mz_gpu_closure *cl, mz_gpu_out **out) {
+                      const mz_gpu_closure *cl, int consolidate_out,
+                      mz_gpu_out **out) {
   ctx->scratch.reset();
   auto &S = ctx->scratch;
   u32 kw = lookup->schema.kw, lvb = lookup->schema.vb;
@@ -1308,7 +1396,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     if (bl.n >= 12) {
       // too many batches: merge first (bounded by push-time policy)
       merge_range(ctx, lookup, 0, lookup->batches.size());
-      return probe_impl(ctx, lookup, u, stream_vb, mode, swap, cl, out);
+      return probe_impl(ctx, lookup, u, stream_vb, mode, swap, cl,
+                        consolidate_out, out);
     }
     bl.b[bl.n++] = b;
   }
@@ -1352,6 +1441,11 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
         n * (8ull * kw + stream_vb + 16) + n * 128ull * (u64)bl.n +
         M * (lvb + 16ull) + M * (8ull * okw + ovb + 16);
   }
+  if (!consolidate_out) {
+    // raw emitted pairs (deterministic order; consumer consolidates)
+    *out = make_out(pk, pv, pt, pd, M, okw, ovb);
+    return 0;
+  }
   // consolidate the emitted pairs
   DevUpdates pin{pk, pv, pt, pd, M};
   u64 *ok;
@@ -1374,14 +1468,23 @@ int mz_gpu_join_push(mz_gpu_ctx *c, mz_gpu_join *op, int side,
   // swap=1 when the delta is input 2 (closure args are (v1, v2) by input
   // number — mz_join_core.rs:69)
   return probe_impl(&c->impl, opp, delta, own->schema.vb, PM_JOIN,
-                    side == 2 ? 1 : 0, &op->cl, out);
+                    side == 2 ? 1 : 0, &op->cl, 1, out);
 }
 
 int mz_gpu_halfjoin(mz_gpu_ctx *c, mz_gpu_arr *lookup,
                     const mz_gpu_updates *delta, uint32_t stream_val_bytes,
                     int le, const mz_gpu_closure *cl, mz_gpu_out **out) {
   return probe_impl(&c->impl, lookup, delta, stream_val_bytes,
-                    le ? PM_HALF_LE : PM_HALF_LT, 0, cl, out);
+                    le ? PM_HALF_LE : PM_HALF_LT, 0, cl, 1, out);
+}
+
+// Raw variant: output left unconsolidated (consumer consolidates).
+int mz_gpu_halfjoin_raw(mz_gpu_ctx *c, mz_gpu_arr *lookup,
+                        const mz_gpu_updates *delta,
+                        uint32_t stream_val_bytes, int le,
+                        const mz_gpu_closure *cl, mz_gpu_out **out) {
+  return probe_impl(&c->impl, lookup, delta, stream_val_bytes,
+                    le ? PM_HALF_LE : PM_HALF_LT, 0, cl, 0, out);
 }
 
 mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
@@ -1437,17 +1540,25 @@ int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
                      d.times, perm, stm, n);
   hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
                      d.diffs, perm, sd, n);
-  // time slice boundaries (host; few distinct times per batch)
-  std::vector<u64> htimes(n);
-  HIP_CHECK(hipMemcpyAsync(htimes.data(), stm, n * 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  // time slice boundaries (host; few distinct times per batch).
+  // Single-timestamp batches (upper == lower+1, the steady-state churn
+  // shape) skip the boundary copy entirely.
+  std::vector<u64> htimes;
   std::vector<std::pair<u64, u64>> slices;  // (start, end)
-  for (u64 i = 0; i < n;) {
-    u64 j = i;
-    while (j < n && htimes[j] == htimes[i]) j++;
-    slices.push_back({i, j});
-    i = j;
+  if (u->upper <= u->lower + 1) {
+    htimes.assign(1, u->lower);
+    slices.push_back({0, n});
+  } else {
+    htimes.resize(n);
+    HIP_CHECK(hipMemcpyAsync(htimes.data(), stm, n * 8,
+                             hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    for (u64 i = 0; i < n;) {
+      u64 j = i;
+      while (j < n && htimes[j] == htimes[i]) j++;
+      slices.push_back({i, j});
+      i = j;
+    }
   }
   // output buffer: capacity 2 * n corrections max (each input row can
   // change at most one key per slice; 2 rows per changed key per slice)
@@ -1492,7 +1603,8 @@ int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
                          misspos, op->n_rows, op->st);
     hipLaunchKernelGGL(k_reduce_apply, dim3(ngrid(G)), dim3(BLK), 0,
                        ctx->stream, sk + lo * kw, sv + lo * vb, kw, vb,
-                       sd + lo, starts, (u64)G, m, htimes[lo], op->st, found,
+                       sd + lo, starts, (u64)G, m,
+                       htimes[htimes.size() == 1 ? 0 : lo], op->st, found,
                        miss, misspos, op->n_rows, op->spec, pk, pv, pt, pd,
                        ocount);
     op->n_rows += Mn;

@@ -81,8 +81,10 @@ struct Cols {
   }
 };
 
-// Key order: i64-tuple ascending. Val order: lexicographic
-// bytes (Row byte-lex order analog, row-spine batch ordering).
+// Key order: i64-tuple ascending. Val order: the val bytes
+// as a tuple of zero-padded little-endian u64 words, unsigned ascending —
+// the engine's canonical order (an arbitrary-but-fixed total order in the
+// role of the reference's Row byte-lex order; identical on oracle and GPU).
 inline int cmp_key(const u64 *a, const u64 *b, u32 kw) {
   for (u32 i = 0; i < kw; i++) {
     i64 x = (i64)a[i], y = (i64)b[i];
@@ -92,7 +94,15 @@ inline int cmp_key(const u64 *a, const u64 *b, u32 kw) {
   return 0;
 }
 inline int cmp_val(const u8 *a, const u8 *b, u32 vb) {
-  return vb ? std::memcmp(a, b, vb) : 0;
+  for (u32 off = 0; off < vb; off += 8) {
+    u64 x = 0, y = 0;
+    u32 m = vb - off < 8 ? vb - off : 8;
+    std::memcpy(&x, a + off, m);
+    std::memcpy(&y, b + off, m);
+    if (x < y) return -1;
+    if (x > y) return 1;
+  }
+  return 0;
 }
 
 // consolidate_updates semantics: sort by (key, val, time), sum diffs

@@ -285,8 +285,12 @@ class TestFloatSum:
         res = ctx.reduce_push(op, abi.make_updates(
             np.zeros(1, np.int64), xs[:1].view(np.uint8),
             np.zeros(1, np.uint64), -np.ones(1, np.int64), 1, 2))
-        # last correction row is the insertion of the new value
-        v = np.frombuffer(res[1][-16:-8].tobytes(), np.float64)[0]
+        # find the insertion (+1) correction row
+        keys, vals, times, diffs = res
+        vals = vals.reshape(len(times), 24)
+        ins = [i for i in range(len(times)) if diffs[i] == 1]
+        assert len(ins) == 1
+        v = np.frombuffer(vals[ins[0]][8:16].tobytes(), np.float64)[0]
         assert v == 1.5
 
 
