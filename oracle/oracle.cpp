@@ -191,13 +191,14 @@ struct Arr {
   u64 logical_compaction = 0;  // times advance to this on merge
   u64 upper = 0;               // acknowledged frontier = max pushed upper
 
-  // Merge all batches into one, advancing times to the logical compaction
-  // frontier (add/retract pairs then cancel) — Spine merge + logical
-  // compaction semantics (mz_join_core.rs:458-465 contract; manager.rs:54).
-  void merge_all() {
+  // Merge batches [from, to) into one, advancing times to the logical
+  // compaction frontier (add/retract pairs then cancel) — Spine merge +
+  // logical compaction semantics (mz_join_core.rs:458-465; manager.rs:54).
+  void merge_span(size_t from, size_t to) {
     Cols all;
     u64 lo = UINT64_MAX, hi = 0;
-    for (auto &b : batches) {
+    for (size_t bi = from; bi < to; bi++) {
+      auto &b = batches[bi];
       lo = std::min(lo, b->lower);
       hi = std::max(hi, b->upper);
       size_t n = b->cols.size();
@@ -209,14 +210,15 @@ struct Arr {
       }
     }
     consolidate(schema, all);
-    batches.clear();
     auto nb = std::make_unique<Batch>();
     nb->cols = std::move(all);
     nb->lower = lo == UINT64_MAX ? 0 : lo;
     nb->upper = hi;
     nb->build_index(schema);
-    batches.push_back(std::move(nb));
+    batches.erase(batches.begin() + from, batches.begin() + to);
+    batches.insert(batches.begin() + from, std::move(nb));
   }
+  void merge_all() { merge_span(0, batches.size()); }
 };
 
 // ------------------------------------------------------------- closures
@@ -679,6 +681,15 @@ int orc_arr_push_batch(orc_ctx *c, Arr *a, const mz_gpu_updates *u) {
   b->build_index(a->schema);
   a->batches.push_back(std::move(b));
   a->upper = std::max(a->upper, u->upper);
+  // geometric tail merging (same amortized policy as the GPU engine):
+  // keeps probe fan-out logarithmic in steady state
+  while (a->batches.size() >= 2) {
+    size_t nb = a->batches.size();
+    if (a->batches[nb - 2]->cols.size() <= 2 * a->batches[nb - 1]->cols.size())
+      a->merge_span(nb - 2, nb);
+    else
+      break;
+  }
   return 0;
 }
 
