@@ -43,7 +43,36 @@ def parse_args():
     p.add_argument("--batch-rows", type=int, default=0, help="0 = auto")
     p.add_argument("--no-cpu-baseline", action="store_true")
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--call-profile", action="store_true",
+                   help="print per-engine-call wall time breakdown")
     return p.parse_args()
+
+
+class CallProfiler:
+    """Wraps an engine ctx's methods with wall-time accumulation."""
+
+    METHODS = ["consolidate_dev", "arr_push", "halfjoin_dev",
+               "reduce_push_dev", "arr_maintain"]
+
+    def __init__(self, ctx):
+        self.t = {m: [0.0, 0] for m in self.METHODS}
+        for m in self.METHODS:
+            orig = getattr(ctx, m)
+
+            def wrap(orig=orig, m=m):
+                def f(*a, **kw):
+                    t0 = time.perf_counter()
+                    r = orig(*a, **kw)
+                    self.t[m][0] += time.perf_counter() - t0
+                    self.t[m][1] += 1
+                    return r
+                return f
+            setattr(ctx, m, wrap())
+
+    def report(self, steps):
+        for m, (tt, calls) in sorted(self.t.items(), key=lambda x: -x[1][0]):
+            print(f"#   {m:18s} {tt*1e3:9.2f} ms total  {calls:5d} calls  "
+                  f"{tt*1e3/max(steps,1):7.2f} ms/step", file=sys.stderr)
 
 
 def stage_churn(churn, t, device):
@@ -178,6 +207,7 @@ def main():
     if dist:
         dist.barrier()
 
+    prof = CallProfiler(ctx) if args.call_profile else None
     ctx.set_kernel_timing(1)
     t0 = time.perf_counter()
     for i in range(W, W + K):
@@ -190,6 +220,10 @@ def main():
         dist.barrier()
     elapsed = time.perf_counter() - t0
     ctx.set_kernel_timing(0)
+    if prof:
+        print(f"# step wall {elapsed/K*1e3:.2f} ms; breakdown:",
+              file=sys.stderr)
+        prof.report(K)
 
     if dist:
         e = torch.tensor([elapsed], device=device)

@@ -74,15 +74,25 @@ enum {
    * offsets are given by arg0/arg1 on the out field. */
   MZ_COMPUTE_REVENUE = 0,
   /* 8 zero bytes (re-key to a constant, e.g. cross-join stages). */
-  MZ_COMPUTE_CONST0 = 1
+  MZ_COMPUTE_CONST0 = 1,
+  /* Q17's correlated-average filter, exact in integers:
+   * quantity < 0.2 * sum/count  <=>  5*q*count < sum  (count > 0).
+   * As a FILTER compute: arg0 = i64 quantity offset; arg1 = the offset of
+   * a SUM_I64 aggregate slot's i128 value, with the COUNT slot's i64 at
+   * arg1+24 (the 24-byte aggregate slot layout). NULL count (count==0 in
+   * the reference's CASE, tpch_create_index.slt:1470) fails the filter. */
+  MZ_COMPUTE_Q17_QTYLT = 2
 };
 
 typedef struct {
-  uint8_t  src;     /* MZ_SRC_* (filters: KEY/VAL_STREAM/VAL_LOOKUP)   */
-  uint16_t off;     /* byte offset into src                            */
+  uint8_t  src;     /* MZ_SRC_* (KEY/VAL_STREAM/VAL_LOOKUP), or
+                       MZ_SRC_COMPUTE with `off` = compute filter id    */
+  uint16_t off;     /* byte offset into src (or compute id)            */
   uint8_t  width;   /* 4 or 8 (signed little-endian integer)           */
   uint8_t  cmp;     /* MZ_CMP_*                                        */
   int64_t  imm;     /* literal operand                                 */
+  uint16_t arg0, arg1;          /* compute-filter operand offsets      */
+  uint8_t  arg0_src, arg1_src;
 } mz_gpu_filter;
 
 typedef struct {

@@ -17,6 +17,7 @@ MZ_SRC_COMPUTE = 3
 MZ_CMP_LT, MZ_CMP_LE, MZ_CMP_GT, MZ_CMP_GE, MZ_CMP_EQ, MZ_CMP_NE = range(6)
 MZ_COMPUTE_REVENUE = 0
 MZ_COMPUTE_CONST0 = 1
+MZ_COMPUTE_Q17_QTYLT = 2
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 
 MZ_GPU_MAX_FILTERS = 4
@@ -51,6 +52,10 @@ class Filter(C.Structure):
         ("width", C.c_uint8),
         ("cmp", C.c_uint8),
         ("imm", C.c_int64),
+        ("arg0", C.c_uint16),
+        ("arg1", C.c_uint16),
+        ("arg0_src", C.c_uint8),
+        ("arg1_src", C.c_uint8),
     ]
 
 
@@ -187,8 +192,11 @@ def out_to_numpy(out, copy=True):
     return keys, vals, times, diffs
 
 
-def filt(src, off, width, cmp, imm):
-    return Filter(src=src, off=off, width=width, cmp=cmp, imm=imm)
+def filt(src, off, width, cmp, imm, arg0=0, arg1=0, arg0_src=0,
+         arg1_src=0):
+    return Filter(src=src, off=off, width=width, cmp=cmp, imm=imm,
+                  arg0=arg0, arg1=arg1, arg0_src=arg0_src,
+                  arg1_src=arg1_src)
 
 
 def field(src, off, width=8, arg0=0, arg1=0, arg0_src=0, arg1_src=0):

@@ -313,6 +313,18 @@ __device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
                                 u8 *oval) {
   for (u32 i = 0; i < cl->n_filters; i++) {
     const mz_gpu_filter f = cl->filters[i];
+    if (f.src == MZ_SRC_COMPUTE) {
+      if (f.off == MZ_COMPUTE_Q17_QTYLT) {
+        // 5*q*count < sum, count > 0 (Q17 correlated average; see header)
+        i64 q = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+        const u8 *slot = d_cl_src(key, v1, v2, f.arg1_src) + f.arg1;
+        i128 S;
+        memcpy(&S, slot, 16);
+        i64 C = d_read_int(slot + 24, 8);
+        if (!(C > 0 && (i128)5 * q * C < S)) return false;
+      }
+      continue;
+    }
     i64 x = d_read_int(d_cl_src(key, v1, v2, f.src) + f.off, f.width);
     bool ok;
     switch (f.cmp) {

@@ -251,6 +251,20 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
                    const u8 *v2, u64 *out_key, u8 *out_val) {
   for (u32 i = 0; i < cl->n_filters; i++) {
     const auto &f = cl->filters[i];
+    if (f.src == MZ_SRC_COMPUTE) {
+      if (f.off == MZ_COMPUTE_Q17_QTYLT) {
+        // quantity < 0.2*sum/count <=> 5*q*count < sum, count > 0
+        // (reference: reduce output CASE WHEN count=0 THEN NULL;
+        //  tpch_create_index.slt:1470)
+        i64 q = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+        const u8 *slot = cl_src(key, v1, v2, f.arg1_src) + f.arg1;
+        i128 S;
+        std::memcpy(&S, slot, 16);
+        i64 C = read_int(slot + 24, 8);
+        if (!(C > 0 && (i128)5 * q * C < S)) return false;
+      }
+      continue;
+    }
     i64 x = read_int(cl_src(key, v1, v2, f.src) + f.off, f.width);
     bool ok;
     switch (f.cmp) {
