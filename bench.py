@@ -81,10 +81,13 @@ def stage_churn(churn, t, device):
     import torch
     from materialize_amd import _abi as abi
     out = {}
+    name_map = {"orders_by_cust": "orders_by_custkey",
+                "orders": "orders_by_orderkey",
+                "lineitem": "lineitem"}
     for name, (keys, vals, diffs) in churn.items():
-        tgt = {"orders_by_cust": "orders_by_custkey",
-               "orders": "orders_by_orderkey",
-               "lineitem": "lineitem"}[name]
+        if name not in name_map:  # relations outside this workload
+            continue
+        tgt = name_map[name]
         n = len(keys)
         kt = torch.from_numpy(np.ascontiguousarray(keys, np.int64)
                               ).to(device)

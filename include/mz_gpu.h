@@ -166,6 +166,10 @@ void        mz_gpu_arr_drop(mz_gpu_ctx *ctx, mz_gpu_arr *arr);
 /* Push a sealed, sorted, consolidated batch (see mz_gpu_updates docs). */
 int  mz_gpu_arr_push_batch(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
                            const mz_gpu_updates *batch);
+/* Consolidate raw updates and push the sealed batch in one call (the
+ * MergeBatcher + arrange step fused — no intermediate out-batch). */
+int  mz_gpu_arr_insert(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                       const mz_gpu_updates *raw);
 /* Advance the logical compaction frontier (times advance to it on merge) —
  * cf. set_logical_compaction, mz_join_core.rs:461. */
 int  mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *ctx, mz_gpu_arr *arr,

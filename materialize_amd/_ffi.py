@@ -55,6 +55,8 @@ def load():
     lib.mz_gpu_arr_drop.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_arr_push_batch.argtypes = [C.c_void_p, C.c_void_p,
                                           C.POINTER(Updates)]
+    lib.mz_gpu_arr_insert.argtypes = [C.c_void_p, C.c_void_p,
+                                      C.POINTER(Updates)]
     lib.mz_gpu_arr_set_logical_compaction.argtypes = [C.c_void_p, C.c_void_p,
                                                       C.c_uint64]
     lib.mz_gpu_arr_maintain.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
@@ -147,6 +149,11 @@ class GpuCtx:
     def arr_push(self, arr, upd):
         self._check(self.lib.mz_gpu_arr_push_batch(self.ctx, arr,
                                                    C.byref(upd)))
+
+    def arr_insert(self, arr, upd):
+        """Consolidate raw updates + push, fused (one call)."""
+        self._check(self.lib.mz_gpu_arr_insert(self.ctx, arr,
+                                               C.byref(upd)))
 
     def arr_set_logical_compaction(self, arr, frontier):
         self.lib.mz_gpu_arr_set_logical_compaction(self.ctx, arr, frontier)

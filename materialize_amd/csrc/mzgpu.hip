@@ -1268,6 +1268,23 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
   a->batches.clear();
 }
 
+// Geometric spine maintenance (amortized merging — scheduling policy per
+// DESIGN.md §2.4; semantics = DD Spine merges with logical compaction):
+// keep batch sizes decreasing by >=2x tail-to-head; merging the tail
+// whenever the invariant breaks costs O(log) amortized merge work per
+// update and keeps the probe fan-out at ~log(arrangement/batch).
+static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
+  while (a->batches.size() >= 2) {
+    size_t nb = a->batches.size();
+    if (a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds)
+      merge_range(ctx, a, nb - 2, nb);
+    else
+      break;
+  }
+  while (a->batches.size() > 10)  // hard cap (probe BatchList capacity)
+    merge_range(ctx, a, 0, a->batches.size());
+}
+
 int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                           const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
@@ -1295,20 +1312,29 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                   u->upper);
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
-  // Geometric spine maintenance (amortized merging — scheduling policy per
-  // DESIGN.md §2.4; semantics = DD Spine merges with logical compaction):
-  // keep batch sizes decreasing by >=2x tail-to-head; merging the tail
-  // whenever the invariant breaks costs O(log) amortized merge work per
-  // update and keeps the probe fan-out at ~log(arrangement/batch).
-  while (a->batches.size() >= 2) {
-    size_t nb = a->batches.size();
-    if (a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds)
-      merge_range(ctx, a, nb - 2, nb);
-    else
-      break;
-  }
-  while (a->batches.size() > 10)  // hard cap (probe BatchList capacity)
-    merge_range(ctx, a, 0, a->batches.size());
+  spine_policy(ctx, a);
+  return 0;
+}
+
+// Consolidate raw updates + build + push, in one call (no intermediate
+// out-batch, copies, or extra syncs).
+int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
+                      const mz_gpu_updates *u) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  u32 kw = a->schema.kw, vb = a->schema.vb;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 M;
+  consolidate_dev(ctx, kw, vb, d, &ok, &ov, &ot, &od, &M);
+  DevBatch b =
+      build_batch(ctx, kw, vb, ok, ov, ot, od, M, u->lower, u->upper);
+  a->batches.push_back(b);
+  a->upper = std::max(a->upper, u->upper);
+  spine_policy(ctx, a);
   return 0;
 }
 

@@ -49,7 +49,8 @@ class TpchGen:
 
     def _lineitem_cols(self, order_idx, rng):
         """Generate lineitem columns for the given order indices (one row
-        per entry of order_idx)."""
+        per entry of order_idx): extprice, discount, shipdate, partkey,
+        quantity."""
         n = len(order_idx)
         quantity = rng.integers(1, 51, n).astype(np.int64)
         retail_cents = rng.integers(90_000, 200_001, n).astype(np.int64)
@@ -57,14 +58,26 @@ class TpchGen:
         discount_bp = rng.integers(0, 9, n).astype(np.int64) * 100
         shipdate = (self.o_orderdate[order_idx].astype(np.int64) +
                     rng.integers(1, 122, n)).astype(np.int32)
-        return extprice, discount_bp, shipdate
+        partkey = rng.integers(1, max(self.n_part, 2), n).astype(np.int64)
+        return extprice, discount_bp, shipdate, partkey, quantity
+
+    def _gen_part(self):
+        """part table (Q17): brand 0..24 (Brand#23 = 23), container 0..39
+        (MED BOX = 17) — 1/1000 selectivity like the reference's
+        predicates."""
+        rng = self.rng
+        self.p_partkey = np.arange(1, self.n_part + 1, dtype=np.int64)
+        self.p_brand = rng.integers(0, 25, self.n_part).astype(np.int64)
+        self.p_container = rng.integers(0, 40, self.n_part).astype(np.int64)
 
     def _gen_lineitems(self):
+        self._gen_part()
         order_idx = np.repeat(np.arange(self.n_orders), self.l_count)
         self.l_order_idx = order_idx
         self.l_orderkey = self.o_orderkey[order_idx]
-        (self.l_extendedprice, self.l_discount,
-         self.l_shipdate) = self._lineitem_cols(order_idx, self.rng)
+        (self.l_extendedprice, self.l_discount, self.l_shipdate,
+         self.l_partkey, self.l_quantity) = \
+            self._lineitem_cols(order_idx, self.rng)
         # per-order start offsets into lineitem arrays
         self.l_offs = np.zeros(self.n_orders + 1, np.int64)
         np.cumsum(self.l_count, out=self.l_offs[1:])
@@ -96,8 +109,8 @@ class TpchGen:
 
     @staticmethod
     def lineitem_vals(extprice, discount, shipdate):
-        """lineitem val: [extprice i64][discount i64][shipdate i32][pad]
-        = 24B."""
+        """lineitem val (Q3, by orderkey): [extprice i64][discount i64]
+        [shipdate i32][pad] = 24B."""
         n = len(extprice)
         v = np.zeros((n, 24), np.uint8)
         v[:, 0:8] = extprice.view(np.uint8).reshape(n, 8)
@@ -108,6 +121,27 @@ class TpchGen:
     def lineitem_updates(self):
         return self.l_orderkey, self.lineitem_vals(
             self.l_extendedprice, self.l_discount, self.l_shipdate)
+
+    @staticmethod
+    def lineitem_bypart_vals(quantity, extprice):
+        """lineitem val (Q17, by partkey): [quantity i64][extprice i64]."""
+        n = len(quantity)
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = quantity.view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = extprice.view(np.uint8).reshape(n, 8)
+        return v
+
+    def lineitem_bypart_updates(self):
+        return self.l_partkey, self.lineitem_bypart_vals(
+            self.l_quantity, self.l_extendedprice)
+
+    def part_updates(self):
+        """part val: [brand i64][container i64]."""
+        n = self.n_part
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = self.p_brand.view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = self.p_container.view(np.uint8).reshape(n, 8)
+        return self.p_partkey, v
 
     # ------------------------------------------------------------ churn
 
@@ -132,6 +166,9 @@ class TpchGen:
         r_l_vals = self.lineitem_vals(self.l_extendedprice[pos].copy(),
                                       self.l_discount[pos].copy(),
                                       self.l_shipdate[pos].copy())
+        r_lp_keys = self.l_partkey[pos].copy()
+        r_lp_vals = self.lineitem_bypart_vals(
+            self.l_quantity[pos].copy(), self.l_extendedprice[pos].copy())
         o_retract_vals = self.orders_vals(idx)
         o_retract_bycust_vals = self.orders_bycust_vals(idx)
         o_retract_bycust_keys = self.o_custkey[idx].copy()
@@ -145,10 +182,12 @@ class TpchGen:
         # the flat arrays stable; the reference redraws 1..7, a shape
         # detail that does not change the maintained row rate)
         order_idx_rep = np.repeat(idx, counts)
-        ep, disc, sd = self._lineitem_cols(order_idx_rep, rng)
+        ep, disc, sd, pk, qty = self._lineitem_cols(order_idx_rep, rng)
         self.l_extendedprice[pos] = ep
         self.l_discount[pos] = disc
         self.l_shipdate[pos] = sd
+        self.l_partkey[pos] = pk
+        self.l_quantity[pos] = qty
         n_keys = self.l_orderkey[pos].copy()
         n_vals = self.lineitem_vals(ep, disc, sd)
         l_keys = np.concatenate([r_l_keys, n_keys])
@@ -156,6 +195,10 @@ class TpchGen:
         nr = len(r_l_keys)
         l_diffs = np.concatenate([-np.ones(nr, np.int64),
                                   np.ones(len(l_keys) - nr, np.int64)])
+        lp_keys = np.concatenate([r_lp_keys, pk])
+        lp_vals = np.concatenate([r_lp_vals,
+                                  self.lineitem_bypart_vals(qty, ep)])
+        lp_diffs = l_diffs.copy()
         o_keys = np.concatenate([self.o_orderkey[idx], self.o_orderkey[idx]])
         o_vals = np.concatenate([o_retract_vals, self.orders_vals(idx)])
         o_diffs = np.concatenate([-np.ones(k, np.int64),
@@ -165,6 +208,7 @@ class TpchGen:
                                   self.orders_bycust_vals(idx)])
         return {
             "lineitem": (l_keys, l_vals, l_diffs),
+            "lineitem_by_part": (lp_keys, lp_vals, lp_diffs),
             "orders": (o_keys, o_vals, o_diffs),
             "orders_by_cust": (oc_keys, oc_vals, o_diffs.copy()),
         }

@@ -126,12 +126,9 @@ class Q3Dataflow:
 
     def _seal_push(self, name, keys, vals, diffs, t):
         ctx = self.ctx
-        kw, vb = self.SCHEMAS[name]
         u = abi.make_updates(keys, vals, np.full(len(keys), t, np.uint64),
                              diffs, t, t + 1)
-        sealed = ctx.consolidate_dev(abi.schema(kw, vb), u)
-        ctx.arr_push(self.arrs[name], sealed.updates(t, t + 1))
-        sealed.release()
+        ctx.arr_insert(self.arrs[name], u)
 
     def load(self, gen):
         """Load the base snapshot at t=0 and seed the reduce with the
@@ -210,11 +207,7 @@ class Q3Dataflow:
         DevOut (or None)."""
         ctx = self.ctx
         for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
-            u = upd[name]
-            kw, vb = self.SCHEMAS[name]
-            sealed = ctx.consolidate_dev(abi.schema(kw, vb), u)
-            ctx.arr_push(self.arrs[name], sealed.updates(t, t + 1))
-            sealed.release()
+            ctx.arr_insert(self.arrs[name], upd[name])
         outs = []
         for rel, src in (("orders", "orders_by_custkey"),
                          ("lineitem", "lineitem")):
@@ -345,3 +338,158 @@ class ShardedQ3Dataflow(Q3Dataflow):
         for out in outs:
             out.release()
         return rows, corr
+
+
+# ===================================================================== Q17
+
+def q17_closures():
+    """Closures of Q17's pinned plan (tpch_create_index.slt:1449-1505):
+    l1 = lineitem ⋈ part (brand/container filters); per-partkey
+    sum(quantity)/count over Distinct(l1.partkey) ⋈ lineitem; final
+    l1 ⋈ avg with quantity < 0.2*avg (exact integer form),
+    then a global SUM(extendedprice). Brand#23 = code 23, MED BOX = 17."""
+    # join1: lineitem(by partkey) ⋈ part — input1 = lineitem
+    #   lineitem val: [quantity i64][extprice i64]; part val: [brand][cont]
+    cl_j1 = abi.closure(
+        [FL(VL, 0, 8, abi.MZ_CMP_EQ, 23),        # p_brand = Brand#23
+         FL(VL, 8, 8, abi.MZ_CMP_EQ, 17)],       # p_container = MED BOX
+        [F(KEY, 0, 8)],                          # key := partkey
+        [F(VS, 0, 16)],                          # val := (qty, extprice)
+        abi.schema(1, 16))
+    # join2: distinct(partkey) ⋈ lineitem(by partkey) — input1 = distinct
+    cl_j2 = abi.closure(
+        [],
+        [F(KEY, 0, 8)],                          # key := partkey
+        [F(VL, 0, 8)],                           # val := quantity
+        abi.schema(1, 8))
+    # join3: l1 ⋈ avg-reduce output — input1 = l1 (qty, extprice),
+    #   input2 val = 48B reduce row: slot0 = SUM_I64(qty), slot1 = COUNT
+    cl_j3 = abi.closure(
+        [FL(CP, abi.MZ_COMPUTE_Q17_QTYLT, 8, abi.MZ_CMP_LT, 0,
+            arg0=0, arg1=8, arg0_src=VS, arg1_src=VL)],
+        [F(CP, abi.MZ_COMPUTE_CONST0, 8)],       # key := 0 (global sum)
+        [F(VS, 8, 8)],                           # val := extprice
+        abi.schema(1, 8))
+    return cl_j1, cl_j2, cl_j3
+
+
+class Q17Dataflow:
+    """TPC-H Q17 maintained incrementally (config 5 shape): two linear
+    joins, a distinct, a per-partkey SUM/COUNT reduce feeding an
+    arrangement, the correlated-average filter join, and a global SUM.
+    Exactly-once across the concurrent per-step deltas follows the
+    mz_join_core drain discipline: for each binary join,
+    one side's delta probes the other side BEFORE that side's delta is
+    pushed, and vice versa."""
+
+    def __init__(self, ctx):
+        self.ctx = ctx
+        self.arr_l0 = ctx.arr_create(abi.schema(1, 16))    # lineitem by pk
+        self.arr_part = ctx.arr_create(abi.schema(1, 16))
+        self.arr_l1 = ctx.arr_create(abi.schema(1, 16))    # filtered rows
+        self.arr_dist = ctx.arr_create(abi.schema(1, 0))   # distinct pk
+        self.arr_avg = ctx.arr_create(abi.schema(1, 48))   # sum/count rows
+        cl_j1, cl_j2, cl_j3 = q17_closures()
+        self.j1 = ctx.join_create(self.arr_l0, self.arr_part, cl_j1)
+        self.j2 = ctx.join_create(self.arr_dist, self.arr_l0, cl_j2)
+        self.j3 = ctx.join_create(self.arr_l1, self.arr_avg, cl_j3)
+        self.distinct = ctx.reduce_create(
+            abi.reduce_spec([], abi.schema(1, 0)))
+        self.avg = ctx.reduce_create(abi.reduce_spec(
+            [abi.Aggregate(func=abi.MZ_AGG_SUM_I64, off=0, width=8,
+                           is_float=0, nullable=0),
+             abi.Aggregate(func=abi.MZ_AGG_COUNT, off=0, width=8,
+                           is_float=0, nullable=0)],
+            abi.schema(1, 8)))
+        self.total = ctx.reduce_create(abi.reduce_spec(
+            [abi.Aggregate(func=abi.MZ_AGG_SUM_I64, off=0, width=8,
+                           is_float=0, nullable=0)],
+            abi.schema(1, 8)))
+        self.result = {}  # maintained {0: sum_extendedprice}
+
+    def _apply_total(self, corr):
+        keys, vals, times, diffs = corr
+        n = len(times)
+        vals = vals.reshape(n, 24) if n else vals
+        # apply retractions before insertions (consolidated output orders
+        # rows by val bytes, not by diff sign)
+        order = sorted(range(n), key=lambda i: (int(times[i]),
+                                                int(diffs[i])))
+        for i in order:
+            lo = int(vals[i][8:16].view(np.uint64)[0])
+            hi = int(vals[i][16:24].view(np.int64)[0])
+            v = hi * 2**64 + lo
+            if int(diffs[i]) == 1:
+                self.result[int(keys[i])] = v
+            else:
+                assert self.result.pop(int(keys[i])) == v
+
+    def avg_yearly(self):
+        """sum(l_extendedprice)/7.0 in cents (the reference's final Map)."""
+        s = self.result.get(0)
+        return None if s is None else s / 7.0
+
+    def _push(self, t, lp=None, part=None):
+        """One timestamp's worth of updates (lp = lineitem-by-partkey
+        columns, part = part columns); pass None for an idle input."""
+        ctx = self.ctx
+        empty = (np.empty(0, np.int64), np.empty((0, 16), np.uint8),
+                 np.empty(0, np.int64))
+        lp_k, lp_v, lp_d = lp if lp is not None else empty
+        p_k, p_v, p_d = part if part is not None else empty
+
+        def updates(k, v, d, vb):
+            return abi.make_updates(
+                np.ascontiguousarray(k, np.int64),
+                np.ascontiguousarray(v, np.uint8).reshape(-1) if vb else None,
+                np.full(len(k), t, np.uint64),
+                np.ascontiguousarray(d, np.int64), t, t + 1)
+
+        def seal(out_cols, vb):
+            k, v, tm, d = out_cols
+            return abi.make_updates(k, v, tm, d, t, t + 1)
+
+        lp_u = updates(lp_k, lp_v, lp_d, 16)
+        p_u = updates(p_k, p_v, p_d, 16)
+        # --- join1 drain: side1 (lineitem) first
+        ctx.arr_insert(self.arr_l0, lp_u)
+        l1d = ctx.join_push(self.j1, 1, lp_u)        # probes part (old)
+        ctx.arr_insert(self.arr_part, p_u)
+        l1d2 = ctx.join_push(self.j1, 2, p_u)        # probes l0 (new)
+        l1_cols = [np.concatenate([a, b]) for a, b in zip(l1d, l1d2)]
+        l1_u = seal(l1_cols, 16)
+        # --- join2 drain: side2 (lineitem) BEFORE the distinct delta
+        j2b = ctx.join_push(self.j2, 2, lp_u)        # probes dist (old)
+        # distinct of l1's partkeys
+        dk, dv, dt_, dd = l1_cols
+        dist_u = abi.make_updates(dk, None, dt_, dd, t, t + 1)
+        dcorr = ctx.reduce_push(self.distinct, dist_u)
+        dcorr_u = abi.make_updates(*dcorr, t, t + 1)
+        ctx.arr_insert(self.arr_dist, dcorr_u)
+        j2a = ctx.join_push(self.j2, 1, dcorr_u)     # probes l0 (new)
+        q_cols = [np.concatenate([a, b]) for a, b in zip(j2a, j2b)]
+        # --- per-partkey sum(quantity)/count
+        q_u = abi.make_updates(*q_cols, t, t + 1)
+        acorr = ctx.reduce_push(self.avg, q_u)
+        acorr_u = abi.make_updates(*acorr, t, t + 1)
+        # --- join3 drain: side1 (l1 delta) BEFORE the avg corrections
+        ctx.arr_insert(self.arr_l1, l1_u)
+        j3a = ctx.join_push(self.j3, 1, l1_u)        # probes avg (old)
+        ctx.arr_insert(self.arr_avg, acorr_u)
+        j3b = ctx.join_push(self.j3, 2, acorr_u)     # probes l1 (new)
+        e_cols = [np.concatenate([a, b]) for a, b in zip(j3a, j3b)]
+        e_u = abi.make_updates(*e_cols, t, t + 1)
+        tcorr = ctx.reduce_push(self.total, e_u)
+        self._apply_total(tcorr)
+        return len(lp_k) + len(p_k)
+
+    def load(self, gen):
+        lp_k, lp_v = gen.lineitem_bypart_updates()
+        p_k, p_v = gen.part_updates()
+        ones = lambda n: np.ones(n, np.int64)
+        return self._push(0, (lp_k, lp_v, ones(len(lp_k))),
+                          (p_k, p_v, ones(len(p_k))))
+
+    def step(self, churn, t):
+        lp = churn["lineitem_by_part"]
+        return self._push(t, lp, None)

@@ -707,6 +707,27 @@ int orc_arr_push_batch(orc_ctx *c, Arr *a, const mz_gpu_updates *u) {
   return 0;
 }
 
+// fused consolidate + push (mirror of mz_gpu_arr_insert)
+int orc_arr_insert(orc_ctx *c, Arr *a, const mz_gpu_updates *u) {
+  auto b = std::make_unique<Batch>();
+  cols_from_updates(u, a->schema, b->cols);
+  consolidate(a->schema, b->cols);
+  b->lower = u->lower;
+  b->upper = u->upper;
+  b->build_index(a->schema);
+  a->batches.push_back(std::move(b));
+  a->upper = std::max(a->upper, u->upper);
+  while (a->batches.size() >= 2) {
+    size_t nb = a->batches.size();
+    if (a->batches[nb - 2]->cols.size() <=
+        2 * a->batches[nb - 1]->cols.size())
+      a->merge_span(nb - 2, nb);
+    else
+      break;
+  }
+  return 0;
+}
+
 int orc_arr_set_logical_compaction(orc_ctx *c, Arr *a, u64 f) {
   a->logical_compaction = f;
   return 0;

@@ -33,6 +33,8 @@ def _load():
     lib.orc_arr_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
     lib.orc_arr_push_batch.argtypes = [C.c_void_p, C.c_void_p,
                                        C.POINTER(Updates)]
+    lib.orc_arr_insert.argtypes = [C.c_void_p, C.c_void_p,
+                                   C.POINTER(Updates)]
     lib.orc_arr_set_logical_compaction.argtypes = [C.c_void_p, C.c_void_p,
                                                    C.c_uint64]
     lib.orc_arr_maintain.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
@@ -85,6 +87,10 @@ class OracleCtx:
 
     def arr_push(self, arr, upd):
         rc = self.lib.orc_arr_push_batch(self.ctx, arr, C.byref(upd))
+        assert rc == 0
+
+    def arr_insert(self, arr, upd):
+        rc = self.lib.orc_arr_insert(self.ctx, arr, C.byref(upd))
         assert rc == 0
     def arr_set_logical_compaction(self, arr, frontier):
         self.lib.orc_arr_set_logical_compaction(self.ctx, arr, frontier)
