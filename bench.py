@@ -51,8 +51,9 @@ def parse_args():
 class CallProfiler:
     """Wraps an engine ctx's methods with wall-time accumulation."""
 
-    METHODS = ["consolidate_dev", "arr_push", "halfjoin_dev",
-               "reduce_push_dev", "arr_maintain"]
+    METHODS = ["consolidate_dev", "arr_push", "arr_insert",
+               "halfjoin_dev", "reduce_push_dev", "join_push",
+               "arr_maintain"]
 
     def __init__(self, ctx):
         self.t = {m: [0.0, 0] for m in self.METHODS}

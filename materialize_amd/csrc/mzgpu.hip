@@ -168,11 +168,18 @@ __global__ void k_group_starts(const u32 *flags, const u32 *gid, u32 *starts,
   GRID_STRIDE(i, n) if (flags[i]) starts[gid[i] - 1] = (u32)i;
 }
 
-// per-group wrapping diff sums via prefix differences; emit nonzero flags
-__global__ void k_group_sums(const u32 *starts, u64 G, u64 n,
+// per-group wrapping diff sums via prefix differences; emit nonzero flags.
+// G (group count) is read on-device from gid[n-1] so the host never has to
+// synchronize for it; the grid is sized by the capacity n.
+__global__ void k_group_sums(const u32 *starts, const u32 *gid, u64 n,
                              const u64 *diff_prefix /* inclusive, permuted */,
                              i64 *gsum, u32 *nz) {
-  GRID_STRIDE(g, G) {
+  u64 G = n ? gid[n - 1] : 0;
+  GRID_STRIDE(g, n) {
+    if (g >= G) {
+      nz[g] = 0;  // zero the padding so the nz scan over n is exact
+      continue;
+    }
     u64 lo = starts[g];
     u64 hi = (g + 1 < G) ? starts[g + 1] : n;
     u64 s = diff_prefix[hi - 1] - (lo ? diff_prefix[lo - 1] : 0);
@@ -184,9 +191,12 @@ __global__ void k_group_sums(const u32 *starts, u64 G, u64 n,
 __global__ void k_emit_consolidated(const u64 *keys, u32 kw, const u8 *vals,
                                     u32 vb, const u64 *times, const u32 *perm,
                                     const u32 *starts, const i64 *gsum,
-                                    const u32 *nz, const u32 *nzpos, u64 G,
-                                    u64 *okeys, u8 *ovals, u64 *otimes,
-                                    i64 *odiffs) {
+                                    const u32 *nz, const u32 *nzpos,
+                                    const u32 *gid, u64 n, u64 *okeys,
+                                    u8 *ovals, u64 *otimes, i64 *odiffs,
+                                    u64 *dcounts /* [0] = M out */) {
+  u64 G = n ? gid[n - 1] : 0;
+  if (blockIdx.x == 0 && threadIdx.x == 0) dcounts[0] = nzpos[G];
   GRID_STRIDE(g, G) {
     if (!nz[g]) continue;
     u64 o = nzpos[g];  // exclusive scan of nz
@@ -204,10 +214,19 @@ __global__ void k_advance_times(u64 *times, u64 n, u64 frontier) {
 
 // ----------------------------------------------- batch structure build
 
-// flags over SEALED (already sorted) arrays
+// flags over SEALED (already sorted) arrays. The actual row count M is
+// read on-device from dcounts[0] (set by k_emit_consolidated); positions
+// beyond it get zero flags so capacity-sized scans stay exact.
 __global__ void k_change_flags(const u64 *keys, u32 kw, const u8 *vals,
-                               u32 vb, u32 *kc, u32 *vc, u64 n) {
-  GRID_STRIDE(i, n) {
+                               u32 vb, u32 *kc, u32 *vc, u64 cap,
+                               const u64 *dcounts) {
+  u64 n = dcounts ? dcounts[0] : cap;
+  GRID_STRIDE(i, cap) {
+    if (i >= n) {
+      kc[i] = 0;
+      vc[i] = 0;
+      continue;
+    }
     if (i == 0) {
       kc[0] = 1;
       vc[0] = 1;
@@ -228,9 +247,18 @@ __global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
                                     u32 vb, const u32 *kc, const u32 *vc,
                                     const u32 *kid /*inclusive*/,
                                     const u32 *vid /*inclusive*/, u64 n,
-                                    u64 n_keys, u64 n_vals, u64 *bkeys,
-                                    u32 *kv_off, u8 *bvals, u32 *vu_off,
-                                    u32 *val_key, u32 *upd_val) {
+                                    u64 *bkeys, u32 *kv_off, u8 *bvals,
+                                    u32 *vu_off, u32 *val_key, u32 *upd_val,
+                                    u64 *dcounts /* [0]=M in; [1]=nk,
+                                                    [2]=nv out */) {
+  u64 cap = n;
+  n = dcounts[0];  // actual consolidated rows
+  u64 n_keys = n ? kid[cap - 1] : 0;  // zero flags beyond n keep this exact
+  u64 n_vals = n ? vid[cap - 1] : 0;
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    dcounts[1] = n_keys;
+    dcounts[2] = n_vals;
+  }
   GRID_STRIDE(i, n) {
     u32 k = kid[i] - 1, v = vid[i] - 1;
     upd_val[i] = v;
@@ -253,7 +281,8 @@ __global__ void k_hash_init(u64 *hash, u64 slots, u32 kw) {
 }
 
 __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
-                             u64 n_keys) {
+                             const u32 *kid, u64 cap, const u64 *dcounts) {
+  u64 n_keys = (cap && dcounts[0]) ? kid[cap - 1] : 0;
   GRID_STRIDE(i, n_keys) {
     u64 h = route_hash(keys + i * kw, kw) & (slots - 1);
     for (;;) {
@@ -969,6 +998,23 @@ u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   return total;
 }
 
+// enqueue-only exclusive scan (out has n+1 slots; no readback)
+void exclusive_scan_u32_ns(Ctx *c, const u32 *in, u32 *out, u64 n) {
+  auto &S = c->scratch;
+  u32 *pad = (u32 *)S.get((n + 1) * 4);
+  HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
+                           c->stream));
+  HIP_CHECK(hipMemsetAsync(pad + n, 0, 4, c->stream));
+  size_t need = 0;
+  (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
+                                rocprim::plus<u32>(), c->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::exclusive_scan(tmp, need, pad, out, 0u, n + 1,
+                                rocprim::plus<u32>(), c->stream);
+}
+
+__global__ void k_write_u64(u64 *p, u64 v) { *p = v; }
+
 void inclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   auto &S = c->scratch;
   size_t need = 0;
@@ -1057,16 +1103,16 @@ mz_gpu_out *make_out(u64 *k, u8 *v, u64 *t, i64 *d, u64 n, u32 kw, u32 vb) {
 
 // Core consolidation: sort + group + sum + compact. Returns owned device
 // arrays (exact-size). Input must be device-resident.
-void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
-                     u8 **ovals, u64 **otimes, i64 **odiffs, u64 *out_n) {
+// Enqueue-only consolidation into CAPACITY-sized outputs (allocated by
+// the caller at n rows); dcounts[0] receives the consolidated row count
+// on device. The only host sync is the column-min/max read inside
+// sort_updates.
+void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
+                      u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts) {
   auto &S = c->scratch;
   u64 n = in.n;
   if (n == 0) {
-    *okeys = dnew<u64>(c, 1);
-    *ovals = (u8 *)dmalloc(c, 1);
-    *otimes = dnew<u64>(c, 1);
-    *odiffs = dnew<i64>(c, 1);
-    *out_n = 0;
+    HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
     return;
   }
   u32 *perm = (u32 *)S.get(n * 4);
@@ -1076,11 +1122,7 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
                      in.keys, kw, in.vals, vb, in.times, perm, flags, n, 1);
   u32 *gid = (u32 *)S.get(n * 4);
   inclusive_scan_u32(c, flags, gid, n);
-  u32 G;
-  HIP_CHECK(hipMemcpyAsync(&G, gid + n - 1, 4, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
-  u32 *starts = (u32 *)S.get((u64)G * 4);
+  u32 *starts = (u32 *)S.get(n * 4);
   hipLaunchKernelGGL(k_group_starts, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
                      flags, gid, starts, n);
   // wrapping inclusive prefix of permuted diffs
@@ -1089,21 +1131,36 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
                      (const u64 *)in.diffs, perm, pdiff, n);
   u64 *pref = (u64 *)S.get(n * 8);
   inclusive_scan_u64(c, pdiff, pref, n);
-  i64 *gsum = (i64 *)S.get((u64)G * 8);
-  u32 *nz = (u32 *)S.get((u64)G * 4 + 4);
-  hipLaunchKernelGGL(k_group_sums, dim3(ngrid(G)), dim3(BLK), 0, c->stream,
-                     starts, (u64)G, n, pref, gsum, nz);
-  u32 *nzpos = (u32 *)S.get(((u64)G + 1) * 4);
-  u64 M = exclusive_scan_u32(c, nz, nzpos, G);
-  *okeys = dnew<u64>(c, std::max<u64>(M, 1) * kw);
-  *ovals = (u8 *)dmalloc(c, std::max<u64>(M, 1) * vb);
-  *otimes = dnew<u64>(c, std::max<u64>(M, 1));
-  *odiffs = dnew<i64>(c, std::max<u64>(M, 1));
-  if (M)
-    hipLaunchKernelGGL(k_emit_consolidated, dim3(ngrid(G)), dim3(BLK), 0,
-                       c->stream, in.keys, kw, in.vals, vb, in.times, perm,
-                       starts, gsum, nz, nzpos, (u64)G, *okeys, *ovals,
-                       *otimes, *odiffs);
+  i64 *gsum = (i64 *)S.get(n * 8);
+  u32 *nz = (u32 *)S.get((n + 1) * 4);
+  hipLaunchKernelGGL(k_group_sums, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     starts, gid, n, pref, gsum, nz);
+  u32 *nzpos = (u32 *)S.get((n + 1) * 4);
+  exclusive_scan_u32_ns(c, nz, nzpos, n);
+  hipLaunchKernelGGL(k_emit_consolidated, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, in.keys, kw, in.vals, vb, in.times, perm,
+                     starts, gsum, nz, nzpos, gid, n, okeys, ovals, otimes,
+                     odiffs, dcounts);
+}
+
+// Synchronous wrapper (ABI-level mz_gpu_consolidate and the consolidated
+// probe outputs): returns owned capacity-sized arrays + the exact count.
+void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
+                     u8 **ovals, u64 **otimes, i64 **odiffs, u64 *out_n) {
+  auto &S = c->scratch;
+  u64 n = in.n;
+  u64 capn = std::max<u64>(n, 1);
+  *okeys = dnew<u64>(c, capn * kw);
+  *ovals = (u8 *)dmalloc(c, std::max<u64>(capn * vb, 1));
+  *otimes = dnew<u64>(c, capn);
+  *odiffs = dnew<i64>(c, capn);
+  u64 *dcounts = (u64 *)S.get(3 * 8);
+  consolidate_core(c, kw, vb, in, *okeys, *ovals, *otimes, *odiffs,
+                   dcounts);
+  u64 M = 0;
+  HIP_CHECK(hipMemcpyAsync(&M, dcounts, 8, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
   *out_n = M;
 }
 
@@ -1115,17 +1172,18 @@ void free_batch(Ctx *c, DevBatch &b) {
   b = DevBatch();
 }
 
-// Build a sealed DevBatch from sealed (sorted+consolidated) device arrays.
-// Takes ownership of the arrays (they become the batch's times/diffs after
-// compaction into structure arrays; keys/vals are re-packed).
-DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
-                     i64 *diffs, u64 n, u64 lower, u64 upper) {
+// Enqueue-only batch build over capacity-sized sealed arrays whose actual
+// row count lives in dcounts[0]; dcounts[1]/[2] receive n_keys/n_vals.
+// Takes ownership of the flat arrays. Caller must sync and fill the
+// batch's host-side counts from dcounts.
+DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
+                          u64 *times, i64 *diffs, u64 cap, u64 lower,
+                          u64 upper, u64 *dcounts) {
   auto &S = c->scratch;
   DevBatch b;
   b.lower = lower;
   b.upper = upper;
-  b.n_upds = n;
-  if (n == 0) {
+  if (cap == 0) {
     b.keys = keys;
     b.vals = vals;
     b.times = times;
@@ -1138,47 +1196,57 @@ DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
     HIP_CHECK(hipMemsetAsync(b.vu_off, 0, 4, c->stream));
     return b;
   }
-  u32 *kc = (u32 *)S.get(n * 4);
-  u32 *vc = (u32 *)S.get(n * 4);
-  hipLaunchKernelGGL(k_change_flags, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
-                     keys, kw, vals, vb, kc, vc, n);
-  u32 *kid = (u32 *)S.get(n * 4);
-  u32 *vid = (u32 *)S.get(n * 4);
-  inclusive_scan_u32(c, kc, kid, n);
-  inclusive_scan_u32(c, vc, vid, n);
-  u32 nk, nv;
-  HIP_CHECK(hipMemcpyAsync(&nk, kid + n - 1, 4, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipMemcpyAsync(&nv, vid + n - 1, 4, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
-  b.n_keys = nk;
-  b.n_vals = nv;
-  b.keys = dnew<u64>(c, (u64)nk * kw);
-  b.kv_off = dnew<u32>(c, (u64)nk + 1);
-  b.vals = (u8 *)dmalloc(c, (u64)nv * vb);
-  b.vu_off = dnew<u32>(c, (u64)nv + 1);
-  b.val_key = dnew<u32>(c, nv);
-  b.upd_val = dnew<u32>(c, n);
-  hipLaunchKernelGGL(k_scatter_structure, dim3(ngrid(n)), dim3(BLK), 0,
-                     c->stream, keys, kw, vals, vb, kc, vc, kid, vid, n,
-                     (u64)nk, (u64)nv, b.keys, b.kv_off, b.vals, b.vu_off,
-                     b.val_key, b.upd_val);
+  u32 *kc = (u32 *)S.get(cap * 4);
+  u32 *vc = (u32 *)S.get(cap * 4);
+  hipLaunchKernelGGL(k_change_flags, dim3(ngrid(cap)), dim3(BLK), 0,
+                     c->stream, keys, kw, vals, vb, kc, vc, cap, dcounts);
+  u32 *kid = (u32 *)S.get(cap * 4);
+  u32 *vid = (u32 *)S.get(cap * 4);
+  inclusive_scan_u32(c, kc, kid, cap);
+  inclusive_scan_u32(c, vc, vid, cap);
+  b.keys = dnew<u64>(c, cap * kw);
+  b.kv_off = dnew<u32>(c, cap + 1);
+  b.vals = (u8 *)dmalloc(c, std::max<u64>(cap * vb, 1));
+  b.vu_off = dnew<u32>(c, cap + 1);
+  b.val_key = dnew<u32>(c, cap);
+  b.upd_val = dnew<u32>(c, cap);
+  hipLaunchKernelGGL(k_scatter_structure, dim3(ngrid(cap)), dim3(BLK), 0,
+                     c->stream, keys, kw, vals, vb, kc, vc, kid, vid, cap,
+                     b.keys, b.kv_off, b.vals, b.vu_off, b.val_key,
+                     b.upd_val, dcounts);
   b.times = times;
   b.diffs = diffs;
-  // hash index: pow2 >= 2*nk
   u64 slots = 16;
-  while (slots < 2 * (u64)nk) slots <<= 1;
+  while (slots < 2 * cap) slots <<= 1;
   b.hash_slots = slots;
   b.hash = dnew<u64>(c, slots * (kw + 1));
-  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0, c->stream,
-                     b.hash, slots, kw);
-  hipLaunchKernelGGL(k_hash_build, dim3(ngrid(nk)), dim3(BLK), 0, c->stream,
-                     b.hash, slots, b.keys, kw, (u64)nk);
-  // keys/vals flat arrays were re-packed into structure; free originals
+  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
+                     c->stream, b.hash, slots, kw);
+  hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
+                     c->stream, b.hash, slots, b.keys, kw, kid, cap,
+                     dcounts);
+  // flat key/val arrays were re-packed; stream-ordered free is safe
+  dfree(c, keys);
+  dfree(c, vals);
+  return b;
+}
+
+// Synchronous wrapper for sealed inputs with host-known count.
+DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
+                     i64 *diffs, u64 n, u64 lower, u64 upper) {
+  auto &S = c->scratch;
+  u64 *dcounts = (u64 *)S.get(3 * 8);
+  hipLaunchKernelGGL(k_write_u64, dim3(1), dim3(1), 0, c->stream, dcounts,
+                     n);
+  DevBatch b = build_batch_core(c, kw, vb, keys, vals, times, diffs, n,
+                                lower, upper, dcounts);
+  u64 cnt[3] = {n, 0, 0};
+  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
+                           c->stream));
   HIP_CHECK(hipStreamSynchronize(c->stream));
-  dfree(c, (keys));
-  dfree(c, (vals));
+  b.n_upds = cnt[0];
+  b.n_keys = cnt[1];
+  b.n_vals = cnt[2];
   return b;
 }
 
@@ -1209,14 +1277,23 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
     base += b.n_upds;
   }
   DevUpdates in{keys, vals, times, diffs, total};
-  u64 *ok;
-  u8 *ov;
-  u64 *ot;
-  i64 *od;
-  u64 M;
-  consolidate_dev(c, kw, vb, in, &ok, &ov, &ot, &od, &M);
-  DevBatch merged = build_batch(c, kw, vb, ok, ov, ot, od, M,
-                                lo == UINT64_MAX ? 0 : lo, hi);
+  u64 capn = std::max<u64>(total, 1);
+  u64 *ok = dnew<u64>(c, capn * kw);
+  u8 *ov = (u8 *)dmalloc(c, std::max<u64>(capn * vb, 1));
+  u64 *ot = dnew<u64>(c, capn);
+  i64 *od = dnew<i64>(c, capn);
+  u64 *dcounts = (u64 *)S.get(3 * 8);
+  consolidate_core(c, kw, vb, in, ok, ov, ot, od, dcounts);
+  DevBatch merged =
+      build_batch_core(c, kw, vb, ok, ov, ot, od, total,
+                       lo == UINT64_MAX ? 0 : lo, hi, dcounts);
+  u64 cnt[3] = {0, 0, 0};
+  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  merged.n_upds = cnt[0];
+  merged.n_keys = cnt[1];
+  merged.n_vals = cnt[2];
   for (size_t i = from; i < to; i++) free_batch(c, a->batches[i]);
   a->batches.erase(a->batches.begin() + from, a->batches.begin() + to);
   a->batches.insert(a->batches.begin() + from, merged);
@@ -1322,16 +1399,25 @@ int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
                       const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
   ctx->scratch.reset();
+  auto &S = ctx->scratch;
   u32 kw = a->schema.kw, vb = a->schema.vb;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
-  u64 *ok;
-  u8 *ov;
-  u64 *ot;
-  i64 *od;
-  u64 M;
-  consolidate_dev(ctx, kw, vb, d, &ok, &ov, &ot, &od, &M);
-  DevBatch b =
-      build_batch(ctx, kw, vb, ok, ov, ot, od, M, u->lower, u->upper);
+  u64 capn = std::max<u64>(d.n, 1);
+  u64 *ok = dnew<u64>(ctx, capn * kw);
+  u8 *ov = (u8 *)dmalloc(ctx, std::max<u64>(capn * vb, 1));
+  u64 *ot = dnew<u64>(ctx, capn);
+  i64 *od = dnew<i64>(ctx, capn);
+  u64 *dcounts = (u64 *)S.get(3 * 8);
+  consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, u->lower,
+                                u->upper, dcounts);
+  u64 cnt[3] = {0, 0, 0};
+  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  b.n_upds = cnt[0];
+  b.n_keys = cnt[1];
+  b.n_vals = cnt[2];
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
   spine_policy(ctx, a);
