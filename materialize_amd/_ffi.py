@@ -85,6 +85,10 @@ def load():
     lib.mz_gpu_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
                                        C.POINTER(Updates),
                                        C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_reduce_push2.argtypes = [C.c_void_p, C.c_void_p,
+                                        C.POINTER(Updates),
+                                        C.POINTER(Updates),
+                                        C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_partition.argtypes = [
         C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
@@ -247,6 +251,12 @@ class GpuCtx:
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
                                                 C.byref(outp)))
+        return self._dev_out(outp)
+
+    def reduce_push2_dev(self, op, u1, u2):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_reduce_push2(
+            self.ctx, op, C.byref(u1), C.byref(u2), C.byref(outp)))
         return self._dev_out(outp)
 
     def set_kernel_timing(self, on):

@@ -484,6 +484,27 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
   }
 }
 
+// Strict-weak ordering over expanded rows by global row id:
+// (key i64-tuple, val LE-u64-tuple, time) — the engine's canonical order.
+struct RowLess {
+  const u64 *keys;
+  const u8 *vals;
+  const u64 *times;
+  u32 kw, vb;
+  __device__ bool operator()(u64 a, u64 b) const {
+    for (u32 w = 0; w < kw; w++) {
+      i64 x = (i64)keys[a * kw + w], y = (i64)keys[b * kw + w];
+      if (x != y) return x < y;
+    }
+    for (u32 off = 0; off < vb; off += 8) {
+      u64 x = le_val_word(vals + a * vb + off, vb - off);
+      u64 y = le_val_word(vals + b * vb + off, vb - off);
+      if (x != y) return x < y;
+    }
+    return times[a] < times[b];
+  }
+};
+
 // expansion: flatten a DevBatch back to per-update (key,val,time,diff)
 __global__ void k_expand_batch(DevBatch b, u32 kw, u32 vb, u64 frontier,
                                u64 *okeys, u8 *ovals, u64 *otimes,
@@ -1107,6 +1128,10 @@ mz_gpu_out *make_out(u64 *k, u8 *v, u64 *t, i64 *d, u64 n, u32 kw, u32 vb) {
 // the caller at n rows); dcounts[0] receives the consolidated row count
 // on device. The only host sync is the column-min/max read inside
 // sort_updates.
+void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
+                           const u32 *perm, u64 *okeys, u8 *ovals,
+                           u64 *otimes, i64 *odiffs, u64 *dcounts);
+
 void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
                       u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts) {
   auto &S = c->scratch;
@@ -1117,6 +1142,20 @@ void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
   }
   u32 *perm = (u32 *)S.get(n * 4);
   sort_updates(c, in.keys, kw, in.vals, vb, in.times, n, perm, false);
+  consolidate_with_perm(c, kw, vb, in, perm, okeys, ovals, otimes, odiffs,
+                        dcounts);
+}
+
+// Consolidation given a ready ordering permutation (group + sum + compact).
+void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
+                           const u32 *perm, u64 *okeys, u8 *ovals,
+                           u64 *otimes, i64 *odiffs, u64 *dcounts) {
+  auto &S = c->scratch;
+  u64 n = in.n;
+  if (n == 0) {
+    HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
+    return;
+  }
   u32 *flags = (u32 *)S.get(n * 4);
   hipLaunchKernelGGL(k_head_flags, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
                      in.keys, kw, in.vals, vb, in.times, perm, flags, n, 1);
@@ -1283,7 +1322,26 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   u64 *ot = dnew<u64>(c, capn);
   i64 *od = dnew<i64>(c, capn);
   u64 *dcounts = (u64 *)S.get(3 * 8);
-  consolidate_core(c, kw, vb, in, ok, ov, ot, od, dcounts);
+  if (to - from == 2 && total) {
+    // Both inputs are sorted runs (logical-compaction advance is monotone,
+    // so expansion preserves order): a single merge-path pass replaces the
+    // full radix re-sort — this is the Spine's pairwise merge proper
+    // (ColInternalMerger::merge semantics, columnation.rs:653-713).
+    u64 n1 = a->batches[from].n_upds, n2 = a->batches[from + 1].n_upds;
+    u32 *perm = (u32 *)S.get(total * 4);
+    RowLess cmp{keys, vals, times, kw, vb};
+    auto it1 = rocprim::make_counting_iterator<u32>(0u);
+    auto it2 = rocprim::make_counting_iterator<u32>((u32)n1);
+    size_t need = 0;
+    (void)rocprim::merge(nullptr, need, it1, it2, perm, (size_t)n1,
+                         (size_t)n2, cmp, c->stream);
+    void *tmp = S.get(need);
+    (void)rocprim::merge(tmp, need, it1, it2, perm, (size_t)n1, (size_t)n2,
+                         cmp, c->stream);
+    consolidate_with_perm(c, kw, vb, in, perm, ok, ov, ot, od, dcounts);
+  } else {
+    consolidate_core(c, kw, vb, in, ok, ov, ot, od, dcounts);
+  }
   DevBatch merged =
       build_batch_core(c, kw, vb, ok, ov, ot, od, total,
                        lo == UINT64_MAX ? 0 : lo, hi, dcounts);
@@ -1640,14 +1698,62 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
 
 void mz_gpu_reduce_drop(mz_gpu_ctx *c, mz_gpu_red *r) { (void)c; (void)r; }
 
+// Internal: reduce over already-staged device updates.
+static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
+                                u64 lower, u64 upper, mz_gpu_out **out);
+
 int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
                        const mz_gpu_updates *u, mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
   ctx->scratch.reset();
+  u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  return reduce_push_dev_impl(ctx, op, d, u->lower, u->upper, out);
+}
+
+// Concatenate two update streams (device-side) and reduce them as ONE
+// logical batch — the render layer's path-output concat without a host
+// round trip (reduce corrections depend on the combined batch, so the
+// two streams must enter one push).
+int mz_gpu_reduce_push2(mz_gpu_ctx *c, mz_gpu_red *op,
+                        const mz_gpu_updates *u1, const mz_gpu_updates *u2,
+                        mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
+  DevUpdates d1 = stage_updates(ctx, u1, kw, vb);
+  DevUpdates d2 = stage_updates(ctx, u2, kw, vb);
+  u64 n = d1.n + d2.n;
+  u64 capn = std::max<u64>(n, 1);
+  u64 *k = (u64 *)S.get(capn * kw * 8);
+  u8 *v = (u8 *)S.get(std::max<u64>(capn * vb, 1));
+  u64 *t = (u64 *)S.get(capn * 8);
+  i64 *df = (i64 *)S.get(capn * 8);
+  auto cat = [&](void *dst, const void *a, u64 abytes, const void *b,
+                 u64 bbytes) {
+    if (abytes)
+      HIP_CHECK(hipMemcpyAsync(dst, a, abytes, hipMemcpyDeviceToDevice,
+                               ctx->stream));
+    if (bbytes)
+      HIP_CHECK(hipMemcpyAsync((char *)dst + abytes, b, bbytes,
+                               hipMemcpyDeviceToDevice, ctx->stream));
+  };
+  cat(k, d1.keys, d1.n * kw * 8, d2.keys, d2.n * kw * 8);
+  if (vb) cat(v, d1.vals, d1.n * vb, d2.vals, d2.n * vb);
+  cat(t, d1.times, d1.n * 8, d2.times, d2.n * 8);
+  cat(df, d1.diffs, d1.n * 8, d2.diffs, d2.n * 8);
+  DevUpdates d{k, v, t, df, n};
+  return reduce_push_dev_impl(ctx, op, d,
+                              std::min(u1->lower, u2->lower),
+                              std::max(u1->upper, u2->upper), out);
+}
+
+static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
+                                u64 lower, u64 upper, mz_gpu_out **out) {
   auto &S = ctx->scratch;
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   u32 okw = op->spec.out.key_words, ovb = op->spec.out.val_bytes;
-  DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 n = d.n;
   if (n == 0) {
     *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1), dnew<u64>(ctx, 1),
@@ -1676,8 +1782,8 @@ int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
   // shape) skip the boundary copy entirely.
   std::vector<u64> htimes;
   std::vector<std::pair<u64, u64>> slices;  // (start, end)
-  if (u->upper <= u->lower + 1) {
-    htimes.assign(1, u->lower);
+  if (upper <= lower + 1) {
+    htimes.assign(1, lower);
     slices.push_back({0, n});
   } else {
     htimes.resize(n);

@@ -221,13 +221,9 @@ class Q3Dataflow:
         if len(outs) == 1:
             corr = self.reduce.push(outs[0].updates(t, t + 1))
         else:
-            cols = [out.to_host() for out in outs]
-            keys = np.concatenate([c[0] for c in cols])
-            vals = np.concatenate([c[1] for c in cols])
-            times = np.concatenate([c[2] for c in cols])
-            diffs = np.concatenate([c[3] for c in cols])
-            u = abi.make_updates(keys, vals, times, diffs, t, t + 1)
-            corr = self.reduce.push(u)
+            corr = self.ctx.reduce_push2_dev(self.reduce.op,
+                                             outs[0].updates(t, t + 1),
+                                             outs[1].updates(t, t + 1))
         for out in outs:
             out.release()
         return corr

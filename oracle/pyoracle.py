@@ -122,8 +122,14 @@ class OracleCtx:
         assert rc == 0
         return self._take(outp)
 
+    _red_specs = None
+
     def reduce_create(self, spec):
-        return self.lib.orc_reduce_create(self.ctx, C.byref(spec))
+        op = self.lib.orc_reduce_create(self.ctx, C.byref(spec))
+        if self._red_specs is None:
+            self._red_specs = {}
+        self._red_specs[op] = (spec.in_.key_words, spec.in_.val_bytes)
+        return op
 
     def reduce_push(self, op, upd):
         outp = C.POINTER(OutBatch)()
@@ -157,6 +163,33 @@ class OracleCtx:
     def reduce_push_dev(self, op, upd):
         res = self.reduce_push(op, upd)
         return self._host_out(res, 0, 0)
+
+    def reduce_push2_dev(self, op, u1, u2):
+        import numpy as np
+        from materialize_amd import _abi as abi
+
+        def cols(u, kw, vb):
+            n = int(u.n)
+            k = np.ctypeslib.as_array(u.keys, shape=(n * kw,)) if n else \
+                np.empty(0, np.uint64)
+            v = np.ctypeslib.as_array(u.vals, shape=(n * vb,)) if n and vb \
+                else np.empty(0, np.uint8)
+            t = np.ctypeslib.as_array(u.times, shape=(n,)) if n else \
+                np.empty(0, np.uint64)
+            d = np.ctypeslib.as_array(u.diffs, shape=(n,)) if n else \
+                np.empty(0, np.int64)
+            return k, v, t, d
+        # concat on host (schema from the op's spec is not exposed here;
+        # infer strides from the updates' n and array shapes via the spec
+        # stored at create time)
+        kw, vb = self._red_specs[op]
+        a = cols(u1, kw, vb)
+        b = cols(u2, kw, vb)
+        cat = [np.concatenate([x, y]) for x, y in zip(a, b)]
+        u = abi.make_updates(cat[0].view(np.int64), cat[1], cat[2], cat[3],
+                             min(u1.lower, u2.lower),
+                             max(u1.upper, u2.upper))
+        return self.reduce_push_dev(op, u)
 
     def route_hash(self, words):
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
