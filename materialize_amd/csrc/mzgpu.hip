@@ -1370,11 +1370,26 @@ extern "C" {
 mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   int ndev = 0;
   if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) return nullptr;
-  if (cfg) HIP_CHECK(hipSetDevice((int)cfg->device_index));
+  int dev = cfg ? (int)cfg->device_index : 0;
+  HIP_CHECK(hipSetDevice(dev));
   mz_gpu_ctx *c = new mz_gpu_ctx();
   HIP_CHECK(hipStreamCreate(&c->impl.stream));
   HIP_CHECK(hipEventCreate(&c->impl.ev_a));
   HIP_CHECK(hipEventCreate(&c->impl.ev_b));
+  // Keep freed stream-ordered allocations in the pool forever (288 GB of
+  // HBM — never hand memory back to the OS mid-run; pool misses showed up
+  // as multi-ms host stalls before large allocations).
+  hipMemPool_t pool = nullptr;
+  if (hipDeviceGetDefaultMemPool(&pool, dev) == hipSuccess && pool) {
+    uint64_t thresh = UINT64_MAX;
+    (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold,
+                                 &thresh);
+  }
+  // pre-grow the scratch arena so per-call growth never stalls the step
+  u64 scratch0 = cfg && cfg->hbm_pool_bytes ? cfg->hbm_pool_bytes
+                                            : (4ull << 30);
+  c->impl.scratch.get(scratch0);
+  c->impl.scratch.reset();
   return c;
 }
 
