@@ -45,6 +45,9 @@ def parse_args():
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--call-profile", action="store_true",
                    help="print per-engine-call wall time breakdown")
+    p.add_argument("--backend", default="nccl",
+                   help="torch.distributed backend for N>1 (nccl = RCCL; "
+                        "gloo only for single-box validation)")
     return p.parse_args()
 
 
@@ -170,17 +173,19 @@ def main():
 
     sf = args.sf or (1.0 if N == 1 else 1.25 * N)
     batch_rows = args.batch_rows or 100_000 * N
-    device = f"cuda:{local_rank}"
-    torch.cuda.set_device(local_rank)
+    ndev = max(torch.cuda.device_count(), 1)
+    dev_idx = local_rank % ndev  # gloo validation: several ranks, one GPU
+    device = f"cuda:{dev_idx}"
+    torch.cuda.set_device(dev_idx)
 
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
-        dist.init_process_group("nccl")
+        dist.init_process_group(args.backend)
         from materialize_amd.dist import TorchExchange
-        ex = TorchExchange(device)
-    ctx = GpuCtx(device=local_rank)
+        ex = TorchExchange("cpu" if args.backend == "gloo" else device)
+    ctx = GpuCtx(device=dev_idx)
 
     gen = TpchGen(sf=sf, seed=args.seed)
     if world > 1:

@@ -109,8 +109,15 @@ __global__ void k_sortkey_val(const u8 *vals, u32 vb, u32 word,
 __global__ void k_pass_minmax(const u64 *keys, u32 kw, const u8 *vals,
                               u32 vb, const u64 *times, u64 n, u32 vwords,
                               int with_time, u64 *mins, u64 *maxs) {
+  // per-thread register accumulation; one LDS merge + one global atomic
+  // per block per pass (per-element LDS atomics serialized badly)
   __shared__ u64 smin[MAX_PASSES], smax[MAX_PASSES];
   u32 np = (with_time ? 1 : 0) + vwords + kw;
+  u64 lmin[MAX_PASSES], lmax[MAX_PASSES];
+  for (u32 p = 0; p < np; p++) {
+    lmin[p] = ~0ull;
+    lmax[p] = 0;
+  }
   if (threadIdx.x < np) {
     smin[threadIdx.x] = ~0ull;
     smax[threadIdx.x] = 0;
@@ -119,20 +126,25 @@ __global__ void k_pass_minmax(const u64 *keys, u32 kw, const u8 *vals,
   GRID_STRIDE(i, n) {
     u32 p = 0;
     if (with_time) {
-      atomicMin((unsigned long long *)&smin[p], times[i]);
-      atomicMax((unsigned long long *)&smax[p], times[i]);
+      u64 x = times[i];
+      lmin[p] = x < lmin[p] ? x : lmin[p];
+      lmax[p] = x > lmax[p] ? x : lmax[p];
       p++;
     }
     for (u32 w = 0; w < vwords; w++, p++) {
       u64 x = le_val_word(vals + i * vb + w * 8, vb - w * 8);
-      atomicMin((unsigned long long *)&smin[p], x);
-      atomicMax((unsigned long long *)&smax[p], x);
+      lmin[p] = x < lmin[p] ? x : lmin[p];
+      lmax[p] = x > lmax[p] ? x : lmax[p];
     }
     for (u32 w = 0; w < kw; w++, p++) {
       u64 x = keys[i * kw + w] ^ 0x8000000000000000ULL;
-      atomicMin((unsigned long long *)&smin[p], x);
-      atomicMax((unsigned long long *)&smax[p], x);
+      lmin[p] = x < lmin[p] ? x : lmin[p];
+      lmax[p] = x > lmax[p] ? x : lmax[p];
     }
+  }
+  for (u32 p = 0; p < np; p++) {
+    atomicMin((unsigned long long *)&smin[p], lmin[p]);
+    atomicMax((unsigned long long *)&smax[p], lmax[p]);
   }
   __syncthreads();
   if (threadIdx.x < np) {
