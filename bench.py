@@ -261,6 +261,19 @@ def main():
     # normalize achieved/peak into GB/s for the printed unit
     roofline["achieved"] = achieved / 1e9
     roofline["peak"] = HBM_PEAK / 1e9
+    # measured HBM traffic per probe launch-pair (separate PMC passes; see
+    # profiles/traffic_*.json for provenance) when this exact workload was
+    # the one profiled
+    try:
+        import glob as _glob
+        for tf in _glob.glob(os.path.join(REPO, "profiles",
+                                          "traffic_*.json")):
+            td = json.load(open(tf))
+            if td.get("workload") == (f"tpch_q3_sf{sf:g}_churn{batch_rows}"
+                                      if N == 1 else None):
+                roofline["traffic"] =                     td["probe_pair_traffic_bytes_per_launch"]
+    except Exception:
+        pass
 
     cpu_baseline = None
     if rank == 0 and N == 1 and not args.no_cpu_baseline:

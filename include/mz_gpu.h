@@ -254,6 +254,15 @@ int  mz_gpu_partition(mz_gpu_ctx *ctx, const mz_gpu_schema *schema,
                       uint64_t *out_times, int64_t *out_diffs,
                       uint64_t *counts);
 
+/* ------------------------------------------------------------------ peek
+ * Replaces the peek path (handle_peek/process_peeks,
+ * src/compute/src/compute_state.rs:763,1155): read, for each requested
+ * key, the arrangement's (val, summed diff) pairs as of `time` (updates at
+ * t' <= time accumulate; zero-sum vals are dropped). `keys` are n_keys
+ * host key rows. Out rows: (key, val, time, diff). */
+int  mz_gpu_peek(mz_gpu_ctx *ctx, mz_gpu_arr *arr, const uint64_t *keys,
+                 uint64_t n_keys, uint64_t time, mz_gpu_out **out);
+
 /* The routing hash itself (host helper; device code uses the same). */
 uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words);
 

@@ -93,6 +93,9 @@ def load():
         C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
         C.POINTER(C.c_int64), C.POINTER(C.c_uint64)]
+    lib.mz_gpu_peek.argtypes = [C.c_void_p, C.c_void_p,
+                                C.POINTER(C.c_uint64), C.c_uint64,
+                                C.c_uint64, C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_route_hash.restype = C.c_uint64
     lib.mz_gpu_route_hash.argtypes = [C.POINTER(C.c_uint64), C.c_uint32]
     lib.mz_gpu_set_kernel_timing.argtypes = [C.c_void_p, C.c_int]
@@ -206,6 +209,17 @@ class GpuCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.mz_gpu_route_hash(arr, len(words))
+
+    def peek(self, arr, keys, time, kw=1):
+        """Read (val, summed diff) per requested key as of `time`."""
+        import numpy as np
+        keys = np.ascontiguousarray(keys, np.int64).ravel()
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_peek(
+            self.ctx, arr,
+            keys.view(np.uint64).ctypes.data_as(C.POINTER(C.c_uint64)),
+            len(keys) // kw, time, C.byref(outp)))
+        return self._take(outp)
 
     # --- device-resident variants (outputs stay on the GPU; used by the
     # render layer to chain stages without host round trips) ---

@@ -1694,6 +1694,36 @@ int mz_gpu_halfjoin(mz_gpu_ctx *c, mz_gpu_arr *lookup,
                     le ? PM_HALF_LE : PM_HALF_LT, 0, cl, 1, out);
 }
 
+// Peek: read each requested key's (val, summed diff) as of `time` —
+// a half-join (le) of the key list against the arrangement with the
+// identity closure (handle_peek/process_peeks analog, compute_state.rs).
+int mz_gpu_peek(mz_gpu_ctx *c, mz_gpu_arr *arr, const uint64_t *keys,
+                uint64_t n_keys, uint64_t time, mz_gpu_out **out) {
+  u32 kw = arr->schema.kw, vb = arr->schema.vb;
+  std::vector<u64> times(n_keys, time);
+  std::vector<i64> diffs(n_keys, 1);
+  mz_gpu_updates u{};
+  u.keys = keys;
+  u.vals = nullptr;
+  u.times = times.data();
+  u.diffs = diffs.data();
+  u.n = n_keys;
+  u.lower = time;
+  u.upper = time + 1;
+  u.on_device = 0;
+  mz_gpu_closure cl{};
+  cl.n_filters = 0;
+  cl.n_key_fields = 1;
+  cl.key_fields[0] = mz_gpu_field{MZ_SRC_KEY, 0, (u8)(8 * kw), 0, 0, 0, 0};
+  cl.n_val_fields = vb ? 1u : 0u;
+  if (vb)
+    cl.val_fields[0] =
+        mz_gpu_field{MZ_SRC_VAL_LOOKUP, 0, (u8)vb, 0, 0, 0, 0};
+  cl.out.key_words = kw;
+  cl.out.val_bytes = vb;
+  return probe_impl(&c->impl, arr, &u, 0, PM_HALF_LE, 0, &cl, 1, out);
+}
+
 // Raw variant: output left unconsolidated (consumer consolidates).
 int mz_gpu_halfjoin_raw(mz_gpu_ctx *c, mz_gpu_arr *lookup,
                         const mz_gpu_updates *delta,

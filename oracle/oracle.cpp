@@ -862,6 +862,35 @@ int orc_halfjoin(orc_ctx *c, Arr *lookup, const mz_gpu_updates *u,
   return 0;
 }
 
+// peek mirror (handle_peek analog): per requested key, (val, summed
+// diff) as of `time` via the le half-join with the identity closure.
+int orc_peek(orc_ctx *c, Arr *arr, const u64 *keys, u64 n_keys, u64 time,
+             mz_gpu_out **out) {
+  u32 kw = arr->schema.kw, vb = arr->schema.vb;
+  std::vector<u64> times(n_keys, time);
+  std::vector<i64> diffs(n_keys, 1);
+  mz_gpu_updates u{};
+  u.keys = keys;
+  u.vals = nullptr;
+  u.times = times.data();
+  u.diffs = diffs.data();
+  u.n = n_keys;
+  u.lower = time;
+  u.upper = time + 1;
+  u.on_device = 0;
+  mz_gpu_closure cl{};
+  cl.n_filters = 0;
+  cl.n_key_fields = 1;
+  cl.key_fields[0] = mz_gpu_field{MZ_SRC_KEY, 0, (u8)(8 * kw), 0, 0, 0, 0};
+  cl.n_val_fields = vb ? 1u : 0u;
+  if (vb)
+    cl.val_fields[0] =
+        mz_gpu_field{MZ_SRC_VAL_LOOKUP, 0, (u8)vb, 0, 0, 0, 0};
+  cl.out.key_words = kw;
+  cl.out.val_bytes = vb;
+  return orc_halfjoin(c, arr, &u, 0, 1, &cl, out);
+}
+
 RedOp *orc_reduce_create(orc_ctx *c, const mz_gpu_reduce_spec *spec) {
   auto r = std::make_unique<RedOp>();
   r->spec = *spec;

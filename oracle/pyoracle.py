@@ -58,6 +58,9 @@ def _load():
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
     lib.orc_out_release.argtypes = [C.c_void_p, C.POINTER(OutBatch)]
+    lib.orc_peek.argtypes = [C.c_void_p, C.c_void_p,
+                             C.POINTER(C.c_uint64), C.c_uint64, C.c_uint64,
+                             C.POINTER(C.POINTER(OutBatch))]
     lib.orc_route_hash.restype = C.c_uint64
     lib.orc_route_hash.argtypes = [C.POINTER(C.c_uint64), C.c_uint32]
     _LIB = lib
@@ -195,3 +198,14 @@ class OracleCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.orc_route_hash(arr, len(words))
+
+    def peek(self, arr, keys, time, kw=1):
+        import numpy as np
+        keys = np.ascontiguousarray(keys, np.int64).ravel()
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_peek(
+            self.ctx, arr,
+            keys.view(np.uint64).ctypes.data_as(C.POINTER(C.c_uint64)),
+            len(keys) // kw, time, C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
