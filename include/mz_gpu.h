@@ -254,6 +254,22 @@ int  mz_gpu_partition(mz_gpu_ctx *ctx, const mz_gpu_schema *schema,
                       uint64_t *out_times, int64_t *out_diffs,
                       uint64_t *counts);
 
+/* --------------------------------------------------- hierarchical reduce
+ * Replaces build_bucketed/build_monotonic + ReductionMonoid
+ * (src/compute/src/render/reduce.rs:850-1224, :2273): MIN/MAX maintained
+ * through a val-hash bucket reduction tree (plan/reduce.rs:319-326);
+ * retracting the current extremum recomputes only the affected buckets.
+ * Input vals are single i64 datums; pushes are single-timestamp.
+ * `buckets` e.g. {4096, 256, 16, 1} (last level = per key). */
+typedef struct mz_gpu_minmax mz_gpu_minmax;
+mz_gpu_minmax *mz_gpu_minmax_create(mz_gpu_ctx *ctx,
+                                    const mz_gpu_schema *in, int is_max,
+                                    const uint32_t *buckets,
+                                    uint32_t n_levels);
+int  mz_gpu_minmax_push(mz_gpu_ctx *ctx, mz_gpu_minmax *op,
+                        const mz_gpu_updates *delta, mz_gpu_out **out);
+void mz_gpu_minmax_drop(mz_gpu_ctx *ctx, mz_gpu_minmax *op);
+
 /* ------------------------------------------------------------------ peek
  * Replaces the peek path (handle_peek/process_peeks,
  * src/compute/src/compute_state.rs:763,1155): read, for each requested

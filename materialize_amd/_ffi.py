@@ -93,6 +93,14 @@ def load():
         C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
         C.POINTER(C.c_int64), C.POINTER(C.c_uint64)]
+    lib.mz_gpu_minmax_create.restype = C.c_void_p
+    lib.mz_gpu_minmax_create.argtypes = [C.c_void_p, C.POINTER(Schema),
+                                         C.c_int, C.POINTER(C.c_uint32),
+                                         C.c_uint32]
+    lib.mz_gpu_minmax_push.argtypes = [C.c_void_p, C.c_void_p,
+                                       C.POINTER(Updates),
+                                       C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_minmax_drop.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_peek.argtypes = [C.c_void_p, C.c_void_p,
                                 C.POINTER(C.c_uint64), C.c_uint64,
                                 C.c_uint64, C.POINTER(C.POINTER(OutBatch))]
@@ -209,6 +217,18 @@ class GpuCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.mz_gpu_route_hash(arr, len(words))
+
+    def minmax_create(self, in_schema, is_max, buckets):
+        arr = (C.c_uint32 * len(buckets))(*buckets)
+        return self.lib.mz_gpu_minmax_create(self.ctx, C.byref(in_schema),
+                                             1 if is_max else 0, arr,
+                                             len(buckets))
+
+    def minmax_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_minmax_push(self.ctx, op, C.byref(upd),
+                                                C.byref(outp)))
+        return self._take(outp)
 
     def peek(self, arr, keys, time, kw=1):
         """Read (val, summed diff) per requested key as of `time`."""

@@ -819,6 +819,109 @@ __global__ void k_key_flags_sorted(const u64 *keys, u32 kw, const u64 *times,
   }
 }
 
+// -------------------------------------------------- hierarchical min/max
+// build_bucketed + ReductionMonoid restatement (reduce.rs:850-1224,:2273):
+// levels of (key, val-hash bucket) arrangements; a changed group's
+// extremum is recomputed from the level arrangement (retraction-safe) and
+// corrections cascade to the next level.
+
+__global__ void k_bucket_keys(const u64 *keys, u32 kw, const u8 *vals,
+                              u64 n, u32 b0, u64 *okeys) {
+  GRID_STRIDE(i, n) {
+    for (u32 w = 0; w < kw; w++) okeys[i * (kw + 1) + w] = keys[i * kw + w];
+    u64 vword;
+    memcpy(&vword, vals + i * 8, 8);
+    okeys[i * (kw + 1) + kw] = route_hash(&vword, 1) % b0;
+  }
+}
+
+// per changed group: recompute the extremum over the level arrangement
+// (multi-batch val merge with net-diff accumulation), compare with the
+// state row, emit corrections keyed for the next level.
+__global__ void k_minmax_apply(const u64 *gkeys, u64 G, u32 kw2,
+                               BatchList bl, int is_max, RedState st,
+                               const u32 *found, const u32 *miss,
+                               const u32 *misspos, u64 base, u32 out_kw,
+                               u32 bucket_next, u64 t, u64 *okeys, u8 *ovals,
+                               u64 *otimes, i64 *odiffs,
+                               unsigned long long *ocount) {
+  GRID_STRIDE(g, G) {
+    const u64 *key = gkeys + g * kw2;
+    // per-batch cursors over the key's val range
+    u32 cur[12], end[12];
+    int ki[12];
+    for (int b = 0; b < bl.n; b++) {
+      ki[b] = hash_lookup(bl.b[b].hash, bl.b[b].hash_slots, key, kw2);
+      if (ki[b] >= 0) {
+        cur[b] = bl.b[b].kv_off[ki[b]];
+        end[b] = bl.b[b].kv_off[ki[b] + 1];
+      } else {
+        cur[b] = end[b] = 0;
+      }
+    }
+    bool exists = false;
+    i64 m = 0;
+    for (;;) {
+      // smallest current val across batches (vals are 8-byte i64 datums;
+      // within-batch order is the canonical LE-u64 order — equality is
+      // what matters here, u64 order is a valid merge order)
+      bool any = false;
+      u64 vmin = 0;
+      for (int b = 0; b < bl.n; b++) {
+        if (cur[b] >= end[b]) continue;
+        u64 v;
+        memcpy(&v, bl.b[b].vals + (u64)cur[b] * 8, 8);
+        if (!any || v < vmin) vmin = v;
+        any = true;
+      }
+      if (!any) break;
+      i64 net = 0;
+      for (int b = 0; b < bl.n; b++) {
+        while (cur[b] < end[b]) {
+          u64 v;
+          memcpy(&v, bl.b[b].vals + (u64)cur[b] * 8, 8);
+          if (v != vmin) break;
+          for (u32 q = bl.b[b].vu_off[cur[b]]; q < bl.b[b].vu_off[cur[b] + 1];
+               q++)
+            net = wadd(net, bl.b[b].diffs[q]);
+          cur[b]++;
+        }
+      }
+      if (net != 0) {
+        i64 v = (i64)vmin;
+        if (!exists || (is_max ? v > m : v < m)) m = v;
+        exists = true;
+      }
+    }
+    // state row: [key kw2][exists][value]
+    u64 idx = miss[g] ? base + misspos[g] : found[g];
+    u64 *row = st.rows + idx * st.stride_w;
+    u64 old_exists = row[kw2];
+    i64 old_m = (i64)row[kw2 + 1];
+    if ((old_exists != 0) == exists && (!exists || old_m == m)) continue;
+    if (old_exists) {
+      u64 o = atomicAdd(ocount, 1ull);
+      for (u32 w = 0; w < out_kw && w < kw2; w++)
+        okeys[o * out_kw + w] = key[w];
+      if (out_kw == kw2) okeys[o * out_kw + (kw2 - 1)] = key[kw2 - 1] % bucket_next;
+      memcpy(ovals + o * 8, &old_m, 8);
+      otimes[o] = t;
+      odiffs[o] = -1;
+    }
+    if (exists) {
+      u64 o = atomicAdd(ocount, 1ull);
+      for (u32 w = 0; w < out_kw && w < kw2; w++)
+        okeys[o * out_kw + w] = key[w];
+      if (out_kw == kw2) okeys[o * out_kw + (kw2 - 1)] = key[kw2 - 1] % bucket_next;
+      memcpy(ovals + o * 8, &m, 8);
+      otimes[o] = t;
+      odiffs[o] = 1;
+    }
+    row[kw2] = exists ? 1 : 0;
+    row[kw2 + 1] = (u64)m;
+  }
+}
+
 // ================================================================== host
 
 namespace {
@@ -1480,6 +1583,34 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
 
 // Consolidate raw updates + build + push, in one call (no intermediate
 // out-batch, copies, or extra syncs).
+// consolidate + build + push over already-staged device updates; returns
+// a pointer to the pushed batch (valid until the next spine merge).
+static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
+                                u64 lower, u64 upper) {
+  auto &S = ctx->scratch;
+  u32 kw = a->schema.kw, vb = a->schema.vb;
+  u64 capn = std::max<u64>(d.n, 1);
+  u64 *ok = dnew<u64>(ctx, capn * kw);
+  u8 *ov = (u8 *)dmalloc(ctx, std::max<u64>(capn * vb, 1));
+  u64 *ot = dnew<u64>(ctx, capn);
+  i64 *od = dnew<i64>(ctx, capn);
+  u64 *dcounts = (u64 *)S.get(3 * 8);
+  consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, lower,
+                                upper, dcounts);
+  u64 cnt[3] = {0, 0, 0};
+  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  b.n_upds = cnt[0];
+  b.n_keys = cnt[1];
+  b.n_vals = cnt[2];
+  a->batches.push_back(b);
+  a->upper = std::max(a->upper, upper);
+  spine_policy(ctx, a);
+  return &a->batches.back();
+}
+
 int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
                       const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
@@ -1982,6 +2113,153 @@ int mz_gpu_partition(mz_gpu_ctx *c, const mz_gpu_schema *s,
 
 uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words) {
   return route_hash(key_words, n_words);
+}
+
+// ----------------------------------------------- hierarchical min/max op
+struct mz_gpu_minmax {
+  u32 kw;
+  int is_max;
+  std::vector<u32> buckets;
+  std::vector<mz_gpu_arr> levels;
+  std::vector<RedState> states;
+  std::vector<u64> n_rows;
+};
+
+mz_gpu_minmax *mz_gpu_minmax_create(mz_gpu_ctx *c, const mz_gpu_schema *in,
+                                    int is_max, const uint32_t *buckets,
+                                    uint32_t n_levels) {
+  Ctx *ctx = &c->impl;
+  auto *op = new mz_gpu_minmax();
+  op->kw = in->key_words;
+  op->is_max = is_max;
+  op->buckets.assign(buckets, buckets + n_levels);
+  op->levels.resize(n_levels);
+  op->states.resize(n_levels);
+  op->n_rows.assign(n_levels, 0);
+  u32 kw2 = op->kw + 1;
+  for (u32 l = 0; l < n_levels; l++) {
+    op->levels[l].schema = {kw2, 8};
+    RedState &st = op->states[l];
+    u64 cap = 1ull << 21;
+    u64 slots = 2 * cap;
+    st.hash = dnew<u64>(ctx, slots * (kw2 + 1));
+    st.slots = slots;
+    st.stride_w = kw2 + 2;  // [key][exists][value]
+    st.rows = dnew<u64>(ctx, cap * st.stride_w);
+    st.capacity = cap;
+    hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
+                       ctx->stream, st.hash, slots, kw2);
+  }
+  return op;
+}
+
+int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
+                       const mz_gpu_updates *u, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  if (u->upper > u->lower + 1) {
+    ctx->err = "minmax_push: single-timestamp batches only";
+    return -1;
+  }
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u64 t = u->lower;
+  u32 kw = op->kw, kw2 = kw + 1;
+  u32 L = (u32)op->buckets.size();
+  DevUpdates d0 = stage_updates(ctx, u, kw, 8);
+  // level-0 input keys: (key, b0(val))
+  u64 n = d0.n;
+  u64 *bkeys = (u64 *)S.get(std::max<u64>(n, 1) * kw2 * 8);
+  if (n)
+    hipLaunchKernelGGL(k_bucket_keys, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d0.keys, kw, d0.vals, n,
+                       op->buckets[0], bkeys);
+  DevUpdates cur{bkeys, d0.vals, d0.times, d0.diffs, n};
+  // ping-pong correction buffers (owned, freed at the end)
+  std::vector<void *> owned;
+  u64 *fk = nullptr;
+  u8 *fv = nullptr;
+  u64 *ft = nullptr;
+  i64 *fd = nullptr;
+  u64 fM = 0;
+  for (u32 l = 0; l < L && cur.n; l++) {
+    mz_gpu_arr *A = &op->levels[l];
+    DevBatch *nb = arr_insert_dev(ctx, A, cur, t, t + 1);
+    u64 G = nb->n_keys;
+    if (G == 0) {
+      cur.n = 0;
+      break;
+    }
+    // 3-phase state upsert over the changed groups (batch keys = the
+    // sorted distinct changed keys)
+    u32 *gstart = (u32 *)S.get(G * 4);
+    hipLaunchKernelGGL(k_iota, dim3(ngrid(G)), dim3(BLK), 0, ctx->stream,
+                       gstart, G);
+    u32 *found = (u32 *)S.get(G * 4);
+    u32 *miss = (u32 *)S.get(G * 4);
+    hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, nb->keys, kw2, gstart, G,
+                       op->states[l], found, miss);
+    u32 *misspos = (u32 *)S.get((G + 1) * 4);
+    u64 Mn = exclusive_scan_u32(ctx, miss, misspos, G);
+    if (op->n_rows[l] + Mn > op->states[l].capacity) {
+      ctx->err = "minmax state capacity exceeded";
+      return -1;
+    }
+    if (Mn)
+      hipLaunchKernelGGL(k_red_insert, dim3(ngrid(G)), dim3(BLK), 0,
+                         ctx->stream, nb->keys, kw2, gstart, G, miss,
+                         misspos, op->n_rows[l], op->states[l]);
+    // output buffer for this level's corrections
+    u32 out_kw = (l + 1 < L) ? kw2 : kw;
+    u32 bnext = (l + 1 < L) ? op->buckets[l + 1] : 1;
+    u64 cap_out = 2 * G + 16;
+    u64 *pk = dnew<u64>(ctx, cap_out * out_kw);
+    u8 *pv = (u8 *)dmalloc(ctx, cap_out * 8);
+    u64 *pt = dnew<u64>(ctx, cap_out);
+    i64 *pd = dnew<i64>(ctx, cap_out);
+    owned.insert(owned.end(), {(void *)pk, (void *)pv, (void *)pt,
+                               (void *)pd});
+    unsigned long long *ocount = (unsigned long long *)S.get(8);
+    HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+    BatchList bl;
+    bl.n = 0;
+    for (auto &b : A->batches)
+      if (b.n_upds && bl.n < 12) bl.b[bl.n++] = b;
+    hipLaunchKernelGGL(k_minmax_apply, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, nb->keys, G, kw2, bl, op->is_max,
+                       op->states[l], found, miss, misspos, op->n_rows[l],
+                       out_kw, bnext, t, pk, pv, pt, pd, ocount);
+    op->n_rows[l] += Mn;
+    unsigned long long M;
+    HIP_CHECK(hipMemcpyAsync(&M, ocount, 8, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    cur = DevUpdates{pk, pv, pt, pd, M};
+    if (l + 1 == L) {
+      fk = pk;
+      fv = pv;
+      ft = pt;
+      fd = pd;
+      fM = M;
+    }
+  }
+  // consolidate the top-level corrections
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 Mc = 0;
+  DevUpdates fin{fk, fv, ft, fd, fM};
+  consolidate_dev(ctx, kw, 8, fin, &ok, &ov, &ot, &od, &Mc);
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  for (void *p : owned) dfree(ctx, p);
+  *out = make_out(ok, ov, ot, od, Mc, kw, 8);
+  return 0;
+}
+
+void mz_gpu_minmax_drop(mz_gpu_ctx *c, mz_gpu_minmax *op) {
+  (void)c;
+  (void)op;
 }
 
 // ---- numeric debug probes (test support; not part of the drop-in surface)

@@ -56,6 +56,7 @@ using u64 = uint64_t;
 using i64 = int64_t;
 using i128 = __int128;
 using u128 = unsigned __int128;
+static constexpr int MAX_KW = 2;
 
 namespace {
 
@@ -647,6 +648,22 @@ struct RedOp {
   std::map<std::vector<u64>, AccumRow> state;
 };
 
+// ------------------------------------------------- hierarchical min/max
+// Restates build_bucketed/build_bucketed_stage (reduce.rs:850-1224) +
+// ReductionMonoid (:2273): a reduction tree over val-hash buckets; each
+// stage recomputes the min/max of a changed (key, bucket) group from its
+// input arrangement (retraction-safe), emitting corrections that feed the
+// next stage. Buckets follow plan/reduce.rs:319-326 (val hash % buckets).
+// Vals are single i64 datums (8 bytes); single-timestamp pushes.
+struct MinMaxOp {
+  u32 kw;        // user key words
+  bool is_max;
+  std::vector<u32> buckets;         // e.g. {4096, 256, 16, 1}
+  std::vector<Arr> levels;          // level l keyed (key, bucket_l)
+  // last emitted value per (key..., bucket) per level
+  std::vector<std::map<std::vector<u64>, i64>> state;
+};
+
 // ------------------------------------------------------------- context
 
 struct Out {
@@ -665,6 +682,8 @@ struct orc_ctx {
 };
 
 extern "C" {
+
+u64 orc_route_hash(const u64 *kw, u32 n);
 
 orc_ctx *orc_init() { return new orc_ctx(); }
 void orc_fini(orc_ctx *c) { delete c; }
@@ -981,6 +1000,117 @@ void orc_out_release(orc_ctx *c, mz_gpu_out *o) {
 
 // Routing hash: splitmix64 over the key words (substitute for fixed-seed
 // ahash, used identically by oracle and GPU — DESIGN.md §2.2).
+MinMaxOp *orc_minmax_create(orc_ctx *c, const mz_gpu_schema *in, int is_max,
+                            const u32 *buckets, u32 n_levels) {
+  auto *op = new MinMaxOp();
+  op->kw = in->key_words;
+  op->is_max = is_max;
+  op->buckets.assign(buckets, buckets + n_levels);
+  op->levels.resize(n_levels);
+  op->state.resize(n_levels);
+  for (u32 l = 0; l < n_levels; l++)
+    op->levels[l].schema = {op->kw + 1, 8};  // (key,bucket) -> val i64
+  return op;
+}
+
+// One push of (key, val i64) updates at a single timestamp; returns the
+// top-level corrections (key, minmax val, +/-1).
+int orc_minmax_push(orc_ctx *c, MinMaxOp *op, const mz_gpu_updates *u,
+                    mz_gpu_out **out) {
+  u32 kw = op->kw;
+  if (u->upper > u->lower + 1) return -1;  // single-timestamp contract
+  u64 t = u->lower;
+  u32 L = (u32)op->buckets.size();
+  // build level-0 input: (key, b0(val)) -> val
+  Cols cur;
+  for (u64 i = 0; i < u->n; i++) {
+    u64 kb[MAX_KW + 1];
+    for (u32 w = 0; w < kw; w++) kb[w] = u->keys[i * kw + w];
+    u64 vword;
+    std::memcpy(&vword, u->vals + i * 8, 8);
+    kb[kw] = orc_route_hash(&vword, 1) % op->buckets[0];
+    cur.push(kb, kw + 1, u->vals + i * 8, 8, t, u->diffs[i]);
+  }
+  Cols final_out;
+  for (u32 l = 0; l < L; l++) {
+    Arr &A = op->levels[l];
+    Schema ls = A.schema;
+    consolidate(ls, cur);
+    if (cur.size() == 0) break;
+    // insert into the level arrangement
+    auto b = std::make_unique<Batch>();
+    b->cols = cur;  // copy
+    b->lower = t;
+    b->upper = t + 1;
+    b->build_index(ls);
+    A.batches.push_back(std::move(b));
+    while (A.batches.size() >= 2) {
+      size_t nb = A.batches.size();
+      if (A.batches[nb - 2]->cols.size() <= 2 * A.batches[nb - 1]->cols.size())
+        A.merge_span(nb - 2, nb);
+      else
+        break;
+    }
+    // changed groups = distinct keys of cur; recompute each from A
+    Cols next;
+    std::map<std::vector<u64>, i64> &st = op->state[l];
+    size_t n = cur.size();
+    for (size_t i = 0; i < n;) {
+      size_t j = i;
+      while (j < n && cmp_key(cur.keys.data() + i * ls.kw,
+                              cur.keys.data() + j * ls.kw, ls.kw) == 0)
+        j++;
+      const u64 *gkey = cur.keys.data() + i * ls.kw;
+      // group scan over all batches: per val net diff
+      std::map<i64, i64> valsum;
+      for (auto &bb : A.batches) {
+        auto r = bb->seek(ls, gkey);
+        for (size_t q = r.first; q < r.second; q++) {
+          i64 v;
+          std::memcpy(&v, bb->cols.vals.data() + q * 8, 8);
+          valsum[v] = (i64)((u64)valsum[v] + (u64)bb->cols.diffs[q]);
+        }
+      }
+      bool exists = false;
+      i64 m = 0;
+      for (auto &[v, d] : valsum) {
+        if (d == 0) continue;
+        if (!exists || (op->is_max ? v > m : v < m)) m = v;
+        exists = true;
+      }
+      std::vector<u64> skey(gkey, gkey + ls.kw);
+      auto it = st.find(skey);
+      bool old_exists = it != st.end();
+      i64 old_m = old_exists ? it->second : 0;
+      if (old_exists != exists || (exists && old_m != m)) {
+        // corrections keyed for the next level (or the user key at top)
+        u64 okb[MAX_KW + 1];
+        for (u32 w = 0; w < kw; w++) okb[w] = gkey[w];
+        u32 okw = (l + 1 < L) ? kw + 1 : kw;
+        if (l + 1 < L) okb[kw] = gkey[kw] % op->buckets[l + 1];
+        if (old_exists)
+          next.push(okb, okw, (const u8 *)&old_m, 8, t, -1);
+        if (exists) {
+          next.push(okb, okw, (const u8 *)&m, 8, t, 1);
+          st[skey] = m;
+        } else {
+          st.erase(skey);
+        }
+      }
+      i = j;
+    }
+    if (l + 1 < L) {
+      cur = std::move(next);
+    } else {
+      final_out = std::move(next);
+    }
+  }
+  Schema os = {kw, 8};
+  consolidate(os, final_out);
+  *out = &make_out(std::move(final_out), os)->pub_;
+  return 0;
+}
+
 u64 orc_route_hash(const u64 *kw, u32 n) {
   u64 h = 0x9E3779B97F4A7C15ULL;
   for (u32 i = 0; i < n; i++) {

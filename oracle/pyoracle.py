@@ -58,6 +58,13 @@ def _load():
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
     lib.orc_out_release.argtypes = [C.c_void_p, C.POINTER(OutBatch)]
+    lib.orc_minmax_create.restype = C.c_void_p
+    lib.orc_minmax_create.argtypes = [C.c_void_p, C.POINTER(Schema),
+                                      C.c_int, C.POINTER(C.c_uint32),
+                                      C.c_uint32]
+    lib.orc_minmax_push.argtypes = [C.c_void_p, C.c_void_p,
+                                    C.POINTER(Updates),
+                                    C.POINTER(C.POINTER(OutBatch))]
     lib.orc_peek.argtypes = [C.c_void_p, C.c_void_p,
                              C.POINTER(C.c_uint64), C.c_uint64, C.c_uint64,
                              C.POINTER(C.POINTER(OutBatch))]
@@ -198,6 +205,19 @@ class OracleCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.orc_route_hash(arr, len(words))
+
+    def minmax_create(self, in_schema, is_max, buckets):
+        arr = (C.c_uint32 * len(buckets))(*buckets)
+        return self.lib.orc_minmax_create(self.ctx, C.byref(in_schema),
+                                          1 if is_max else 0, arr,
+                                          len(buckets))
+
+    def minmax_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_minmax_push(self.ctx, op, C.byref(upd),
+                                      C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
 
     def peek(self, arr, keys, time, kw=1):
         import numpy as np
