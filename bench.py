@@ -119,39 +119,71 @@ def filter_shard(churn, world, rank):
     return out
 
 
-def run_cpu_baseline(seed):
+def run_cpu_baseline(seed, cores=0, batch=100_000, steps=4, sf=1.0):
     """Oracle (the CPU restatement, kind 'port') on a bounded sample of the
-    SAME workload shape as the GPU run: TPC-H SF1 Q3, 100k-row churn
-    batches, single thread. The sample is a handful of batches (~10-30s of
-    CPU work including per-batch arrangement maintenance)."""
+    SAME workload shape as the GPU run: TPC-H SF1 Q3 churn batches on the
+    box's host cores — sharded oracle dataflows on worker threads with an
+    in-process exchange (the thread analog of timely workers; the oracle's
+    C calls release the GIL). Reports the better of 1 thread and N
+    threads (small batches are fixed-cost bound and favor 1)."""
+    import copy
+    import threading
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     from materialize_amd.tpch import TpchGen
-    from materialize_amd.workloads import Q3Dataflow
+    from materialize_amd.thread_exchange import ThreadExchangeGroup
+    from materialize_amd.workloads import ShardedQ3Dataflow
     from pyoracle import OracleCtx
-    sf, batch, steps = 1.0, 100_000, 4
+
     gen = TpchGen(sf=sf, seed=seed)
-    df = Q3Dataflow(OracleCtx())
-    df.load(gen)
+    base = copy.deepcopy(gen.__dict__)  # t=0 snapshot for load()
     churns = [gen.churn(batch) for _ in range(steps + 1)]
-    rows, corr = df.step(churns[0], 1)  # warmup
-    if corr is not None:
-        corr.release()
-    t0 = time.perf_counter()
-    total = 0
-    for i in range(1, steps + 1):
-        r, corr = df.step(churns[i], i + 1)
-        if corr is not None:
-            corr.release()
-        total += r
-    dt = time.perf_counter() - t0
+    gen.__dict__.update(base)  # back to the t=0 state the churns start from
+
+    def run(W):
+        group = ThreadExchangeGroup(W)
+        results = [None] * W
+
+        def worker(rank):
+            df = ShardedQ3Dataflow(OracleCtx(), group.member(rank))
+            df.load(gen)
+            r, corr = df.step(churns[0], 1)  # warmup
+            if corr is not None:
+                corr.release()
+            group.barrier.wait()
+            t0 = time.perf_counter()
+            total = 0
+            for i in range(1, steps + 1):
+                r, corr = df.step(churns[i], i + 1)
+                if corr is not None:
+                    corr.release()
+                total += r
+            group.barrier.wait()
+            results[rank] = (total, time.perf_counter() - t0)
+
+        threads = [threading.Thread(target=worker, args=(r,))
+                   for r in range(W)]
+        for th in threads:
+            th.start()
+        for th in threads:
+            th.join()
+        total = results[0][0]
+        dt = max(r[1] for r in results)
+        return total / dt, dt
+
+    W = cores or min(os.cpu_count() or 1, 8)
+    v1, d1 = run(1)
+    vw, dw = (v1, d1) if W == 1 else run(W)
+    best, used, dt = (vw, W, dw) if vw >= v1 else (v1, 1, d1)
     return {
-        "value": total / dt,
+        "value": best,
         "unit": "rows/s",
-        "cores": 1,
+        "cores": used,
         "kind": "port",
         "sample": (f"oracle C++ restatement, TPC-H SF{sf:g} Q3, {steps} "
-                   f"churn batches of ~{batch} rows, single thread "
-                   f"({dt:.1f}s of CPU work in the timed sample)"),
+                   f"churn batches of ~{batch} rows; best of 1 thread "
+                   f"({v1:.0f} rows/s) and {W} sharded worker threads "
+                   f"({vw:.0f} rows/s) with in-process exchange "
+                   f"({dt:.1f}s timed)"),
     }
 
 

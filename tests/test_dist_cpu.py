@@ -89,3 +89,64 @@ def test_sharded_q3_gloo_world2():
     merged = {**s0, **s1}
     assert merged == want
     assert len(want) > 0
+
+
+def test_sharded_q3_threads_world2():
+    """Same invariant via the in-process ThreadExchange (the CPU-baseline
+    transport): union of 2 thread-shards == unsharded."""
+    import threading
+    from materialize_amd.thread_exchange import ThreadExchangeGroup
+    from materialize_amd.workloads import Q3Dataflow, ShardedQ3Dataflow
+    from materialize_amd.tpch import TpchGen
+    from pyoracle import OracleCtx
+    df = Q3Dataflow(OracleCtx())
+    gen = TpchGen(sf=0.002, seed=7)
+    want = _final_state(df, gen, steps=3, batch_rows=300)
+
+    import copy
+    gen2 = TpchGen(sf=0.002, seed=7)
+    base = copy.deepcopy(gen2.__dict__)
+    churns = [gen2.churn(300) for _ in range(3)]
+    gen2.__dict__.update(base)
+    group = ThreadExchangeGroup(2)
+    states = [None, None]
+
+    def worker(rank):
+        ctx = OracleCtx()
+        dfr = ShardedQ3Dataflow(ctx, group.member(rank))
+        holder = []
+        orig = dfr.reduce.push
+
+        def capture(u):
+            o = orig(u)
+            holder.append(o.to_host())
+            return o
+
+        dfr.reduce.push = capture
+        dfr.load(gen2)
+        for t in range(1, 4):
+            dfr.step(churns[t - 1], t)
+        st = {}
+        for keys, vals, times, diffs in holder:
+            n = len(times)
+            vals = vals.reshape(n, 24) if n else vals
+            for i in range(n):
+                k = (int(keys[2 * i]), int(np.uint64(keys[2 * i + 1])))
+                lo = int(vals[i][8:16].view(np.uint64)[0])
+                hi = int(vals[i][16:24].view(np.int64)[0])
+                v = hi * 2**64 + lo
+                if int(diffs[i]) == 1:
+                    st[k] = v
+                else:
+                    assert st.pop(k) == v
+        states[rank] = st
+
+    ths = [threading.Thread(target=worker, args=(r,)) for r in range(2)]
+    for t in ths:
+        t.start()
+    for t in ths:
+        t.join(timeout=300)
+    s0, s1 = states
+    assert s0 is not None and s1 is not None
+    assert not (set(s0) & set(s1))
+    assert {**s0, **s1} == want
