@@ -254,6 +254,14 @@ int  mz_gpu_partition(mz_gpu_ctx *ctx, const mz_gpu_schema *schema,
                       uint64_t *out_times, int64_t *out_diffs,
                       uint64_t *counts);
 
+/* -------------------------------------------------------------- flat map
+ * FlatMap / key-preparation analog (src/compute/src/render/flat_map.rs;
+ * DeltaJoinKeyPreparation, delta_join.rs:444-464): apply a closure to a
+ * stream without a lookup (VAL_STREAM = the input val). */
+int  mz_gpu_map(mz_gpu_ctx *ctx, const mz_gpu_schema *in,
+                const mz_gpu_updates *updates, const mz_gpu_closure *cl,
+                mz_gpu_out **out);
+
 /* --------------------------------------------------- hierarchical reduce
  * Replaces build_bucketed/build_monotonic + ReductionMonoid
  * (src/compute/src/render/reduce.rs:850-1224, :2273): MIN/MAX maintained

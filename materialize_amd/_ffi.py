@@ -93,6 +93,9 @@ def load():
         C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
         C.POINTER(C.c_int64), C.POINTER(C.c_uint64)]
+    lib.mz_gpu_map.argtypes = [C.c_void_p, C.POINTER(Schema),
+                               C.POINTER(Updates), C.POINTER(Closure),
+                               C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_minmax_create.restype = C.c_void_p
     lib.mz_gpu_minmax_create.argtypes = [C.c_void_p, C.POINTER(Schema),
                                          C.c_int, C.POINTER(C.c_uint32),
@@ -217,6 +220,13 @@ class GpuCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.mz_gpu_route_hash(arr, len(words))
+
+    def map(self, in_schema, upd, cl):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_map(self.ctx, C.byref(in_schema),
+                                        C.byref(upd), C.byref(cl),
+                                        C.byref(outp)))
+        return self._take(outp)
 
     def minmax_create(self, in_schema, is_max, buckets):
         arr = (C.c_uint32 * len(buckets))(*buckets)

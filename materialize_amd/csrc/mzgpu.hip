@@ -2115,6 +2115,75 @@ uint64_t mz_gpu_route_hash(const uint64_t *key_words, uint32_t n_words) {
   return route_hash(key_words, n_words);
 }
 
+// ------------------------------------------------------------------ map
+// FlatMap/key-preparation analog (render/flat_map.rs; the
+// DeltaJoinKeyPreparation map at delta_join.rs:444-464): apply a closure
+// (filters + field map) to each update row, no lookup. VAL_STREAM = the
+// input val; VAL_LOOKUP is absent.
+__global__ void k_map_count(const u64 *keys, u32 kw, const u8 *vals, u32 vb,
+                            u64 n, const mz_gpu_closure cl, u32 *flags) {
+  GRID_STRIDE(i, n) {
+    flags[i] = d_closure_apply(&cl, keys + i * kw,
+                               vals ? vals + i * vb : nullptr, nullptr,
+                               nullptr, nullptr)
+                   ? 1u
+                   : 0u;
+  }
+}
+
+__global__ void k_map_emit(const u64 *keys, u32 kw, const u8 *vals, u32 vb,
+                           const u64 *times, const i64 *diffs, u64 n,
+                           const mz_gpu_closure cl, const u32 *flags,
+                           const u32 *pos, u64 *okeys, u8 *ovals,
+                           u64 *otimes, i64 *odiffs) {
+  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
+  GRID_STRIDE(i, n) {
+    if (!flags[i]) continue;
+    u64 o = pos[i];
+    u64 okey[MAX_KW];
+    u8 oval[MAX_VB];
+    d_closure_apply(&cl, keys + i * kw, vals ? vals + i * vb : nullptr,
+                    nullptr, okey, oval);
+    for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
+    for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oval[c];
+    otimes[o] = times[i];
+    odiffs[o] = diffs[i];
+  }
+}
+
+int mz_gpu_map(mz_gpu_ctx *c, const mz_gpu_schema *in,
+               const mz_gpu_updates *u, const mz_gpu_closure *cl,
+               mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = in->key_words, vb = in->val_bytes;
+  u32 okw = cl->out.key_words, ovb = cl->out.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 n = d.n;
+  if (n == 0) {
+    *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                    dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, okw, ovb);
+    return 0;
+  }
+  u32 *flags = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_map_count, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.keys, kw, d.vals, vb, n, *cl, flags);
+  u32 *pos = (u32 *)S.get((n + 1) * 4);
+  u64 M = exclusive_scan_u32(ctx, flags, pos, n);
+  u64 *pk = dnew<u64>(ctx, std::max<u64>(M, 1) * okw);
+  u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(M * ovb, 1));
+  u64 *pt = dnew<u64>(ctx, std::max<u64>(M, 1));
+  i64 *pd = dnew<i64>(ctx, std::max<u64>(M, 1));
+  if (M)
+    hipLaunchKernelGGL(k_map_emit, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                       d.keys, kw, d.vals, vb, d.times, d.diffs, n, *cl,
+                       flags, pos, pk, pv, pt, pd);
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  *out = make_out(pk, pv, pt, pd, M, okw, ovb);
+  return 0;
+}
+
 // ----------------------------------------------- hierarchical min/max op
 struct mz_gpu_minmax {
   u32 kw;

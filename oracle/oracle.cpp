@@ -881,6 +881,25 @@ int orc_halfjoin(orc_ctx *c, Arr *lookup, const mz_gpu_updates *u,
   return 0;
 }
 
+// flat-map mirror: closure over a stream, no lookup
+int orc_map(orc_ctx *c, const mz_gpu_schema *in, const mz_gpu_updates *u,
+            const mz_gpu_closure *cl, mz_gpu_out **out) {
+  Schema s = {in->key_words, in->val_bytes};
+  Schema os = {cl->out.key_words, cl->out.val_bytes};
+  Cols result;
+  std::vector<u64> okey(os.kw);
+  std::vector<u8> oval(os.vb ? os.vb : 1);
+  for (u64 i = 0; i < u->n; i++) {
+    const u64 *key = u->keys + i * s.kw;
+    const u8 *val = s.vb ? u->vals + i * s.vb : nullptr;
+    if (closure_apply(cl, key, val, nullptr, okey.data(), oval.data()))
+      result.push(okey.data(), os.kw, oval.data(), os.vb, u->times[i],
+                  u->diffs[i]);
+  }
+  *out = &make_out(std::move(result), os)->pub_;
+  return 0;
+}
+
 // peek mirror (handle_peek analog): per requested key, (val, summed
 // diff) as of `time` via the le half-join with the identity closure.
 int orc_peek(orc_ctx *c, Arr *arr, const u64 *keys, u64 n_keys, u64 time,

@@ -58,6 +58,9 @@ def _load():
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
     lib.orc_out_release.argtypes = [C.c_void_p, C.POINTER(OutBatch)]
+    lib.orc_map.argtypes = [C.c_void_p, C.POINTER(Schema),
+                            C.POINTER(Updates), C.POINTER(Closure),
+                            C.POINTER(C.POINTER(OutBatch))]
     lib.orc_minmax_create.restype = C.c_void_p
     lib.orc_minmax_create.argtypes = [C.c_void_p, C.POINTER(Schema),
                                       C.c_int, C.POINTER(C.c_uint32),
@@ -205,6 +208,13 @@ class OracleCtx:
         arr = (C.c_uint64 * len(words))(*[w & 0xFFFFFFFFFFFFFFFF
                                           for w in words])
         return self.lib.orc_route_hash(arr, len(words))
+
+    def map(self, in_schema, upd, cl):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_map(self.ctx, C.byref(in_schema), C.byref(upd),
+                              C.byref(cl), C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
 
     def minmax_create(self, in_schema, is_max, buckets):
         arr = (C.c_uint32 * len(buckets))(*buckets)
