@@ -1539,13 +1539,23 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
 // whenever the invariant breaks costs O(log) amortized merge work per
 // update and keeps the probe fan-out at ~log(arrangement/batch).
 static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
-  while (a->batches.size() >= 2) {
+  // Large batches keep the geometric pair rule (merge-path pair merges
+  // are O(n)); small batches pool lazily and merge k-way when the pool
+  // exceeds 6 — per-merge fixed overhead (~25 kernel launches + syncs)
+  // made per-step pair merges of 100k-row batches the dominant step cost.
+  constexpr u64 SMALL = 1u << 20;
+  for (;;) {
     size_t nb = a->batches.size();
-    if (a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds)
+    if (nb >= 2 && a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds
+        && a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL)
       merge_range(ctx, a, nb - 2, nb);
     else
       break;
   }
+  size_t nb = a->batches.size();
+  size_t i = nb;
+  while (i > 0 && a->batches[i - 1].n_upds < SMALL) i--;
+  if (nb - i > 6) merge_range(ctx, a, i, nb);
   while (a->batches.size() > 10)  // hard cap (probe BatchList capacity)
     merge_range(ctx, a, 0, a->batches.size());
 }
