@@ -309,7 +309,10 @@ def main():
 
     cpu_baseline = None
     if rank == 0 and N == 1 and not args.no_cpu_baseline:
-        cpu_baseline = run_cpu_baseline(args.seed)
+        # same batch size as the GPU run (bounded step count)
+        bsteps = 4 if batch_rows <= 200_000 else 2
+        cpu_baseline = run_cpu_baseline(args.seed, batch=batch_rows,
+                                        steps=bsteps, sf=sf)
 
     if rank == 0:
         line = {
