@@ -407,9 +407,13 @@ __device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
 // --------------------------------------------------------------- probe
 
 // A probe target list (snapshot of an arrangement's batches).
+// allpass[i]: every update time in batch i is strictly below the delta's
+// lower bound, so le/lt time filters are vacuously true and the output
+// time is the stream time — the batch's time column is never read.
 struct BatchList {
   int n;
   DevBatch b[12];
+  u8 allpass[12];
 };
 
 enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
@@ -417,10 +421,13 @@ enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 // Phase 1: count emitted pairs per delta update.
 // delta: n updates (keys/vals/times/diffs columns); swap: delta is input 2
 // of a linear join (closure arg order is (key, v1=input1, v2=input2)).
+// ki_cache[n * bl.n] stores each (row, batch) hash-lookup result so the
+// emit phase never re-reads the hash tables (≈half the probe traffic).
 __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
                               const u64 *dtimes, u64 n, u32 kw, u32 lvb,
                               BatchList bl, int mode, int swap,
-                              const mz_gpu_closure cl, u32 *count) {
+                              const mz_gpu_closure cl, u32 *count,
+                              int *ki_cache) {
   GRID_STRIDE(i, n) {
     const u64 *key = dkeys + i * kw;
     const u8 *dv = dvals ? dvals + i * dvb : nullptr;
@@ -429,6 +436,7 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
     for (int bi = 0; bi < bl.n; bi++) {
       const DevBatch &b = bl.b[bi];
       int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
+      ki_cache[i * bl.n + bi] = ki;
       if (ki < 0) continue;
       for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
@@ -436,7 +444,7 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
         const u8 *v2 = swap ? dv : lv;
         if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
         u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
-        if (mode == PM_JOIN) {
+        if (mode == PM_JOIN || bl.allpass[bi]) {
           c += hi - lo;
         } else {
           for (u32 u = lo; u < hi; u++) {
@@ -456,8 +464,9 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              const u64 *dtimes, const i64 *ddiffs, u64 n,
                              u32 kw, u32 lvb, BatchList bl, int mode,
                              int swap, const mz_gpu_closure cl,
-                             const u32 *offs /*exclusive*/, u64 *okeys,
-                             u8 *ovals, u64 *otimes, i64 *odiffs) {
+                             const u32 *offs /*exclusive*/,
+                             const int *ki_cache, u64 *okeys, u8 *ovals,
+                             u64 *otimes, i64 *odiffs) {
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   GRID_STRIDE(i, n) {
     const u64 *key = dkeys + i * kw;
@@ -467,7 +476,7 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
     u64 o = offs[i];
     for (int bi = 0; bi < bl.n; bi++) {
       const DevBatch &b = bl.b[bi];
-      int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
+      int ki = ki_cache[i * bl.n + bi];
       if (ki < 0) continue;
       for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
@@ -477,11 +486,14 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
         u8 oval[MAX_VB];
         if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
         for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
-          u64 t2 = b.times[u];
           u64 tout;
-          if (mode == PM_JOIN) {
+          if (bl.allpass[bi]) {
+            tout = t;  // t2 < delta lower <= t on every update
+          } else if (mode == PM_JOIN) {
+            u64 t2 = b.times[u];
             tout = t2 > t ? t2 : t;
           } else {
+            u64 t2 = b.times[u];
             if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
             tout = t;
           }
@@ -1756,6 +1768,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
       return probe_impl(ctx, lookup, u, stream_vb, mode, swap, cl,
                         consolidate_out, out);
     }
+    u64 tmax_excl = std::max(b.upper, lookup->logical_compaction + 1);
+    bl.allpass[bl.n] = tmax_excl <= u->lower ? 1 : 0;
     bl.b[bl.n++] = b;
   }
   u64 n = d.n;
@@ -1765,10 +1779,11 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     return 0;
   }
   u32 *count = (u32 *)S.get((n + 1) * 4);
+  int *ki_cache = (int *)S.get(n * (u64)bl.n * 4);
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   hipLaunchKernelGGL(k_probe_count, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
                      d.keys, d.vals, stream_vb, d.times, n, kw, lvb, bl, mode,
-                     swap, *cl, count);
+                     swap, *cl, count, ki_cache);
   u32 *offs = (u32 *)S.get((n + 1) * 4);
   u64 M = exclusive_scan_u32(ctx, count, offs, n);
   u64 *pk = dnew<u64>(ctx, std::max<u64>(M, 1) * okw);
@@ -1778,8 +1793,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   if (M)
     hipLaunchKernelGGL(k_probe_emit, dim3(ngrid(n)), dim3(BLK), 0,
                        ctx->stream, d.keys, d.vals, stream_vb, d.times,
-                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, offs, pk, pv,
-                       pt, pd);
+                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, offs,
+                       ki_cache, pk, pv, pt, pd);
   if (ctx->time_kernels) {
     HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
@@ -2303,7 +2318,10 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
     BatchList bl;
     bl.n = 0;
     for (auto &b : A->batches)
-      if (b.n_upds && bl.n < 12) bl.b[bl.n++] = b;
+      if (b.n_upds && bl.n < 12) {
+        bl.allpass[bl.n] = 0;
+        bl.b[bl.n++] = b;
+      }
     hipLaunchKernelGGL(k_minmax_apply, dim3(ngrid(G)), dim3(BLK), 0,
                        ctx->stream, nb->keys, G, kw2, bl, op->is_max,
                        op->states[l], found, miss, misspos, op->n_rows[l],
