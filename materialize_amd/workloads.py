@@ -156,10 +156,21 @@ class Q3Dataflow:
             out.release()
         return n_corr
 
+    def _advance_compaction(self, t):
+        """Logical compaction to one step back of the frontier
+        (mz_join_core.rs:461 via the delta path's step_back held frontier,
+        delta_join.rs:614-616): merges then cancel retract/insert pairs at
+        compacted times, keeping steady-state arrangements bounded, while
+        lt tie-breaks at data times >= t stay exact (frontier-1 < t)."""
+        if t >= 1:
+            for arr in self.arrs.values():
+                self.ctx.arr_set_logical_compaction(arr, t - 1)
+
     def step(self, churn, t):
         """Maintain one churn batch at time t. Returns (input_rows,
         corrections DevOut or None)."""
         ctx = self.ctx
+        self._advance_compaction(t)
         l_keys, l_vals, l_diffs = churn["lineitem"]
         o_keys, o_vals, o_diffs = churn["orders"]
         oc_keys, oc_vals, oc_diffs = churn["orders_by_cust"]
