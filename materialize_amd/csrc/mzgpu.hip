@@ -25,6 +25,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -1556,6 +1557,10 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // exceeds 6 — per-merge fixed overhead (~25 kernel launches + syncs)
   // made per-step pair merges of 100k-row batches the dominant step cost.
   constexpr u64 SMALL = 1u << 20;
+  static const long POOL = [] {
+    const char *e = getenv("MZ_GPU_SMALL_POOL");
+    return e ? atol(e) : 6;
+  }();
   for (;;) {
     size_t nb = a->batches.size();
     if (nb >= 2 && a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds
@@ -1567,7 +1572,7 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   size_t nb = a->batches.size();
   size_t i = nb;
   while (i > 0 && a->batches[i - 1].n_upds < SMALL) i--;
-  if (nb - i > 6) merge_range(ctx, a, i, nb);
+  if ((long)(nb - i) > POOL) merge_range(ctx, a, i, nb);
   while (a->batches.size() > 10)  // hard cap (probe BatchList capacity)
     merge_range(ctx, a, 0, a->batches.size());
 }
