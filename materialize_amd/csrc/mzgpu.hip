@@ -698,8 +698,11 @@ struct RedState {
 };
 
 // Phase A: lookup each key group in the (fully published) table.
+// G may be host-known (gidn == nullptr) or device-derived from gidn[m-1].
 __global__ void k_red_lookup(const u64 *keys, u32 kw, const u32 *gstart,
-                             u64 G, RedState st, u32 *found, u32 *miss) {
+                             u64 G, const u32 *gidn, u64 m, RedState st,
+                             u32 *found, u32 *miss) {
+  if (gidn) G = m ? gidn[m - 1] : 0;
   GRID_STRIDE(g, G) {
     const u64 *key = keys + (u64)gstart[g] * kw;
     int idx = hash_lookup(st.hash, st.slots, key, kw);
@@ -709,13 +712,22 @@ __global__ void k_red_lookup(const u64 *keys, u32 kw, const u32 *gstart,
 }
 
 // Phase B: insert missing keys; row idx = base + rank among misses.
+// base comes from *d_nrows when provided (bumped AFTER this launch);
+// overflow sets *d_err instead of writing out of bounds.
 __global__ void k_red_insert(const u64 *keys, u32 kw, const u32 *gstart,
-                             u64 G, const u32 *miss, const u32 *misspos,
-                             u64 base, RedState st) {
+                             u64 G, const u32 *gidn, u64 m,
+                             const u32 *miss, const u32 *misspos, u64 base,
+                             const u64 *d_nrows, u64 *d_err, RedState st) {
+  if (gidn) G = m ? gidn[m - 1] : 0;
+  if (d_nrows) base = *d_nrows;
   GRID_STRIDE(g, G) {
     if (!miss[g]) continue;
     const u64 *key = keys + (u64)gstart[g] * kw;
     u64 idx = base + misspos[g];
+    if (idx >= st.capacity) {
+      if (d_err) *d_err = 1;
+      continue;
+    }
     u64 *row = st.rows + idx * st.stride_w;
     for (u32 w = 0; w < kw; w++) row[w] = key[w];
     for (u32 w = kw; w < st.stride_w; w++) row[w] = 0;
@@ -738,19 +750,23 @@ __global__ void k_red_insert(const u64 *keys, u32 kw, const u32 *gstart,
 // corrections (new minus old finalized rows).
 __global__ void k_reduce_apply(const u64 *keys, const u8 *vals, u32 kw,
                                u32 vb, const i64 *diffs, const u32 *gstart,
-                               u64 G, u64 hi_row, u64 t, RedState st,
-                               const u32 *found, const u32 *miss,
-                               const u32 *misspos, u64 base,
+                               u64 G, const u32 *gidn, u64 hi_row, u64 t,
+                               RedState st, const u32 *found,
+                               const u32 *miss, const u32 *misspos,
+                               u64 base, const u64 *d_nrows,
                                mz_gpu_reduce_spec spec, u64 *okeys, u8 *ovals,
                                u64 *otimes, i64 *odiffs,
                                unsigned long long *ocount) {
   u32 na = spec.n_aggs;
   u32 ovb = spec.out.val_bytes;
+  if (gidn) G = hi_row ? gidn[hi_row - 1] : 0;
+  if (d_nrows) base = *d_nrows;
   GRID_STRIDE(g, G) {
     u64 lo = gstart[g];
     u64 end = (g + 1 < G) ? gstart[g + 1] : hi_row;
     const u64 *key = keys + lo * kw;
     u64 idx = miss[g] ? base + misspos[g] : found[g];
+    if (idx >= st.capacity) continue;  // overflow flagged by insert
     u64 *row = st.rows + idx * st.stride_w;
     i64 *total_p = (i64 *)(row + kw);
     Acc5 *accs = (Acc5 *)(row + kw + 1);
@@ -995,7 +1011,9 @@ struct mz_gpu_red {
   mz_gpu_reduce_spec spec;
   RedState st;
   u64 capacity;
-  u64 n_rows = 0;  // host-tracked row count (insert order is deterministic)
+  u64 n_rows = 0;   // host mirror (minmax path); the reduce path keeps the
+  u64 *d_nrows = nullptr;  // authoritative count on device (no readback)
+  u64 *d_err = nullptr;
 };
 
 namespace {
@@ -1163,6 +1181,14 @@ void exclusive_scan_u32_ns(Ctx *c, const u32 *in, u32 *out, u64 n) {
 }
 
 __global__ void k_write_u64(u64 *p, u64 v) { *p = v; }
+// bump a device counter by pos[G] (the miss total) — G from gidn[m-1]
+__global__ void k_bump_ctr(u64 *ctr, const u32 *pos, const u32 *gidn,
+                           u64 m) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    u64 G = m ? gidn[m - 1] : 0;
+    *ctr += pos[G];
+  }
+}
 
 void inclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   auto &S = c->scratch;
@@ -1908,6 +1934,10 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
   r->st.stride_w = kw + 1 + 6 * spec->n_aggs;
   r->st.rows = dnew<u64>(ctx, cap * r->st.stride_w);
   r->st.capacity = cap;
+  r->d_nrows = dnew<u64>(ctx, 1);
+  r->d_err = dnew<u64>(ctx, 1);
+  HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
   hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
                      ctx->stream, r->st.hash, slots, kw);
   c->impl.reds.push_back(r);
@@ -2026,59 +2056,61 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
   HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
   u32 *flags = (u32 *)S.get(n * 4);
   u32 *gid = (u32 *)S.get(n * 4);
+  // zero-fill the corrections buffer: unwritten capacity rows carry diff 0
+  // and drop out in the consolidation — no emitted-count readback needed
+  HIP_CHECK(hipMemsetAsync(pk, 0, cap_out * okw * 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(cap_out * ovb, 1),
+                           ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pt, 0, cap_out * 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pd, 0, cap_out * 8, ctx->stream));
   for (auto [lo, hi] : slices) {
     u64 m = hi - lo;
-    // key-group starts within the slice
+    // key-group starts within the slice (group count stays on device)
     hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(m)), dim3(BLK), 0,
                        ctx->stream, sk + lo * kw, kw, stm + lo, flags, m);
     inclusive_scan_u32(ctx, flags, gid, m);
-    u32 G;
-    HIP_CHECK(hipMemcpyAsync(&G, gid + m - 1, 4, hipMemcpyDeviceToHost,
-                             ctx->stream));
-    HIP_CHECK(hipStreamSynchronize(ctx->stream));
-    u32 *starts = (u32 *)S.get((u64)G * 4);
+    u32 *starts = (u32 *)S.get(m * 4);
     hipLaunchKernelGGL(k_group_starts, dim3(ngrid(m)), dim3(BLK), 0,
                        ctx->stream, flags, gid, starts, m);
     // Phase A: lookup; Phase B: insert misses (separate launches for
-    // coherence — see RedState docs); Phase C: apply + emit.
-    u32 *found = (u32 *)S.get((u64)G * 4);
-    u32 *miss = (u32 *)S.get((u64)G * 4);
-    hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(G)), dim3(BLK), 0,
-                       ctx->stream, sk + lo * kw, kw, starts, (u64)G, op->st,
-                       found, miss);
-    u32 *misspos = (u32 *)S.get(((u64)G + 1) * 4);
-    u64 Mn = exclusive_scan_u32(ctx, miss, misspos, G);
-    if (op->n_rows + Mn > op->capacity) {
-      ctx->err = "reduce state capacity exceeded";
-      return -1;
-    }
-    if (Mn)
-      hipLaunchKernelGGL(k_red_insert, dim3(ngrid(G)), dim3(BLK), 0,
-                         ctx->stream, sk + lo * kw, kw, starts, (u64)G, miss,
-                         misspos, op->n_rows, op->st);
-    hipLaunchKernelGGL(k_reduce_apply, dim3(ngrid(G)), dim3(BLK), 0,
+    // coherence — see RedState docs); Phase C: apply + emit; bump counter.
+    u32 *found = (u32 *)S.get(m * 4);
+    u32 *miss = (u32 *)S.get(m * 4);
+    hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, starts, 0, gid, m,
+                       op->st, found, miss);
+    u32 *misspos = (u32 *)S.get((m + 1) * 4);
+    exclusive_scan_u32_ns(ctx, miss, misspos, m);
+    hipLaunchKernelGGL(k_red_insert, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, starts, 0, gid, m,
+                       miss, misspos, 0, op->d_nrows, op->d_err, op->st);
+    hipLaunchKernelGGL(k_reduce_apply, dim3(ngrid(m)), dim3(BLK), 0,
                        ctx->stream, sk + lo * kw, sv + lo * vb, kw, vb,
-                       sd + lo, starts, (u64)G, m,
+                       sd + lo, starts, 0, gid, m,
                        htimes[htimes.size() == 1 ? 0 : lo], op->st, found,
-                       miss, misspos, op->n_rows, op->spec, pk, pv, pt, pd,
-                       ocount);
-    op->n_rows += Mn;
+                       miss, misspos, 0, op->d_nrows, op->spec, pk, pv, pt,
+                       pd, ocount);
+    hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
+                       op->d_nrows, misspos, gid, m);
   }
-  unsigned long long M;
-  HIP_CHECK(hipMemcpyAsync(&M, ocount, 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
-  // consolidate corrections (deterministic order)
-  DevUpdates pin{pk, pv, pt, pd, M};
+  // consolidate the capacity-sized corrections (zero-diff rows drop out)
+  DevUpdates pin{pk, pv, pt, pd, cap_out};
   u64 *ok;
   u8 *ov;
   u64 *ot;
   i64 *od;
   u64 Mc;
   consolidate_dev(ctx, okw, ovb, pin, &ok, &ov, &ot, &od, &Mc);
+  u64 errflag = 0;
+  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
     dfree(ctx, (p));
+  if (errflag) {
+    ctx->err = "reduce state capacity exceeded";
+    return -1;
+  }
   *out = make_out(ok, ov, ot, od, Mc, okw, ovb);
   return 0;
 }
@@ -2297,7 +2329,7 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
     u32 *miss = (u32 *)S.get(G * 4);
     hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(G)), dim3(BLK), 0,
                        ctx->stream, nb->keys, kw2, gstart, G,
-                       op->states[l], found, miss);
+                       (const u32 *)nullptr, 0, op->states[l], found, miss);
     u32 *misspos = (u32 *)S.get((G + 1) * 4);
     u64 Mn = exclusive_scan_u32(ctx, miss, misspos, G);
     if (op->n_rows[l] + Mn > op->states[l].capacity) {
@@ -2306,8 +2338,10 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
     }
     if (Mn)
       hipLaunchKernelGGL(k_red_insert, dim3(ngrid(G)), dim3(BLK), 0,
-                         ctx->stream, nb->keys, kw2, gstart, G, miss,
-                         misspos, op->n_rows[l], op->states[l]);
+                         ctx->stream, nb->keys, kw2, gstart, G,
+                         (const u32 *)nullptr, 0, miss, misspos,
+                         op->n_rows[l], (const u64 *)nullptr,
+                         (u64 *)nullptr, op->states[l]);
     // output buffer for this level's corrections
     u32 out_kw = (l + 1 < L) ? kw2 : kw;
     u32 bnext = (l + 1 < L) ? op->buckets[l + 1] : 1;
