@@ -241,6 +241,23 @@ void mz_gpu_reduce_drop(mz_gpu_ctx *ctx, mz_gpu_red *op);
 int  mz_gpu_reduce_push(mz_gpu_ctx *ctx, mz_gpu_red *op,
                         const mz_gpu_updates *delta, mz_gpu_out **out);
 
+/* ----------------------------------------------------------- threshold
+ * Replaces build_threshold_basic / threshold_local
+ * (src/compute/src/render/threshold.rs:34-51,75-97): a reduce over the
+ * row-keyed arrangement that keeps each record whose accumulated count is
+ * positive, with that count as the output multiplicity
+ * (count.is_positive() filter, threshold.rs:42). The operator owns the
+ * resident per-(key,val) net-count table; each push returns corrections
+ * with diff = pos(new_count) - pos(old_count) per changed record
+ * (reduce_abelian contract, src/compute/src/extensions/reduce.rs:131).
+ * Output schema equals the input schema. */
+typedef struct mz_gpu_thr mz_gpu_thr;
+mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *ctx,
+                                    const mz_gpu_schema *schema);
+int  mz_gpu_threshold_push(mz_gpu_ctx *ctx, mz_gpu_thr *op,
+                           const mz_gpu_updates *delta, mz_gpu_out **out);
+void mz_gpu_threshold_drop(mz_gpu_ctx *ctx, mz_gpu_thr *op);
+
 /* ------------------------------------------------------------ exchange
  * Replaces the Exchange pact routing (linear_join.rs:390,
  * extensions/arrange.rs:134): shard = splitmix64(key words) % nshards

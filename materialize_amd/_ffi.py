@@ -89,6 +89,12 @@ def load():
                                         C.POINTER(Updates),
                                         C.POINTER(Updates),
                                         C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_threshold_create.restype = C.c_void_p
+    lib.mz_gpu_threshold_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
+    lib.mz_gpu_threshold_push.argtypes = [C.c_void_p, C.c_void_p,
+                                          C.POINTER(Updates),
+                                          C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_threshold_drop.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_partition.argtypes = [
         C.c_void_p, C.POINTER(Schema), C.POINTER(Updates), C.c_uint32,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint8), C.POINTER(C.c_uint64),
@@ -210,6 +216,16 @@ class GpuCtx:
                                                 C.byref(outp)))
         return self._take(outp)
 
+    def threshold_create(self, sch):
+        return self.lib.mz_gpu_threshold_create(self.ctx, C.byref(sch))
+
+    def threshold_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_threshold_push(self.ctx, op,
+                                                   C.byref(upd),
+                                                   C.byref(outp)))
+        return self._take(outp)
+
     def consolidate(self, sch, upd):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_consolidate(self.ctx, C.byref(sch),
@@ -289,6 +305,13 @@ class GpuCtx:
         self._check(self.lib.mz_gpu_halfjoin_raw(
             self.ctx, lookup, C.byref(upd), stream_vb, 1 if le else 0,
             C.byref(cl), C.byref(outp)))
+        return self._dev_out(outp)
+
+    def threshold_push_dev(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_threshold_push(self.ctx, op,
+                                                   C.byref(upd),
+                                                   C.byref(outp)))
         return self._dev_out(outp)
 
     def reduce_push_dev(self, op, upd):

@@ -829,6 +829,63 @@ __global__ void k_reduce_apply(const u64 *keys, const u8 *vals, u32 kw,
 }
 
 // time-change flags over sorted times
+// Pack (key words, zero-padded val words) into one combined key row —
+// the threshold's grouping key (its "key" is the whole record:
+// render/threshold.rs:38-50 iterates (record, count) within a key group;
+// state is per (key, record) pair).
+__global__ void k_pack_combined(const u64 *keys, u32 kw, const u8 *vals,
+                                u32 vb, u64 n, u64 *ck, u32 kw2) {
+  GRID_STRIDE(i, n) {
+    u64 *dst = ck + i * kw2;
+    for (u32 w = 0; w < kw; w++) dst[w] = keys[i * kw + w];
+    for (u32 w = kw; w < kw2; w++) {
+      u64 word = 0;
+      u32 off = (w - kw) * 8;
+      u32 m = vb - off < 8 ? vb - off : 8;
+      for (u32 b = 0; b < m; b++)
+        word |= (u64)vals[i * vb + off + b] << (8 * b);
+      dst[w] = word;
+    }
+  }
+}
+
+// Threshold apply: one thread per distinct (key,val) group of a time
+// slice. Keeps the wrapping net count resident; emits one correction row
+// (the record itself) with diff = pos(new) - pos(old)
+// (threshold_local's count.is_positive() filter, threshold.rs:42-47).
+__global__ void k_threshold_apply(const u64 *ckeys, u32 kw2,
+                                  const i64 *diffs, const u32 *gstart,
+                                  const u32 *gidn, u64 m, u64 t,
+                                  RedState st, const u32 *found,
+                                  const u32 *miss, const u32 *misspos,
+                                  const u64 *d_nrows, u32 kw, u32 vb,
+                                  u64 *okeys, u8 *ovals, u64 *otimes,
+                                  i64 *odiffs, unsigned long long *ocount) {
+  u64 G = m ? gidn[m - 1] : 0;
+  u64 base = *d_nrows;
+  GRID_STRIDE(g, G) {
+    u64 lo = gstart[g];
+    u64 end = (g + 1 < G) ? gstart[g + 1] : m;
+    const u64 *key = ckeys + lo * kw2;
+    u64 idx = miss[g] ? base + misspos[g] : found[g];
+    if (idx >= st.capacity) continue;  // overflow flagged by insert
+    u64 *row = st.rows + idx * st.stride_w;
+    i64 *cnt = (i64 *)(row + kw2);
+    i64 old = *cnt, nw = old;
+    for (u64 r = lo; r < end; r++) nw = wadd(nw, diffs[r]);
+    *cnt = nw;
+    i64 po = old > 0 ? old : 0, pn = nw > 0 ? nw : 0;
+    i64 delta = (i64)((u64)pn - (u64)po);
+    if (!delta) continue;
+    u64 o = atomicAdd(ocount, 1ull);
+    for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = key[w];
+    const u8 *vsrc = (const u8 *)(key + kw);
+    for (u32 b = 0; b < vb; b++) ovals[o * vb + b] = vsrc[b];
+    otimes[o] = t;
+    odiffs[o] = delta;
+  }
+}
+
 __global__ void k_time_flags(const u64 *times, const u32 *perm, u32 *flags,
                              u64 n) {
   GRID_STRIDE(i, n)
@@ -1013,6 +1070,18 @@ struct mz_gpu_red {
   u64 capacity;
   u64 n_rows = 0;   // host mirror (minmax path); the reduce path keeps the
   u64 *d_nrows = nullptr;  // authoritative count on device (no readback)
+  u64 *d_err = nullptr;
+};
+
+// Threshold operator (render/threshold.rs:34-51): net count per (key,val)
+// pair in a resident RedState table keyed by the combined
+// (key-words || val-words) row; corrections delta = pos(new) - pos(old).
+struct mz_gpu_thr {
+  mz_gpu_schema s;
+  u32 kw2;  // combined key words: key_words + ceil(val_bytes/8)
+  RedState st;
+  u64 capacity;
+  u64 *d_nrows = nullptr;
   u64 *d_err = nullptr;
 };
 
@@ -2112,6 +2181,141 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
     return -1;
   }
   *out = make_out(ok, ov, ot, od, Mc, okw, ovb);
+  return 0;
+}
+
+mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  mz_gpu_thr *r = new mz_gpu_thr();
+  r->s = *s;
+  r->kw2 = s->key_words + (s->val_bytes + 7) / 8;
+  u64 cap = 1ull << 21;  // 2M (key,val) pairs default
+  r->capacity = cap;
+  u64 slots = 2 * cap;
+  r->st.hash = dnew<u64>(ctx, slots * (r->kw2 + 1));
+  r->st.slots = slots;
+  r->st.stride_w = r->kw2 + 1;
+  r->st.rows = dnew<u64>(ctx, cap * r->st.stride_w);
+  r->st.capacity = cap;
+  r->d_nrows = dnew<u64>(ctx, 1);
+  r->d_err = dnew<u64>(ctx, 1);
+  HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
+  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
+                     ctx->stream, r->st.hash, slots, r->kw2);
+  return r;
+}
+
+void mz_gpu_threshold_drop(mz_gpu_ctx *c, mz_gpu_thr *r) {
+  (void)c;
+  (void)r;
+}
+
+int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
+                          const mz_gpu_updates *u, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  auto &S = ctx->scratch;
+  u32 kw = op->s.key_words, vb = op->s.val_bytes, kw2 = op->kw2;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 n = d.n;
+  if (n == 0) {
+    *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                    dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, kw, vb);
+    return 0;
+  }
+  u64 *ck = (u64 *)S.get(n * kw2 * 8);
+  hipLaunchKernelGGL(k_pack_combined, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, d.keys, kw, vb ? d.vals : nullptr, vb, n,
+                     ck, kw2);
+  // sort by (time, combined key)
+  u32 *perm = (u32 *)S.get(n * 4);
+  sort_updates(ctx, ck, kw2, nullptr, 0, d.times, n, perm, true);
+  u64 *sk = (u64 *)S.get(n * kw2 * 8);
+  u64 *stm = (u64 *)S.get(n * 8);
+  i64 *sd = (i64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, ck, kw2, perm, sk, n);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.times, perm, stm, n);
+  hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.diffs, perm, sd, n);
+  std::vector<u64> htimes;
+  std::vector<std::pair<u64, u64>> slices;
+  if (u->upper <= u->lower + 1) {
+    htimes.assign(1, u->lower);
+    slices.push_back({0, n});
+  } else {
+    htimes.resize(n);
+    HIP_CHECK(hipMemcpyAsync(htimes.data(), stm, n * 8,
+                             hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    for (u64 i = 0; i < n;) {
+      u64 j = i;
+      while (j < n && htimes[j] == htimes[i]) j++;
+      slices.push_back({i, j});
+      i = j;
+    }
+  }
+  u64 cap_out = n + 16;  // ≤1 correction per group per slice
+  u64 *pk = dnew<u64>(ctx, cap_out * kw);
+  u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap_out * vb, 1));
+  u64 *pt = dnew<u64>(ctx, cap_out);
+  i64 *pd = dnew<i64>(ctx, cap_out);
+  unsigned long long *ocount = (unsigned long long *)S.get(8);
+  HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pk, 0, cap_out * kw * 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(cap_out * vb, 1),
+                           ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pt, 0, cap_out * 8, ctx->stream));
+  HIP_CHECK(hipMemsetAsync(pd, 0, cap_out * 8, ctx->stream));
+  u32 *flags = (u32 *)S.get(n * 4);
+  u32 *gid = (u32 *)S.get(n * 4);
+  for (auto [lo, hi] : slices) {
+    u64 m = hi - lo;
+    hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw2, kw2, stm + lo, flags, m);
+    inclusive_scan_u32(ctx, flags, gid, m);
+    u32 *starts = (u32 *)S.get(m * 4);
+    hipLaunchKernelGGL(k_group_starts, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, flags, gid, starts, m);
+    u32 *found = (u32 *)S.get(m * 4);
+    u32 *miss = (u32 *)S.get(m * 4);
+    hipLaunchKernelGGL(k_red_lookup, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw2, kw2, starts, 0, gid, m,
+                       op->st, found, miss);
+    u32 *misspos = (u32 *)S.get((m + 1) * 4);
+    exclusive_scan_u32_ns(ctx, miss, misspos, m);
+    hipLaunchKernelGGL(k_red_insert, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw2, kw2, starts, 0, gid, m,
+                       miss, misspos, 0, op->d_nrows, op->d_err, op->st);
+    hipLaunchKernelGGL(k_threshold_apply, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw2, kw2, sd + lo, starts, gid,
+                       m, htimes[htimes.size() == 1 ? 0 : lo], op->st, found,
+                       miss, misspos, op->d_nrows, kw, vb, pk, pv, pt, pd,
+                       ocount);
+    hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
+                       op->d_nrows, misspos, gid, m);
+  }
+  DevUpdates pin{pk, pv, pt, pd, cap_out};
+  u64 *ok;
+  u8 *ov;
+  u64 *ot;
+  i64 *od;
+  u64 Mc;
+  consolidate_dev(ctx, kw, vb, pin, &ok, &ov, &ot, &od, &Mc);
+  u64 errflag = 0;
+  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
+    dfree(ctx, p);
+  if (errflag) {
+    ctx->err = "threshold state capacity exceeded";
+    return -1;
+  }
+  *out = make_out(ok, ov, ot, od, Mc, kw, vb);
   return 0;
 }
 

@@ -224,9 +224,34 @@ class ReduceOp:
         return self.ctx.reduce_push_dev(self.op, updates)
 
 
+class ThresholdPlan:
+    """render_threshold / BasicThresholdPlan surface
+    (src/compute/src/render/threshold.rs:100-116): the ensure_arrangement
+    key is the full record in our fixed-width model."""
+
+    def __init__(self, schema):
+        self.schema = schema
+
+
+class ThresholdOp:
+    """build_threshold_basic (threshold.rs:75-97): keeps rows with a
+    positive accumulated count, at that count."""
+
+    def __init__(self, ctx, plan: ThresholdPlan):
+        self.ctx = ctx
+        self.op = ctx.threshold_create(plan.schema)
+
+    def push(self, updates) -> DevOut:
+        return self.ctx.threshold_push_dev(self.op, updates)
+
+
 def render_delta_join(ctx, arrangements, plan, exchange=None) -> DeltaJoinOp:
     return DeltaJoinOp(ctx, arrangements, plan, exchange=exchange)
 
 
 def render_reduce(ctx, plan) -> ReduceOp:
     return ReduceOp(ctx, plan)
+
+
+def render_threshold(ctx, plan) -> ThresholdOp:
+    return ThresholdOp(ctx, plan)

@@ -664,6 +664,20 @@ struct MinMaxOp {
   std::vector<std::map<std::vector<u64>, i64>> state;
 };
 
+// ------------------------------------------------------------ threshold
+// Restates threshold_local (src/compute/src/render/threshold.rs:34-51):
+// a reduce over the row-keyed arrangement that keeps each record whose
+// accumulated count is positive, with that count as the output
+// multiplicity. Maintained incrementally per time slice with
+// new-minus-old corrections (reduce_abelian contract,
+// src/compute/src/extensions/reduce.rs:131). State is the net count per
+// (key, val) pair; output diff delta = pos(new) - pos(old), wrapping i64
+// like every Diff (src/ore/src/overflowing.rs:24-31).
+struct ThrOp {
+  Schema s;
+  std::map<std::vector<u8>, i64> state;  // key-words bytes || val bytes
+};
+
 // ------------------------------------------------------------- context
 
 struct Out {
@@ -679,6 +693,7 @@ struct orc_ctx {
   std::vector<std::unique_ptr<Arr>> arrs;
   std::vector<std::unique_ptr<JoinOp>> joins;
   std::vector<std::unique_ptr<RedOp>> reds;
+  std::vector<std::unique_ptr<ThrOp>> thrs;
 };
 
 extern "C" {
@@ -935,6 +950,53 @@ RedOp *orc_reduce_create(orc_ctx *c, const mz_gpu_reduce_spec *spec) {
   RedOp *p = r.get();
   c->reds.push_back(std::move(r));
   return p;
+}
+
+ThrOp *orc_threshold_create(orc_ctx *c, const mz_gpu_schema *s) {
+  auto r = std::make_unique<ThrOp>();
+  r->s = {s->key_words, s->val_bytes};
+  ThrOp *p = r.get();
+  c->thrs.push_back(std::move(r));
+  return p;
+}
+
+int orc_threshold_push(orc_ctx *c, ThrOp *op, const mz_gpu_updates *u,
+                       mz_gpu_out **out) {
+  const Schema s = op->s;
+  Cols result;
+  std::vector<u64> order(u->n);
+  for (u64 i = 0; i < u->n; i++) order[i] = i;
+  std::stable_sort(order.begin(), order.end(),
+                   [&](u64 a, u64 b) { return u->times[a] < u->times[b]; });
+  std::vector<u8> ck(s.kw * 8 + s.vb);
+  size_t p = 0;
+  while (p < order.size()) {
+    u64 t = u->times[order[p]];
+    std::map<std::vector<u8>, i64> olds;
+    while (p < order.size() && u->times[order[p]] == t) {
+      u64 i = order[p++];
+      std::memcpy(ck.data(), u->keys + i * s.kw, s.kw * 8);
+      if (s.vb) std::memcpy(ck.data() + s.kw * 8, u->vals + i * s.vb, s.vb);
+      auto it = op->state.find(ck);
+      i64 cur = it == op->state.end() ? 0 : it->second;
+      if (olds.find(ck) == olds.end()) olds[ck] = cur;
+      op->state[ck] = (i64)((u64)cur + (u64)u->diffs[i]);
+    }
+    for (auto &[k, old] : olds) {
+      i64 nw = op->state[k];
+      i64 po = old > 0 ? old : 0, pn = nw > 0 ? nw : 0;
+      i64 delta = (i64)((u64)pn - (u64)po);
+      if (delta) {
+        u64 kws[MAX_KW];
+        std::memcpy(kws, k.data(), s.kw * 8);
+        result.push(kws, s.kw, k.data() + s.kw * 8, s.vb, t, delta);
+      }
+      if (nw == 0) op->state.erase(k);
+    }
+  }
+  consolidate(s, result);
+  *out = &make_out(std::move(result), s)->pub_;
+  return 0;
 }
 
 // build_accumulable push: move datums into accumulators scaled by diff

@@ -54,6 +54,11 @@ def _load():
     lib.orc_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_threshold_create.restype = C.c_void_p
+    lib.orc_threshold_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
+    lib.orc_threshold_push.argtypes = [C.c_void_p, C.c_void_p,
+                                       C.POINTER(Updates),
+                                       C.POINTER(C.POINTER(OutBatch))]
     lib.orc_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
@@ -148,6 +153,16 @@ class OracleCtx:
         outp = C.POINTER(OutBatch)()
         rc = self.lib.orc_reduce_push(self.ctx, op, C.byref(upd),
                                       C.byref(outp))
+        assert rc == 0
+        return self._take(outp)
+
+    def threshold_create(self, sch):
+        return self.lib.orc_threshold_create(self.ctx, C.byref(sch))
+
+    def threshold_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_threshold_push(self.ctx, op, C.byref(upd),
+                                         C.byref(outp))
         assert rc == 0
         return self._take(outp)
 

@@ -260,6 +260,68 @@ class TestReduceParity:
                     "multi-timestamp reduce")
 
 
+class TestThresholdParity:
+    """build_threshold_basic semantics (render/threshold.rs:34-51):
+    corrections with diff = pos(new count) - pos(old count) per record."""
+
+    def test_random_churn(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(53)
+        sch = abi.schema(1, 8)
+        gop, oop = g.threshold_create(sch), o.threshold_create(sch)
+        for step in range(5):
+            n = 600
+            keys = rng.integers(-5, 40, n).astype(np.int64)
+            vals = rng.integers(0, 3, n).astype(np.int64)
+            diffs = rng.integers(-2, 3, n).astype(np.int64)
+            u = abi.make_updates(keys, vals.view(np.uint8),
+                                 np.full(n, step, np.uint64), diffs, step,
+                                 step + 1)
+            assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
+                        f"threshold step {step}")
+
+    def test_two_word_keys_wide_vals(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(59)
+        kw, vb = 2, 12
+        sch = abi.schema(kw, vb)
+        gop, oop = g.threshold_create(sch), o.threshold_create(sch)
+        for step in range(3):
+            n = 400
+            keys = rng.integers(-4, 10, (n, kw)).astype(np.int64)
+            vals = rng.integers(0, 4, (n, vb)).astype(np.uint8)
+            diffs = rng.integers(-3, 4, n).astype(np.int64)
+            u = abi.make_updates(keys, vals, np.full(n, step, np.uint64),
+                                 diffs, step, step + 1)
+            assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
+                        f"threshold wide step {step}")
+
+    def test_multi_timestamp(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(61)
+        sch = abi.schema(1, 8)
+        gop, oop = g.threshold_create(sch), o.threshold_create(sch)
+        n = 500
+        keys = rng.integers(0, 25, n).astype(np.int64)
+        vals = rng.integers(0, 2, n).astype(np.int64)
+        times = rng.integers(0, 4, n).astype(np.uint64)
+        diffs = rng.integers(-2, 3, n).astype(np.int64)
+        u = abi.make_updates(keys, vals.view(np.uint8), times, diffs, 0, 4)
+        assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
+                    "threshold multi-timestamp")
+
+    def test_empty(self, ctxs):
+        g, o = ctxs
+        sch = abi.schema(1, 8)
+        gop, oop = g.threshold_create(sch), o.threshold_create(sch)
+        u = abi.make_updates(np.zeros((0, 1), np.int64),
+                             np.zeros((0, 8), np.uint8),
+                             np.zeros(0, np.uint64), np.zeros(0, np.int64),
+                             0, 1)
+        assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
+                    "threshold empty")
+
+
 class TestRouteHashParity:
     def test_hash_agrees(self, ctxs):
         g, o = ctxs
