@@ -258,6 +258,48 @@ int  mz_gpu_threshold_push(mz_gpu_ctx *ctx, mz_gpu_thr *op,
                            const mz_gpu_updates *delta, mz_gpu_out **out);
 void mz_gpu_threshold_drop(mz_gpu_ctx *ctx, mz_gpu_thr *op);
 
+/* ----------------------------------------------------------------- topk
+ * Replaces render_topk's Basic plan path — build_topk /
+ * build_topk_negated_stage (src/compute/src/render/top_k.rs:322-418,
+ * 614-770): per group-key, order records by the order columns
+ * (ColumnOrder asc/desc; compare_columns at :733-739, ties broken in the
+ * engine's canonical val order, standing in for the reference's Row-order
+ * tie-break :738), then keep the multiplicity window [offset,
+ * offset+limit) of the running prefix (:743-766). The operator owns the
+ * resident group-contents state; each push returns corrections (new
+ * minus old kept rows per changed group, reduce_abelian contract).
+ * Restrictions vs the reference: literal limits only (no
+ * per-key limit expressions, :659-688); order columns are non-null signed
+ * little-endian integers of width 4 or 8; the monotonic plan variants
+ * (MonotonicTop1/TopK, :157-288) are streaming-input optimizations whose
+ * output equals Basic's and are not separate entry points; the bucketed
+ * stage hierarchy (:380-398) is a work-thinning policy that leaves the
+ * final modulus-1 stage's output unchanged, so groups are evaluated
+ * directly. Negative input multiplicities return an error
+ * ("Negative multiplicities in TopK", :494). */
+typedef struct {
+  uint16_t off;    /* byte offset of the order datum in the val bytes */
+  uint8_t  width;  /* 4 or 8 (signed little-endian integer)           */
+  uint8_t  desc;   /* 1 = descending                                  */
+} mz_gpu_order_col;
+
+#define MZ_GPU_MAX_ORDER 4
+
+typedef struct {
+  mz_gpu_schema in;  /* group key words + record val bytes            */
+  uint64_t offset;   /* rows to skip per group (TopKPlan::offset)     */
+  int64_t  limit;    /* rows to keep after offset; < 0 = no limit     */
+  uint32_t n_order;
+  mz_gpu_order_col order[MZ_GPU_MAX_ORDER];
+} mz_gpu_topk_spec;
+
+typedef struct mz_gpu_topk mz_gpu_topk;
+mz_gpu_topk *mz_gpu_topk_create(mz_gpu_ctx *ctx,
+                                const mz_gpu_topk_spec *spec);
+int  mz_gpu_topk_push(mz_gpu_ctx *ctx, mz_gpu_topk *op,
+                      const mz_gpu_updates *delta, mz_gpu_out **out);
+void mz_gpu_topk_drop(mz_gpu_ctx *ctx, mz_gpu_topk *op);
+
 /* ------------------------------------------------------------ exchange
  * Replaces the Exchange pact routing (linear_join.rs:390,
  * extensions/arrange.rs:134): shard = splitmix64(key words) % nshards

@@ -102,6 +102,27 @@ class ReduceSpec(C.Structure):
     ]
 
 
+MZ_GPU_MAX_ORDER = 4
+
+
+class OrderCol(C.Structure):
+    _fields_ = [
+        ("off", C.c_uint16),
+        ("width", C.c_uint8),
+        ("desc", C.c_uint8),
+    ]
+
+
+class TopKSpec(C.Structure):
+    _fields_ = [
+        ("in_", Schema),
+        ("offset", C.c_uint64),
+        ("limit", C.c_int64),
+        ("n_order", C.c_uint32),
+        ("order", OrderCol * MZ_GPU_MAX_ORDER),
+    ]
+
+
 class OutBatch(C.Structure):
     _fields_ = [
         ("keys", C.POINTER(C.c_uint64)),
@@ -217,6 +238,18 @@ def closure(filters, key_fields, val_fields, out_schema):
         cl.val_fields[i] = f
     cl.out = out_schema
     return cl
+
+
+def topk_spec(in_schema, order, offset=0, limit=-1):
+    """order: list of (off, width, desc) over the val bytes."""
+    sp = TopKSpec()
+    sp.in_ = in_schema
+    sp.offset = offset
+    sp.limit = limit
+    sp.n_order = len(order)
+    for i, (off, width, desc) in enumerate(order):
+        sp.order[i] = OrderCol(off=off, width=width, desc=1 if desc else 0)
+    return sp
 
 
 def reduce_spec(aggs, in_schema):

@@ -678,6 +678,73 @@ struct ThrOp {
   std::map<std::vector<u8>, i64> state;  // key-words bytes || val bytes
 };
 
+// ----------------------------------------------------------------- topk
+// Restates build_topk / build_topk_negated_stage
+// (src/compute/src/render/top_k.rs:322-418,614-770): per group, order the
+// records by the order columns (ColumnOrder asc/desc over signed
+// little-endian integers; compare_columns, :733-739), break ties in the
+// engine's canonical val order (the role of the reference's Row-order
+// tie-break `left.cmp(right)`, :738 — deviation: our packed encoding's
+// order, identical on oracle and GPU), then keep multiplicities clipped
+// to [offset, offset+limit) of the running prefix (:743-766). The
+// reference's bucketed stage hierarchy (:380-398) thins work but never
+// changes the final modulus-1 stage's output, so we evaluate groups
+// directly (policy deviation like the spine-merge schedule).
+// Negative input multiplicities are an error (:476,"Negative
+// multiplicities in TopK", :494).
+using TopKOrderCol = mz_gpu_order_col;
+
+struct TopKOp {
+  Schema s;
+  u64 offset;
+  i64 limit;  // < 0 = none
+  std::vector<TopKOrderCol> order;
+  // group key -> (val bytes -> net count); inner map in canonical val
+  // order via explicit sort at eval time
+  std::map<std::vector<u64>, std::map<std::vector<u8>, i64>> state;
+};
+
+static i64 read_order_datum(const u8 *val, const TopKOrderCol &c) {
+  return read_int(val + c.off, c.width);
+}
+
+// Current TopK output of one group: vector of (val, kept multiplicity).
+static int topk_eval(const TopKOp *op,
+                     const std::map<std::vector<u8>, i64> &grp,
+                     std::vector<std::pair<std::vector<u8>, i64>> &out) {
+  out.clear();
+  std::vector<std::pair<const std::vector<u8> *, i64>> items;
+  for (auto &[v, c] : grp) {
+    if (c == 0) continue;
+    if (c < 0) return -1;  // top_k.rs:690-697 validating stage
+    items.push_back({&v, c});
+  }
+  std::stable_sort(items.begin(), items.end(), [&](auto &a, auto &b) {
+    for (auto &c : op->order) {
+      i64 x = read_order_datum(a.first->data(), c);
+      i64 y = read_order_datum(b.first->data(), c);
+      if (x != y) return c.desc ? x > y : x < y;
+    }
+    return cmp_val(a.first->data(), b.first->data(),
+                   (u32)a.first->size()) < 0;
+  });
+  u64 O = op->offset;
+  i64 L = op->limit;
+  u64 running = 0;
+  for (auto &[v, c] : items) {
+    u64 lo = running, hi = running + (u64)c;
+    running = hi;
+    u64 wlo = lo > O ? lo : O;
+    u64 whi = hi;
+    if (L >= 0) {
+      u64 cap = O + (u64)L;
+      if (whi > cap) whi = cap;
+    }
+    if (whi > wlo) out.push_back({*v, (i64)(whi - wlo)});
+  }
+  return 0;
+}
+
 // ------------------------------------------------------------- context
 
 struct Out {
@@ -694,6 +761,7 @@ struct orc_ctx {
   std::vector<std::unique_ptr<JoinOp>> joins;
   std::vector<std::unique_ptr<RedOp>> reds;
   std::vector<std::unique_ptr<ThrOp>> thrs;
+  std::vector<std::unique_ptr<TopKOp>> topks;
 };
 
 extern "C" {
@@ -1059,6 +1127,64 @@ int orc_reduce_push(orc_ctx *c, RedOp *op, const mz_gpu_updates *u,
   }
   consolidate(os, result);
   *out = &make_out(std::move(result), os)->pub_;
+  return 0;
+}
+
+TopKOp *orc_topk_create(orc_ctx *c, const mz_gpu_topk_spec *spec) {
+  auto r = std::make_unique<TopKOp>();
+  r->s = {spec->in.key_words, spec->in.val_bytes};
+  r->offset = spec->offset;
+  r->limit = spec->limit;
+  for (u32 i = 0; i < spec->n_order; i++) r->order.push_back(spec->order[i]);
+  TopKOp *p = r.get();
+  c->topks.push_back(std::move(r));
+  return p;
+}
+
+int orc_topk_push(orc_ctx *c, TopKOp *op, const mz_gpu_updates *u,
+                  mz_gpu_out **out) {
+  const Schema s = op->s;
+  Cols result;
+  std::vector<u64> order(u->n);
+  for (u64 i = 0; i < u->n; i++) order[i] = i;
+  std::stable_sort(order.begin(), order.end(),
+                   [&](u64 a, u64 b) { return u->times[a] < u->times[b]; });
+  std::vector<std::pair<std::vector<u8>, i64>> oldo, newo;
+  size_t p = 0;
+  while (p < order.size()) {
+    u64 t = u->times[order[p]];
+    // group contents before this slice, for changed groups
+    std::map<std::vector<u64>, std::map<std::vector<u8>, i64>> olds;
+    while (p < order.size() && u->times[order[p]] == t) {
+      u64 i = order[p++];
+      std::vector<u64> key(u->keys + i * s.kw, u->keys + (i + 1) * s.kw);
+      std::vector<u8> val(u->vals + i * s.vb, u->vals + (i + 1) * s.vb);
+      auto it = op->state.find(key);
+      if (olds.find(key) == olds.end())
+        olds[key] = it != op->state.end()
+                        ? it->second
+                        : std::map<std::vector<u8>, i64>();
+      auto &grp = op->state[key];
+      grp[val] = (i64)((u64)grp[val] + (u64)u->diffs[i]);
+      if (grp[val] == 0) grp.erase(val);
+    }
+    for (auto &[key, old] : olds) {
+      auto it = op->state.find(key);
+      static const std::map<std::vector<u8>, i64> kEmpty;
+      const auto &nw = it != op->state.end() ? it->second : kEmpty;
+      if (topk_eval(op, old, oldo) || topk_eval(op, nw, newo)) {
+        c->err = "negative multiplicities in TopK";
+        return -1;
+      }
+      for (auto &[v, d] : oldo)
+        result.push(key.data(), s.kw, v.data(), s.vb, t, -d);
+      for (auto &[v, d] : newo)
+        result.push(key.data(), s.kw, v.data(), s.vb, t, d);
+      if (it != op->state.end() && it->second.empty()) op->state.erase(it);
+    }
+  }
+  consolidate(s, result);
+  *out = &make_out(std::move(result), s)->pub_;
   return 0;
 }
 

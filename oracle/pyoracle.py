@@ -9,7 +9,7 @@ import os
 import subprocess
 
 from materialize_amd._abi import (  # type defs of the shared boundary
-    Closure, OutBatch, ReduceSpec, Schema, Updates, out_to_numpy,
+    Closure, OutBatch, ReduceSpec, Schema, TopKSpec, Updates, out_to_numpy,
 )
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
@@ -59,6 +59,11 @@ def _load():
     lib.orc_threshold_push.argtypes = [C.c_void_p, C.c_void_p,
                                        C.POINTER(Updates),
                                        C.POINTER(C.POINTER(OutBatch))]
+    lib.orc_topk_create.restype = C.c_void_p
+    lib.orc_topk_create.argtypes = [C.c_void_p, C.POINTER(TopKSpec)]
+    lib.orc_topk_push.argtypes = [C.c_void_p, C.c_void_p,
+                                  C.POINTER(Updates),
+                                  C.POINTER(C.POINTER(OutBatch))]
     lib.orc_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
                                     C.POINTER(Updates),
                                     C.POINTER(C.POINTER(OutBatch))]
@@ -154,6 +159,17 @@ class OracleCtx:
         rc = self.lib.orc_reduce_push(self.ctx, op, C.byref(upd),
                                       C.byref(outp))
         assert rc == 0
+        return self._take(outp)
+
+    def topk_create(self, spec):
+        return self.lib.orc_topk_create(self.ctx, C.byref(spec))
+
+    def topk_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        rc = self.lib.orc_topk_push(self.ctx, op, C.byref(upd),
+                                    C.byref(outp))
+        if rc != 0:
+            raise RuntimeError(self.lib.orc_last_error(self.ctx).decode())
         return self._take(outp)
 
     def threshold_create(self, sch):
