@@ -8,7 +8,7 @@ import ctypes as C
 import os
 import subprocess
 
-from ._abi import (Closure, OutBatch, ReduceSpec, Schema, Updates,
+from ._abi import (Closure, OutBatch, ReduceSpec, Schema, TopKSpec, Updates,
                    out_to_numpy)
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
@@ -89,6 +89,12 @@ def load():
                                         C.POINTER(Updates),
                                         C.POINTER(Updates),
                                         C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_topk_create.restype = C.c_void_p
+    lib.mz_gpu_topk_create.argtypes = [C.c_void_p, C.POINTER(TopKSpec)]
+    lib.mz_gpu_topk_push.argtypes = [C.c_void_p, C.c_void_p,
+                                     C.POINTER(Updates),
+                                     C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_topk_drop.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_threshold_create.restype = C.c_void_p
     lib.mz_gpu_threshold_create.argtypes = [C.c_void_p, C.POINTER(Schema)]
     lib.mz_gpu_threshold_push.argtypes = [C.c_void_p, C.c_void_p,
@@ -214,6 +220,15 @@ class GpuCtx:
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
                                                 C.byref(outp)))
+        return self._take(outp)
+
+    def topk_create(self, spec):
+        return self.lib.mz_gpu_topk_create(self.ctx, C.byref(spec))
+
+    def topk_push(self, op, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_topk_push(self.ctx, op, C.byref(upd),
+                                              C.byref(outp)))
         return self._take(outp)
 
     def threshold_create(self, sch):

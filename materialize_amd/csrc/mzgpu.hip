@@ -886,6 +886,72 @@ __global__ void k_threshold_apply(const u64 *ckeys, u32 kw2,
   }
 }
 
+// ------------------------------------------------------------- topk
+// Compact one row per distinct group of a sorted slice (group key list
+// for the eval probes).
+__global__ void k_gather_group_keys(const u64 *keys, u32 kw,
+                                    const u32 *starts, const u32 *gidn,
+                                    u64 m, u64 *dg) {
+  u64 G = m ? gidn[m - 1] : 0;
+  GRID_STRIDE(g, G) {
+    for (u32 w = 0; w < kw; w++)
+      dg[g * kw + w] = keys[(u64)starts[g] * kw + w];
+  }
+}
+
+__global__ void k_fill_u64(u64 *p, u64 n, u64 v) { GRID_STRIDE(i, n) p[i] = v; }
+__global__ void k_fill_i64(i64 *p, u64 n, i64 v) { GRID_STRIDE(i, n) p[i] = v; }
+
+__global__ void k_check_pos(const i64 *d, u64 n, u64 *err) {
+  GRID_STRIDE(i, n) if (d[i] < 0) *err = 1;
+}
+
+// Order-column radix key: signed little-endian int biased to unsigned
+// order; descending columns invert (compare_columns restatement,
+// top_k.rs:733-739).
+__global__ void k_order_sortkey(const u8 *vals, u32 vb, const u32 *perm,
+                                u64 *skey, u64 n, u32 off, u32 width,
+                                u32 desc) {
+  GRID_STRIDE(i, n) {
+    const u8 *v = vals + (u64)perm[i] * vb + off;
+    u64 raw = 0;
+    for (u32 b = 0; b < width; b++) raw |= (u64)v[b] << (8 * b);
+    i64 x = width == 8 ? (i64)raw : (i64)(int32_t)(u32)raw;
+    u64 s = (u64)x ^ 0x8000000000000000ull;
+    skey[i] = desc ? ~s : s;
+  }
+}
+
+// Kept multiplicity of each ordered row: the overlap of its running
+// prefix window [lo,hi) with [offset, offset+limit) (top_k.rs:743-766);
+// emit with the given sign (old rows negated, new rows positive).
+__global__ void k_topk_kept(const u64 *keys, u32 kw, const u8 *vals, u32 vb,
+                            const i64 *diffs, const u64 *pre,
+                            const u32 *starts, const u32 *gid, u64 n,
+                            u64 off, i64 lim, u64 t, i64 sign, u64 *okeys,
+                            u8 *ovals, u64 *otimes, i64 *odiffs,
+                            unsigned long long *ocount) {
+  GRID_STRIDE(i, n) {
+    u32 g = gid[i] - 1;
+    u64 s = starts[g];
+    u64 base = s ? pre[s - 1] : 0;
+    u64 lo = (i == s) ? 0 : pre[i - 1] - base;
+    u64 hi = pre[i] - base;
+    u64 wlo = lo > off ? lo : off;
+    u64 whi = hi;
+    if (lim >= 0) {
+      u64 cap = off + (u64)lim;
+      if (whi > cap) whi = cap;
+    }
+    if (whi <= wlo) continue;
+    u64 o = atomicAdd(ocount, 1ull);
+    for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = keys[i * kw + w];
+    for (u32 b = 0; b < vb; b++) ovals[o * vb + b] = vals[i * vb + b];
+    otimes[o] = t;
+    odiffs[o] = sign * (i64)(whi - wlo);
+  }
+}
+
 __global__ void k_time_flags(const u64 *times, const u32 *perm, u32 *flags,
                              u64 n) {
   GRID_STRIDE(i, n)
@@ -1082,6 +1148,16 @@ struct mz_gpu_thr {
   RedState st;
   u64 capacity;
   u64 *d_nrows = nullptr;
+  u64 *d_err = nullptr;
+};
+
+// TopK operator (top_k.rs:322-418): the operator owns its input
+// arrangement (the reference's "Arranged TopK input", :647-656); each push
+// evaluates changed groups before and after the slice insert and emits
+// new-minus-old kept rows. See include/mz_gpu.h for the semantics note.
+struct mz_gpu_topk {
+  mz_gpu_topk_spec spec;
+  mz_gpu_arr *arr = nullptr;  // owned group-contents arrangement
   u64 *d_err = nullptr;
 };
 
@@ -1651,7 +1727,10 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // are O(n)); small batches pool lazily and merge k-way when the pool
   // exceeds 6 — per-merge fixed overhead (~25 kernel launches + syncs)
   // made per-step pair merges of 100k-row batches the dominant step cost.
-  constexpr u64 SMALL = 1u << 20;
+  static const u64 SMALL = [] {
+    const char *e = getenv("MZ_GPU_SMALL");
+    return e ? (u64)atoll(e) : (u64)(1u << 20);
+  }();
   static const long POOL = [] {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
     return e ? atol(e) : 6;
@@ -2316,6 +2395,285 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
     return -1;
   }
   *out = make_out(ok, ov, ot, od, Mc, kw, vb);
+  return 0;
+}
+
+// Pass-through closure: the group-eval probe reads each (key, val, net)
+// of the probed groups unchanged (the peek shape).
+static void topk_pass_cl(u32 kw, u32 vb, mz_gpu_closure &cl) {
+  std::memset(&cl, 0, sizeof(cl));
+  cl.n_key_fields = 1;
+  cl.key_fields[0] = mz_gpu_field{MZ_SRC_KEY, 0, (u8)(8 * kw), 0, 0, 0, 0};
+  cl.n_val_fields = vb ? 1u : 0u;
+  if (vb)
+    cl.val_fields[0] =
+        mz_gpu_field{MZ_SRC_VAL_LOOKUP, 0, (u8)vb, 0, 0, 0, 0};
+  cl.out.key_words = kw;
+  cl.out.val_bytes = vb;
+}
+
+// Order one eval set (a consolidated (key,val,net) snapshot of the
+// changed groups) by (group, order columns, canonical val tie-break),
+// compute each row's kept multiplicity window and append corrections
+// with the given sign. Stable LSD radix passes: order columns least-
+// significant first, then key words to regroup (compare_columns +
+// Row-order tie-break restatement, top_k.rs:733-766).
+static void topk_eval_emit(Ctx *ctx, mz_gpu_topk *op, mz_gpu_out *pe,
+                           u64 t, i64 sign, u64 *pk, u8 *pv, u64 *pt,
+                           i64 *pd, unsigned long long *ocount) {
+  u64 n = pe->n;
+  if (!n) return;
+  u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
+  auto &S = ctx->scratch;
+  const u64 *keys = (const u64 *)pe->keys;
+  const u8 *vals = (const u8 *)pe->vals;
+  const i64 *diffs = (const i64 *)pe->diffs;
+  hipLaunchKernelGGL(k_check_pos, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     diffs, n, op->d_err);
+  u32 *perm = (u32 *)S.get(n * 4);
+  u32 *perm_out = (u32 *)S.get(n * 4);
+  u64 *skey = (u64 *)S.get(n * 8);
+  u64 *skey_out = (u64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     perm, n);
+  size_t tmp_bytes = 0;
+  (void)rocprim::radix_sort_pairs(nullptr, tmp_bytes, skey, skey_out, perm,
+                                  perm_out, (unsigned)n, 0, 64, ctx->stream);
+  void *tmp = S.get(tmp_bytes);
+  auto radix_pass = [&]() {
+    size_t nb = tmp_bytes;
+    (void)rocprim::radix_sort_pairs(tmp, nb, skey, skey_out, perm, perm_out,
+                                    (unsigned)n, 0, 64, ctx->stream);
+    std::swap(perm, perm_out);
+  };
+  for (int j = (int)op->spec.n_order - 1; j >= 0; j--) {
+    const mz_gpu_order_col &oc = op->spec.order[j];
+    hipLaunchKernelGGL(k_order_sortkey, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, vals, vb, perm, skey, n, (u32)oc.off,
+                       (u32)oc.width, (u32)oc.desc);
+    radix_pass();
+  }
+  for (int w = (int)kw - 1; w >= 0; w--) {
+    hipLaunchKernelGGL(k_sortkey_key, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, keys, kw, (u32)w, perm, skey, n,
+                       (u64)0);
+    radix_pass();
+  }
+  u64 *sk2 = (u64 *)S.get(n * kw * 8);
+  u8 *sv2 = (u8 *)S.get(std::max<u64>(n * vb, 1));
+  i64 *sd2 = (i64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, keys, kw, perm, sk2, n);
+  if (vb)
+    hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, vals, vb, perm, sv2, n);
+  hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, diffs, perm, sd2, n);
+  u32 *flags = (u32 *)S.get(n * 4);
+  u32 *gid = (u32 *)S.get(n * 4);
+  // eval rows all carry time t (the peek probe's join time)
+  hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, sk2, kw, (const u64 *)pe->times, flags, n);
+  inclusive_scan_u32(ctx, flags, gid, n);
+  u32 *starts = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_group_starts, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, flags, gid, starts, n);
+  u64 *pre = (u64 *)S.get(n * 8);
+  inclusive_scan_u64(ctx, (const u64 *)sd2, pre, n);
+  hipLaunchKernelGGL(k_topk_kept, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     sk2, kw, sv2, vb, sd2, pre, starts, gid, n,
+                     op->spec.offset, (i64)op->spec.limit, t, sign, pk, pv,
+                     pt, pd, ocount);
+}
+
+mz_gpu_topk *mz_gpu_topk_create(mz_gpu_ctx *c, const mz_gpu_topk_spec *spec) {
+  Ctx *ctx = &c->impl;
+  mz_gpu_topk *r = new mz_gpu_topk();
+  r->spec = *spec;
+  r->arr = mz_gpu_arr_create(c, &spec->in);
+  r->d_err = dnew<u64>(ctx, 1);
+  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
+  return r;
+}
+
+void mz_gpu_topk_drop(mz_gpu_ctx *c, mz_gpu_topk *op) {
+  (void)c;
+  (void)op;
+}
+
+int mz_gpu_topk_push(mz_gpu_ctx *c, mz_gpu_topk *op,
+                     const mz_gpu_updates *u, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  ctx->scratch.reset();
+  u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  u64 n = d.n;
+  if (n == 0) {
+    *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                    dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, kw, vb);
+    return 0;
+  }
+  // Owned sorted columns: they outlive spine merges' scratch resets.
+  u32 *perm = dnew<u32>(ctx, n);
+  sort_updates(ctx, d.keys, kw, d.vals, vb, d.times, n, perm, true);
+  u64 *sk = dnew<u64>(ctx, n * kw);
+  u8 *sv = (u8 *)dmalloc(ctx, std::max<u64>(n * vb, 1));
+  u64 *stm = dnew<u64>(ctx, n);
+  i64 *sd = dnew<i64>(ctx, n);
+  hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                     ctx->stream, d.keys, kw, perm, sk, n);
+  if (vb)
+    hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.vals, vb, perm, sv, n);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.times, perm, stm, n);
+  hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                     d.diffs, perm, sd, n);
+  std::vector<u64> htimes;
+  std::vector<std::pair<u64, u64>> slices;
+  if (u->upper <= u->lower + 1) {
+    htimes.assign(1, u->lower);
+    slices.push_back({0, n});
+  } else {
+    htimes.resize(n);
+    HIP_CHECK(hipMemcpyAsync(htimes.data(), stm, n * 8,
+                             hipMemcpyDeviceToHost, ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    for (u64 i = 0; i < n;) {
+      u64 j = i;
+      while (j < n && htimes[j] == htimes[i]) j++;
+      slices.push_back({i, j});
+      i = j;
+    }
+  }
+  std::vector<std::array<void *, 4>> segs;
+  std::vector<u64> segn;
+  int rc = 0;
+  for (auto [lo, hi] : slices) {
+    u64 m = hi - lo;
+    u64 t = htimes[htimes.size() == 1 ? 0 : lo];
+    auto &S = ctx->scratch;
+    u32 *flags = (u32 *)S.get(m * 4);
+    u32 *gid = (u32 *)S.get(m * 4);
+    hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, stm + lo, flags, m);
+    inclusive_scan_u32(ctx, flags, gid, m);
+    u32 *starts = (u32 *)S.get(m * 4);
+    hipLaunchKernelGGL(k_group_starts, dim3(ngrid(m)), dim3(BLK), 0,
+                       ctx->stream, flags, gid, starts, m);
+    u32 G32 = 0;
+    HIP_CHECK(hipMemcpyAsync(&G32, gid + (m - 1), 4, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    u64 G = G32;
+    u64 *dg = dnew<u64>(ctx, G * kw);
+    u64 *gt = dnew<u64>(ctx, G);
+    i64 *gd = dnew<i64>(ctx, G);
+    hipLaunchKernelGGL(k_gather_group_keys, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, sk + lo * kw, kw, starts, gid, m, dg);
+    hipLaunchKernelGGL(k_fill_u64, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, gt, G, t);
+    hipLaunchKernelGGL(k_fill_i64, dim3(ngrid(G)), dim3(BLK), 0,
+                       ctx->stream, gd, G, (i64)1);
+    mz_gpu_updates su{};
+    su.keys = dg;
+    su.vals = nullptr;
+    su.times = gt;
+    su.diffs = gd;
+    su.n = G;
+    su.lower = t;
+    su.upper = t + 1;
+    su.on_device = 1;
+    mz_gpu_closure cl;
+    topk_pass_cl(kw, vb, cl);
+    mz_gpu_out *oldp = nullptr, *newp = nullptr;
+    rc = probe_impl(ctx, op->arr, &su, 0, PM_HALF_LE, 0, &cl, 1, &oldp);
+    if (rc) break;
+    DevUpdates sl{sk + lo * kw, vb ? sv + lo * vb : sv, stm + lo, sd + lo,
+                  m};
+    arr_insert_dev(ctx, op->arr, sl, t, t + 1);
+    rc = probe_impl(ctx, op->arr, &su, 0, PM_HALF_LE, 0, &cl, 1, &newp);
+    if (rc) break;
+    u64 capn = oldp->n + newp->n;
+    if (capn) {
+      u64 *pk = dnew<u64>(ctx, capn * kw);
+      u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(capn * vb, 1));
+      u64 *pt = dnew<u64>(ctx, capn);
+      i64 *pd = dnew<i64>(ctx, capn);
+      HIP_CHECK(hipMemsetAsync(pk, 0, capn * kw * 8, ctx->stream));
+      HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(capn * vb, 1),
+                               ctx->stream));
+      HIP_CHECK(hipMemsetAsync(pt, 0, capn * 8, ctx->stream));
+      HIP_CHECK(hipMemsetAsync(pd, 0, capn * 8, ctx->stream));
+      unsigned long long *ocount = (unsigned long long *)S.get(8);
+      HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+      topk_eval_emit(ctx, op, oldp, t, -1, pk, pv, pt, pd, ocount);
+      topk_eval_emit(ctx, op, newp, t, +1, pk, pv, pt, pd, ocount);
+      segs.push_back({pk, pv, pt, pd});
+      segn.push_back(capn);
+    }
+    mz_gpu_out_release(c, oldp);
+    mz_gpu_out_release(c, newp);
+    dfree(ctx, dg);
+    dfree(ctx, gt);
+    dfree(ctx, gd);
+  }
+  u64 total = 0;
+  for (u64 x : segn) total += x;
+  u64 errflag = 0;
+  if (!rc) {
+    if (total == 0) {
+      HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8,
+                               hipMemcpyDeviceToHost, ctx->stream));
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      if (!errflag)
+        *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                        dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, kw, vb);
+    } else {
+      u64 *ck2 = dnew<u64>(ctx, total * kw);
+      u8 *cv2 = (u8 *)dmalloc(ctx, std::max<u64>(total * vb, 1));
+      u64 *ct2 = dnew<u64>(ctx, total);
+      i64 *cd2 = dnew<i64>(ctx, total);
+      u64 base = 0;
+      for (size_t i = 0; i < segs.size(); i++) {
+        u64 m2 = segn[i];
+        HIP_CHECK(hipMemcpyAsync(ck2 + base * kw, segs[i][0],
+                                 m2 * kw * 8, hipMemcpyDeviceToDevice,
+                                 ctx->stream));
+        if (vb)
+          HIP_CHECK(hipMemcpyAsync(cv2 + base * vb, segs[i][1], m2 * vb,
+                                   hipMemcpyDeviceToDevice, ctx->stream));
+        HIP_CHECK(hipMemcpyAsync(ct2 + base, segs[i][2], m2 * 8,
+                                 hipMemcpyDeviceToDevice, ctx->stream));
+        HIP_CHECK(hipMemcpyAsync(cd2 + base, segs[i][3], m2 * 8,
+                                 hipMemcpyDeviceToDevice, ctx->stream));
+        base += m2;
+      }
+      DevUpdates pin{ck2, cv2, ct2, cd2, total};
+      u64 *ok;
+      u8 *ov;
+      u64 *ot;
+      i64 *od;
+      u64 Mc;
+      consolidate_dev(ctx, kw, vb, pin, &ok, &ov, &ot, &od, &Mc);
+      HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8,
+                               hipMemcpyDeviceToHost, ctx->stream));
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      for (void *p : {(void *)ck2, (void *)cv2, (void *)ct2, (void *)cd2})
+        dfree(ctx, p);
+      if (!errflag) *out = make_out(ok, ov, ot, od, Mc, kw, vb);
+    }
+  }
+  for (auto &sg : segs)
+    for (void *p : sg) dfree(ctx, p);
+  for (void *p : {(void *)perm, (void *)sk, (void *)sv, (void *)stm,
+                  (void *)sd})
+    dfree(ctx, p);
+  if (rc) return rc;
+  if (errflag) {
+    ctx->err = "negative multiplicities in TopK";
+    return -1;
+  }
   return 0;
 }
 

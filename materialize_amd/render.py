@@ -224,6 +224,30 @@ class ReduceOp:
         return self.ctx.reduce_push_dev(self.op, updates)
 
 
+class TopKPlan:
+    """render_topk / BasicTopKPlan surface (top_k.rs:289-311,
+    plan/top_k.rs): group key = the arrangement key; order columns over
+    the record val bytes; literal offset/limit."""
+
+    def __init__(self, spec):
+        self.spec = spec
+
+
+class TopKOp:
+    """build_topk (top_k.rs:322-418): maintains per-group kept windows."""
+
+    def __init__(self, ctx, plan: TopKPlan):
+        self.ctx = ctx
+        self.op = ctx.topk_create(plan.spec)
+
+    def push(self, updates):
+        return self.ctx.topk_push(self.op, updates)
+
+
+def render_topk(ctx, plan) -> "TopKOp":
+    return TopKOp(ctx, plan)
+
+
 class ThresholdPlan:
     """render_threshold / BasicThresholdPlan surface
     (src/compute/src/render/threshold.rs:100-116): the ensure_arrangement

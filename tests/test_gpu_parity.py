@@ -322,6 +322,93 @@ class TestThresholdParity:
                     "threshold empty")
 
 
+class TestTopKParity:
+    """render_topk Basic plan (top_k.rs:322-418): corrections parity under
+    churn across offset/limit/desc shapes."""
+
+    def _churn(self, rng, net, n, nkeys=10, vb=8):
+        ks, vs, ds = [], [], []
+        for _ in range(n):
+            k = int(rng.integers(0, nkeys))
+            v = int(rng.integers(-15, 15))
+            vbytes = v.to_bytes(8, "little", signed=True)[:vb]
+            cur = net.get((k, vbytes), 0)
+            if cur > 0 and rng.random() < 0.45:
+                d = -int(rng.integers(1, cur + 1))
+            else:
+                d = int(rng.integers(1, 3))
+            net[(k, vbytes)] = cur + d
+            ks.append(k)
+            vs.append(list(vbytes))
+            ds.append(d)
+        return (np.array(ks, np.int64).reshape(-1, 1),
+                np.array(vs, np.uint8), np.array(ds, np.int64))
+
+    @pytest.mark.parametrize("offset,limit,desc", [
+        (0, 3, 0), (0, 1, 1), (2, 4, 0), (0, -1, 0), (3, 2, 1),
+    ])
+    def test_churn(self, ctxs, offset, limit, desc):
+        g, o = ctxs
+        rng = np.random.default_rng(67 + offset + limit + desc)
+        spec = abi.topk_spec(abi.schema(1, 8), [(0, 8, desc)],
+                             offset=offset, limit=limit)
+        gop, oop = g.topk_create(spec), o.topk_create(spec)
+        net = {}
+        for step in range(4):
+            ks, vs, ds = self._churn(rng, net, 300)
+            u = abi.make_updates(ks, vs, np.full(300, step, np.uint64), ds,
+                                 step, step + 1)
+            assert_same(g.topk_push(gop, u), o.topk_push(oop, u),
+                        f"topk o{offset} l{limit} d{desc} step {step}")
+
+    def test_two_order_cols(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(71)
+        spec = abi.topk_spec(abi.schema(1, 8),
+                             [(0, 4, 1), (4, 4, 0)], offset=0, limit=3)
+        gop, oop = g.topk_create(spec), o.topk_create(spec)
+        for step in range(3):
+            n = 250
+            ks = rng.integers(0, 8, n).astype(np.int64).reshape(-1, 1)
+            a = rng.integers(0, 3, n).astype(np.int32)
+            b = rng.integers(-6, 6, n).astype(np.int32)
+            vals = np.zeros((n, 8), np.uint8)
+            vals[:, :4] = a.view(np.uint8).reshape(-1, 4)
+            vals[:, 4:] = b.view(np.uint8).reshape(-1, 4)
+            u = abi.make_updates(ks, vals, np.full(n, step, np.uint64),
+                                 np.ones(n, np.int64), step, step + 1)
+            assert_same(g.topk_push(gop, u), o.topk_push(oop, u),
+                        f"topk 2-col step {step}")
+
+    def test_multi_timestamp_push(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(73)
+        spec = abi.topk_spec(abi.schema(1, 8), [(0, 8, 0)], offset=1,
+                             limit=2)
+        gop, oop = g.topk_create(spec), o.topk_create(spec)
+        net = {}
+        cols = [self._churn(rng, net, 200) for _ in range(3)]
+        ks = np.concatenate([c[0] for c in cols])
+        vs = np.concatenate([c[1] for c in cols])
+        ds = np.concatenate([c[2] for c in cols])
+        times = np.repeat(np.arange(3, dtype=np.uint64), 200)
+        u = abi.make_updates(ks, vs, times, ds, 0, 3)
+        assert_same(g.topk_push(gop, u), o.topk_push(oop, u),
+                    "topk multi-timestamp")
+
+    def test_negative_multiplicity_errors(self, ctxs):
+        g, o = ctxs
+        spec = abi.topk_spec(abi.schema(1, 8), [(0, 8, 0)], offset=0,
+                             limit=2)
+        gop = g.topk_create(spec)
+        u = abi.make_updates(np.array([[1]], np.int64),
+                             np.zeros((1, 8), np.uint8),
+                             np.zeros(1, np.uint64),
+                             np.array([-1], np.int64), 0, 1)
+        with pytest.raises(Exception, match="[Nn]egative multiplicities"):
+            g.topk_push(gop, u)
+
+
 class TestRouteHashParity:
     def test_hash_agrees(self, ctxs):
         g, o = ctxs
