@@ -55,8 +55,8 @@ class CallProfiler:
     """Wraps an engine ctx's methods with wall-time accumulation."""
 
     METHODS = ["consolidate_dev", "arr_push", "arr_insert",
-               "halfjoin_dev", "reduce_push_dev", "join_push",
-               "arr_maintain"]
+               "halfjoin_dev", "reduce_push_dev", "reduce_push2_dev",
+               "join_push", "arr_maintain"]
 
     def __init__(self, ctx):
         self.t = {m: [0.0, 0] for m in self.METHODS}
@@ -189,6 +189,9 @@ def run_cpu_baseline(seed, cores=0, batch=100_000, steps=4, sf=1.0):
 
 def main():
     args = parse_args()
+    if args.call_profile:
+        # engine-side sub-phase event profiling (mz_gpu_prof_dump)
+        os.environ.setdefault("MZ_GPU_PROF", "1")
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
@@ -249,6 +252,9 @@ def main():
         dist.barrier()
 
     prof = CallProfiler(ctx) if args.call_profile else None
+    if prof:
+        print("# MZPROF setup+warmup (discard):", file=sys.stderr)
+        ctx.prof_dump()  # reset engine sub-phase counters before timing
     ctx.set_kernel_timing(1)
     t0 = time.perf_counter()
     for i in range(W, W + K):
@@ -265,6 +271,7 @@ def main():
         print(f"# step wall {elapsed/K*1e3:.2f} ms; breakdown:",
               file=sys.stderr)
         prof.report(K)
+        ctx.prof_dump()  # MZPROF sub-phase lines (stderr)
 
     if dist:
         e = torch.tensor([elapsed], device=device)

@@ -241,6 +241,11 @@ void mz_gpu_reduce_drop(mz_gpu_ctx *ctx, mz_gpu_red *op);
 int  mz_gpu_reduce_push(mz_gpu_ctx *ctx, mz_gpu_red *op,
                         const mz_gpu_updates *delta, mz_gpu_out **out);
 
+/* Diagnostics: when the environment sets MZ_GPU_PROF=1, the engine
+ * records per-phase HIP event pairs; this prints and resets the sums
+ * ("MZPROF <phase> <ms> <count>" lines on stdout). No-op otherwise. */
+void mz_gpu_prof_dump(mz_gpu_ctx *ctx);
+
 /* ----------------------------------------------------------- threshold
  * Replaces build_threshold_basic / threshold_local
  * (src/compute/src/render/threshold.rs:34-51,75-97): a reduce over the

@@ -89,6 +89,7 @@ def load():
                                         C.POINTER(Updates),
                                         C.POINTER(Updates),
                                         C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_prof_dump.argtypes = [C.c_void_p]
     lib.mz_gpu_topk_create.restype = C.c_void_p
     lib.mz_gpu_topk_create.argtypes = [C.c_void_p, C.POINTER(TopKSpec)]
     lib.mz_gpu_topk_push.argtypes = [C.c_void_p, C.c_void_p,
@@ -221,6 +222,9 @@ class GpuCtx:
         self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
                                                 C.byref(outp)))
         return self._take(outp)
+
+    def prof_dump(self):
+        self.lib.mz_gpu_prof_dump(self.ctx)
 
     def topk_create(self, spec):
         return self.lib.mz_gpu_topk_create(self.ctx, C.byref(spec))

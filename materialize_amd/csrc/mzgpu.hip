@@ -27,6 +27,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <map>
 #include <string>
 #include <vector>
 
@@ -1163,10 +1164,20 @@ struct mz_gpu_topk {
 
 namespace {
 
+// Sub-phase profiler (MZ_GPU_PROF=1): HIP event pairs per category,
+// summed at mz_gpu_prof_dump. Events are recorded on the ctx stream, so
+// sums are device-busy time per section (overlap-free on one stream).
+struct Prof {
+  bool enabled = false;
+  std::map<std::string,
+           std::vector<std::pair<hipEvent_t, hipEvent_t>>> cats;
+};
+
 struct Ctx {
   hipStream_t stream = nullptr;
   std::string err;
   Scratch scratch;
+  Prof prof;
   std::vector<mz_gpu_arr *> arrs;
   std::vector<mz_gpu_join *> joins;
   std::vector<mz_gpu_red *> reds;
@@ -1181,6 +1192,28 @@ struct Ctx {
 // Stream-ordered allocation: hipMallocAsync/hipFreeAsync on the ctx
 // stream avoid the synchronizing hipMalloc/hipFree (which dominated the
 // step time before this change — ~100 allocations per step).
+struct ProfScope {
+  Ctx *c = nullptr;
+  hipEvent_t a = nullptr, b = nullptr;
+  const char *name;
+  ProfScope(Ctx *ctx, const char *n) : name(n) {
+    if (!ctx->prof.enabled) return;
+    c = ctx;
+    (void)hipEventCreate(&a);
+    (void)hipEventCreate(&b);
+    (void)hipEventRecord(a, c->stream);
+  }
+  ~ProfScope() {
+    if (!c) return;
+    (void)hipEventRecord(b, c->stream);
+    c->prof.cats[name].push_back({a, b});
+  }
+};
+
+#define MZ_PROF_CAT2(a, b) a##b
+#define MZ_PROF_CAT(a, b) MZ_PROF_CAT2(a, b)
+#define MZ_PROF(ctx, name) ProfScope MZ_PROF_CAT(_ps, __LINE__)(ctx, name)
+
 void *dmalloc(Ctx *c, size_t bytes) {
   void *p = nullptr;
   if (bytes == 0) bytes = 16;
@@ -1202,6 +1235,7 @@ T *dnew(Ctx *c, u64 n) {
 // Sort passes use rocprim radix_sort_pairs (AMD-native primitive).
 void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                   const u64 *times, u64 n, u32 *perm, bool time_major) {
+  MZ_PROF(c, "sort_updates");
   auto &S = c->scratch;
   u64 *skey = (u64 *)S.get(n * 8);
   u64 *skey_out = (u64 *)S.get(n * 8);
@@ -1433,6 +1467,7 @@ void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
 
 void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
                       u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts) {
+  MZ_PROF(c, "consolidate_core");
   auto &S = c->scratch;
   u64 n = in.n;
   if (n == 0) {
@@ -1517,6 +1552,7 @@ void free_batch(Ctx *c, DevBatch &b) {
 DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
                           u64 *times, i64 *diffs, u64 cap, u64 lower,
                           u64 upper, u64 *dcounts) {
+  MZ_PROF(c, "build_batch");
   auto &S = c->scratch;
   DevBatch b;
   b.lower = lower;
@@ -1592,6 +1628,7 @@ DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
 // applied). Policy is the host's; semantics = concat + advance + consolidate.
 void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   if (to - from <= 1) return;
+  MZ_PROF(c, "merge_range");
   auto &S = c->scratch;
   S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
@@ -1605,14 +1642,17 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   u8 *vals = (u8 *)S.get(std::max<u64>(total * vb, 1));
   u64 *times = (u64 *)S.get(total * 8);
   i64 *diffs = (i64 *)S.get(total * 8);
-  u64 base = 0;
-  for (size_t i = from; i < to; i++) {
-    DevBatch &b = a->batches[i];
-    if (b.n_upds)
-      hipLaunchKernelGGL(k_expand_batch, dim3(ngrid(b.n_upds)), dim3(BLK), 0,
-                         c->stream, b, kw, vb, a->logical_compaction, keys,
-                         vals, times, diffs, base);
-    base += b.n_upds;
+  {
+    MZ_PROF(c, "merge_expand");
+    u64 base = 0;
+    for (size_t i = from; i < to; i++) {
+      DevBatch &b = a->batches[i];
+      if (b.n_upds)
+        hipLaunchKernelGGL(k_expand_batch, dim3(ngrid(b.n_upds)), dim3(BLK),
+                           0, c->stream, b, kw, vb, a->logical_compaction,
+                           keys, vals, times, diffs, base);
+      base += b.n_upds;
+    }
   }
   DevUpdates in{keys, vals, times, diffs, total};
   u64 capn = std::max<u64>(total, 1);
@@ -1628,16 +1668,22 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
     // (ColInternalMerger::merge semantics, columnation.rs:653-713).
     u64 n1 = a->batches[from].n_upds, n2 = a->batches[from + 1].n_upds;
     u32 *perm = (u32 *)S.get(total * 4);
-    RowLess cmp{keys, vals, times, kw, vb};
-    auto it1 = rocprim::make_counting_iterator<u32>(0u);
-    auto it2 = rocprim::make_counting_iterator<u32>((u32)n1);
-    size_t need = 0;
-    (void)rocprim::merge(nullptr, need, it1, it2, perm, (size_t)n1,
-                         (size_t)n2, cmp, c->stream);
-    void *tmp = S.get(need);
-    (void)rocprim::merge(tmp, need, it1, it2, perm, (size_t)n1, (size_t)n2,
-                         cmp, c->stream);
-    consolidate_with_perm(c, kw, vb, in, perm, ok, ov, ot, od, dcounts);
+    {
+      MZ_PROF(c, "merge_path");
+      RowLess cmp{keys, vals, times, kw, vb};
+      auto it1 = rocprim::make_counting_iterator<u32>(0u);
+      auto it2 = rocprim::make_counting_iterator<u32>((u32)n1);
+      size_t need = 0;
+      (void)rocprim::merge(nullptr, need, it1, it2, perm, (size_t)n1,
+                           (size_t)n2, cmp, c->stream);
+      void *tmp = S.get(need);
+      (void)rocprim::merge(tmp, need, it1, it2, perm, (size_t)n1,
+                           (size_t)n2, cmp, c->stream);
+    }
+    {
+      MZ_PROF(c, "merge_consol_perm");
+      consolidate_with_perm(c, kw, vb, in, perm, ok, ov, ot, od, dcounts);
+    }
   } else {
     consolidate_core(c, kw, vb, in, ok, ov, ot, od, dcounts);
   }
@@ -1689,7 +1735,31 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
                                             : (4ull << 30);
   c->impl.scratch.get(scratch0);
   c->impl.scratch.reset();
+  const char *prof = getenv("MZ_GPU_PROF");
+  c->impl.prof.enabled = prof && prof[0] && prof[0] != '0';
   return c;
+}
+
+// Sub-phase profile dump (MZ_GPU_PROF=1): per category, total device ms
+// over the recorded event pairs since the last dump. Printed to stdout as
+// "MZPROF <cat> <ms> <count>" lines; events are released.
+void mz_gpu_prof_dump(mz_gpu_ctx *c) {
+  Ctx *ctx = &c->impl;
+  if (!ctx->prof.enabled) return;
+  (void)hipStreamSynchronize(ctx->stream);
+  for (auto &[name, evs] : ctx->prof.cats) {
+    double total = 0;
+    for (auto &[a, b] : evs) {
+      float ms = 0;
+      if (hipEventElapsedTime(&ms, a, b) == hipSuccess) total += ms;
+      (void)hipEventDestroy(a);
+      (void)hipEventDestroy(b);
+    }
+    fprintf(stderr, "MZPROF %s %.3f %zu\n", name.c_str(), total,
+            evs.size());
+  }
+  fflush(stderr);
+  ctx->prof.cats.clear();
 }
 
 void mz_gpu_fini(mz_gpu_ctx *c) {
@@ -1788,6 +1858,7 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
 // a pointer to the pushed batch (valid until the next spine merge).
 static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
                                 u64 lower, u64 upper) {
+  MZ_PROF(ctx, "arr_insert");
   auto &S = ctx->scratch;
   u32 kw = a->schema.kw, vb = a->schema.vb;
   u64 capn = std::max<u64>(d.n, 1);
@@ -1815,6 +1886,7 @@ static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
 int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
                       const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
+  MZ_PROF(ctx, "arr_insert");
   ctx->scratch.reset();
   auto &S = ctx->scratch;
   u32 kw = a->schema.kw, vb = a->schema.vb;
