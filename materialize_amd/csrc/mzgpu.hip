@@ -55,6 +55,10 @@ __device__ __forceinline__ i64 wmul(i64 a, i64 b) {
 
 __global__ void k_iota(u32 *p, u64 n) { GRID_STRIDE(i, n) p[i] = (u32)i; }
 
+__global__ void k_iota_off(u32 *p, u64 n, u32 base) {
+  GRID_STRIDE(i, n) p[i] = base + (u32)i;
+}
+
 __global__ void k_gather_u64(const u64 *src, const u32 *perm, u64 *dst,
                              u64 n) {
   GRID_STRIDE(i, n) dst[i] = src[perm[i]];
@@ -1661,24 +1665,80 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   u64 *ot = dnew<u64>(c, capn);
   i64 *od = dnew<i64>(c, capn);
   u64 *dcounts = (u64 *)S.get(3 * 8);
-  if (to - from == 2 && total) {
-    // Both inputs are sorted runs (logical-compaction advance is monotone,
-    // so expansion preserves order): a single merge-path pass replaces the
-    // full radix re-sort — this is the Spine's pairwise merge proper
-    // (ColInternalMerger::merge semantics, columnation.rs:653-713).
-    u64 n1 = a->batches[from].n_upds, n2 = a->batches[from + 1].n_upds;
+  if (total) {
+    // All inputs are sorted runs (logical-compaction advance is monotone,
+    // so expansion preserves order): a merge-path tournament over the
+    // expanded index ranges replaces the full radix re-sort — this is the
+    // Spine's pairwise merge proper (ColInternalMerger::merge semantics,
+    // columnation.rs:653-713), generalized to k runs in ceil(log2 k)
+    // passes. First-pass inputs are counting ranges; later passes merge
+    // the sorted index arrays with the same row comparator.
     u32 *perm = (u32 *)S.get(total * 4);
     {
       MZ_PROF(c, "merge_path");
       RowLess cmp{keys, vals, times, kw, vb};
-      auto it1 = rocprim::make_counting_iterator<u32>(0u);
-      auto it2 = rocprim::make_counting_iterator<u32>((u32)n1);
-      size_t need = 0;
-      (void)rocprim::merge(nullptr, need, it1, it2, perm, (size_t)n1,
-                           (size_t)n2, cmp, c->stream);
-      void *tmp = S.get(need);
-      (void)rocprim::merge(tmp, need, it1, it2, perm, (size_t)n1,
-                           (size_t)n2, cmp, c->stream);
+      u32 *bufB = (u32 *)S.get(total * 4);
+      size_t tmpsz = 0;
+      {
+        size_t n1 = 0, n2 = 0;
+        auto it0 = rocprim::make_counting_iterator<u32>(0u);
+        (void)rocprim::merge(nullptr, n1, it0, it0, perm, (size_t)1,
+                             (size_t)1, cmp, c->stream);
+        (void)rocprim::merge(nullptr, n2, (const u32 *)perm,
+                             (const u32 *)perm, bufB, (size_t)1, (size_t)1,
+                             cmp, c->stream);
+        tmpsz = std::max(n1, n2);
+      }
+      void *tmp = S.get(tmpsz);
+      // runs as (start, len) index ranges over the expanded arrays
+      std::vector<std::pair<u64, u64>> runs;
+      u64 base2 = 0;
+      for (size_t i = from; i < to; i++) {
+        if (a->batches[i].n_upds)
+          runs.push_back({base2, a->batches[i].n_upds});
+        base2 += a->batches[i].n_upds;
+      }
+      bool first = true;
+      u32 *cur = nullptr;  // buffer holding the current runs' indices
+      while (first || runs.size() > 1) {
+        u32 *dst = (cur == perm) ? bufB : perm;
+        std::vector<std::pair<u64, u64>> next;
+        u64 outbase = 0;
+        for (size_t i = 0; i + 1 < runs.size(); i += 2) {
+          auto [s1, n1] = runs[i];
+          auto [s2, n2] = runs[i + 1];
+          size_t nb = tmpsz;
+          if (first) {
+            auto it1 = rocprim::make_counting_iterator<u32>((u32)s1);
+            auto it2 = rocprim::make_counting_iterator<u32>((u32)s2);
+            (void)rocprim::merge(tmp, nb, it1, it2, dst + outbase,
+                                 (size_t)n1, (size_t)n2, cmp, c->stream);
+          } else {
+            (void)rocprim::merge(tmp, nb, (const u32 *)(cur + s1),
+                                 (const u32 *)(cur + s2), dst + outbase,
+                                 (size_t)n1, (size_t)n2, cmp, c->stream);
+          }
+          next.push_back({outbase, n1 + n2});
+          outbase += n1 + n2;
+        }
+        if (runs.size() % 2) {
+          auto [s, n1] = runs.back();
+          if (first)
+            hipLaunchKernelGGL(k_iota_off, dim3(ngrid(n1)), dim3(BLK), 0,
+                               c->stream, dst + outbase, n1, (u32)s);
+          else
+            HIP_CHECK(hipMemcpyAsync(dst + outbase, cur + s, n1 * 4,
+                                     hipMemcpyDeviceToDevice, c->stream));
+          next.push_back({outbase, n1});
+          outbase += n1;
+        }
+        runs = std::move(next);
+        cur = dst;
+        first = false;
+      }
+      if (cur != perm)
+        HIP_CHECK(hipMemcpyAsync(perm, cur, total * 4,
+                                 hipMemcpyDeviceToDevice, c->stream));
     }
     {
       MZ_PROF(c, "merge_consol_perm");
@@ -1797,9 +1857,12 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // are O(n)); small batches pool lazily and merge k-way when the pool
   // exceeds 6 — per-merge fixed overhead (~25 kernel launches + syncs)
   // made per-step pair merges of 100k-row batches the dominant step cost.
+  // 4M default: measured 2x on the 1M-row churn config (per-step pair
+  // merges of 1M batches into the resident run were the dominant cost;
+  // pooling amortizes the big merge over POOL steps).
   static const u64 SMALL = [] {
     const char *e = getenv("MZ_GPU_SMALL");
-    return e ? (u64)atoll(e) : (u64)(1u << 20);
+    return e ? (u64)atoll(e) : (u64)(4u << 20);
   }();
   static const long POOL = [] {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
