@@ -207,3 +207,51 @@ class TestReduceGoldens:
         got = self._run(fx)
         want = sorted([(enc(k), list(v)) for k, v in fx["expect"]])
         assert got == want
+
+
+class TestThresholdGoldens:
+    def test_except_all_values(self):
+        """cockroach/union.slt EXCEPT ALL golden: Threshold(A + Negate(B))
+        keeps rows at their positive net multiplicity."""
+        fx = BY_NAME["threshold_except_all_values"]
+        ctx = OracleCtx()
+        op = ctx.threshold_create(abi.schema(1, 0))
+        rows = fx["input"]
+        keys = np.array([r[0] for r in rows], np.int64)
+        diffs = np.array([r[1] for r in rows], np.int64)
+        u = abi.make_updates(keys, None, np.zeros(len(rows), np.uint64),
+                             diffs, 0, 1)
+        k, v, t, d = ctx.threshold_push(op, u)
+        got = sorted((int(k[i]), int(d[i])) for i in range(len(t)))
+        assert got == sorted((a, b) for a, b in fx["expect"])
+
+
+class TestTopKGoldens:
+    @pytest.mark.parametrize("name", ["topk_cities_desc_nulls_last",
+                                      "topk_cities_desc_nulls_first"])
+    def test_cities(self, name):
+        """topk.slt per-state top-3 by pop DESC; NULL pop encoded at the
+        fixture's stated extreme (note field)."""
+        fx = BY_NAME[name]
+        null_pop = (-2**63) if fx["null_enc"] == "min" else (2**63 - 1)
+        rows = fx["cities"]
+        keys = np.array([enc(st) for (_, st, _) in rows], np.int64)
+        vals = np.zeros((len(rows), 16), np.uint8)
+        for i, (nm, _, pop) in enumerate(rows):
+            vals[i, :8] = np.array(
+                [null_pop if pop is None else pop], np.int64).view(np.uint8)
+            vals[i, 8:] = np.array([enc(nm)], np.int64).view(np.uint8)
+        spec = abi.topk_spec(abi.schema(1, 16), [(0, 8, 1)], offset=0,
+                             limit=fx["limit"])
+        ctx = OracleCtx()
+        op = ctx.topk_create(spec)
+        u = abi.make_updates(keys, vals, np.zeros(len(rows), np.uint64),
+                             np.ones(len(rows), np.int64), 0, 1)
+        k, v, t, d = ctx.topk_push(op, u)
+        n = len(t)
+        vv = v.reshape(n, 16)
+        got = sorted((int(k[i]),
+                      int(vv[i, 8:].copy().view(np.int64)[0]), int(d[i]))
+                     for i in range(n))
+        want = sorted((enc(st), enc(nm), 1) for st, nm in fx["expect"])
+        assert got == want
