@@ -1678,15 +1678,17 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
       MZ_PROF(c, "merge_path");
       RowLess cmp{keys, vals, times, kw, vb};
       u32 *bufB = (u32 *)S.get(total * 4);
+      // temp storage scales with input size (merge-path partitions):
+      // query an upper bound at (total, total) for both iterator shapes
       size_t tmpsz = 0;
       {
         size_t n1 = 0, n2 = 0;
         auto it0 = rocprim::make_counting_iterator<u32>(0u);
-        (void)rocprim::merge(nullptr, n1, it0, it0, perm, (size_t)1,
-                             (size_t)1, cmp, c->stream);
+        (void)rocprim::merge(nullptr, n1, it0, it0, perm, (size_t)total,
+                             (size_t)total, cmp, c->stream);
         (void)rocprim::merge(nullptr, n2, (const u32 *)perm,
-                             (const u32 *)perm, bufB, (size_t)1, (size_t)1,
-                             cmp, c->stream);
+                             (const u32 *)perm, bufB, (size_t)total,
+                             (size_t)total, cmp, c->stream);
         tmpsz = std::max(n1, n2);
       }
       void *tmp = S.get(tmpsz);
