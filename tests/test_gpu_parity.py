@@ -91,6 +91,34 @@ class TestConsolidateParity:
         assert_same(rg, ro, "cancel")
 
 
+class TestSpineMergeParity:
+    def test_pool_and_pair_merges(self, ctxs):
+        """Many small arr_inserts force the lazy pool's k-way tournament
+        merges and geometric pair merges; a halfjoin probe after every
+        insert checks the merged spine (hash tables, offsets,
+        consolidation) bit-exactly against the oracle."""
+        g, o = ctxs
+        rng = np.random.default_rng(83)
+        sch = abi.schema(1, 8)
+        ga, oa = g.arr_create(sch), o.arr_create(sch)
+        cl = concat_cl(0, 8)
+        for t in range(16):
+            n = 1500
+            keys = rng.integers(0, 400, (n, 1)).astype(np.int64)
+            vals = rng.integers(0, 4, n).astype(np.int64)
+            diffs = rng.integers(-2, 3, n).astype(np.int64)
+            u = abi.make_updates(keys, vals.view(np.uint8),
+                                 np.full(n, t, np.uint64), diffs, t, t + 1)
+            g.arr_insert(ga, u)
+            o.arr_insert(oa, u)
+            pk = rng.integers(0, 400, (200, 1)).astype(np.int64)
+            pu = abi.make_updates(pk, None, np.full(200, t, np.uint64),
+                                  np.ones(200, np.int64), t, t + 1)
+            assert_same(g.halfjoin(ga, pu, 0, True, cl),
+                        o.halfjoin(oa, pu, 0, True, cl),
+                        f"spine probe t={t}")
+
+
 class TestJoinParity:
     def _stream(self, ctxs, steps, kw, vb1, vb2, seed, nkeys=30):
         """Run the same per-step batches through GPU and oracle join ops;
