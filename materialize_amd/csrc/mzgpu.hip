@@ -2341,13 +2341,6 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
   HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
   u32 *flags = (u32 *)S.get(n * 4);
   u32 *gid = (u32 *)S.get(n * 4);
-  // zero-fill the corrections buffer: unwritten capacity rows carry diff 0
-  // and drop out in the consolidation — no emitted-count readback needed
-  HIP_CHECK(hipMemsetAsync(pk, 0, cap_out * okw * 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(cap_out * ovb, 1),
-                           ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pt, 0, cap_out * 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pd, 0, cap_out * 8, ctx->stream));
   for (auto [lo, hi] : slices) {
     u64 m = hi - lo;
     // key-group starts within the slice (group count stays on device)
@@ -2378,18 +2371,22 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
     hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
                        op->d_nrows, misspos, gid, m);
   }
-  // consolidate the capacity-sized corrections (zero-diff rows drop out)
-  DevUpdates pin{pk, pv, pt, pd, cap_out};
+  // consolidate the actual emitted corrections (one count readback —
+  // sorting the 2n-capacity zero-padded buffer dominated this path)
+  unsigned long long emitted = 0;
+  u64 errflag = 0;
+  HIP_CHECK(hipMemcpyAsync(&emitted, ocount, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
   u8 *ov;
   u64 *ot;
   i64 *od;
   u64 Mc;
   consolidate_dev(ctx, okw, ovb, pin, &ok, &ov, &ot, &od, &Mc);
-  u64 errflag = 0;
-  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
   for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
     dfree(ctx, (p));
   if (errflag) {
@@ -2481,11 +2478,6 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
   i64 *pd = dnew<i64>(ctx, cap_out);
   unsigned long long *ocount = (unsigned long long *)S.get(8);
   HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pk, 0, cap_out * kw * 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(cap_out * vb, 1),
-                           ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pt, 0, cap_out * 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(pd, 0, cap_out * 8, ctx->stream));
   u32 *flags = (u32 *)S.get(n * 4);
   u32 *gid = (u32 *)S.get(n * 4);
   for (auto [lo, hi] : slices) {
@@ -2514,17 +2506,20 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
     hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
                        op->d_nrows, misspos, gid, m);
   }
-  DevUpdates pin{pk, pv, pt, pd, cap_out};
+  unsigned long long emitted = 0;
+  u64 errflag = 0;
+  HIP_CHECK(hipMemcpyAsync(&emitted, ocount, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
+                           ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
   u8 *ov;
   u64 *ot;
   i64 *od;
   u64 Mc;
   consolidate_dev(ctx, kw, vb, pin, &ok, &ov, &ot, &od, &Mc);
-  u64 errflag = 0;
-  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
   for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
     dfree(ctx, p);
   if (errflag) {
