@@ -424,26 +424,32 @@ struct BatchList {
 
 enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 
-// Phase 1: count emitted pairs per delta update.
+// Phase 1: count emitted pairs per (delta update, batch) — one thread per
+// pair, idx = bi*n + i, so consecutive threads read consecutive delta
+// rows (coalesced) and each thread does ONE independent hash lookup
+// (10x the memory-level parallelism of the per-row batch loop on a
+// pooled spine; the probe is random-line bound, MI355X_MICROARCH §HBM).
 // delta: n updates (keys/vals/times/diffs columns); swap: delta is input 2
 // of a linear join (closure arg order is (key, v1=input1, v2=input2)).
-// ki_cache[n * bl.n] stores each (row, batch) hash-lookup result so the
+// ki_cache[bl.n * n] stores each (batch, row) hash-lookup result so the
 // emit phase never re-reads the hash tables (≈half the probe traffic).
 __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
                               const u64 *dtimes, u64 n, u32 kw, u32 lvb,
                               BatchList bl, int mode, int swap,
                               const mz_gpu_closure cl, u32 *count,
                               int *ki_cache) {
-  GRID_STRIDE(i, n) {
+  u64 total = n * (u64)bl.n;
+  GRID_STRIDE(idx, total) {
+    u64 i = idx % n;
+    int bi = (int)(idx / n);
     const u64 *key = dkeys + i * kw;
     const u8 *dv = dvals ? dvals + i * dvb : nullptr;
     u64 t = dtimes[i];
     u32 c = 0;
-    for (int bi = 0; bi < bl.n; bi++) {
-      const DevBatch &b = bl.b[bi];
-      int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
-      ki_cache[i * bl.n + bi] = ki;
-      if (ki < 0) continue;
+    const DevBatch &b = bl.b[bi];
+    int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
+    ki_cache[idx] = ki;
+    if (ki >= 0) {
       for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
         const u8 *v1 = swap ? lv : dv;
@@ -460,12 +466,12 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
         }
       }
     }
-    count[i] = c;
+    count[idx] = c;
   }
 }
 
 // Phase 2: emit pairs at offsets (deterministic: offset is a function of
-// the input index).
+// the (batch, row) index).
 __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              const u64 *dtimes, const i64 *ddiffs, u64 n,
                              u32 kw, u32 lvb, BatchList bl, int mode,
@@ -474,41 +480,42 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              const int *ki_cache, u64 *okeys, u8 *ovals,
                              u64 *otimes, i64 *odiffs) {
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
-  GRID_STRIDE(i, n) {
+  u64 total = n * (u64)bl.n;
+  GRID_STRIDE(idx, total) {
+    int ki = ki_cache[idx];
+    if (ki < 0) continue;
+    u64 i = idx % n;
+    int bi = (int)(idx / n);
     const u64 *key = dkeys + i * kw;
     const u8 *dv = dvals ? dvals + i * dvb : nullptr;
     u64 t = dtimes[i];
     i64 d1 = ddiffs[i];
-    u64 o = offs[i];
-    for (int bi = 0; bi < bl.n; bi++) {
-      const DevBatch &b = bl.b[bi];
-      int ki = ki_cache[i * bl.n + bi];
-      if (ki < 0) continue;
-      for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
-        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-        const u8 *v1 = swap ? lv : dv;
-        const u8 *v2 = swap ? dv : lv;
-        u64 okey[MAX_KW];
-        u8 oval[MAX_VB];
-        if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
-        for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
-          u64 tout;
-          if (bl.allpass[bi]) {
-            tout = t;  // t2 < delta lower <= t on every update
-          } else if (mode == PM_JOIN) {
-            u64 t2 = b.times[u];
-            tout = t2 > t ? t2 : t;
-          } else {
-            u64 t2 = b.times[u];
-            if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
-            tout = t;
-          }
-          for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
-          for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oval[c];
-          otimes[o] = tout;
-          odiffs[o] = wmul(d1, b.diffs[u]);
-          o++;
+    u64 o = offs[idx];
+    const DevBatch &b = bl.b[bi];
+    for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
+      const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+      const u8 *v1 = swap ? lv : dv;
+      const u8 *v2 = swap ? dv : lv;
+      u64 okey[MAX_KW];
+      u8 oval[MAX_VB];
+      if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
+      for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
+        u64 tout;
+        if (bl.allpass[bi]) {
+          tout = t;  // t2 < delta lower <= t on every update
+        } else if (mode == PM_JOIN) {
+          u64 t2 = b.times[u];
+          tout = t2 > t ? t2 : t;
+        } else {
+          u64 t2 = b.times[u];
+          if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
+          tout = t;
         }
+        for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
+        for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oval[c];
+        otimes[o] = tout;
+        odiffs[o] = wmul(d1, b.diffs[u]);
+        o++;
       }
     }
   }
@@ -1862,13 +1869,17 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // 4M default: measured 2x on the 1M-row churn config (per-step pair
   // merges of 1M batches into the resident run were the dominant cost;
   // pooling amortizes the big merge over POOL steps).
+  // 8M/8 defaults: measured best on the 1M-row churn config (231M rows/s
+  // vs 178M at 4M/6 and 90M at per-step pair merges; pooling amortizes
+  // the big-run rewrite over POOL steps, probe fan-out stays under the
+  // 10-batch cap).
   static const u64 SMALL = [] {
     const char *e = getenv("MZ_GPU_SMALL");
-    return e ? (u64)atoll(e) : (u64)(4u << 20);
+    return e ? (u64)atoll(e) : (u64)(8u << 20);
   }();
   static const long POOL = [] {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
-    return e ? atol(e) : 6;
+    return e ? atol(e) : 8;
   }();
   for (;;) {
     size_t nb = a->batches.size();
@@ -2094,20 +2105,21 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                     dnew<i64>(ctx, 1), 0, okw, ovb);
     return 0;
   }
-  u32 *count = (u32 *)S.get((n + 1) * 4);
-  int *ki_cache = (int *)S.get(n * (u64)bl.n * 4);
+  u64 nb2 = n * (u64)bl.n;
+  u32 *count = (u32 *)S.get((nb2 + 1) * 4);
+  int *ki_cache = (int *)S.get(nb2 * 4);
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
-  hipLaunchKernelGGL(k_probe_count, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
-                     d.keys, d.vals, stream_vb, d.times, n, kw, lvb, bl, mode,
-                     swap, *cl, count, ki_cache);
-  u32 *offs = (u32 *)S.get((n + 1) * 4);
-  u64 M = exclusive_scan_u32(ctx, count, offs, n);
+  hipLaunchKernelGGL(k_probe_count, dim3(ngrid(nb2)), dim3(BLK), 0,
+                     ctx->stream, d.keys, d.vals, stream_vb, d.times, n, kw,
+                     lvb, bl, mode, swap, *cl, count, ki_cache);
+  u32 *offs = (u32 *)S.get((nb2 + 1) * 4);
+  u64 M = exclusive_scan_u32(ctx, count, offs, nb2);
   u64 *pk = dnew<u64>(ctx, std::max<u64>(M, 1) * okw);
   u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(M * ovb, 1));
   u64 *pt = dnew<u64>(ctx, std::max<u64>(M, 1));
   i64 *pd = dnew<i64>(ctx, std::max<u64>(M, 1));
   if (M)
-    hipLaunchKernelGGL(k_probe_emit, dim3(ngrid(n)), dim3(BLK), 0,
+    hipLaunchKernelGGL(k_probe_emit, dim3(ngrid(nb2)), dim3(BLK), 0,
                        ctx->stream, d.keys, d.vals, stream_vb, d.times,
                        d.diffs, n, kw, lvb, bl, mode, swap, *cl, offs,
                        ki_cache, pk, pv, pt, pd);
