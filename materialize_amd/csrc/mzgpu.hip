@@ -163,6 +163,38 @@ __global__ void k_sortkey_u64(const u64 *src, const u32 *perm, u64 *out,
   GRID_STRIDE(i, n) out[i] = src[perm[i]] - sub;
 }
 
+// Composite radix key: several narrowed columns packed into one u64
+// (least-significant column at shift 0 == the LSD pass order), so one
+// rocprim sort replaces up to MAX_COMP per-column passes.
+#define MAX_COMP 6
+struct CompSpec {
+  u32 n;
+  u32 kind[MAX_COMP];  // 0 = time, 1 = val word, 2 = key word
+  u32 word[MAX_COMP];
+  u64 sub[MAX_COMP];
+  u32 shift[MAX_COMP];
+};
+
+__global__ void k_sortkey_comp(const u64 *keys, u32 kw, const u8 *vals,
+                               u32 vb, const u64 *times, const u32 *perm,
+                               u64 *out, u64 n, CompSpec sp) {
+  GRID_STRIDE(i, n) {
+    u64 p = perm[i];
+    u64 acc = 0;
+    for (u32 c = 0; c < sp.n; c++) {
+      u64 x;
+      if (sp.kind[c] == 0)
+        x = times[p];
+      else if (sp.kind[c] == 1)
+        x = le_val_word(vals + p * vb + sp.word[c] * 8, vb - sp.word[c] * 8);
+      else
+        x = keys[p * kw + sp.word[c]] ^ 0x8000000000000000ULL;
+      acc |= (x - sp.sub[c]) << sp.shift[c];
+    }
+    out[i] = acc;
+  }
+}
+
 // head flags over the permuted (key,val,time) order
 __global__ void k_head_flags(const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                              const u64 *times, const u32 *perm, u32 *flags,
@@ -1303,31 +1335,70 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
     passes.push_back({kind, word, hmin[s],
                       64 - (int)__builtin_clzll(range)});
   }
+  // Pack consecutive LSD passes into <=64-bit composite chunks: one
+  // k_sortkey_comp + one rocprim sort per chunk replaces per-column
+  // passes (typical TPC-H columns are 14-30 live bits, so 3-5 columns
+  // collapse into 1-2 sorts).
+  std::vector<std::vector<Pass>> chunks;
+  u32 curbits = 0;
   for (auto &p : passes) {
-    if (p.kind == 0)
-      hipLaunchKernelGGL(k_sortkey_u64, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, times, perm, skey, n, p.sub);
-    else if (p.kind == 1)
-      hipLaunchKernelGGL(k_sortkey_val, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, vals, vb, p.word, perm, skey, n, p.sub);
-    else
-      hipLaunchKernelGGL(k_sortkey_key, dim3(ngrid(n)), dim3(BLK), 0,
-                         c->stream, keys, kw, p.word, perm, skey, n, p.sub);
+    if (chunks.empty() || curbits + (u32)p.bits > 64 ||
+        chunks.back().size() == MAX_COMP) {
+      chunks.push_back({});
+      curbits = 0;
+    }
+    chunks.back().push_back(p);
+    curbits += (u32)p.bits;
+  }
+  size_t nchunks = 0;
+  for (auto &ch : chunks) {
+    int totbits = 0;
+    if (ch.size() == 1) {
+      const Pass &p = ch[0];
+      totbits = p.bits;
+      if (p.kind == 0)
+        hipLaunchKernelGGL(k_sortkey_u64, dim3(ngrid(n)), dim3(BLK), 0,
+                           c->stream, times, perm, skey, n, p.sub);
+      else if (p.kind == 1)
+        hipLaunchKernelGGL(k_sortkey_val, dim3(ngrid(n)), dim3(BLK), 0,
+                           c->stream, vals, vb, p.word, perm, skey, n,
+                           p.sub);
+      else
+        hipLaunchKernelGGL(k_sortkey_key, dim3(ngrid(n)), dim3(BLK), 0,
+                           c->stream, keys, kw, p.word, perm, skey, n,
+                           p.sub);
+    } else {
+      CompSpec sp{};
+      sp.n = (u32)ch.size();
+      u32 shift = 0;
+      for (size_t i2 = 0; i2 < ch.size(); i2++) {
+        sp.kind[i2] = (u32)ch[i2].kind;
+        sp.word[i2] = ch[i2].word;
+        sp.sub[i2] = ch[i2].sub;
+        sp.shift[i2] = shift;
+        shift += (u32)ch[i2].bits;
+      }
+      totbits = (int)shift;
+      hipLaunchKernelGGL(k_sortkey_comp, dim3(ngrid(n)), dim3(BLK), 0,
+                         c->stream, keys, kw, vals, vb, times, perm, skey,
+                         n, sp);
+    }
     size_t need = 0;
     (void)rocprim::radix_sort_pairs(nullptr, need, skey, skey_out, perm,
-                                    perm_out, (unsigned)n, 0, p.bits,
+                                    perm_out, (unsigned)n, 0, totbits,
                                     c->stream);
     if (need > tmp_bytes) {
       tmp = S.get(need);
       tmp_bytes = need;
     }
     (void)rocprim::radix_sort_pairs(tmp, tmp_bytes, skey, skey_out, perm,
-                                    perm_out, (unsigned)n, 0, p.bits,
+                                    perm_out, (unsigned)n, 0, totbits,
                                     c->stream);
     std::swap(perm, perm_out);
+    nchunks++;
   }
   // ensure the result lands in the caller's buffer
-  if (passes.size() % 2 == 1) {
+  if (nchunks % 2 == 1) {
     HIP_CHECK(hipMemcpyAsync(perm_out, perm, n * 4, hipMemcpyDeviceToDevice,
                              c->stream));
     std::swap(perm, perm_out);
