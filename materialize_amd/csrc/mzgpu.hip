@@ -469,7 +469,7 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
                               const u64 *dtimes, u64 n, u32 kw, u32 lvb,
                               BatchList bl, int mode, int swap,
                               const mz_gpu_closure cl, u32 *count,
-                              int *ki_cache) {
+                              u64 *ki_cache) {
   u64 total = n * (u64)bl.n;
   GRID_STRIDE(idx, total) {
     u64 i = idx % n;
@@ -480,9 +480,13 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
     u32 c = 0;
     const DevBatch &b = bl.b[bi];
     int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
-    ki_cache[idx] = ki;
+    // cache the val range (not the key index): the emit pass re-reads
+    // neither the hash tables nor the kv_off lines
+    u64 kvr = ki < 0 ? ~0ull
+                     : ((u64)b.kv_off[ki] | ((u64)b.kv_off[ki + 1] << 32));
+    ki_cache[idx] = kvr;
     if (ki >= 0) {
-      for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
+      for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
         const u8 *v1 = swap ? lv : dv;
         const u8 *v2 = swap ? dv : lv;
@@ -509,13 +513,13 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              u32 kw, u32 lvb, BatchList bl, int mode,
                              int swap, const mz_gpu_closure cl,
                              const u32 *offs /*exclusive*/,
-                             const int *ki_cache, u64 *okeys, u8 *ovals,
+                             const u64 *ki_cache, u64 *okeys, u8 *ovals,
                              u64 *otimes, i64 *odiffs) {
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   u64 total = n * (u64)bl.n;
   GRID_STRIDE(idx, total) {
-    int ki = ki_cache[idx];
-    if (ki < 0) continue;
+    u64 kvr = ki_cache[idx];
+    if (kvr == ~0ull) continue;
     u64 i = idx % n;
     int bi = (int)(idx / n);
     const u64 *key = dkeys + i * kw;
@@ -524,7 +528,7 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
     i64 d1 = ddiffs[i];
     u64 o = offs[idx];
     const DevBatch &b = bl.b[bi];
-    for (u32 j = b.kv_off[ki]; j < b.kv_off[ki + 1]; j++) {
+    for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
       const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
       const u8 *v1 = swap ? lv : dv;
       const u8 *v2 = swap ? dv : lv;
@@ -2178,7 +2182,7 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   }
   u64 nb2 = n * (u64)bl.n;
   u32 *count = (u32 *)S.get((nb2 + 1) * 4);
-  int *ki_cache = (int *)S.get(nb2 * 4);
+  u64 *ki_cache = (u64 *)S.get(nb2 * 8);
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   hipLaunchKernelGGL(k_probe_count, dim3(ngrid(nb2)), dim3(BLK), 0,
                      ctx->stream, d.keys, d.vals, stream_vb, d.times, n, kw,
