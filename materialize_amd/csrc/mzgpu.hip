@@ -115,47 +115,47 @@ __global__ void k_sortkey_val(const u8 *vals, u32 vb, u32 word,
 __global__ void k_pass_minmax(const u64 *keys, u32 kw, const u8 *vals,
                               u32 vb, const u64 *times, u64 n, u32 vwords,
                               int with_time, u64 *mins, u64 *maxs) {
-  // per-thread register accumulation; one LDS merge + one global atomic
-  // per block per pass (per-element LDS atomics serialized badly)
-  __shared__ u64 smin[MAX_PASSES], smax[MAX_PASSES];
-  u32 np = (with_time ? 1 : 0) + vwords + kw;
-  u64 lmin[MAX_PASSES], lmax[MAX_PASSES];
-  for (u32 p = 0; p < np; p++) {
-    lmin[p] = ~0ull;
-    lmax[p] = 0;
+  // one column per blockIdx.y: two registers per thread (the
+  // dynamically-indexed per-column register arrays of the 1-D version
+  // spilled to scratch and ran at 0.4 TB/s), one LDS merge + one global
+  // atomic pair per block.
+  u32 p = blockIdx.y;
+  u32 t0 = with_time ? 1u : 0u;
+  int kind;
+  u32 word = 0;
+  if (with_time && p == 0) {
+    kind = 0;
+  } else if (p < t0 + vwords) {
+    kind = 1;
+    word = p - t0;
+  } else {
+    kind = 2;
+    word = p - t0 - vwords;
   }
-  if (threadIdx.x < np) {
-    smin[threadIdx.x] = ~0ull;
-    smax[threadIdx.x] = 0;
-  }
-  __syncthreads();
+  u64 lmin = ~0ull, lmax = 0;
   GRID_STRIDE(i, n) {
-    u32 p = 0;
-    if (with_time) {
-      u64 x = times[i];
-      lmin[p] = x < lmin[p] ? x : lmin[p];
-      lmax[p] = x > lmax[p] ? x : lmax[p];
-      p++;
-    }
-    for (u32 w = 0; w < vwords; w++, p++) {
-      u64 x = le_val_word(vals + i * vb + w * 8, vb - w * 8);
-      lmin[p] = x < lmin[p] ? x : lmin[p];
-      lmax[p] = x > lmax[p] ? x : lmax[p];
-    }
-    for (u32 w = 0; w < kw; w++, p++) {
-      u64 x = keys[i * kw + w] ^ 0x8000000000000000ULL;
-      lmin[p] = x < lmin[p] ? x : lmin[p];
-      lmax[p] = x > lmax[p] ? x : lmax[p];
-    }
+    u64 x;
+    if (kind == 0)
+      x = times[i];
+    else if (kind == 1)
+      x = le_val_word(vals + i * vb + word * 8, vb - word * 8);
+    else
+      x = keys[i * kw + word] ^ 0x8000000000000000ULL;
+    lmin = x < lmin ? x : lmin;
+    lmax = x > lmax ? x : lmax;
   }
-  for (u32 p = 0; p < np; p++) {
-    atomicMin((unsigned long long *)&smin[p], lmin[p]);
-    atomicMax((unsigned long long *)&smax[p], lmax[p]);
+  __shared__ u64 smin, smax;
+  if (threadIdx.x == 0) {
+    smin = ~0ull;
+    smax = 0;
   }
   __syncthreads();
-  if (threadIdx.x < np) {
-    atomicMin((unsigned long long *)&mins[threadIdx.x], smin[threadIdx.x]);
-    atomicMax((unsigned long long *)&maxs[threadIdx.x], smax[threadIdx.x]);
+  atomicMin((unsigned long long *)&smin, lmin);
+  atomicMax((unsigned long long *)&smax, lmax);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicMin((unsigned long long *)&mins[p], smin);
+    atomicMax((unsigned long long *)&maxs[p], smax);
   }
 }
 __global__ void k_sortkey_u64(const u64 *src, const u32 *perm, u64 *out,
@@ -326,9 +326,6 @@ __global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
   }
 }
 
-__global__ void k_hash_init(u64 *hash, u64 slots, u32 kw) {
-  GRID_STRIDE(i, slots) hash[i * (kw + 1) + kw] = ~0ull;
-}
 
 __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
                              const u32 *kid, u64 cap, const u64 *dcounts) {
@@ -1300,7 +1297,8 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
   HIP_CHECK(hipMemsetAsync(dmin, 0xFF, MAX_PASSES * 8, c->stream));
   HIP_CHECK(hipMemsetAsync(dmax, 0, MAX_PASSES * 8, c->stream));
   if (n)
-    hipLaunchKernelGGL(k_pass_minmax, dim3(ngrid(n)), dim3(BLK), 0,
+    hipLaunchKernelGGL(k_pass_minmax,
+                       dim3(ngrid(n), 1 + vwords + kw), dim3(BLK), 0,
                        c->stream, keys, kw, vals, vb, times, n, vwords, 1,
                        dmin, dmax);
   u64 hmin[MAX_PASSES], hmax[MAX_PASSES];
@@ -1680,8 +1678,9 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   while (slots < 2 * cap) slots <<= 1;
   b.hash_slots = slots;
   b.hash = dnew<u64>(c, slots * (kw + 1));
-  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
-                     c->stream, b.hash, slots, kw);
+  // full-line 0xFF fill: hash_lookup checks the idx-word sentinel
+  // before key compares, so poisoned key words are never read
+  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 1) * 8, c->stream));
   hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
                      c->stream, b.hash, slots, b.keys, kw, kid, cap,
                      dcounts);
@@ -2310,8 +2309,8 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
   r->d_err = dnew<u64>(ctx, 1);
   HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
-  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
-                     ctx->stream, r->st.hash, slots, kw);
+  HIP_CHECK(hipMemsetAsync(r->st.hash, 0xFF, slots * (kw + 1) * 8,
+                           ctx->stream));
   c->impl.reds.push_back(r);
   return r;
 }
@@ -2502,8 +2501,8 @@ mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
   r->d_err = dnew<u64>(ctx, 1);
   HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
   HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
-  hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
-                     ctx->stream, r->st.hash, slots, r->kw2);
+  HIP_CHECK(hipMemsetAsync(r->st.hash, 0xFF, slots * (r->kw2 + 1) * 8,
+                           ctx->stream));
   return r;
 }
 
@@ -3059,8 +3058,8 @@ mz_gpu_minmax *mz_gpu_minmax_create(mz_gpu_ctx *c, const mz_gpu_schema *in,
     st.stride_w = kw2 + 2;  // [key][exists][value]
     st.rows = dnew<u64>(ctx, cap * st.stride_w);
     st.capacity = cap;
-    hipLaunchKernelGGL(k_hash_init, dim3(ngrid(slots)), dim3(BLK), 0,
-                       ctx->stream, st.hash, slots, kw2);
+    HIP_CHECK(hipMemsetAsync(st.hash, 0xFF, slots * (kw2 + 1) * 8,
+                             ctx->stream));
   }
   return op;
 }
