@@ -328,17 +328,19 @@ __global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
 
 
 __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
-                             const u32 *kid, u64 cap, const u64 *dcounts) {
+                             const u32 *kid, const u32 *kv_off, u64 cap,
+                             const u64 *dcounts) {
   u64 n_keys = (cap && dcounts[0]) ? kid[cap - 1] : 0;
   GRID_STRIDE(i, n_keys) {
     u64 h = route_hash(keys + i * kw, kw) & (slots - 1);
     for (;;) {
-      u64 *slot = hash + h * (kw + 1);
+      u64 *slot = hash + h * (kw + 2);
       unsigned long long expected = ~0ull;
       unsigned long long got = atomicCAS(
           (unsigned long long *)(slot + kw), expected, (unsigned long long)i);
       if (got == ~0ull) {
         for (u32 w = 0; w < kw; w++) slot[w] = keys[i * kw + w];
+        slot[kw + 1] = (u64)kv_off[i] | ((u64)kv_off[i + 1] << 32);
         break;
       }
       h = (h + 1) & (slots - 1);
@@ -347,16 +349,35 @@ __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
 }
 
 __device__ __forceinline__ int hash_lookup(const u64 *hash, u64 slots,
-                                           const u64 *key, u32 kw) {
+                                           const u64 *key, u32 kw,
+                                           u32 sw) {
   if (slots == 0) return -1;
   u64 h = route_hash(key, kw) & (slots - 1);
   for (;;) {
-    const u64 *slot = hash + h * (kw + 1);
+    const u64 *slot = hash + h * sw;
     u64 iw = slot[kw];
     if (iw == ~0ull) return -1;
     bool eq = true;
     for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
     if (eq) return (int)(u32)iw;
+    h = (h + 1) & (slots - 1);
+  }
+}
+
+// Batch-table lookup returning the key's packed val range
+// (kv_lo | kv_hi<<32) straight from the widened slot — one random line
+// instead of two (no kv_off read). ~0 = miss.
+__device__ __forceinline__ u64 hash_lookup_range(const u64 *hash, u64 slots,
+                                                 const u64 *key, u32 kw) {
+  if (slots == 0) return ~0ull;
+  u64 h = route_hash(key, kw) & (slots - 1);
+  for (;;) {
+    const u64 *slot = hash + h * (kw + 2);
+    u64 iw = slot[kw];
+    if (iw == ~0ull) return ~0ull;
+    bool eq = true;
+    for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
+    if (eq) return slot[kw + 1];
     h = (h + 1) & (slots - 1);
   }
 }
@@ -476,13 +497,10 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
     u64 t = dtimes[i];
     u32 c = 0;
     const DevBatch &b = bl.b[bi];
-    int ki = hash_lookup(b.hash, b.hash_slots, key, kw);
-    // cache the val range (not the key index): the emit pass re-reads
-    // neither the hash tables nor the kv_off lines
-    u64 kvr = ki < 0 ? ~0ull
-                     : ((u64)b.kv_off[ki] | ((u64)b.kv_off[ki + 1] << 32));
+    // the val range rides in the widened slot: one random line per probe
+    u64 kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
     ki_cache[idx] = kvr;
-    if (ki >= 0) {
+    if (kvr != ~0ull) {
       for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
         const u8 *v1 = swap ? lv : dv;
@@ -750,7 +768,7 @@ __global__ void k_red_lookup(const u64 *keys, u32 kw, const u32 *gstart,
   if (gidn) G = m ? gidn[m - 1] : 0;
   GRID_STRIDE(g, G) {
     const u64 *key = keys + (u64)gstart[g] * kw;
-    int idx = hash_lookup(st.hash, st.slots, key, kw);
+    int idx = hash_lookup(st.hash, st.slots, key, kw, kw + 1);
     found[g] = idx < 0 ? ~0u : (u32)idx;
     miss[g] = idx < 0 ? 1u : 0u;
   }
@@ -1046,12 +1064,12 @@ __global__ void k_minmax_apply(const u64 *gkeys, u64 G, u32 kw2,
     const u64 *key = gkeys + g * kw2;
     // per-batch cursors over the key's val range
     u32 cur[12], end[12];
-    int ki[12];
     for (int b = 0; b < bl.n; b++) {
-      ki[b] = hash_lookup(bl.b[b].hash, bl.b[b].hash_slots, key, kw2);
-      if (ki[b] >= 0) {
-        cur[b] = bl.b[b].kv_off[ki[b]];
-        end[b] = bl.b[b].kv_off[ki[b] + 1];
+      u64 kvr = hash_lookup_range(bl.b[b].hash, bl.b[b].hash_slots, key,
+                                  kw2);
+      if (kvr != ~0ull) {
+        cur[b] = (u32)kvr;
+        end[b] = (u32)(kvr >> 32);
       } else {
         cur[b] = end[b] = 0;
       }
@@ -1677,13 +1695,13 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   u64 slots = 16;
   while (slots < 2 * cap) slots <<= 1;
   b.hash_slots = slots;
-  b.hash = dnew<u64>(c, slots * (kw + 1));
-  // full-line 0xFF fill: hash_lookup checks the idx-word sentinel
-  // before key compares, so poisoned key words are never read
-  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 1) * 8, c->stream));
+  b.hash = dnew<u64>(c, slots * (kw + 2));
+  // full-line 0xFF fill: lookups check the idx-word sentinel before key
+  // compares, so poisoned key words are never read
+  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 2) * 8, c->stream));
   hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
-                     c->stream, b.hash, slots, b.keys, kw, kid, cap,
-                     dcounts);
+                     c->stream, b.hash, slots, b.keys, kw, kid, b.kv_off,
+                     cap, dcounts);
   // flat key/val arrays were re-packed; stream-ordered free is safe
   dfree(c, keys);
   dfree(c, vals);
@@ -2083,7 +2101,7 @@ int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
   for (auto &b : a->batches) {
     n += b.n_upds;
     by += b.n_keys * kw * 8 + b.n_vals * vb + b.n_upds * 16 +
-          b.hash_slots * (kw + 1) * 8 + (b.n_keys + b.n_vals) * 4 +
+          b.hash_slots * (kw + 2) * 8 + (b.n_keys + b.n_vals) * 4 +
           b.n_upds * 4 + b.n_vals * 4;
   }
   *n_updates = n;
