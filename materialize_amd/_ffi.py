@@ -326,6 +326,12 @@ class GpuCtx:
             C.byref(cl), C.byref(outp)))
         return self._dev_out(outp)
 
+    def join_push_dev(self, op, side, upd):
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_join_push(self.ctx, op, side,
+                                              C.byref(upd), C.byref(outp)))
+        return self._dev_out(outp)
+
     def threshold_push_dev(self, op, upd):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_threshold_push(self.ctx, op,

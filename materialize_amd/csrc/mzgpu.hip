@@ -1279,6 +1279,14 @@ struct ProfScope {
 void *dmalloc(Ctx *c, size_t bytes) {
   void *p = nullptr;
   if (bytes == 0) bytes = 16;
+  // Coarse size classes for large blocks: spine merges allocate
+  // slightly different multi-hundred-MB sizes every cycle, and an exact
+  // pool can never reuse them — each miss carves fresh memory from the
+  // driver (tens of ms per big merge). Rounding >16 MB requests to
+  // 32 MB multiples makes the classes recur (288 GB HBM absorbs the
+  // slack).
+  if (bytes > (16u << 20))
+    bytes = (bytes + (32u << 20) - 1) & ~((size_t)(32u << 20) - 1);
   HIP_CHECK(hipMallocAsync(&p, bytes, c->stream));
   return p;
 }
@@ -1961,17 +1969,17 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // 4M default: measured 2x on the 1M-row churn config (per-step pair
   // merges of 1M batches into the resident run were the dominant cost;
   // pooling amortizes the big merge over POOL steps).
-  // 8M/8 defaults: measured best on the 1M-row churn config (231M rows/s
-  // vs 178M at 4M/6 and 90M at per-step pair merges; pooling amortizes
-  // the big-run rewrite over POOL steps, probe fan-out stays under the
-  // 10-batch cap).
+  // 4M/6 defaults: measured best at honest steady state (20-step runs
+  // spanning full pool cycles: 4.06 ms/step vs 12.6 at 8M/8 — deep pools
+  // inflate probe fan-out and the amortized big-run rewrite; per-step
+  // pair merges without pooling ran 10+ ms/step).
   static const u64 SMALL = [] {
     const char *e = getenv("MZ_GPU_SMALL");
-    return e ? (u64)atoll(e) : (u64)(8u << 20);
+    return e ? (u64)atoll(e) : (u64)(4u << 20);
   }();
   static const long POOL = [] {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
-    return e ? atol(e) : 8;
+    return e ? atol(e) : 6;
   }();
   for (;;) {
     size_t nb = a->batches.size();

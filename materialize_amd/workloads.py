@@ -500,3 +500,38 @@ class Q17Dataflow:
     def step(self, churn, t):
         lp = churn["lineitem_by_part"]
         return self._push(t, lp, None)
+
+    def step_dev(self, lp_u, t):
+        """Bench path: one churn step whose lineitem-by-partkey update
+        columns are ALREADY staged (device Updates at time t; the part
+        input is idle in the churn workload). Same drain order as
+        `_push`, with interior streams staying device-resident: the two
+        halves of each join's delta push separately (exactly-once per
+        DESIGN.md §5 holds — both probe the not-yet-updated other side),
+        and two-stream reduce inputs go through reduce_push2."""
+        ctx = self.ctx
+        # --- join1 drain: side1 (lineitem) first; part idle this step
+        ctx.arr_insert(self.arr_l0, lp_u)
+        l1a = ctx.join_push_dev(self.j1, 1, lp_u)   # probes part
+        # --- join2 drain: side2 (lineitem) BEFORE the distinct delta
+        j2b = ctx.join_push_dev(self.j2, 2, lp_u)   # probes dist (old)
+        dcorr = ctx.reduce_push_dev(self.distinct, l1a.updates(t, t + 1))
+        dcorr_u = dcorr.updates(t, t + 1)
+        self.ctx.arr_insert(self.arr_dist, dcorr_u)
+        j2a = ctx.join_push_dev(self.j2, 1, dcorr_u)  # probes l0 (new)
+        # --- per-partkey sum(quantity)/count over both join2 halves
+        acorr = ctx.reduce_push2_dev(self.avg, j2a.updates(t, t + 1),
+                                     j2b.updates(t, t + 1))
+        acorr_u = acorr.updates(t, t + 1)
+        # --- join3 drain: side1 (l1 delta) BEFORE the avg corrections
+        l1_u = l1a.updates(t, t + 1)
+        ctx.arr_insert(self.arr_l1, l1_u)
+        j3a = ctx.join_push_dev(self.j3, 1, l1_u)     # probes avg (old)
+        ctx.arr_insert(self.arr_avg, acorr_u)
+        j3b = ctx.join_push_dev(self.j3, 2, acorr_u)  # probes l1 (new)
+        tcorr = ctx.reduce_push2_dev(self.total, j3a.updates(t, t + 1),
+                                     j3b.updates(t, t + 1))
+        cols = tcorr.to_host()
+        self._apply_total(cols)
+        for o in (l1a, j2a, j2b, dcorr, acorr, j3a, j3b, tcorr):
+            o.release()
