@@ -485,10 +485,17 @@ class Q17Dataflow:
         ctx.arr_insert(self.arr_avg, acorr_u)
         j3b = ctx.join_push(self.j3, 2, acorr_u)     # probes l1 (new)
         e_cols = [np.concatenate([a, b]) for a, b in zip(j3a, j3b)]
+        e_cols = self._route_total(e_cols)
         e_u = abi.make_updates(*e_cols, t, t + 1)
         tcorr = ctx.reduce_push(self.total, e_u)
         self._apply_total(tcorr)
         return len(lp_k) + len(p_k)
+
+    def _route_total(self, cols):
+        """Hook: the global-SUM input is keyed by a constant, so the
+        sharded variant exchanges it to the key's owner rank (timely
+        Exchange for a keyed-by-unit reduce); unsharded = identity."""
+        return cols
 
     def load(self, gen):
         lp_k, lp_v = gen.lineitem_bypart_updates()
@@ -534,4 +541,81 @@ class Q17Dataflow:
         cols = tcorr.to_host()
         self._apply_total(cols)
         for o in (l1a, j2a, j2b, dcorr, acorr, j3a, j3b, tcorr):
+            o.release()
+
+
+class ShardedQ17Dataflow(Q17Dataflow):
+    """Q17 sharded by l_partkey hash (config 5 at N GPUs): every join
+    and the per-partkey AVG key on partkey, so the dataflow is
+    shard-local end-to-end; only the final global SUM's input stream is
+    exchanged to the constant key's owner rank (route_hash(0) % world).
+    Weak scaling with one tiny all-to-all per step."""
+
+    def __init__(self, ctx, exchange):
+        from .dist import shard_of
+        self.shard_of = shard_of
+        self.exchange = exchange
+        super().__init__(ctx)
+
+    def _filter_shard(self, keys, vals, diffs):
+        W, r = self.exchange.world, self.exchange.rank
+        keys = np.ascontiguousarray(keys, np.int64)
+        vals = np.ascontiguousarray(vals, np.uint8).reshape(len(keys), -1)
+        diffs = np.ascontiguousarray(diffs, np.int64)
+        if W == 1:
+            return keys, vals, diffs
+        m = self.shard_of(keys, 1, W) == r
+        return keys[m], vals[m], diffs[m]
+
+    def load(self, gen):
+        lp_k, lp_v = gen.lineitem_bypart_updates()
+        p_k, p_v = gen.part_updates()
+        lp_k, lp_v, lp_d = self._filter_shard(lp_k, lp_v,
+                                              np.ones(len(lp_k), np.int64))
+        p_k, p_v, p_d = self._filter_shard(p_k, p_v,
+                                           np.ones(len(p_k), np.int64))
+        return self._push(0, (lp_k, lp_v, lp_d), (p_k, p_v, p_d))
+
+    def step(self, churn, t):
+        lp_k, lp_v, lp_d = churn["lineitem_by_part"]
+        lp = self._filter_shard(lp_k, lp_v, lp_d)
+        return self._push(t, lp, None)
+
+    def _route_total(self, cols):
+        k, v, tm, d = cols
+        return self.exchange.exchange(
+            np.ascontiguousarray(k, np.int64),
+            np.ascontiguousarray(v, np.uint8),
+            np.ascontiguousarray(tm, np.uint64),
+            np.ascontiguousarray(d, np.int64), 1, 8)
+
+    def step_dev(self, lp_u, t):
+        """Device bench step: identical to the base until join3's
+        outputs, whose (tiny) constant-key stream leaves the device for
+        the owner-rank exchange."""
+        ctx = self.ctx
+        ctx.arr_insert(self.arr_l0, lp_u)
+        l1a = ctx.join_push_dev(self.j1, 1, lp_u)
+        j2b = ctx.join_push_dev(self.j2, 2, lp_u)
+        dcorr = ctx.reduce_push_dev(self.distinct, l1a.updates(t, t + 1))
+        dcorr_u = dcorr.updates(t, t + 1)
+        ctx.arr_insert(self.arr_dist, dcorr_u)
+        j2a = ctx.join_push_dev(self.j2, 1, dcorr_u)
+        acorr = ctx.reduce_push2_dev(self.avg, j2a.updates(t, t + 1),
+                                     j2b.updates(t, t + 1))
+        acorr_u = acorr.updates(t, t + 1)
+        l1_u = l1a.updates(t, t + 1)
+        ctx.arr_insert(self.arr_l1, l1_u)
+        j3a = ctx.join_push_dev(self.j3, 1, l1_u)
+        ctx.arr_insert(self.arr_avg, acorr_u)
+        j3b = ctx.join_push_dev(self.j3, 2, acorr_u)
+        ka, va, ta, da = j3a.to_host()
+        kb, vb_, tb, db = j3b.to_host()
+        cols = [np.concatenate([ka, kb]), np.concatenate([va, vb_]),
+                np.concatenate([ta, tb]), np.concatenate([da, db])]
+        cols = self._route_total(cols)
+        e_u = abi.make_updates(*cols, t, t + 1)
+        tcorr = ctx.reduce_push(self.total, e_u)
+        self._apply_total(tcorr)
+        for o in (l1a, j2a, j2b, dcorr, acorr, j3a, j3b):
             o.release()

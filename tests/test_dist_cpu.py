@@ -150,3 +150,52 @@ def test_sharded_q3_threads_world2():
     assert s0 is not None and s1 is not None
     assert not (set(s0) & set(s1))
     assert {**s0, **s1} == want
+
+
+def _q17_worker(rank, world, port, ret):
+    sys.path.insert(0, REPO)
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from materialize_amd.dist import TorchExchange
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import ShardedQ17Dataflow
+    from pyoracle import OracleCtx
+    df = ShardedQ17Dataflow(OracleCtx(), TorchExchange("cpu"))
+    gen = TpchGen(sf=0.02, seed=13)
+    df.load(gen)
+    for t in range(1, 4):
+        df.step(gen.churn(800), t)
+    ret[rank] = dict(df.result)
+    dist.destroy_process_group()
+
+
+def test_sharded_q17_gloo_world2():
+    """Q17 shards by partkey with only the final global-SUM exchange: the
+    const-key owner rank's maintained result equals the unsharded run."""
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import Q17Dataflow
+    from materialize_amd.dist import shard_of
+    from pyoracle import OracleCtx
+    df = Q17Dataflow(OracleCtx())
+    gen = TpchGen(sf=0.02, seed=13)
+    df.load(gen)
+    for t in range(1, 4):
+        df.step(gen.churn(800), t)
+    want = dict(df.result)
+    ctx = mp.get_context("spawn")
+    mgr = ctx.Manager()
+    ret = mgr.dict()
+    ps = [ctx.Process(target=_q17_worker, args=(r, 2, 29517, ret))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(timeout=300)
+        assert p.exitcode == 0
+    owner = int(shard_of(np.zeros(1, np.int64), 1, 2)[0])
+    assert ret[owner] == want
+    assert ret[1 - owner] == {}
+    assert len(want) > 0
