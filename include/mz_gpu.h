@@ -170,6 +170,16 @@ int  mz_gpu_arr_push_batch(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
  * MergeBatcher + arrange step fused — no intermediate out-batch). */
 int  mz_gpu_arr_insert(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
                        const mz_gpu_updates *raw);
+
+/* Overlapped maintenance: the async form enqueues the insert's
+ * consolidation+build on the arrangement's own HIP stream and returns
+ * without a device sync; the sealed batch joins the spine at
+ * mz_gpu_arr_flush (any probe of the arrangement flushes implicitly, as
+ * does mz_gpu_sync). Independent arrangements' inserts overlap this way.
+ * mz_gpu_arr_insert == insert_async + flush. */
+int  mz_gpu_arr_insert_async(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                             const mz_gpu_updates *updates);
+int  mz_gpu_arr_flush(mz_gpu_ctx *ctx, mz_gpu_arr *arr);
 /* Advance the logical compaction frontier (times advance to it on merge) —
  * cf. set_logical_compaction, mz_join_core.rs:461. */
 int  mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *ctx, mz_gpu_arr *arr,

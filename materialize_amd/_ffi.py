@@ -89,6 +89,9 @@ def load():
                                         C.POINTER(Updates),
                                         C.POINTER(Updates),
                                         C.POINTER(C.POINTER(OutBatch))]
+    lib.mz_gpu_arr_insert_async.argtypes = [C.c_void_p, C.c_void_p,
+                                            C.POINTER(Updates)]
+    lib.mz_gpu_arr_flush.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_prof_dump.argtypes = [C.c_void_p]
     lib.mz_gpu_topk_create.restype = C.c_void_p
     lib.mz_gpu_topk_create.argtypes = [C.c_void_p, C.POINTER(TopKSpec)]
@@ -222,6 +225,13 @@ class GpuCtx:
         self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
                                                 C.byref(outp)))
         return self._take(outp)
+
+    def arr_insert_async(self, arr, upd):
+        self._check(self.lib.mz_gpu_arr_insert_async(self.ctx, arr,
+                                                     C.byref(upd)))
+
+    def arr_flush(self, arr):
+        self._check(self.lib.mz_gpu_arr_flush(self.ctx, arr))
 
     def prof_dump(self):
         self.lib.mz_gpu_prof_dump(self.ctx)

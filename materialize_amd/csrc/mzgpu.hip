@@ -1186,6 +1186,20 @@ struct mz_gpu_arr {
   u64 logical_compaction = 0;
   u64 upper = 0;
   Ctx *ctx = nullptr;
+  // Per-arrangement lane: inserts/merges run on this stream with this
+  // scratch arena so independent arrangements' maintenance overlaps;
+  // ev_done orders downstream probes after the last lane enqueue,
+  // ev_gate orders lane work after the main stream's prior enqueues.
+  hipStream_t stream = nullptr;
+  struct Scratch *lane_scr = nullptr;
+  hipEvent_t ev_done = nullptr, ev_gate = nullptr;
+  // deferred insert (arr_insert_async): counts land here asynchronously
+  struct Pending {
+    int active = 0;
+    u64 cnt[3] = {0, 0, 0};
+    DevBatch batch;
+    u64 upper = 0;
+  } pending;
 };
 
 struct mz_gpu_join {
@@ -1236,9 +1250,13 @@ struct Prof {
 };
 
 struct Ctx {
-  hipStream_t stream = nullptr;
+  hipStream_t stream = nullptr;   // the CURRENT lane's stream (arrangement
+                                  // inserts/merges temporarily swap in
+                                  // their own — see LaneGuard)
+  hipStream_t main_stream = nullptr;
+  Scratch *scr = nullptr;         // the current lane's scratch arena
   std::string err;
-  Scratch scratch;
+  Scratch scratch;                // the main lane's arena
   Prof prof;
   std::vector<mz_gpu_arr *> arrs;
   std::vector<mz_gpu_join *> joins;
@@ -1276,6 +1294,39 @@ struct ProfScope {
 #define MZ_PROF_CAT(a, b) MZ_PROF_CAT2(a, b)
 #define MZ_PROF(ctx, name) ProfScope MZ_PROF_CAT(_ps, __LINE__)(ctx, name)
 
+// Swap the ctx onto an arrangement's lane (stream + scratch) for the
+// guard's scope. The lane first waits for everything already enqueued on
+// the current stream (probes of earlier batches, prior frees), so lane
+// work can never overtake readers of the state it mutates; ev_done is
+// recorded on exit for downstream probes to wait on.
+struct LaneGuard {
+  Ctx *c;
+  mz_gpu_arr *a;
+  hipStream_t ps;
+  Scratch *pscr;
+  LaneGuard(Ctx *ctx, mz_gpu_arr *arr) : c(ctx), a(arr) {
+    ps = c->stream;
+    pscr = c->scr;
+    if (!a->stream) {
+      HIP_CHECK(hipStreamCreate(&a->stream));
+      a->lane_scr = new Scratch();
+      HIP_CHECK(hipEventCreate(&a->ev_done));
+      HIP_CHECK(hipEventCreate(&a->ev_gate));
+    }
+    if (c->stream != a->stream) {
+      (void)hipEventRecord(a->ev_gate, c->stream);
+      (void)hipStreamWaitEvent(a->stream, a->ev_gate, 0);
+    }
+    c->stream = a->stream;
+    c->scr = a->lane_scr;
+  }
+  ~LaneGuard() {
+    (void)hipEventRecord(a->ev_done, a->stream);
+    c->stream = ps;
+    c->scr = pscr;
+  }
+};
+
 void *dmalloc(Ctx *c, size_t bytes) {
   void *p = nullptr;
   if (bytes == 0) bytes = 16;
@@ -1306,7 +1357,7 @@ T *dnew(Ctx *c, u64 n) {
 void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                   const u64 *times, u64 n, u32 *perm, bool time_major) {
   MZ_PROF(c, "sort_updates");
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 *skey = (u64 *)S.get(n * 8);
   u64 *skey_out = (u64 *)S.get(n * 8);
   u32 *perm_out = (u32 *)S.get(n * 4);
@@ -1436,7 +1487,7 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
 u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   // returns total; out = exclusive prefix (out has n+1 slots).
   // The n+1-sized scan would read in[n]; pad a copy to stay defined.
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u32 *pad = (u32 *)S.get((n + 1) * 4);
   HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
                            c->stream));
@@ -1456,7 +1507,7 @@ u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
 
 // enqueue-only exclusive scan (out has n+1 slots; no readback)
 void exclusive_scan_u32_ns(Ctx *c, const u32 *in, u32 *out, u64 n) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u32 *pad = (u32 *)S.get((n + 1) * 4);
   HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
                            c->stream));
@@ -1480,7 +1531,7 @@ __global__ void k_bump_ctr(u64 *ctr, const u32 *pos, const u32 *gidn,
 }
 
 void inclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   size_t need = 0;
   (void)rocprim::inclusive_scan(nullptr, need, in, out, n, rocprim::plus<u32>(),
                           c->stream);
@@ -1490,7 +1541,7 @@ void inclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
 }
 
 void inclusive_scan_u64(Ctx *c, const u64 *in, u64 *out, u64 n) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   size_t need = 0;
   (void)rocprim::inclusive_scan(nullptr, need, in, out, n, rocprim::plus<u64>(),
                           c->stream);
@@ -1518,7 +1569,7 @@ DevUpdates stage_updates(Ctx *c, const mz_gpu_updates *u, u32 kw, u32 vb) {
     d.diffs = u->diffs;
     return d;
   }
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 *k = (u64 *)S.get(u->n * kw * 8);
   u8 *v = vb ? (u8 *)S.get(u->n * vb) : nullptr;
   u64 *t = (u64 *)S.get(u->n * 8);
@@ -1578,7 +1629,7 @@ void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
 void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
                       u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts) {
   MZ_PROF(c, "consolidate_core");
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 n = in.n;
   if (n == 0) {
     HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
@@ -1594,7 +1645,7 @@ void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
 void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
                            const u32 *perm, u64 *okeys, u8 *ovals,
                            u64 *otimes, i64 *odiffs, u64 *dcounts) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 n = in.n;
   if (n == 0) {
     HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
@@ -1630,7 +1681,7 @@ void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
 // probe outputs): returns owned capacity-sized arrays + the exact count.
 void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
                      u8 **ovals, u64 **otimes, i64 **odiffs, u64 *out_n) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 n = in.n;
   u64 capn = std::max<u64>(n, 1);
   *okeys = dnew<u64>(c, capn * kw);
@@ -1663,7 +1714,7 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
                           u64 *times, i64 *diffs, u64 cap, u64 lower,
                           u64 upper, u64 *dcounts) {
   MZ_PROF(c, "build_batch");
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   DevBatch b;
   b.lower = lower;
   b.upper = upper;
@@ -1719,7 +1770,7 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
 // Synchronous wrapper for sealed inputs with host-known count.
 DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
                      i64 *diffs, u64 n, u64 lower, u64 upper) {
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   u64 *dcounts = (u64 *)S.get(3 * 8);
   hipLaunchKernelGGL(k_write_u64, dim3(1), dim3(1), 0, c->stream, dcounts,
                      n);
@@ -1740,7 +1791,7 @@ DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
 void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   if (to - from <= 1) return;
   MZ_PROF(c, "merge_range");
-  auto &S = c->scratch;
+  auto &S = (*c->scr);
   S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
   u64 total = 0, lo = UINT64_MAX, hi = 0;
@@ -1879,6 +1930,8 @@ struct mz_gpu_ctx {
   Ctx impl;
 };
 
+static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a);
+
 extern "C" {
 
 mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
@@ -1888,6 +1941,8 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   HIP_CHECK(hipSetDevice(dev));
   mz_gpu_ctx *c = new mz_gpu_ctx();
   HIP_CHECK(hipStreamCreate(&c->impl.stream));
+  c->impl.main_stream = c->impl.stream;
+  c->impl.scr = &c->impl.scratch;
   HIP_CHECK(hipEventCreate(&c->impl.ev_a));
   HIP_CHECK(hipEventCreate(&c->impl.ev_b));
   // Keep freed stream-ordered allocations in the pool forever (288 GB of
@@ -1902,8 +1957,8 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   // pre-grow the scratch arena so per-call growth never stalls the step
   u64 scratch0 = cfg && cfg->hbm_pool_bytes ? cfg->hbm_pool_bytes
                                             : (4ull << 30);
-  c->impl.scratch.get(scratch0);
-  c->impl.scratch.reset();
+  (*c->impl.scr).get(scratch0);
+  (*c->impl.scr).reset();
   const char *prof = getenv("MZ_GPU_PROF");
   c->impl.prof.enabled = prof && prof[0] && prof[0] != '0';
   return c;
@@ -1940,6 +1995,10 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
 const char *mz_gpu_last_error(mz_gpu_ctx *c) { return c->impl.err.c_str(); }
 
 int mz_gpu_sync(mz_gpu_ctx *c) {
+  for (mz_gpu_arr *a : c->impl.arrs) {
+    arr_flush_impl(&c->impl, a);
+    if (a->stream) HIP_CHECK(hipStreamSynchronize(a->stream));
+  }
   HIP_CHECK(hipStreamSynchronize(c->impl.stream));
   return 0;
 }
@@ -1952,6 +2011,7 @@ mz_gpu_arr *mz_gpu_arr_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
 }
 
 void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
+  arr_flush_impl(&c->impl, a);
   for (auto &b : a->batches) free_batch(&c->impl, b);
   a->batches.clear();
 }
@@ -2000,7 +2060,9 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
 int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                           const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
+  arr_flush_impl(ctx, a);
+  LaneGuard lane(ctx, a);
+  (*ctx->scr).reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   // copy into owned arrays (batch owns its storage)
@@ -2035,7 +2097,7 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
 static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
                                 u64 lower, u64 upper) {
   MZ_PROF(ctx, "arr_insert");
-  auto &S = ctx->scratch;
+  auto &S = (*ctx->scr);
   u32 kw = a->schema.kw, vb = a->schema.vb;
   u64 capn = std::max<u64>(d.n, 1);
   u64 *ok = dnew<u64>(ctx, capn * kw);
@@ -2059,12 +2121,17 @@ static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
   return &a->batches.back();
 }
 
-int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
-                      const mz_gpu_updates *u) {
-  Ctx *ctx = &c->impl;
+// Enqueue-only half of an insert: consolidate + build run on the
+// arrangement's own lane; the counts readback is issued asynchronously
+// and the sealed batch joins the spine at mz_gpu_arr_flush. Independent
+// arrangements' inserts overlap this way (one per GPU stream).
+static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
+                                  const mz_gpu_updates *u) {
   MZ_PROF(ctx, "arr_insert");
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  if (a->pending.active) arr_flush_impl(ctx, a);
+  LaneGuard lane(ctx, a);
+  auto &S = (*ctx->scr);
+  S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 capn = std::max<u64>(d.n, 1);
@@ -2076,16 +2143,43 @@ int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
   consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
   DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, u->lower,
                                 u->upper, dcounts);
-  u64 cnt[3] = {0, 0, 0};
-  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
+  a->pending.active = 1;
+  a->pending.batch = b;
+  a->pending.upper = u->upper;
+  HIP_CHECK(hipMemcpyAsync(a->pending.cnt, dcounts, 3 * 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+}
+
+static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
+  if (!a->pending.active) return;
+  MZ_PROF(ctx, "arr_flush");
+  LaneGuard lane(ctx, a);
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
-  b.n_upds = cnt[0];
-  b.n_keys = cnt[1];
-  b.n_vals = cnt[2];
+  DevBatch b = a->pending.batch;
+  b.n_upds = a->pending.cnt[0];
+  b.n_keys = a->pending.cnt[1];
+  b.n_vals = a->pending.cnt[2];
   a->batches.push_back(b);
-  a->upper = std::max(a->upper, u->upper);
+  a->upper = std::max(a->upper, a->pending.upper);
+  a->pending.active = 0;
   spine_policy(ctx, a);
+}
+
+int mz_gpu_arr_insert_async(mz_gpu_ctx *c, mz_gpu_arr *a,
+                            const mz_gpu_updates *u) {
+  arr_insert_async_impl(&c->impl, a, u);
+  return 0;
+}
+
+int mz_gpu_arr_flush(mz_gpu_ctx *c, mz_gpu_arr *a) {
+  arr_flush_impl(&c->impl, a);
+  return 0;
+}
+
+int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
+                      const mz_gpu_updates *u) {
+  arr_insert_async_impl(&c->impl, a, u);
+  arr_flush_impl(&c->impl, a);
   return 0;
 }
 
@@ -2103,6 +2197,7 @@ int mz_gpu_arr_maintain(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t fuel) {
 
 int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
                      uint64_t *n_updates, uint64_t *hbm_bytes) {
+  arr_flush_impl(&c->impl, a);
   *n_batches = a->batches.size();
   u64 n = 0, by = 0;
   u32 kw = a->schema.kw, vb = a->schema.vb;
@@ -2147,7 +2242,7 @@ int mz_gpu_out_to_host(mz_gpu_ctx *c, const mz_gpu_out *o, uint64_t *keys,
 int mz_gpu_consolidate(mz_gpu_ctx *c, const mz_gpu_schema *s,
                        const mz_gpu_updates *u, mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
+  (*ctx->scr).reset();
   u32 kw = s->key_words, vb = s->val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 *ok;
@@ -2180,8 +2275,14 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                       u32 stream_vb, int mode, int swap,
                       const mz_gpu_closure *cl, int consolidate_out,
                       mz_gpu_out **out) {
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  // a probe after insert_async must see the batch (probes of the OLD
+  // state precede the insert call entirely), and must run after the
+  // lookup lane's enqueued maintenance
+  arr_flush_impl(ctx, lookup);
+  if (lookup->stream)
+    (void)hipStreamWaitEvent(ctx->stream, lookup->ev_done, 0);
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u32 kw = lookup->schema.kw, lvb = lookup->schema.vb;
   u32 okw = cl->out.key_words, ovb = cl->out.val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, stream_vb);
@@ -2350,7 +2451,7 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
 int mz_gpu_reduce_push(mz_gpu_ctx *c, mz_gpu_red *op,
                        const mz_gpu_updates *u, mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
+  (*ctx->scr).reset();
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   return reduce_push_dev_impl(ctx, op, d, u->lower, u->upper, out);
@@ -2364,8 +2465,8 @@ int mz_gpu_reduce_push2(mz_gpu_ctx *c, mz_gpu_red *op,
                         const mz_gpu_updates *u1, const mz_gpu_updates *u2,
                         mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   DevUpdates d1 = stage_updates(ctx, u1, kw, vb);
   DevUpdates d2 = stage_updates(ctx, u2, kw, vb);
@@ -2396,7 +2497,7 @@ int mz_gpu_reduce_push2(mz_gpu_ctx *c, mz_gpu_red *op,
 
 static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
                                 u64 lower, u64 upper, mz_gpu_out **out) {
-  auto &S = ctx->scratch;
+  auto &S = (*ctx->scr);
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   u32 okw = op->spec.out.key_words, ovb = op->spec.out.val_bytes;
   u64 n = d.n;
@@ -2511,7 +2612,7 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
 
 mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
+  (*ctx->scr).reset();
   mz_gpu_thr *r = new mz_gpu_thr();
   r->s = *s;
   r->kw2 = s->key_words + (s->val_bytes + 7) / 8;
@@ -2540,8 +2641,8 @@ void mz_gpu_threshold_drop(mz_gpu_ctx *c, mz_gpu_thr *r) {
 int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
                           const mz_gpu_updates *u, mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u32 kw = op->s.key_words, vb = op->s.val_bytes, kw2 = op->kw2;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 n = d.n;
@@ -2668,7 +2769,7 @@ static void topk_eval_emit(Ctx *ctx, mz_gpu_topk *op, mz_gpu_out *pe,
   u64 n = pe->n;
   if (!n) return;
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
-  auto &S = ctx->scratch;
+  auto &S = (*ctx->scr);
   const u64 *keys = (const u64 *)pe->keys;
   const u8 *vals = (const u8 *)pe->vals;
   const i64 *diffs = (const i64 *)pe->diffs;
@@ -2748,7 +2849,7 @@ void mz_gpu_topk_drop(mz_gpu_ctx *c, mz_gpu_topk *op) {
 int mz_gpu_topk_push(mz_gpu_ctx *c, mz_gpu_topk *op,
                      const mz_gpu_updates *u, mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
+  (*ctx->scr).reset();
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 n = d.n;
@@ -2796,7 +2897,7 @@ int mz_gpu_topk_push(mz_gpu_ctx *c, mz_gpu_topk *op,
   for (auto [lo, hi] : slices) {
     u64 m = hi - lo;
     u64 t = htimes[htimes.size() == 1 ? 0 : lo];
-    auto &S = ctx->scratch;
+    auto &S = (*ctx->scr);
     u32 *flags = (u32 *)S.get(m * 4);
     u32 *gid = (u32 *)S.get(m * 4);
     hipLaunchKernelGGL(k_key_flags_sorted, dim3(ngrid(m)), dim3(BLK), 0,
@@ -2927,8 +3028,8 @@ int mz_gpu_partition(mz_gpu_ctx *c, const mz_gpu_schema *s,
                      uint64_t *out_times, int64_t *out_diffs,
                      uint64_t *counts) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u32 kw = s->key_words, vb = s->val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 n = d.n;
@@ -3023,8 +3124,8 @@ int mz_gpu_map(mz_gpu_ctx *c, const mz_gpu_schema *in,
                const mz_gpu_updates *u, const mz_gpu_closure *cl,
                mz_gpu_out **out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u32 kw = in->key_words, vb = in->val_bytes;
   u32 okw = cl->out.key_words, ovb = cl->out.val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
@@ -3097,8 +3198,8 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
     ctx->err = "minmax_push: single-timestamp batches only";
     return -1;
   }
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   u64 t = u->lower;
   u32 kw = op->kw, kw2 = kw + 1;
   u32 L = (u32)op->buckets.size();
@@ -3217,8 +3318,8 @@ void mz_gpu_debug_float_paths(mz_gpu_ctx *c, const double *xs, uint64_t n,
                               uint64_t *fp_out, const uint64_t *fp_in,
                               double *dec_out) {
   Ctx *ctx = &c->impl;
-  ctx->scratch.reset();
-  auto &S = ctx->scratch;
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
   double *dx = (double *)S.get(n * 8);
   u128 *dfp = (u128 *)S.get(n * 16);
   double *dd = (double *)S.get(n * 8);

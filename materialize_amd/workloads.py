@@ -217,8 +217,11 @@ class Q3Dataflow:
         Updates descriptors with times == t. Returns the corrections
         DevOut (or None)."""
         ctx = self.ctx
+        # async: each arrangement's consolidate+build runs on its own HIP
+        # stream; the first probe of each flushes it (overlap of the three
+        # independent maintenance pipelines)
         for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
-            ctx.arr_insert(self.arrs[name], upd[name])
+            ctx.arr_insert_async(self.arrs[name], upd[name])
         outs = []
         for rel, src in (("orders", "orders_by_custkey"),
                          ("lineitem", "lineitem")):

@@ -112,6 +112,13 @@ class OracleCtx:
         rc = self.lib.orc_arr_push_batch(self.ctx, arr, C.byref(upd))
         assert rc == 0
 
+    def arr_insert_async(self, arr, upd):
+        # the oracle has no device lanes: async == sync
+        return self.arr_insert(arr, upd)
+
+    def arr_flush(self, arr):
+        return 0
+
     def arr_insert(self, arr, upd):
         rc = self.lib.orc_arr_insert(self.ctx, arr, C.byref(upd))
         assert rc == 0
