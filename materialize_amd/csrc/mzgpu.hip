@@ -1200,6 +1200,15 @@ struct mz_gpu_arr {
     DevBatch batch;
     u64 upper = 0;
   } pending;
+  // deferred spine merge: the merged batch is computed on the lane while
+  // probes keep using the pre-merge batch list (identical logical
+  // content); it replaces its inputs at the next flush.
+  struct PendingMerge {
+    int active = 0;
+    size_t from = 0, to = 0;
+    u64 cnt[3] = {0, 0, 0};
+    DevBatch merged;
+  } pending_merge;
 };
 
 struct mz_gpu_join {
@@ -1788,7 +1797,12 @@ DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
 
 // Merge an arrangement's batches [from, to) into one (logical compaction
 // applied). Policy is the host's; semantics = concat + advance + consolidate.
-void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
+// deferred=1: enqueue the merge on the current lane and record it in
+// a->pending_merge — the inputs stay in the batch list (probes keep
+// using them; the merged batch holds the same logical updates) until
+// merge_install replaces them at the next flush.
+void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to,
+                 int deferred = 0) {
   if (to - from <= 1) return;
   MZ_PROF(c, "merge_range");
   auto &S = (*c->scr);
@@ -1910,6 +1924,15 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   DevBatch merged =
       build_batch_core(c, kw, vb, ok, ov, ot, od, total,
                        lo == UINT64_MAX ? 0 : lo, hi, dcounts);
+  if (deferred) {
+    a->pending_merge.active = 1;
+    a->pending_merge.from = from;
+    a->pending_merge.to = to;
+    a->pending_merge.merged = merged;
+    HIP_CHECK(hipMemcpyAsync(a->pending_merge.cnt, dcounts, 3 * 8,
+                             hipMemcpyDeviceToHost, c->stream));
+    return;
+  }
   u64 cnt[3] = {0, 0, 0};
   HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
                            c->stream));
@@ -1920,6 +1943,32 @@ void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
   for (size_t i = from; i < to; i++) free_batch(c, a->batches[i]);
   a->batches.erase(a->batches.begin() + from, a->batches.begin() + to);
   a->batches.insert(a->batches.begin() + from, merged);
+}
+
+// Install a deferred merge (the lane is synced by the caller): replace
+// the input range with the merged batch; the inputs' frees are gated on
+// the main stream's enqueues so far (probes that read them).
+void merge_install(Ctx *c, mz_gpu_arr *a) {
+  auto &pm = a->pending_merge;
+  if (!pm.active) return;
+  DevBatch merged = pm.merged;
+  merged.n_upds = pm.cnt[0];
+  merged.n_keys = pm.cnt[1];
+  merged.n_vals = pm.cnt[2];
+  // gate: lane frees must run after main-stream probes of the inputs
+  (void)hipEventRecord(a->ev_gate, c->main_stream);
+  (void)hipStreamWaitEvent(a->stream ? a->stream : c->stream, a->ev_gate,
+                           0);
+  {
+    hipStream_t ps = c->stream;
+    if (a->stream) c->stream = a->stream;
+    for (size_t i = pm.from; i < pm.to; i++) free_batch(c, a->batches[i]);
+    c->stream = ps;
+  }
+  a->batches.erase(a->batches.begin() + pm.from,
+                   a->batches.begin() + pm.to);
+  a->batches.insert(a->batches.begin() + pm.from, merged);
+  pm.active = 0;
 }
 
 }  // namespace
@@ -2021,6 +2070,35 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
 // keep batch sizes decreasing by >=2x tail-to-head; merging the tail
 // whenever the invariant breaks costs O(log) amortized merge work per
 // update and keeps the probe fan-out at ~log(arrangement/batch).
+// Deferred variant: enqueue at most ONE merge (the innermost due) per
+// call; cascades progress one merge per flush, off the probe critical
+// path. The 10-batch hard cap stays synchronous (probe BatchList bound).
+static void spine_policy_deferred(Ctx *ctx, mz_gpu_arr *a) {
+  static const u64 SMALL = [] {
+    const char *e = getenv("MZ_GPU_SMALL");
+    return e ? (u64)atoll(e) : (u64)(4u << 20);
+  }();
+  static const long POOL = [] {
+    const char *e = getenv("MZ_GPU_SMALL_POOL");
+    return e ? atol(e) : 6;
+  }();
+  if (a->pending_merge.active) return;  // one in flight per arrangement
+  size_t nb = a->batches.size();
+  if (nb >= 2 && a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds
+      && a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL) {
+    merge_range(ctx, a, nb - 2, nb, 1);
+    return;
+  }
+  size_t i = nb;
+  while (i > 0 && a->batches[i - 1].n_upds < SMALL) i--;
+  if ((long)(nb - i) > POOL) {
+    merge_range(ctx, a, i, nb, 1);
+    return;
+  }
+  while (a->batches.size() > 10)  // hard cap: synchronous catch-up
+    merge_range(ctx, a, 0, a->batches.size());
+}
+
 static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
   // Large batches keep the geometric pair rule (merge-path pair merges
   // are O(n)); small batches pool lazily and merge k-way when the pool
@@ -2086,7 +2164,7 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                   u->upper);
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
-  spine_policy(ctx, a);
+  spine_policy_deferred(ctx, a);
   return 0;
 }
 
@@ -2151,18 +2229,21 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
 }
 
 static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
-  if (!a->pending.active) return;
+  if (!a->pending.active && !a->pending_merge.active) return;
   MZ_PROF(ctx, "arr_flush");
   LaneGuard lane(ctx, a);
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
-  DevBatch b = a->pending.batch;
-  b.n_upds = a->pending.cnt[0];
-  b.n_keys = a->pending.cnt[1];
-  b.n_vals = a->pending.cnt[2];
-  a->batches.push_back(b);
-  a->upper = std::max(a->upper, a->pending.upper);
-  a->pending.active = 0;
-  spine_policy(ctx, a);
+  merge_install(ctx, a);
+  if (a->pending.active) {
+    DevBatch b = a->pending.batch;
+    b.n_upds = a->pending.cnt[0];
+    b.n_keys = a->pending.cnt[1];
+    b.n_vals = a->pending.cnt[2];
+    a->batches.push_back(b);
+    a->upper = std::max(a->upper, a->pending.upper);
+    a->pending.active = 0;
+  }
+  spine_policy_deferred(ctx, a);
 }
 
 int mz_gpu_arr_insert_async(mz_gpu_ctx *c, mz_gpu_arr *a,
@@ -2291,7 +2372,13 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   for (auto &b : lookup->batches) {
     if (b.n_upds == 0) continue;
     if (bl.n >= 12) {
-      // too many batches: merge first (bounded by push-time policy)
+      // too many batches: catch up synchronously (install any deferred
+      // merge first — its inputs are in the list being merged)
+      if (lookup->pending_merge.active) {
+        LaneGuard l2(ctx, lookup);
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        merge_install(ctx, lookup);
+      }
       merge_range(ctx, lookup, 0, lookup->batches.size());
       return probe_impl(ctx, lookup, u, stream_vb, mode, swap, cl,
                         consolidate_out, out);
