@@ -255,3 +255,29 @@ class TestTopKGoldens:
                      for i in range(n))
         want = sorted((enc(st), enc(nm), 1) for st, nm in fx["expect"])
         assert got == want
+
+
+class TestMinMaxGoldens:
+    @pytest.mark.parametrize("is_max", [0, 1])
+    def test_group_min_max(self, is_max):
+        """aggregates.slt:123-131 — min(b)/max(b) per group through the
+        hierarchical bucket tree."""
+        fx = BY_NAME["agg_min_max_group"]
+        ctx = OracleCtx()
+        op = ctx.minmax_create(abi.schema(1, 8), bool(is_max),
+                               [16, 4, 1])
+        rows = fx["input"]
+        keys = np.array([r[0] for r in rows], np.int64)
+        vals = np.array([r[1] for r in rows], np.int64)
+        u = abi.make_updates(keys, vals.view(np.uint8),
+                             np.zeros(len(rows), np.uint64),
+                             np.ones(len(rows), np.int64), 0, 1)
+        k, v, t, dd = ctx.minmax_push(op, u)
+        n = len(t)
+        got = sorted((int(k[i]),
+                      int(v[i * 8:(i + 1) * 8].copy().view(np.int64)[0]))
+                     for i in range(n) if dd[i] == 1)
+        want = sorted((a, b) for a, b in
+                      fx["expect_max" if is_max else "expect_min"])
+        assert got == want
+        ctx.close()
