@@ -1193,6 +1193,10 @@ struct mz_gpu_arr {
   hipStream_t stream = nullptr;
   struct Scratch *lane_scr = nullptr;
   hipEvent_t ev_done = nullptr, ev_gate = nullptr;
+  // probes wait ev_ready, recorded when the probe-visible batch list is
+  // final for the step (after install+push, BEFORE a deferred merge is
+  // enqueued — the merge must not block the probes it runs under)
+  hipEvent_t ev_ready = nullptr;
   // deferred insert (arr_insert_async): counts land here asynchronously
   struct Pending {
     int active = 0;
@@ -1321,6 +1325,8 @@ struct LaneGuard {
       a->lane_scr = new Scratch();
       HIP_CHECK(hipEventCreate(&a->ev_done));
       HIP_CHECK(hipEventCreate(&a->ev_gate));
+      HIP_CHECK(hipEventCreate(&a->ev_ready));
+      (void)hipEventRecord(a->ev_ready, a->stream);
     }
     if (c->stream != a->stream) {
       (void)hipEventRecord(a->ev_gate, c->stream);
@@ -2164,6 +2170,7 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                   u->upper);
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
+  (void)hipEventRecord(a->ev_ready, ctx->stream);
   spine_policy_deferred(ctx, a);
   return 0;
 }
@@ -2243,6 +2250,9 @@ static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
     a->upper = std::max(a->upper, a->pending.upper);
     a->pending.active = 0;
   }
+  // the probe-visible state is final here; the deferred merge enqueued
+  // below runs concurrently with this step's probes
+  (void)hipEventRecord(a->ev_ready, ctx->stream);
   spine_policy_deferred(ctx, a);
 }
 
@@ -2361,7 +2371,7 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   // lookup lane's enqueued maintenance
   arr_flush_impl(ctx, lookup);
   if (lookup->stream)
-    (void)hipStreamWaitEvent(ctx->stream, lookup->ev_done, 0);
+    (void)hipStreamWaitEvent(ctx->stream, lookup->ev_ready, 0);
   (*ctx->scr).reset();
   auto &S = (*ctx->scr);
   u32 kw = lookup->schema.kw, lvb = lookup->schema.vb;
