@@ -192,6 +192,33 @@ def run_cpu_baseline(seed, cores=0, batch=100_000, steps=4, sf=1.0):
     }
 
 
+def run_cpu_baseline_q17(seed, batch, steps=3, sf=1.0):
+    """Oracle Q17 dataflow on one host core, bounded sample."""
+    import copy
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    from materialize_amd.tpch import TpchGen
+    from materialize_amd.workloads import Q17Dataflow
+    from pyoracle import OracleCtx
+    gen = TpchGen(sf=sf, seed=seed)
+    base = copy.deepcopy(gen.__dict__)
+    churns = [gen.churn(batch) for _ in range(steps + 1)]
+    gen.__dict__.update(base)
+    df = Q17Dataflow(OracleCtx())
+    df.load(gen)
+    df.step(churns[0], 1)  # warmup
+    t0 = time.perf_counter()
+    total = 0
+    for i in range(1, steps + 1):
+        total += df.step(churns[i], i + 1)
+    dt = time.perf_counter() - t0
+    return {
+        "value": total / dt, "unit": "rows/s", "cores": 1, "kind": "port",
+        "sample": (f"oracle C++ restatement, TPC-H SF{sf:g} Q17, {steps} "
+                   f"churn batches of ~{batch} rows, 1 thread "
+                   f"({dt:.1f}s timed)"),
+    }
+
+
 def main_q17(args):
     """BASELINE config 5 shape at N=1: TPC-H Q17 (two linear joins,
     distinct, per-partkey AVG arrangement, correlated filter, global SUM)
@@ -288,7 +315,9 @@ def main_q17(args):
         "roofline": {"bound": "hbm", "achieved": achieved / 1e9,
                      "peak": HBM_PEAK / 1e9, "unit": "GB/s",
                      "frac": achieved / HBM_PEAK, "traffic": None},
-        "cpu_baseline": None,
+        "cpu_baseline": (run_cpu_baseline_q17(args.seed, batch_rows, sf=sf)
+                         if rank == 0 and N == 1
+                         and not args.no_cpu_baseline else None),
         "probe_kernel": {"ms_total": probe_ms,
                          "launch_pairs": probe_launches // 2,
                          "alg_bytes": alg_bytes.value},
