@@ -2170,8 +2170,8 @@ int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                   u->upper);
   a->batches.push_back(b);
   a->upper = std::max(a->upper, u->upper);
+  spine_policy(ctx, a);
   (void)hipEventRecord(a->ev_ready, ctx->stream);
-  spine_policy_deferred(ctx, a);
   return 0;
 }
 
@@ -2250,10 +2250,22 @@ static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
     a->upper = std::max(a->upper, a->pending.upper);
     a->pending.active = 0;
   }
-  // the probe-visible state is final here; the deferred merge enqueued
-  // below runs concurrently with this step's probes
-  (void)hipEventRecord(a->ev_ready, ctx->stream);
-  spine_policy_deferred(ctx, a);
+  // Measured on the 1M churn config: deferring merges off the probe path
+  // LOSES ~10% — probes pay for the deeper pre-merge spine and the
+  // merges are bandwidth-bound anyway (no free overlap). Kept as an
+  // option for latency-sensitive shapes. ev_ready (what probes wait on)
+  // is recorded BEFORE a deferred merge but AFTER synchronous ones.
+  static const bool DEFER = [] {
+    const char *e = getenv("MZ_GPU_DEFER_MERGE");
+    return e && e[0] && e[0] != '0';
+  }();
+  if (DEFER) {
+    (void)hipEventRecord(a->ev_ready, ctx->stream);
+    spine_policy_deferred(ctx, a);
+  } else {
+    spine_policy(ctx, a);
+    (void)hipEventRecord(a->ev_ready, ctx->stream);
+  }
 }
 
 int mz_gpu_arr_insert_async(mz_gpu_ctx *c, mz_gpu_arr *a,
