@@ -2043,7 +2043,18 @@ void mz_gpu_prof_dump(mz_gpu_ctx *c) {
 
 void mz_gpu_fini(mz_gpu_ctx *c) {
   if (!c) return;
-  hipStreamSynchronize(c->impl.stream);
+  for (mz_gpu_arr *a : c->impl.arrs) {
+    if (a->stream) {
+      (void)hipStreamSynchronize(a->stream);
+      (void)hipStreamDestroy(a->stream);
+      (void)hipEventDestroy(a->ev_done);
+      (void)hipEventDestroy(a->ev_gate);
+      (void)hipEventDestroy(a->ev_ready);
+      delete a->lane_scr;
+      a->stream = nullptr;
+    }
+  }
+  (void)hipStreamSynchronize(c->impl.stream);
   delete c;
 }
 
