@@ -349,6 +349,22 @@ class TestThresholdParity:
         assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
                     "threshold empty")
 
+    def test_no_vals(self, ctxs):
+        """vb=0 — the EXCEPT ALL lowering's whole-row-as-key shape."""
+        g, o = ctxs
+        rng = np.random.default_rng(97)
+        sch = abi.schema(1, 0)
+        gop, oop = g.threshold_create(sch), o.threshold_create(sch)
+        for step in range(3):
+            n = 400
+            keys = rng.integers(-8, 20, n).astype(np.int64)
+            diffs = rng.integers(-2, 3, n).astype(np.int64)
+            u = abi.make_updates(keys, None,
+                                 np.full(n, step, np.uint64), diffs, step,
+                                 step + 1)
+            assert_same(g.threshold_push(gop, u), o.threshold_push(oop, u),
+                        f"threshold vb0 step {step}")
+
 
 class TestTopKParity:
     """render_topk Basic plan (top_k.rs:322-418): corrections parity under
@@ -423,6 +439,22 @@ class TestTopKParity:
         u = abi.make_updates(ks, vs, times, ds, 0, 3)
         assert_same(g.topk_push(gop, u), o.topk_push(oop, u),
                     "topk multi-timestamp")
+
+    def test_two_word_group_keys(self, ctxs):
+        g, o = ctxs
+        rng = np.random.default_rng(79)
+        spec = abi.topk_spec(abi.schema(2, 8), [(0, 8, 1)], offset=0,
+                             limit=2)
+        gop, oop = g.topk_create(spec), o.topk_create(spec)
+        for step in range(3):
+            n = 300
+            ks = rng.integers(-3, 5, (n, 2)).astype(np.int64)
+            vs = rng.integers(-10, 10, n).astype(np.int64)
+            u = abi.make_updates(ks, vs.view(np.uint8),
+                                 np.full(n, step, np.uint64),
+                                 np.ones(n, np.int64), step, step + 1)
+            assert_same(g.topk_push(gop, u), o.topk_push(oop, u),
+                        f"topk kw2 step {step}")
 
     def test_negative_multiplicity_errors(self, ctxs):
         g, o = ctxs
