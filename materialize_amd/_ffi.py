@@ -336,6 +336,22 @@ class GpuCtx:
             C.byref(cl), C.byref(outp)))
         return self._dev_out(outp)
 
+    def partition_dev(self, sch, upd, nshards, out_tensors):
+        """mz_gpu_partition into caller-provided torch device tensors
+        (keys, vals-or-None, times, diffs); returns per-shard counts.
+        The call synchronizes: the tensors are fully written on return."""
+        counts = (C.c_uint64 * nshards)()
+        kt, vt, tt, dt = out_tensors
+        self._check(self.lib.mz_gpu_partition(
+            self.ctx, C.byref(sch), C.byref(upd), nshards,
+            C.cast(kt.data_ptr(), C.POINTER(C.c_uint64)),
+            (C.cast(vt.data_ptr(), C.POINTER(C.c_uint8))
+             if vt is not None else None),
+            C.cast(tt.data_ptr(), C.POINTER(C.c_uint64)),
+            C.cast(dt.data_ptr(), C.POINTER(C.c_int64)),
+            counts))
+        return list(counts)
+
     def join_push_dev(self, op, side, upd):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_join_push(self.ctx, op, side,

@@ -76,6 +76,51 @@ class TorchExchange:
         self.rank = dist.get_rank()
         self.device = device
 
+    def exchange_dev(self, gctx, upd, kw, vb, t):
+        """Device-resident all-to-all-v: shard-partition on the GPU
+        (mz_gpu_partition — splitmix64 routing identical to the host
+        path), then column-wise torch.distributed all_to_all_single.
+        On nccl (RCCL over xGMI) the columns never leave HBM; on gloo
+        they bounce through pinned host copies (the CPU-validation
+        transport). Returns a device Updates at [t, t+1)."""
+        import torch
+
+        from . import _abi as abi
+        dist = self.dist
+        W = self.world
+        n = upd.n
+        work_dev = torch.device("cuda", torch.cuda.current_device())
+        kt = torch.empty(max(n, 1) * kw, dtype=torch.int64,
+                         device=work_dev)
+        vt = (torch.empty(max(n * vb, 1), dtype=torch.uint8,
+                          device=work_dev) if vb else None)
+        tt = torch.empty(max(n, 1), dtype=torch.int64, device=work_dev)
+        dt = torch.empty(max(n, 1), dtype=torch.int64, device=work_dev)
+        counts = gctx.partition_dev(abi.schema(kw, vb), upd, W,
+                                    (kt, vt, tt, dt))
+        comm_cpu = self.device == "cpu"
+        cnt_dev = "cpu" if comm_cpu else work_dev
+        cnt_t = torch.tensor(counts, dtype=torch.int64, device=cnt_dev)
+        recv_cnt = torch.zeros(W, dtype=torch.int64, device=cnt_dev)
+        dist.all_to_all_single(recv_cnt, cnt_t)
+        rc = [int(x) for x in recv_cnt.cpu()]
+        outs = []
+        for ten, mult in ((kt, kw), (vt, vb), (tt, 1), (dt, 1)):
+            if mult == 0 or ten is None:
+                outs.append(None)
+                continue
+            send = ten[:n * mult]
+            if comm_cpu:
+                send = send.cpu()
+            recv = torch.empty(sum(rc) * mult, dtype=ten.dtype,
+                               device=send.device)
+            dist.all_to_all_single(recv, send,
+                                   [c * mult for c in rc],
+                                   [int(c) * mult for c in counts])
+            outs.append(recv.to(work_dev) if comm_cpu else recv)
+        ko, vo, to_, do_ = outs
+        return abi.make_updates_from_torch(ko, vo, to_, do_, t, t + 1)
+
     def exchange(self, keys, vals, times, diffs, kw, vb):
         import torch
         dist = self.dist

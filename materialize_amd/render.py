@@ -115,7 +115,16 @@ class DeltaJoinOp:
         """Re-distribute a stage output by its (new) key hash. COLLECTIVE:
         every rank calls this for every stage, with however few local rows
         it has — a rank-dependent skip would desynchronize the
-        all-to-all."""
+        all-to-all. Device outputs take the device-resident exchange
+        (partition in HBM + column all_to_all; no host round trip on
+        RCCL); host outputs and the oracle keep the numpy path."""
+        if (isinstance(cur, DevOut)
+                and hasattr(self.exchange, "exchange_dev")
+                and hasattr(self.ctx, "partition_dev")):
+            u = self.exchange.exchange_dev(self.ctx, cur.updates(t, t + 1),
+                                           kw, vb, t)
+            cur.release()
+            return u
         keys, vals, times, diffs = cur.to_host()
         cur.release()
         keys, vals, times, diffs = self.exchange.exchange(
