@@ -2012,8 +2012,22 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   // pre-grow the scratch arena so per-call growth never stalls the step
   u64 scratch0 = cfg && cfg->hbm_pool_bytes ? cfg->hbm_pool_bytes
                                             : (4ull << 30);
-  (*c->impl.scr).get(scratch0);
+  void *warm = (*c->impl.scr).get(scratch0);
+  // touch the arena once: un-backed pages otherwise fault in lazily
+  // across the first process's timed steps (measured ~2x first-process
+  // slowdown on some pool boxes)
+  (void)hipMemsetAsync(warm, 0, scratch0, c->impl.stream);
   (*c->impl.scr).reset();
+  // pre-back the async mempool the same way (spine batches and merge
+  // outputs allocate from it at 32MB size classes)
+  for (int i = 0; i < 4; i++) {
+    void *p = nullptr;
+    if (hipMallocAsync(&p, 2ull << 30, c->impl.stream) == hipSuccess && p) {
+      (void)hipMemsetAsync(p, 0, 2ull << 30, c->impl.stream);
+      (void)hipFreeAsync(p, c->impl.stream);
+    }
+  }
+  (void)hipStreamSynchronize(c->impl.stream);
   const char *prof = getenv("MZ_GPU_PROF");
   c->impl.prof.enabled = prof && prof[0] && prof[0] != '0';
   return c;
