@@ -398,16 +398,24 @@ def main():
         print("# MZPROF setup+warmup (discard):", file=sys.stderr)
         ctx.prof_dump()  # reset engine sub-phase counters before timing
     ctx.set_kernel_timing(1)
+    step_times = [] if os.environ.get("MZ_BENCH_STEP_TIMES") else None
     t0 = time.perf_counter()
     for i in range(W, W + K):
         corr = df.step_dev(staged[i], i + 1)
         if corr is not None:
             corr.release()
+        if step_times is not None:
+            ctx.lib.mz_gpu_sync(ctx.ctx)
+            step_times.append(time.perf_counter() - t0 -
+                              sum(step_times))
     ctx.lib.mz_gpu_sync(ctx.ctx)
     torch.cuda.synchronize()
     if dist:
         dist.barrier()
     elapsed = time.perf_counter() - t0
+    if step_times is not None:
+        print("# step_ms " + " ".join(f"{s*1e3:.1f}" for s in step_times),
+              file=sys.stderr)
     ctx.set_kernel_timing(0)
     if prof:
         print(f"# step wall {elapsed/K*1e3:.2f} ms; breakdown:",
