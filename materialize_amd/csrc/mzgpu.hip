@@ -1323,6 +1323,12 @@ struct LaneGuard {
     if (!a->stream) {
       HIP_CHECK(hipStreamCreate(&a->stream));
       a->lane_scr = new Scratch();
+      // pre-grow AND touch: lane arenas otherwise carve + fault their
+      // pages inside the first big merges (KFD zeroes fresh VRAM —
+      // measured as a 350ms step on a cold box's first process)
+      void *w = a->lane_scr->get(2ull << 30);
+      (void)hipMemsetAsync(w, 0, 2ull << 30, a->stream);
+      a->lane_scr->reset();
       HIP_CHECK(hipEventCreate(&a->ev_done));
       HIP_CHECK(hipEventCreate(&a->ev_gate));
       HIP_CHECK(hipEventCreate(&a->ev_ready));
@@ -2020,7 +2026,7 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   (*c->impl.scr).reset();
   // pre-back the async mempool the same way (spine batches and merge
   // outputs allocate from it at 32MB size classes)
-  for (int i = 0; i < 4; i++) {
+  for (int i = 0; i < 12; i++) {
     void *p = nullptr;
     if (hipMallocAsync(&p, 2ull << 30, c->impl.stream) == hipSuccess && p) {
       (void)hipMemsetAsync(p, 0, 2ull << 30, c->impl.stream);
