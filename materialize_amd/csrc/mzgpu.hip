@@ -2026,12 +2026,19 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   (*c->impl.scr).reset();
   // pre-back the async mempool the same way (spine batches and merge
   // outputs allocate from it at 32MB size classes)
-  for (int i = 0; i < 12; i++) {
-    void *p = nullptr;
-    if (hipMallocAsync(&p, 2ull << 30, c->impl.stream) == hipSuccess && p) {
-      (void)hipMemsetAsync(p, 0, 2ull << 30, c->impl.stream);
-      (void)hipFreeAsync(p, c->impl.stream);
+  {
+    // allocate all, touch all, then free all — alloc/free pairs would
+    // recycle one block and back only 2GB
+    void *ps[12] = {};
+    for (int i = 0; i < 12; i++) {
+      if (hipMallocAsync(&ps[i], 2ull << 30, c->impl.stream) !=
+          hipSuccess)
+        ps[i] = nullptr;
+      if (ps[i])
+        (void)hipMemsetAsync(ps[i], 0, 2ull << 30, c->impl.stream);
     }
+    for (int i = 0; i < 12; i++)
+      if (ps[i]) (void)hipFreeAsync(ps[i], c->impl.stream);
   }
   (void)hipStreamSynchronize(c->impl.stream);
   const char *prof = getenv("MZ_GPU_PROF");
