@@ -2126,10 +2126,16 @@ static void spine_policy_deferred(Ctx *ctx, mz_gpu_arr *a) {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
     return e ? atol(e) : 6;
   }();
+  static const double GEO = [] {
+    const char *e = getenv("MZ_GPU_GEO");
+    return e ? atof(e) : 2.0;
+  }();
   if (a->pending_merge.active) return;  // one in flight per arrangement
   size_t nb = a->batches.size();
-  if (nb >= 2 && a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds
-      && a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL) {
+  if (nb >= 2 &&
+      (double)a->batches[nb - 2].n_upds <=
+          GEO * (double)a->batches[nb - 1].n_upds &&
+      a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL) {
     merge_range(ctx, a, nb - 2, nb, 1);
     return;
   }
@@ -2163,10 +2169,16 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
     const char *e = getenv("MZ_GPU_SMALL_POOL");
     return e ? atol(e) : 6;
   }();
+  static const double GEO = [] {
+    const char *e = getenv("MZ_GPU_GEO");
+    return e ? atof(e) : 2.0;
+  }();
   for (;;) {
     size_t nb = a->batches.size();
-    if (nb >= 2 && a->batches[nb - 2].n_upds <= 2 * a->batches[nb - 1].n_upds
-        && a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL)
+    if (nb >= 2 &&
+        (double)a->batches[nb - 2].n_upds <=
+            GEO * (double)a->batches[nb - 1].n_upds &&
+        a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL)
       merge_range(ctx, a, nb - 2, nb);
     else
       break;
