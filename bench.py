@@ -37,9 +37,10 @@ METRIC = "input update rows/sec maintained (TPC-H Q3 delta join)"
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    # defaults span >=2 spine pool cycles (POOL=6): short windows
-    # undersample the amortized big-run merge and overstate rows/s
-    p.add_argument("--steps", type=int, default=18)
+    # defaults span the full merge hierarchy (pool merges every ~6 steps,
+    # the giant run-fold every ~30): shorter windows undersample the
+    # amortized merges and overstate rows/s
+    p.add_argument("--steps", type=int, default=36)
     p.add_argument("--warmup", type=int, default=6)
     p.add_argument("--sf", type=float, default=0.0, help="0 = auto by N")
     p.add_argument("--batch-rows", type=int, default=0, help="0 = auto")
