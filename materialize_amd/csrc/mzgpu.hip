@@ -328,25 +328,19 @@ __global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
 
 
 __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
-                             const u32 *kid, const u32 *kv_off,
-                             const u32 *vu_off, u64 cap,
+                             const u32 *kid, const u32 *kv_off, u64 cap,
                              const u64 *dcounts) {
   u64 n_keys = (cap && dcounts[0]) ? kid[cap - 1] : 0;
   GRID_STRIDE(i, n_keys) {
     u64 h = route_hash(keys + i * kw, kw) & (slots - 1);
     for (;;) {
-      u64 *slot = hash + h * (kw + 3);
+      u64 *slot = hash + h * (kw + 2);
       unsigned long long expected = ~0ull;
       unsigned long long got = atomicCAS(
           (unsigned long long *)(slot + kw), expected, (unsigned long long)i);
       if (got == ~0ull) {
         for (u32 w = 0; w < kw; w++) slot[w] = keys[i * kw + w];
-        u32 kv_lo = kv_off[i], kv_hi = kv_off[i + 1];
-        slot[kw + 1] = (u64)kv_lo | ((u64)kv_hi << 32);
-        slot[kw + 2] = (kv_hi - kv_lo == 1)
-                           ? ((u64)vu_off[kv_lo] |
-                              ((u64)vu_off[kv_lo + 1] << 32))
-                           : ~0ull;
+        slot[kw + 1] = (u64)kv_off[i] | ((u64)kv_off[i + 1] << 32);
         break;
       }
       h = (h + 1) & (slots - 1);
@@ -370,31 +364,22 @@ __device__ __forceinline__ int hash_lookup(const u64 *hash, u64 slots,
   }
 }
 
-// Batch-table lookup returning the key's slot — the widened slot packs
-// the val range (kv_lo | kv_hi<<32) at [kw+1] and, for single-val keys,
-// the upd range (vu_lo | vu_hi<<32) at [kw+2] (~0 when multi-val): one
-// random line serves the whole probe bookkeeping. nullptr = miss.
-__device__ __forceinline__ const u64 *hash_lookup_slot(const u64 *hash,
-                                                       u64 slots,
-                                                       const u64 *key,
-                                                       u32 kw) {
-  if (slots == 0) return nullptr;
-  u64 h = route_hash(key, kw) & (slots - 1);
-  for (;;) {
-    const u64 *slot = hash + h * (kw + 3);
-    u64 iw = slot[kw];
-    if (iw == ~0ull) return nullptr;
-    bool eq = true;
-    for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
-    if (eq) return slot;
-    h = (h + 1) & (slots - 1);
-  }
-}
-
+// Batch-table lookup returning the key's packed val range
+// (kv_lo | kv_hi<<32) straight from the widened slot — one random line
+// instead of two (no kv_off read). ~0 = miss.
 __device__ __forceinline__ u64 hash_lookup_range(const u64 *hash, u64 slots,
                                                  const u64 *key, u32 kw) {
-  const u64 *slot = hash_lookup_slot(hash, slots, key, kw);
-  return slot ? slot[kw + 1] : ~0ull;
+  if (slots == 0) return ~0ull;
+  u64 h = route_hash(key, kw) & (slots - 1);
+  for (;;) {
+    const u64 *slot = hash + h * (kw + 2);
+    u64 iw = slot[kw];
+    if (iw == ~0ull) return ~0ull;
+    bool eq = true;
+    for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
+    if (eq) return slot[kw + 1];
+    h = (h + 1) & (slots - 1);
+  }
 }
 
 // ------------------------------------------------------- closure (device)
@@ -512,27 +497,16 @@ __global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
     u64 t = dtimes[i];
     u32 c = 0;
     const DevBatch &b = bl.b[bi];
-    // val (and, for single-val keys, upd) ranges ride in the widened
-    // slot: one random line covers the probe bookkeeping
-    const u64 *slot = hash_lookup_slot(b.hash, b.hash_slots, key, kw);
-    u64 kvr = slot ? slot[kw + 1] : ~0ull;
-    u64 vur = slot ? slot[kw + 2] : ~0ull;
-    ki_cache[2 * idx] = kvr;
-    ki_cache[2 * idx + 1] = vur;
+    // the val range rides in the widened slot: one random line per probe
+    u64 kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
+    ki_cache[idx] = kvr;
     if (kvr != ~0ull) {
       for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
         const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
         const u8 *v1 = swap ? lv : dv;
         const u8 *v2 = swap ? dv : lv;
         if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
-        u32 lo, hi;
-        if (vur != ~0ull) {
-          lo = (u32)vur;
-          hi = (u32)(vur >> 32);
-        } else {
-          lo = b.vu_off[j];
-          hi = b.vu_off[j + 1];
-        }
+        u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
         if (mode == PM_JOIN || bl.allpass[bi]) {
           c += hi - lo;
         } else {
@@ -559,9 +533,8 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   u64 total = n * (u64)bl.n;
   GRID_STRIDE(idx, total) {
-    u64 kvr = ki_cache[2 * idx];
+    u64 kvr = ki_cache[idx];
     if (kvr == ~0ull) continue;
-    u64 vur = ki_cache[2 * idx + 1];
     u64 i = idx % n;
     int bi = (int)(idx / n);
     const u64 *key = dkeys + i * kw;
@@ -577,9 +550,7 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
       u64 okey[MAX_KW];
       u8 oval[MAX_VB];
       if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
-      u32 ulo = vur != ~0ull ? (u32)vur : b.vu_off[j];
-      u32 uhi = vur != ~0ull ? (u32)(vur >> 32) : b.vu_off[j + 1];
-      for (u32 u = ulo; u < uhi; u++) {
+      for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
         u64 tout;
         if (bl.allpass[bi]) {
           tout = t;  // t2 < delta lower <= t on every update
@@ -1804,13 +1775,13 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   u64 slots = 16;
   while (slots < 2 * cap) slots <<= 1;
   b.hash_slots = slots;
-  b.hash = dnew<u64>(c, slots * (kw + 3));
+  b.hash = dnew<u64>(c, slots * (kw + 2));
   // full-line 0xFF fill: lookups check the idx-word sentinel before key
   // compares, so poisoned key words are never read
-  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 3) * 8, c->stream));
+  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 2) * 8, c->stream));
   hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
                      c->stream, b.hash, slots, b.keys, kw, kid, b.kv_off,
-                     b.vu_off, cap, dcounts);
+                     cap, dcounts);
   // flat key/val arrays were re-packed; stream-ordered free is safe
   dfree(c, keys);
   dfree(c, vals);
@@ -2390,7 +2361,7 @@ int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
   for (auto &b : a->batches) {
     n += b.n_upds;
     by += b.n_keys * kw * 8 + b.n_vals * vb + b.n_upds * 16 +
-          b.hash_slots * (kw + 3) * 8 + (b.n_keys + b.n_vals) * 4 +
+          b.hash_slots * (kw + 2) * 8 + (b.n_keys + b.n_vals) * 4 +
           b.n_upds * 4 + b.n_vals * 4;
   }
   *n_updates = n;
@@ -2500,7 +2471,7 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   }
   u64 nb2 = n * (u64)bl.n;
   u32 *count = (u32 *)S.get((nb2 + 1) * 4);
-  u64 *ki_cache = (u64 *)S.get(nb2 * 16);
+  u64 *ki_cache = (u64 *)S.get(nb2 * 8);
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   hipLaunchKernelGGL(k_probe_count, dim3(ngrid(nb2)), dim3(BLK), 0,
                      ctx->stream, d.keys, d.vals, stream_vb, d.times, n, kw,
