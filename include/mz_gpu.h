@@ -184,6 +184,12 @@ int  mz_gpu_arr_flush(mz_gpu_ctx *ctx, mz_gpu_arr *arr);
  * cf. set_logical_compaction, mz_join_core.rs:461. */
 int  mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
                                        uint64_t frontier);
+/* Physical-compaction hint (mz_join_core.rs:465): a floor on batch
+ * merging. This spine merges eagerly by level policy,
+ * so the hint is recorded but imposes nothing — batches at or beyond
+ * the frontier are already merged as the policy reaches them. */
+int  mz_gpu_arr_set_physical_compaction(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                                        uint64_t frontier);
 /* Perform up to `fuel` rows of spine merge work — cf. manager.rs:54. */
 int  mz_gpu_arr_maintain(mz_gpu_ctx *ctx, mz_gpu_arr *arr, uint64_t fuel);
 /* Introspection (arrangement-size logging, extensions/arrange.rs:249). */

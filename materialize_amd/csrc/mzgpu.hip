@@ -1184,6 +1184,7 @@ struct mz_gpu_arr {
   DevSchema schema;
   std::vector<DevBatch> batches;
   u64 logical_compaction = 0;
+  u64 physical_compaction = 0;
   u64 upper = 0;
   Ctx *ctx = nullptr;
   // Per-arrangement lane: inserts/merges run on this stream with this
@@ -2343,6 +2344,13 @@ int mz_gpu_arr_insert(mz_gpu_ctx *c, mz_gpu_arr *a,
 int mz_gpu_arr_set_logical_compaction(mz_gpu_ctx *c, mz_gpu_arr *a,
                                       uint64_t f) {
   a->logical_compaction = f;
+  return 0;
+}
+
+int mz_gpu_arr_set_physical_compaction(mz_gpu_ctx *c, mz_gpu_arr *a,
+                                       uint64_t f) {
+  (void)c;
+  a->physical_compaction = f;  // recorded; eager level merges subsume it
   return 0;
 }
 
