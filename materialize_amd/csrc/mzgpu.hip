@@ -2105,8 +2105,15 @@ mz_gpu_arr *mz_gpu_arr_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
 }
 
 void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
-  arr_flush_impl(&c->impl, a);
-  for (auto &b : a->batches) free_batch(&c->impl, b);
+  Ctx *ctx = &c->impl;
+  arr_flush_impl(ctx, a);
+  if (a->pending_merge.active) {  // in-flight deferred merge reads batches
+    LaneGuard lane(ctx, a);
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    merge_install(ctx, a);
+  }
+  if (a->stream) (void)hipStreamSynchronize(a->stream);
+  for (auto &b : a->batches) free_batch(ctx, b);
   a->batches.clear();
 }
 
@@ -2356,7 +2363,14 @@ int mz_gpu_arr_set_physical_compaction(mz_gpu_ctx *c, mz_gpu_arr *a,
 
 int mz_gpu_arr_maintain(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t fuel) {
   (void)fuel;
-  merge_range(&c->impl, a, 0, a->batches.size());
+  Ctx *ctx = &c->impl;
+  arr_flush_impl(ctx, a);  // pending insert joins; pending merge installs
+  LaneGuard lane(ctx, a);
+  if (a->pending_merge.active) {  // flush's policy may have enqueued one
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    merge_install(ctx, a);
+  }
+  merge_range(ctx, a, 0, a->batches.size());
   return 0;
 }
 
