@@ -176,7 +176,10 @@ int  mz_gpu_arr_insert(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
  * without a device sync; the sealed batch joins the spine at
  * mz_gpu_arr_flush (any probe of the arrangement flushes implicitly, as
  * does mz_gpu_sync). Independent arrangements' inserts overlap this way.
- * mz_gpu_arr_insert == insert_async + flush. */
+ * mz_gpu_arr_insert == insert_async + flush.
+ * Lifetime: host-memory updates are staged inside the call (pageable
+ * async copies complete in-call); DEVICE-pointer updates must stay
+ * valid until the arrangement is flushed (explicitly or by a probe). */
 int  mz_gpu_arr_insert_async(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
                              const mz_gpu_updates *updates);
 int  mz_gpu_arr_flush(mz_gpu_ctx *ctx, mz_gpu_arr *arr);
