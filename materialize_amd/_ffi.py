@@ -146,23 +146,10 @@ class GpuCtx:
             raise MzGpuError(
                 "mz_gpu_init failed: no HIP device available. The product "
                 "path requires an MI355X; there is no CPU fallback.")
-        if not os.environ.get("MZ_NO_WARMUP"):
-            self._warmup()
-
-    def _warmup(self):
-        """One throwaway consolidate: a virgin context's very first sort
-        pipeline has produced garbled permutations (first reduce push on
-        a fresh ctx garbles group accumulation — tests/test_virgin_ctx
-        pins it); any prior op clears it. Root cause under
-        investigation; this mirrors what every long-lived context does
-        anyway (load precedes maintenance)."""
-        import numpy as np
-        from . import _abi as abi
-        u = abi.make_updates(np.array([2, 1], np.int64),
-                             np.arange(16, dtype=np.uint8),
-                             np.zeros(2, np.uint64),
-                             np.ones(2, np.int64), 0, 1)
-        self.consolidate(abi.schema(1, 8), u)
+        # (The round-1 "virgin context" warmup is gone: the fault was
+        # hipMemsetAsync fills being lost on a fresh process's first ops —
+        # all semantic state init now uses compute-kernel fills in the C
+        # layer, so every ABI consumer is covered. DESIGN.md §9.)
 
     def close(self):
         if getattr(self, "ctx", None):

@@ -963,6 +963,25 @@ __global__ void k_gather_group_keys(const u64 *keys, u32 kw,
 }
 
 __global__ void k_fill_u64(u64 *p, u64 n, u64 v) { GRID_STRIDE(i, n) p[i] = v; }
+__global__ void k_fill_u32(u32 *p, u64 n, u32 v) { GRID_STRIDE(i, n) p[i] = v; }
+__global__ void k_fill_u8(u8 *p, u64 n, u8 v) { GRID_STRIDE(i, n) p[i] = v; }
+// mins[i] = ~0, maxs[i] = 0 in one launch (k_pass_minmax init)
+__global__ void k_init_minmax(u64 *mins, u64 *maxs, u64 n) {
+  GRID_STRIDE(i, n) {
+    mins[i] = ~0ull;
+    maxs[i] = 0;
+  }
+}
+
+// diagnostic: kernel's-eye view of the hash slot a key probes first
+__global__ void k_dbg_read_slot(const u64 *hash, u64 slots, const u64 *key,
+                                u32 kw, u64 *out) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    u64 h = route_hash(key, kw) & (slots - 1);
+    for (u32 w = 0; w < kw + 1; w++) out[w] = hash[h * (kw + 1) + w];
+    out[kw + 1] = h;
+  }
+}
 __global__ void k_fill_i64(i64 *p, u64 n, i64 v) { GRID_STRIDE(i, n) p[i] = v; }
 
 __global__ void k_check_pos(const i64 *d, u64 n, u64 *err) {
@@ -1373,6 +1392,28 @@ T *dnew(Ctx *c, u64 n) {
   return (T *)dmalloc(c, n * sizeof(T));
 }
 
+// Kernel-based fills for SEMANTIC device state (hash sentinels, counters,
+// zero-padding that later passes read as content). hipMemsetAsync fills
+// into a freshly-carved hipMallocAsync block were observed to be silently
+// LOST on a process's first operations (virgin-context fault: the reduce
+// hash table kept its pre-memset zeros, so lookups aliased distinct keys
+// onto one row — isolated round 2). Compute-kernel
+// writes share the consuming kernels' ordering and address path and are
+// not affected; memsets remain only for page-touch warming where content
+// is never read.
+static void fill_u64(Ctx *c, u64 *p, u64 n, u64 v) {
+  hipLaunchKernelGGL(k_fill_u64, dim3(ngrid(n)), dim3(BLK), 0, c->stream, p,
+                     n, v);
+}
+static void fill_u32(Ctx *c, u32 *p, u64 n, u32 v) {
+  hipLaunchKernelGGL(k_fill_u32, dim3(ngrid(n)), dim3(BLK), 0, c->stream, p,
+                     n, v);
+}
+static void fill_u8(Ctx *c, u8 *p, u64 n, u8 v) {
+  hipLaunchKernelGGL(k_fill_u8, dim3(ngrid(n)), dim3(BLK), 0, c->stream, p,
+                     n, v);
+}
+
 // composite stable sort: returns perm ordering rows by (key, val, time) —
 // or (time, key) when for_reduce (primary time).
 // Sort passes use rocprim radix_sort_pairs (AMD-native primitive).
@@ -1393,8 +1434,8 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
   u32 vwords = time_major ? 0 : (vb + 7) / 8;
   u64 *dminmax = (u64 *)S.get(2 * MAX_PASSES * 8);
   u64 *dmin = dminmax, *dmax = dminmax + MAX_PASSES;
-  HIP_CHECK(hipMemsetAsync(dmin, 0xFF, MAX_PASSES * 8, c->stream));
-  HIP_CHECK(hipMemsetAsync(dmax, 0, MAX_PASSES * 8, c->stream));
+  hipLaunchKernelGGL(k_init_minmax, dim3(1), dim3(BLK), 0, c->stream, dmin,
+                     dmax, MAX_PASSES);
   if (n)
     hipLaunchKernelGGL(k_pass_minmax,
                        dim3(ngrid(n), 1 + vwords + kw), dim3(BLK), 0,
@@ -1504,6 +1545,67 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                              c->stream));
     std::swap(perm, perm_out);
   }
+  if (getenv("MZ_DBG_SORT") && n) {
+    // diagnostic: host-side validation of the produced permutation
+    HIP_CHECK(hipStreamSynchronize(c->stream));
+    fprintf(stderr, "[dbg_sort] n=%llu kw=%u vb=%u tm=%d vwords=%u "
+            "passes=%zu chunks=%zu\n", (unsigned long long)n, kw, vb,
+            (int)time_major, vwords, passes.size(), chunks.size());
+    for (u32 s = 0; s < 1 + vwords + kw; s++)
+      fprintf(stderr, "[dbg_sort] slot %u min=%llx max=%llx\n", s,
+              (unsigned long long)hmin[s], (unsigned long long)hmax[s]);
+    std::vector<u32> hp(n);
+    std::vector<u64> hk(n * kw), ht(n);
+    std::vector<u8> hv(std::max<u64>(n * vb, 1));
+    HIP_CHECK(hipMemcpy(hp.data(), perm, n * 4, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(hk.data(), keys, n * kw * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(ht.data(), times, n * 8, hipMemcpyDeviceToHost));
+    if (vb)
+      HIP_CHECK(hipMemcpy(hv.data(), vals, n * vb, hipMemcpyDeviceToHost));
+    std::vector<u8> seen(n, 0);
+    bool isperm = true;
+    for (u64 i = 0; i < n; i++) {
+      if (hp[i] >= n || seen[hp[i]]) { isperm = false; break; }
+      seen[hp[i]] = 1;
+    }
+    auto cmp_le = [&](u32 a, u32 b) {  // (time?,key,val) order per mode
+      auto keycmp = [&]() -> int {
+        for (u32 w = 0; w < kw; w++) {
+          i64 x = (i64)hk[(u64)a * kw + w], y = (i64)hk[(u64)b * kw + w];
+          if (x != y) return x < y ? -1 : 1;
+        }
+        return 0;
+      };
+      if (time_major) {
+        if (ht[a] != ht[b]) return ht[a] < ht[b];
+        int kc = keycmp();
+        return kc <= 0;
+      }
+      int kc = keycmp();
+      if (kc) return kc < 0;
+      for (u32 cbyte = 0; cbyte < vb; cbyte += 8) {
+        u64 x = 0, y = 0;
+        u32 rem = vb - cbyte < 8 ? vb - cbyte : 8;
+        memcpy(&x, hv.data() + (u64)a * vb + cbyte, rem);
+        memcpy(&y, hv.data() + (u64)b * vb + cbyte, rem);
+        if (x != y) return x < y;
+      }
+      if (ht[a] != ht[b]) return ht[a] < ht[b];
+      return true;
+    };
+    u64 bad = ~0ull;
+    for (u64 i = 0; i + 1 < n && bad == ~0ull; i++)
+      if (!cmp_le(hp[i], hp[i + 1])) bad = i;
+    fprintf(stderr, "[dbg_sort] perm_valid=%d sorted=%d bad_at=%lld\n",
+            (int)isperm, (int)(bad == ~0ull), (long long)bad);
+    if (bad != ~0ull) {
+      for (u64 i = (bad > 2 ? bad - 2 : 0); i < std::min(n, bad + 3); i++)
+        fprintf(stderr, "[dbg_sort]  i=%llu perm=%u key=%lld t=%llu\n",
+                (unsigned long long)i, hp[i],
+                (long long)hk[(u64)hp[i] * kw], (unsigned long long)ht[hp[i]]);
+    }
+    fflush(stderr);
+  }
 }
 
 u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
@@ -1513,7 +1615,7 @@ u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   u32 *pad = (u32 *)S.get((n + 1) * 4);
   HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
                            c->stream));
-  HIP_CHECK(hipMemsetAsync(pad + n, 0, 4, c->stream));
+  fill_u32(c, pad + n, 1, 0);
   size_t need = 0;
   (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
                           rocprim::plus<u32>(), c->stream);
@@ -1533,7 +1635,7 @@ void exclusive_scan_u32_ns(Ctx *c, const u32 *in, u32 *out, u64 n) {
   u32 *pad = (u32 *)S.get((n + 1) * 4);
   HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
                            c->stream));
-  HIP_CHECK(hipMemsetAsync(pad + n, 0, 4, c->stream));
+  fill_u32(c, pad + n, 1, 0);
   size_t need = 0;
   (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
                                 rocprim::plus<u32>(), c->stream);
@@ -1654,7 +1756,7 @@ void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
   auto &S = (*c->scr);
   u64 n = in.n;
   if (n == 0) {
-    HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
+    fill_u64(c, dcounts, 1, 0);
     return;
   }
   u32 *perm = (u32 *)S.get(n * 4);
@@ -1670,7 +1772,7 @@ void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
   auto &S = (*c->scr);
   u64 n = in.n;
   if (n == 0) {
-    HIP_CHECK(hipMemsetAsync(dcounts, 0, 8, c->stream));
+    fill_u64(c, dcounts, 1, 0);
     return;
   }
   u32 *flags = (u32 *)S.get(n * 4);
@@ -1749,8 +1851,8 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
     b.vu_off = dnew<u32>(c, 1);
     b.val_key = dnew<u32>(c, 1);
     b.upd_val = dnew<u32>(c, 1);
-    HIP_CHECK(hipMemsetAsync(b.kv_off, 0, 4, c->stream));
-    HIP_CHECK(hipMemsetAsync(b.vu_off, 0, 4, c->stream));
+    fill_u32(c, b.kv_off, 1, 0);
+    fill_u32(c, b.vu_off, 1, 0);
     return b;
   }
   u32 *kc = (u32 *)S.get(cap * 4);
@@ -1779,7 +1881,7 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   b.hash = dnew<u64>(c, slots * (kw + 2));
   // full-line 0xFF fill: lookups check the idx-word sentinel before key
   // compares, so poisoned key words are never read
-  HIP_CHECK(hipMemsetAsync(b.hash, 0xFF, slots * (kw + 2) * 8, c->stream));
+  fill_u64(c, b.hash, slots * (kw + 2), ~0ull);
   hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
                      c->stream, b.hash, slots, b.keys, kw, kid, b.kv_off,
                      cap, dcounts);
@@ -2619,10 +2721,9 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
   r->st.capacity = cap;
   r->d_nrows = dnew<u64>(ctx, 1);
   r->d_err = dnew<u64>(ctx, 1);
-  HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(r->st.hash, 0xFF, slots * (kw + 1) * 8,
-                           ctx->stream));
+  fill_u64(ctx, r->d_nrows, 1, 0);
+  fill_u64(ctx, r->d_err, 1, 0);
+  fill_u64(ctx, r->st.hash, slots * (kw + 1), ~0ull);
   c->impl.reds.push_back(r);
   return r;
 }
@@ -2736,7 +2837,7 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
   u64 *pt = dnew<u64>(ctx, cap_out);
   i64 *pd = dnew<i64>(ctx, cap_out);
   unsigned long long *ocount = (unsigned long long *)S.get(8);
-  HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+  fill_u64(ctx, (u64 *)ocount, 1, 0);
   u32 *flags = (u32 *)S.get(n * 4);
   u32 *gid = (u32 *)S.get(n * 4);
   for (auto [lo, hi] : slices) {
@@ -2748,6 +2849,72 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
     u32 *starts = (u32 *)S.get(m * 4);
     hipLaunchKernelGGL(k_group_starts, dim3(ngrid(m)), dim3(BLK), 0,
                        ctx->stream, flags, gid, starts, m);
+    if (getenv("MZ_DBG_SORT")) {
+      // diagnostic: recompute flags/gid on host from the gathered columns
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      std::vector<u64> hsk(m * kw), hstm(m);
+      std::vector<u32> hflags(m), hgid(m);
+      HIP_CHECK(hipMemcpy(hsk.data(), sk + lo * kw, m * kw * 8,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hstm.data(), stm + lo, m * 8,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hflags.data(), flags, m * 4,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hgid.data(), gid, m * 4, hipMemcpyDeviceToHost));
+      u64 flag_bad = ~0ull, gid_bad = ~0ull;
+      u32 run = 0;
+      for (u64 i = 0; i < m; i++) {
+        u32 want = 1;
+        if (i) {
+          bool neq = hstm[i] != hstm[i - 1];
+          for (u32 w = 0; w < kw && !neq; w++)
+            neq |= hsk[i * kw + w] != hsk[(i - 1) * kw + w];
+          want = neq ? 1 : 0;
+        }
+        if (hflags[i] != want && flag_bad == ~0ull) flag_bad = i;
+        run += want;  // recomputed scan uses *want*, so gid_bad isolates
+        if (hgid[i] != run && gid_bad == ~0ull) gid_bad = i;  // the scan
+      }
+      fprintf(stderr, "[dbg_grp] m=%llu flag_bad=%lld gid_bad=%lld "
+              "G_dev=%u G_host=%u sorted_gather_key0=%lld\n",
+              (unsigned long long)m, (long long)flag_bad, (long long)gid_bad,
+              hgid[m - 1], run, (long long)hsk[0]);
+      if (flag_bad != ~0ull || gid_bad != ~0ull) {
+        u64 b0 = std::min(flag_bad, gid_bad);
+        for (u64 i = (b0 > 2 ? b0 - 2 : 0); i < std::min<u64>(m, b0 + 3);
+             i++)
+          fprintf(stderr, "[dbg_grp]  i=%llu flag=%u gid=%u key=%lld "
+                  "t=%llu\n", (unsigned long long)i, hflags[i], hgid[i],
+                  (long long)hsk[i * kw], (unsigned long long)hstm[i]);
+      }
+      fflush(stderr);
+    }
+    if (getenv("MZ_DBG_TBL")) {
+      // two views of the slot the first group's key probes: kernel-read
+      // (through L2) vs hipMemcpy (DMA) — distinguishes "memory truly
+      // not 0xFF" from "kernel sees stale cache"
+      u64 *dslot = (u64 *)S.get((kw + 2) * 8);
+      hipLaunchKernelGGL(k_dbg_read_slot, dim3(1), dim3(1), 0, ctx->stream,
+                         op->st.hash, op->st.slots, sk + lo * kw, kw, dslot);
+      std::vector<u64> kview(kw + 2), key0(kw);
+      HIP_CHECK(hipMemcpy(kview.data(), dslot, (kw + 2) * 8,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(key0.data(), sk + lo * kw, kw * 8,
+                          hipMemcpyDeviceToHost));
+      u64 h = route_hash(key0.data(), kw) & (op->st.slots - 1);
+      std::vector<u64> mview(kw + 1);
+      HIP_CHECK(hipMemcpy(mview.data(), op->st.hash + h * (kw + 1),
+                          (kw + 1) * 8, hipMemcpyDeviceToHost));
+      fprintf(stderr, "[dbg_tbl] key0=%lld h=%llu kernel_view=",
+              (long long)key0[0], (unsigned long long)kview[kw + 1]);
+      for (u32 w = 0; w < kw + 1; w++)
+        fprintf(stderr, "%llx,", (unsigned long long)kview[w]);
+      fprintf(stderr, " memcpy_view=");
+      for (u32 w = 0; w < kw + 1; w++)
+        fprintf(stderr, "%llx,", (unsigned long long)mview[w]);
+      fprintf(stderr, "\n");
+      fflush(stderr);
+    }
     // Phase A: lookup; Phase B: insert misses (separate launches for
     // coherence — see RedState docs); Phase C: apply + emit; bump counter.
     u32 *found = (u32 *)S.get(m * 4);
@@ -2768,6 +2935,51 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
                        pd, ocount);
     hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
                        op->d_nrows, misspos, gid, m);
+    if (getenv("MZ_DBG_SORT")) {
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+      std::vector<u32> hgid(m), hstarts(m), hfound(m), hmiss(m);
+      std::vector<u64> hsk(m * kw);
+      HIP_CHECK(hipMemcpy(hgid.data(), gid, m * 4, hipMemcpyDeviceToHost));
+      u64 G = hgid[m - 1];
+      if (G > m) {
+        fprintf(stderr, "[dbg_red] GARBAGE G=%llu > m=%llu\n",
+                (unsigned long long)G, (unsigned long long)m);
+        fflush(stderr);
+        continue;
+      }
+      hstarts.resize(G ? G : 1);
+      hfound.resize(G ? G : 1);
+      hmiss.resize(G ? G : 1);
+      HIP_CHECK(hipMemcpy(hstarts.data(), starts, G * 4,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hfound.data(), found, G * 4,
+                          hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hmiss.data(), miss, G * 4, hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(hsk.data(), sk + lo * kw, m * kw * 8,
+                          hipMemcpyDeviceToHost));
+      // groups must carry distinct keys (input sorted by key within slice)
+      bool dup = false;
+      for (u64 g1 = 0; g1 + 1 < G && !dup; g1++)
+        for (u64 g2 = g1 + 1; g2 < G && !dup; g2++) {
+          bool eq = true;
+          for (u32 w = 0; w < kw; w++)
+            eq &= hsk[(u64)hstarts[g1] * kw + w] ==
+                  hsk[(u64)hstarts[g2] * kw + w];
+          dup = eq;
+        }
+      u64 nmiss = 0, nfound_bad = 0;
+      for (u64 g1 = 0; g1 < G; g1++) nmiss += hmiss[g1];
+      for (u64 g1 = 0; g1 < G; g1++)
+        if (!hmiss[g1] && hfound[g1] == ~0u) nfound_bad++;
+      unsigned long long hoc = 0;
+      HIP_CHECK(hipMemcpy(&hoc, ocount, 8, hipMemcpyDeviceToHost));
+      fprintf(stderr, "[dbg_red] slice m=%llu G=%llu dup_groups=%d "
+              "misses=%llu found_bad=%llu emitted=%llu\n",
+              (unsigned long long)m, (unsigned long long)G, (int)dup,
+              (unsigned long long)nmiss, (unsigned long long)nfound_bad,
+              hoc);
+      fflush(stderr);
+    }
   }
   // consolidate the actual emitted corrections (one count readback —
   // sorting the 2n-capacity zero-padded buffer dominated this path)
@@ -2811,10 +3023,9 @@ mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
   r->st.capacity = cap;
   r->d_nrows = dnew<u64>(ctx, 1);
   r->d_err = dnew<u64>(ctx, 1);
-  HIP_CHECK(hipMemsetAsync(r->d_nrows, 0, 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
-  HIP_CHECK(hipMemsetAsync(r->st.hash, 0xFF, slots * (r->kw2 + 1) * 8,
-                           ctx->stream));
+  fill_u64(ctx, r->d_nrows, 1, 0);
+  fill_u64(ctx, r->d_err, 1, 0);
+  fill_u64(ctx, r->st.hash, slots * (r->kw2 + 1), ~0ull);
   return r;
 }
 
@@ -2875,7 +3086,7 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
   u64 *pt = dnew<u64>(ctx, cap_out);
   i64 *pd = dnew<i64>(ctx, cap_out);
   unsigned long long *ocount = (unsigned long long *)S.get(8);
-  HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+  fill_u64(ctx, (u64 *)ocount, 1, 0);
   u32 *flags = (u32 *)S.get(n * 4);
   u32 *gid = (u32 *)S.get(n * 4);
   for (auto [lo, hi] : slices) {
@@ -3022,7 +3233,7 @@ mz_gpu_topk *mz_gpu_topk_create(mz_gpu_ctx *c, const mz_gpu_topk_spec *spec) {
   r->spec = *spec;
   r->arr = mz_gpu_arr_create(c, &spec->in);
   r->d_err = dnew<u64>(ctx, 1);
-  HIP_CHECK(hipMemsetAsync(r->d_err, 0, 8, ctx->stream));
+  fill_u64(ctx, r->d_err, 1, 0);
   return r;
 }
 
@@ -3130,13 +3341,12 @@ int mz_gpu_topk_push(mz_gpu_ctx *c, mz_gpu_topk *op,
       u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(capn * vb, 1));
       u64 *pt = dnew<u64>(ctx, capn);
       i64 *pd = dnew<i64>(ctx, capn);
-      HIP_CHECK(hipMemsetAsync(pk, 0, capn * kw * 8, ctx->stream));
-      HIP_CHECK(hipMemsetAsync(pv, 0, std::max<u64>(capn * vb, 1),
-                               ctx->stream));
-      HIP_CHECK(hipMemsetAsync(pt, 0, capn * 8, ctx->stream));
-      HIP_CHECK(hipMemsetAsync(pd, 0, capn * 8, ctx->stream));
+      fill_u64(ctx, pk, capn * kw, 0);
+      fill_u8(ctx, pv, std::max<u64>(capn * vb, 1), 0);
+      fill_u64(ctx, pt, capn, 0);
+      fill_u64(ctx, (u64 *)pd, capn, 0);
       unsigned long long *ocount = (unsigned long long *)S.get(8);
-      HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+      fill_u64(ctx, (u64 *)ocount, 1, 0);
       topk_eval_emit(ctx, op, oldp, t, -1, pk, pv, pt, pd, ocount);
       topk_eval_emit(ctx, op, newp, t, +1, pk, pv, pt, pd, ocount);
       segs.push_back({pk, pv, pt, pd});
@@ -3370,8 +3580,7 @@ mz_gpu_minmax *mz_gpu_minmax_create(mz_gpu_ctx *c, const mz_gpu_schema *in,
     st.stride_w = kw2 + 2;  // [key][exists][value]
     st.rows = dnew<u64>(ctx, cap * st.stride_w);
     st.capacity = cap;
-    HIP_CHECK(hipMemsetAsync(st.hash, 0xFF, slots * (kw2 + 1) * 8,
-                             ctx->stream));
+    fill_u64(ctx, st.hash, slots * (kw2 + 1), ~0ull);
   }
   return op;
 }
@@ -3445,7 +3654,7 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
     owned.insert(owned.end(), {(void *)pk, (void *)pv, (void *)pt,
                                (void *)pd});
     unsigned long long *ocount = (unsigned long long *)S.get(8);
-    HIP_CHECK(hipMemsetAsync(ocount, 0, 8, ctx->stream));
+    fill_u64(ctx, (u64 *)ocount, 1, 0);
     BatchList bl;
     bl.n = 0;
     for (auto &b : A->batches)

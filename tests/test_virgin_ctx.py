@@ -1,7 +1,15 @@
 """Pin the virgin-context first-op scenario: a freshly created engine
 context's FIRST operation being a reduce push must already be correct
 (smoke() runs exactly this shape; the structured suites always warmed
-the context with other ops first and missed it)."""
+the context with other ops first and missed it).
+
+Round-2 root cause: hipMemsetAsync fills into freshly-carved
+hipMallocAsync blocks were silently LOST on a process's first
+operations — the reduce hash table kept its pre-memset zeros, so a
+zeroed slot read as "key 0 at row 0" and distinct keys aliased onto one
+accumulator row. Semantic state init now uses compute-kernel fills
+(mzgpu.hip fill_u64/fill_u32/fill_u8); this test pins that path with no
+warmup crutch."""
 import numpy as np
 import pytest
 
