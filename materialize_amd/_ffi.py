@@ -120,6 +120,8 @@ def load():
                                        C.POINTER(Updates),
                                        C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_minmax_drop.argtypes = [C.c_void_p, C.c_void_p]
+    lib.mz_gpu_join_drop.argtypes = [C.c_void_p, C.c_void_p]
+    lib.mz_gpu_reduce_drop.argtypes = [C.c_void_p, C.c_void_p]
     lib.mz_gpu_peek.argtypes = [C.c_void_p, C.c_void_p,
                                 C.POINTER(C.c_uint64), C.c_uint64,
                                 C.c_uint64, C.POINTER(C.POINTER(OutBatch))]

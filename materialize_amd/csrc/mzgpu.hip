@@ -891,6 +891,46 @@ __global__ void k_reduce_apply(const u64 *keys, const u8 *vals, u32 kw,
   }
 }
 
+// ---- resident-table compaction/growth (long-running churn: zero-count /
+// zero-accum rows are reclaimed and the table doubles when live + incoming
+// exceeds capacity; render/threshold.rs erases zeroed entries, reduce's
+// trace compaction drops empty accums — same effect, batched).
+__global__ void k_row_live_flags(const u64 *rows, u32 stride_w, u32 from,
+                                 u64 nrows, u32 *flags) {
+  GRID_STRIDE(i, nrows) {
+    u32 live = 0;
+    for (u32 w = from; w < stride_w; w++) live |= rows[i * stride_w + w] != 0;
+    flags[i] = live;
+  }
+}
+__global__ void k_compact_rows(const u64 *rows, u32 stride_w, u64 nrows,
+                               const u32 *flags, const u32 *pos,
+                               u64 *newrows) {
+  GRID_STRIDE(i, nrows) {
+    if (!flags[i]) continue;
+    u64 j = pos[i];
+    for (u32 w = 0; w < stride_w; w++)
+      newrows[j * stride_w + w] = rows[i * stride_w + w];
+  }
+}
+__global__ void k_rehash_rows(const u64 *rows, u32 stride_w, u32 kw, u64 n,
+                              u64 *hash, u64 slots) {
+  GRID_STRIDE(i, n) {
+    const u64 *key = rows + i * stride_w;
+    u64 h = route_hash(key, kw) & (slots - 1);
+    for (;;) {
+      u64 *slot = hash + h * (kw + 1);
+      unsigned long long prev = atomicCAS((unsigned long long *)(slot + kw),
+                                          ~0ull, (unsigned long long)i);
+      if (prev == ~0ull) {
+        for (u32 w = 0; w < kw; w++) slot[w] = key[w];
+        break;
+      }
+      h = (h + 1) & (slots - 1);
+    }
+  }
+}
+
 // time-change flags over sorted times
 // Pack (key words, zero-padded val words) into one combined key row —
 // the threshold's grouping key (its "key" is the whole record:
@@ -1257,6 +1297,8 @@ struct mz_gpu_thr {
   u32 kw2;  // combined key words: key_words + ceil(val_bytes/8)
   RedState st;
   u64 capacity;
+  u64 n_rows = 0;  // host mirror of *d_nrows (refreshed at each push's
+                   // closing sync; drives compaction/growth pre-checks)
   u64 *d_nrows = nullptr;
   u64 *d_err = nullptr;
 };
@@ -1412,6 +1454,48 @@ static void fill_u32(Ctx *c, u32 *p, u64 n, u32 v) {
 static void fill_u8(Ctx *c, u8 *p, u64 n, u8 v) {
   hipLaunchKernelGGL(k_fill_u8, dim3(ngrid(n)), dim3(BLK), 0, c->stream, p,
                      n, v);
+}
+
+u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n);
+
+// Reclaim dead rows (all state words zero from `from_w` up) and grow the
+// table so `need` more rows fit: rebuild rows + hash at the new capacity.
+// Synchronizes. h_nrows is the op's host mirror of *d_nrows.
+static void red_compact_grow(Ctx *c, RedState &st, u64 *d_nrows,
+                             u64 &h_nrows, u64 need, u32 key_words,
+                             u32 from_w) {
+  auto &S = (*c->scr);
+  u64 n = h_nrows;
+  u32 *flags = (u32 *)S.get(std::max<u64>(n, 1) * 4);
+  u32 *pos = (u32 *)S.get((std::max<u64>(n, 1) + 1) * 4);
+  u64 live = 0;
+  if (n) {
+    hipLaunchKernelGGL(k_row_live_flags, dim3(ngrid(n)), dim3(BLK), 0,
+                       c->stream, st.rows, st.stride_w, from_w, n, flags);
+    live = exclusive_scan_u32(c, flags, pos, n);  // syncs
+  }
+  u64 newcap = st.capacity;
+  while (live + need > newcap) newcap *= 2;
+  u64 slots = 2 * newcap;
+  u64 *nrows_arr = dnew<u64>(c, newcap * st.stride_w);
+  u64 *nhash = dnew<u64>(c, slots * (key_words + 1));
+  fill_u64(c, nhash, slots * (key_words + 1), ~0ull);
+  if (live) {
+    hipLaunchKernelGGL(k_compact_rows, dim3(ngrid(n)), dim3(BLK), 0,
+                       c->stream, st.rows, st.stride_w, n, flags, pos,
+                       nrows_arr);
+    hipLaunchKernelGGL(k_rehash_rows, dim3(ngrid(live)), dim3(BLK), 0,
+                       c->stream, nrows_arr, st.stride_w, key_words, live,
+                       nhash, slots);
+  }
+  dfree(c, st.rows);
+  dfree(c, st.hash);
+  st.rows = nrows_arr;
+  st.hash = nhash;
+  st.capacity = newcap;
+  st.slots = slots;
+  h_nrows = live;
+  fill_u64(c, d_nrows, 1, live);
 }
 
 // composite stable sort: returns perm ordering rows by (key, val, time) —
@@ -2150,11 +2234,14 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
 }
 
 // Sub-phase profile dump (MZ_GPU_PROF=1): per category, total device ms
-// over the recorded event pairs since the last dump. Printed to stdout as
-// "MZPROF <cat> <ms> <count>" lines; events are released.
+// over the recorded event pairs since the last dump. Printed to stderr as
+// "MZPROF <cat> <ms> <count>" lines; events are released. Event pairs may
+// live on per-arrangement lane streams, so all lanes are synced first.
 void mz_gpu_prof_dump(mz_gpu_ctx *c) {
   Ctx *ctx = &c->impl;
   if (!ctx->prof.enabled) return;
+  for (mz_gpu_arr *a : ctx->arrs)
+    if (a->stream) (void)hipStreamSynchronize(a->stream);
   (void)hipStreamSynchronize(ctx->stream);
   for (auto &[name, evs] : ctx->prof.cats) {
     double total = 0;
@@ -2217,6 +2304,19 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
   if (a->stream) (void)hipStreamSynchronize(a->stream);
   for (auto &b : a->batches) free_batch(ctx, b);
   a->batches.clear();
+  // full teardown: lane resources, registry entry, the struct itself
+  if (a->stream) {
+    (void)hipStreamDestroy(a->stream);
+    (void)hipEventDestroy(a->ev_done);
+    (void)hipEventDestroy(a->ev_gate);
+    (void)hipEventDestroy(a->ev_ready);
+    delete a->lane_scr;
+    a->stream = nullptr;
+    a->lane_scr = nullptr;
+  }
+  auto &v = ctx->arrs;
+  v.erase(std::remove(v.begin(), v.end(), a), v.end());
+  delete a;
 }
 
 // Geometric spine maintenance (amortized merging — scheduling policy per
@@ -2547,7 +2647,11 @@ mz_gpu_join *mz_gpu_join_create(mz_gpu_ctx *c, mz_gpu_arr *a1, mz_gpu_arr *a2,
   return j;
 }
 
-void mz_gpu_join_drop(mz_gpu_ctx *c, mz_gpu_join *j) { (void)c; (void)j; }
+void mz_gpu_join_drop(mz_gpu_ctx *c, mz_gpu_join *j) {
+  auto &v = c->impl.joins;
+  v.erase(std::remove(v.begin(), v.end(), j), v.end());
+  delete j;  // the operator does not own its input arrangements
+}
 
 // shared probe path for linear join and half join; consolidate_out=0
 // skips the output consolidation (legal when the consumer consolidates —
@@ -2711,7 +2815,9 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
   mz_gpu_red *r = new mz_gpu_red();
   r->spec = *spec;
   u32 kw = spec->in.key_words;
-  u64 cap = 1ull << 22;  // 4M keys default
+  u64 cap = 1ull << 22;  // 4M keys default; grows on demand (see
+                         // red_compact_grow); env override for tests/tuning
+  if (const char *e = getenv("MZ_GPU_RED_CAP")) cap = (u64)atoll(e);
   r->capacity = cap;
   u64 slots = 2 * cap;
   r->st.hash = dnew<u64>(ctx, slots * (kw + 1));
@@ -2728,7 +2834,15 @@ mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
   return r;
 }
 
-void mz_gpu_reduce_drop(mz_gpu_ctx *c, mz_gpu_red *r) { (void)c; (void)r; }
+void mz_gpu_reduce_drop(mz_gpu_ctx *c, mz_gpu_red *r) {
+  Ctx *ctx = &c->impl;
+  for (void *p : {(void *)r->st.hash, (void *)r->st.rows,
+                  (void *)r->d_nrows, (void *)r->d_err})
+    dfree(ctx, p);
+  auto &v = ctx->reds;
+  v.erase(std::remove(v.begin(), v.end(), r), v.end());
+  delete r;
+}
 
 // Internal: reduce over already-staged device updates.
 static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
@@ -2787,6 +2901,12 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
   u32 kw = op->spec.in.key_words, vb = op->spec.in.val_bytes;
   u32 okw = op->spec.out.key_words, ovb = op->spec.out.val_bytes;
   u64 n = d.n;
+  if (op->n_rows + n > op->capacity) {
+    // worst case every update opens a new group: reclaim zeroed rows and
+    // grow so the push cannot overflow (n_rows mirror from the last sync)
+    red_compact_grow(ctx, op->st, op->d_nrows, op->n_rows, n, kw, kw);
+    op->capacity = op->st.capacity;
+  }
   if (n == 0) {
     *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1), dnew<u64>(ctx, 1),
                     dnew<i64>(ctx, 1), 0, okw, ovb);
@@ -2989,6 +3109,8 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
                            ctx->stream));
   HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
                            ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(&op->n_rows, op->d_nrows, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
@@ -3013,7 +3135,8 @@ mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
   mz_gpu_thr *r = new mz_gpu_thr();
   r->s = *s;
   r->kw2 = s->key_words + (s->val_bytes + 7) / 8;
-  u64 cap = 1ull << 21;  // 2M (key,val) pairs default
+  u64 cap = 1ull << 21;  // 2M (key,val) pairs default; grows on demand
+  if (const char *e = getenv("MZ_GPU_THR_CAP")) cap = (u64)atoll(e);
   r->capacity = cap;
   u64 slots = 2 * cap;
   r->st.hash = dnew<u64>(ctx, slots * (r->kw2 + 1));
@@ -3030,8 +3153,11 @@ mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
 }
 
 void mz_gpu_threshold_drop(mz_gpu_ctx *c, mz_gpu_thr *r) {
-  (void)c;
-  (void)r;
+  Ctx *ctx = &c->impl;
+  for (void *p : {(void *)r->st.hash, (void *)r->st.rows,
+                  (void *)r->d_nrows, (void *)r->d_err})
+    dfree(ctx, p);
+  delete r;
 }
 
 int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
@@ -3042,6 +3168,10 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
   u32 kw = op->s.key_words, vb = op->s.val_bytes, kw2 = op->kw2;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
   u64 n = d.n;
+  if (op->n_rows + n > op->capacity) {
+    red_compact_grow(ctx, op->st, op->d_nrows, op->n_rows, n, kw2, kw2);
+    op->capacity = op->st.capacity;
+  }
   if (n == 0) {
     *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
                     dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, kw, vb);
@@ -3121,6 +3251,8 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
                            ctx->stream));
   HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
                            ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(&op->n_rows, op->d_nrows, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
@@ -3238,8 +3370,9 @@ mz_gpu_topk *mz_gpu_topk_create(mz_gpu_ctx *c, const mz_gpu_topk_spec *spec) {
 }
 
 void mz_gpu_topk_drop(mz_gpu_ctx *c, mz_gpu_topk *op) {
-  (void)c;
-  (void)op;
+  mz_gpu_arr_drop(c, op->arr);  // owned group-contents arrangement
+  dfree(&c->impl, op->d_err);
+  delete op;
 }
 
 int mz_gpu_topk_push(mz_gpu_ctx *c, mz_gpu_topk *op,
@@ -3695,8 +3828,16 @@ int mz_gpu_minmax_push(mz_gpu_ctx *c, mz_gpu_minmax *op,
 }
 
 void mz_gpu_minmax_drop(mz_gpu_ctx *c, mz_gpu_minmax *op) {
-  (void)c;
-  (void)op;
+  Ctx *ctx = &c->impl;
+  for (auto &lvl : op->levels) {
+    for (auto &b : lvl.batches) free_batch(ctx, b);
+    lvl.batches.clear();
+  }
+  for (auto &st : op->states) {
+    dfree(ctx, st.hash);
+    dfree(ctx, st.rows);
+  }
+  delete op;
 }
 
 // ---- numeric debug probes (test support; not part of the drop-in surface)

@@ -12,9 +12,13 @@ page-locked (pinned) column buffers and a dedicated copy stream:
   land (the engine runs on its own HIP stream, so the hand-off point is
   a host-side event wait — batch t+1's PCIe transfer overlaps batch t's
   compute, which is the part that pays).
-- The ring depth bounds in-flight batches; re-using a slot waits for its
-  previous copy (the engine has consumed the data by then: `stage` →
-  `ready` → push happens in order per slot).
+- Slot lifetime contract: a staged batch's DEVICE buffers stay owned by
+  the feeder slot until `mark_consumed()` is called for it. Synchronous
+  engine pushes (`arr_insert`, `join_push`, ...) consume on return;
+  `arr_insert_async` consumes only at `arr_flush` (mz_gpu.h lifetime
+  note) — call `mark_consumed()` at that point. Re-using a slot whose
+  batch was never marked raises rather than silently overwriting
+  buffers a pending lane insert may still read.
 
 PCIe-inclusive throughput is a different number from the HBM-resident
 bench (`bench.py` stages everything up front per the contract); the
@@ -60,7 +64,9 @@ class PinnedFeeder:
                 dev = {k: torch.empty_like(v, device=device)
                        for k, v in host.items()}
                 self.slots.append({"host": host, "dev": dev,
-                                   "event": torch.cuda.Event()})
+                                   "event": torch.cuda.Event(),
+                                   "inflight": False})
+        self._pending = []  # slot indices staged but not yet consumed
 
     def stage(self, keys, vals, times, diffs, lower, upper):
         """Stage one columnar batch; returns (updates, ready) where
@@ -78,8 +84,17 @@ class PinnedFeeder:
                                  times.view(np.uint64), diffs, lower, upper)
             return u, (lambda: None)
         torch = self.torch
-        s = self.slots[self.slot]
+        idx = self.slot
+        s = self.slots[idx]
         self.slot = (self.slot + 1) % self.depth
+        if s["inflight"]:
+            raise RuntimeError(
+                "PinnedFeeder slot re-use before mark_consumed(): the "
+                "engine may still read this slot's device buffers "
+                "(async inserts hold them until arr_flush — see the "
+                "lifetime contract in the module docstring)")
+        s["inflight"] = True
+        self._pending.append(idx)
         # the slot's previous transfer must have landed before re-packing
         s["event"].synchronize()
         h, d = s["host"], s["dev"]
@@ -102,3 +117,14 @@ class PinnedFeeder:
             d["vals"][:n * self.vb] if self.vb else None,
             d["times"][:n], d["diffs"][:n], lower, upper)
         return u, s["event"].synchronize
+
+    def mark_consumed(self):
+        """The engine has consumed the OLDEST outstanding staged batch:
+        after a synchronous push returns, or after `arr_flush` for
+        batches handed to `arr_insert_async`. Frees that slot for
+        re-use."""
+        if not self.gpu:
+            return
+        if not self._pending:
+            raise RuntimeError("mark_consumed() with no staged batch")
+        self.slots[self._pending.pop(0)]["inflight"] = False

@@ -52,6 +52,7 @@ def test_pinned_path_matches_oracle():
         u, ready = fd.stage(keys, vals, times, diffs, t, t + 1)
         ready()
         g.arr_insert(ga, u)
+        fd.mark_consumed()  # synchronous push: consumed on return
         o.arr_insert(oa, abi.make_updates(keys, vals, times, diffs,
                                           t, t + 1))
         pk = rng.integers(0, 50, (300, 1)).astype(np.int64)
@@ -63,3 +64,24 @@ def test_pinned_path_matches_oracle():
             np.testing.assert_array_equal(x, y)
     g.close()
     o.close()
+
+@pytest.mark.gpu
+def test_slot_reuse_without_consume_raises():
+    """Lifetime contract: staging past the ring depth without
+    mark_consumed() must raise, not overwrite in-flight buffers."""
+    from materialize_amd._ffi import GpuCtx
+    rng = np.random.default_rng(7)
+    sch = abi.schema(1, 8)
+    g = GpuCtx()
+    fd = PinnedFeeder(sch, capacity_rows=1024, device="cuda:0", depth=2)
+    assert fd.gpu
+    for t in range(2):
+        keys, vals, times, diffs = batch(rng, 100, t)
+        fd.stage(keys, vals, times, diffs, t, t + 1)
+    keys, vals, times, diffs = batch(rng, 100, 2)
+    with pytest.raises(RuntimeError):
+        fd.stage(keys, vals, times, diffs, 2, 3)
+    fd.mark_consumed()
+    u, ready = fd.stage(keys, vals, times, diffs, 2, 3)  # now fits
+    ready()
+    g.close()
