@@ -86,13 +86,13 @@ class PinnedFeeder:
         torch = self.torch
         idx = self.slot
         s = self.slots[idx]
-        self.slot = (self.slot + 1) % self.depth
         if s["inflight"]:
             raise RuntimeError(
                 "PinnedFeeder slot re-use before mark_consumed(): the "
                 "engine may still read this slot's device buffers "
                 "(async inserts hold them until arr_flush — see the "
                 "lifetime contract in the module docstring)")
+        self.slot = (self.slot + 1) % self.depth
         s["inflight"] = True
         self._pending.append(idx)
         # the slot's previous transfer must have landed before re-packing
