@@ -474,75 +474,86 @@ struct BatchList {
 
 enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 
-// Phase 1: count emitted pairs per (delta update, batch) — one thread per
-// pair, idx = bi*n + i, so consecutive threads read consecutive delta
-// rows (coalesced) and each thread does ONE independent hash lookup
-// (10x the memory-level parallelism of the per-row batch loop on a
-// pooled spine; the probe is random-line bound, MI355X_MICROARCH §HBM).
-// delta: n updates (keys/vals/times/diffs columns); swap: delta is input 2
-// of a linear join (closure arg order is (key, v1=input1, v2=input2)).
-// ki_cache[bl.n * n] stores each (batch, row) hash-lookup result so the
-// emit phase never re-reads the hash tables (≈half the probe traffic).
-__global__ void k_probe_count(const u64 *dkeys, const u8 *dvals, u32 dvb,
-                              const u64 *dtimes, u64 n, u32 kw, u32 lvb,
-                              BatchList bl, int mode, int swap,
-                              const mz_gpu_closure cl, u32 *count,
-                              u64 *ki_cache) {
+// Single-walk probe (DESIGN §9 candidate (b), the half_join2 probe loop
+// replacement, delta_join.rs:500,544): one kernel probes AND emits —
+// each (row, batch) pair counts its matches from the just-read range
+// (L2-hot), reserves a slice of the output queue with one wave-
+// aggregated atomic, and writes. Halves the probe pair's HBM traffic
+// versus count+scan+emit: the delta tuple, hash line, val range and upd
+// ranges are each touched once cold. Output order is queue order (non-
+// deterministic); every consumer consolidates (canonical sort), so
+// results are unchanged — parity holds on consolidated outputs.
+// *ctr always accumulates the EXACT total match count; writes are
+// skipped when a slice would cross `cap`, and the host relaunches with
+// cap = exact count on overflow (rare; capacity hint kept per
+// arrangement).
+__global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                             const u64 *dtimes, const i64 *ddiffs, u64 n,
+                             u32 kw, u32 lvb, BatchList bl, int mode,
+                             int swap, const mz_gpu_closure cl, u64 cap,
+                             unsigned long long *ctr, u64 *okeys, u8 *ovals,
+                             u64 *otimes, i64 *odiffs) {
+  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   u64 total = n * (u64)bl.n;
-  GRID_STRIDE(idx, total) {
-    u64 i = idx % n;
-    int bi = (int)(idx / n);
-    const u64 *key = dkeys + i * kw;
-    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-    u64 t = dtimes[i];
+  u64 stride = (u64)gridDim.x * blockDim.x;
+  u64 start = blockIdx.x * (u64)blockDim.x + threadIdx.x;
+  u64 iters = (total + stride - 1) / stride;  // uniform across the wave
+  u32 lane = threadIdx.x & 63;
+  for (u64 it = 0; it < iters; it++) {
+    u64 idx = start + it * stride;
+    bool active = idx < total;
+    u64 kvr = ~0ull;
+    u64 i = 0, t = 0;
+    int bi = 0;
+    const u64 *key = nullptr;
+    const u8 *dv = nullptr;
     u32 c = 0;
-    const DevBatch &b = bl.b[bi];
-    // the val range rides in the widened slot: one random line per probe
-    u64 kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
-    ki_cache[idx] = kvr;
-    if (kvr != ~0ull) {
-      for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
-        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-        const u8 *v1 = swap ? lv : dv;
-        const u8 *v2 = swap ? dv : lv;
-        if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
-        u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
-        if (mode == PM_JOIN || bl.allpass[bi]) {
-          c += hi - lo;
-        } else {
-          for (u32 u = lo; u < hi; u++) {
-            u64 t2 = b.times[u];
-            c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+    if (active) {
+      i = idx % n;
+      bi = (int)(idx / n);
+      key = dkeys + i * kw;
+      dv = dvals ? dvals + i * dvb : nullptr;
+      t = dtimes[i];
+      const DevBatch &b = bl.b[bi];
+      kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
+      if (kvr != ~0ull) {
+        for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
+          const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+          const u8 *v1 = swap ? lv : dv;
+          const u8 *v2 = swap ? dv : lv;
+          if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr))
+            continue;
+          u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+          if (mode == PM_JOIN || bl.allpass[bi]) {
+            c += hi - lo;
+          } else {
+            for (u32 u = lo; u < hi; u++) {
+              u64 t2 = b.times[u];
+              c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+            }
           }
         }
       }
     }
-    count[idx] = c;
-  }
-}
-
-// Phase 2: emit pairs at offsets (deterministic: offset is a function of
-// the (batch, row) index).
-__global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
-                             const u64 *dtimes, const i64 *ddiffs, u64 n,
-                             u32 kw, u32 lvb, BatchList bl, int mode,
-                             int swap, const mz_gpu_closure cl,
-                             const u32 *offs /*exclusive*/,
-                             const u64 *ki_cache, u64 *okeys, u8 *ovals,
-                             u64 *otimes, i64 *odiffs) {
-  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
-  u64 total = n * (u64)bl.n;
-  GRID_STRIDE(idx, total) {
-    u64 kvr = ki_cache[idx];
-    if (kvr == ~0ull) continue;
-    u64 i = idx % n;
-    int bi = (int)(idx / n);
-    const u64 *key = dkeys + i * kw;
-    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-    u64 t = dtimes[i];
+    // wave-aggregated queue reservation: exclusive prefix of c across the
+    // wavefront, one atomicAdd per wave (all 64 lanes participate)
+    u32 pre = c;
+    for (int d = 1; d < 64; d <<= 1) {
+      u32 up = __shfl_up(pre, d, 64);
+      if ((int)lane >= d) pre += up;
+    }
+    u32 excl = pre - c;
+    unsigned long long wtotal = (unsigned long long)__shfl((int64_t)pre,
+                                                           63, 64);
+    long long basell = 0;
+    if (lane == 63 && wtotal)
+      basell = (long long)atomicAdd(ctr, wtotal);
+    u64 base = (u64)__shfl((int64_t)basell, 63, 64) + excl;
+    if (!active || c == 0 || base + c > cap) continue;
+    // second walk emits from L2-hot lines
     i64 d1 = ddiffs[i];
-    u64 o = offs[idx];
     const DevBatch &b = bl.b[bi];
+    u64 o = base;
     for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
       const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
       const u8 *v1 = swap ? lv : dv;
@@ -553,7 +564,7 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
       for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
         u64 tout;
         if (bl.allpass[bi]) {
-          tout = t;  // t2 < delta lower <= t on every update
+          tout = t;
         } else if (mode == PM_JOIN) {
           u64 t2 = b.times[u];
           tout = t2 > t ? t2 : t;
@@ -563,7 +574,7 @@ __global__ void k_probe_emit(const u64 *dkeys, const u8 *dvals, u32 dvb,
           tout = t;
         }
         for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
-        for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oval[c];
+        for (u32 cc = 0; cc < ovb; cc++) ovals[o * ovb + cc] = oval[cc];
         otimes[o] = tout;
         odiffs[o] = wmul(d1, b.diffs[u]);
         o++;
@@ -1245,6 +1256,8 @@ struct mz_gpu_arr {
   u64 logical_compaction = 0;
   u64 physical_compaction = 0;
   u64 upper = 0;
+  u64 probe_cap_hint = 0;  // last probe's ceil(matches/row): sizes the
+                           // single-walk output queue (k_probe_walk)
   Ctx *ctx = nullptr;
   // Per-arrangement lane: inserts/merges run on this stream with this
   // scratch arena so independent arrangements' maintenance overlaps;
@@ -2698,37 +2711,56 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     return 0;
   }
   u64 nb2 = n * (u64)bl.n;
-  u32 *count = (u32 *)S.get((nb2 + 1) * 4);
-  u64 *ki_cache = (u64 *)S.get(nb2 * 8);
+  // Single-walk probe: allocate the output queue from the arrangement's
+  // emit-ratio hint, relaunch once with the exact count on overflow.
+  u64 cap = (lookup->probe_cap_hint ? lookup->probe_cap_hint + 1 : 2) * n +
+            1024;
+  unsigned long long *ctr = (unsigned long long *)S.get(8);
+  fill_u64(ctx, (u64 *)ctr, 1, 0);
+  u64 *pk = dnew<u64>(ctx, cap * okw);
+  u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
+  u64 *pt = dnew<u64>(ctx, cap);
+  i64 *pd = dnew<i64>(ctx, cap);
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
-  hipLaunchKernelGGL(k_probe_count, dim3(ngrid(nb2)), dim3(BLK), 0,
-                     ctx->stream, d.keys, d.vals, stream_vb, d.times, n, kw,
-                     lvb, bl, mode, swap, *cl, count, ki_cache);
-  u32 *offs = (u32 *)S.get((nb2 + 1) * 4);
-  u64 M = exclusive_scan_u32(ctx, count, offs, nb2);
-  u64 *pk = dnew<u64>(ctx, std::max<u64>(M, 1) * okw);
-  u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(M * ovb, 1));
-  u64 *pt = dnew<u64>(ctx, std::max<u64>(M, 1));
-  i64 *pd = dnew<i64>(ctx, std::max<u64>(M, 1));
-  if (M)
-    hipLaunchKernelGGL(k_probe_emit, dim3(ngrid(nb2)), dim3(BLK), 0,
+  hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
+                     ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                     d.diffs, n, kw, lvb, bl, mode, swap, *cl, cap, ctr, pk,
+                     pv, pt, pd);
+  if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+  unsigned long long M = 0;
+  HIP_CHECK(hipMemcpyAsync(&M, ctr, 8, hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  u64 launches = 1;
+  if (M > cap) {  // rare: queue overflow — exact-size relaunch
+    for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
+      dfree(ctx, p);
+    cap = M;
+    pk = dnew<u64>(ctx, cap * okw);
+    pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
+    pt = dnew<u64>(ctx, cap);
+    pd = dnew<i64>(ctx, cap);
+    fill_u64(ctx, (u64 *)ctr, 1, 0);
+    if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+    hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
                        ctx->stream, d.keys, d.vals, stream_vb, d.times,
-                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, offs,
-                       ki_cache, pk, pv, pt, pd);
+                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, cap, ctr,
+                       pk, pv, pt, pd);
+    if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+    launches = 2;
+  }
+  lookup->probe_cap_hint = (M + n - 1) / n;
   if (ctx->time_kernels) {
-    HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
     float ms = 0;
     HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_a, ctx->ev_b));
     ctx->probe_ms += ms;
     ctx->probe_rows += n;
-    ctx->probe_launches += 2;
+    ctx->probe_launches += launches;
     ctx->probe_pairs += M;
     ctx->probe_batches += (u64)bl.n;
-    // Algorithmic bytes of the probe pair (SURVEY §8d model): delta tuple
-    // + one hash line per probed batch + matched val+upd read + output
-    // write. Counted ONCE for the count+emit pair — the 2-phase re-read is
-    // implementation overhead and must show as a lower roofline fraction.
+    // Algorithmic bytes of the probe (SURVEY §8d model): delta tuple +
+    // one hash line per probed batch + matched val+upd read + output
+    // write — each touched once cold in the single walk.
     ctx->probe_alg_bytes +=
         n * (8ull * kw + stream_vb + 16) + n * 128ull * (u64)bl.n +
         M * (lvb + 16ull) + M * (8ull * okw + ovb + 16);
