@@ -327,20 +327,24 @@ __global__ void k_scatter_structure(const u64 *keys, u32 kw, const u8 *vals,
 }
 
 
+// Compressed batch slot: [key words][range word], stride kw+1 (16 B for
+// 1-word keys -> 4 slots per 64 B line). The packed val range
+// kv_lo|kv_hi<<32 doubles as the occupancy sentinel (~0 is impossible:
+// kv_hi > kv_lo and n_vals < 2^32), so the CAS publishes the range
+// directly and no separate idx word is needed.
 __global__ void k_hash_build(u64 *hash, u64 slots, const u64 *keys, u32 kw,
                              const u32 *kid, const u32 *kv_off, u64 cap,
                              const u64 *dcounts) {
   u64 n_keys = (cap && dcounts[0]) ? kid[cap - 1] : 0;
   GRID_STRIDE(i, n_keys) {
     u64 h = route_hash(keys + i * kw, kw) & (slots - 1);
+    u64 range = (u64)kv_off[i] | ((u64)kv_off[i + 1] << 32);
     for (;;) {
-      u64 *slot = hash + h * (kw + 2);
-      unsigned long long expected = ~0ull;
-      unsigned long long got = atomicCAS(
-          (unsigned long long *)(slot + kw), expected, (unsigned long long)i);
+      u64 *slot = hash + h * (kw + 1);
+      unsigned long long got = atomicCAS((unsigned long long *)(slot + kw),
+                                         ~0ull, (unsigned long long)range);
       if (got == ~0ull) {
         for (u32 w = 0; w < kw; w++) slot[w] = keys[i * kw + w];
-        slot[kw + 1] = (u64)kv_off[i] | ((u64)kv_off[i + 1] << 32);
         break;
       }
       h = (h + 1) & (slots - 1);
@@ -372,12 +376,12 @@ __device__ __forceinline__ u64 hash_lookup_range(const u64 *hash, u64 slots,
   if (slots == 0) return ~0ull;
   u64 h = route_hash(key, kw) & (slots - 1);
   for (;;) {
-    const u64 *slot = hash + h * (kw + 2);
-    u64 iw = slot[kw];
+    const u64 *slot = hash + h * (kw + 1);
+    u64 iw = slot[kw];  // packed range; ~0 = empty
     if (iw == ~0ull) return ~0ull;
     bool eq = true;
     for (u32 w = 0; w < kw; w++) eq &= slot[w] == key[w];
-    if (eq) return slot[kw + 1];
+    if (eq) return iw;
     h = (h + 1) & (slots - 1);
   }
 }
@@ -487,6 +491,7 @@ enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 // skipped when a slice would cross `cap`, and the host relaunches with
 // cap = exact count on overflow (rare; capacity hint kept per
 // arrangement).
+#define PROBE_TILE 4
 __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              const u64 *dtimes, const i64 *ddiffs, u64 n,
                              u32 kw, u32 lvb, BatchList bl, int mode,
@@ -499,85 +504,109 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
   u64 start = blockIdx.x * (u64)blockDim.x + threadIdx.x;
   u64 iters = (total + stride - 1) / stride;  // uniform across the wave
   u32 lane = threadIdx.x & 63;
-  for (u64 it = 0; it < iters; it++) {
-    u64 idx = start + it * stride;
-    bool active = idx < total;
-    u64 kvr = ~0ull;
-    u64 i = 0, t = 0;
-    int bi = 0;
-    const u64 *key = nullptr;
-    const u8 *dv = nullptr;
-    u32 c = 0;
-    if (active) {
-      i = idx % n;
-      bi = (int)(idx / n);
-      key = dkeys + i * kw;
-      dv = dvals ? dvals + i * dvb : nullptr;
-      t = dtimes[i];
+  // PROBE_TILE pairs per macro-iteration: phase A issues up to 4
+  // independent random hash-line reads per thread (MLP), phases B/C walk
+  // the L2-hot ranges; ONE wave-aggregated reservation per tile keeps
+  // lockstep convergence off the per-pair critical path.
+  for (u64 it0 = 0; it0 < iters; it0 += PROBE_TILE) {
+    u64 kvr[PROBE_TILE];
+    u32 cc[PROBE_TILE];
+    // Phase A: hash lookups (independent random lines in flight)
+#pragma unroll
+    for (int tt = 0; tt < PROBE_TILE; tt++) {
+      u64 idx = start + (it0 + tt) * stride;
+      kvr[tt] = ~0ull;
+      cc[tt] = 0;
+      if (it0 + tt < iters && idx < total) {
+        int bi = (int)(idx / n);
+        const DevBatch &b = bl.b[bi];
+        kvr[tt] = hash_lookup_range(b.hash, b.hash_slots,
+                                    dkeys + (idx % n) * kw, kw);
+      }
+    }
+    // Phase B: count matches per slot (filters applied)
+    u32 csum = 0;
+#pragma unroll
+    for (int tt = 0; tt < PROBE_TILE; tt++) {
+      if (kvr[tt] == ~0ull) continue;
+      u64 idx = start + (it0 + tt) * stride;
+      u64 i = idx % n;
+      int bi = (int)(idx / n);
       const DevBatch &b = bl.b[bi];
-      kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
-      if (kvr != ~0ull) {
-        for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
-          const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-          const u8 *v1 = swap ? lv : dv;
-          const u8 *v2 = swap ? dv : lv;
-          if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr))
-            continue;
-          u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
-          if (mode == PM_JOIN || bl.allpass[bi]) {
-            c += hi - lo;
-          } else {
-            for (u32 u = lo; u < hi; u++) {
-              u64 t2 = b.times[u];
-              c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
-            }
+      const u64 *key = dkeys + i * kw;
+      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+      u64 t = dtimes[i];
+      u32 c = 0;
+      for (u32 j = (u32)kvr[tt]; j < (u32)(kvr[tt] >> 32); j++) {
+        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+        const u8 *v1 = swap ? lv : dv;
+        const u8 *v2 = swap ? dv : lv;
+        if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
+        u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+        if (mode == PM_JOIN || bl.allpass[bi]) {
+          c += hi - lo;
+        } else {
+          for (u32 u = lo; u < hi; u++) {
+            u64 t2 = b.times[u];
+            c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
           }
         }
       }
+      cc[tt] = c;
+      csum += c;
     }
-    // wave-aggregated queue reservation: exclusive prefix of c across the
-    // wavefront, one atomicAdd per wave (all 64 lanes participate)
-    u32 pre = c;
+    // wave-aggregated queue reservation: exclusive prefix of csum across
+    // the wavefront, one atomicAdd per wave (all 64 lanes participate)
+    u32 pre = csum;
     for (int d = 1; d < 64; d <<= 1) {
       u32 up = __shfl_up(pre, d, 64);
       if ((int)lane >= d) pre += up;
     }
-    u32 excl = pre - c;
-    unsigned long long wtotal = (unsigned long long)__shfl((int64_t)pre,
-                                                           63, 64);
+    u32 excl = pre - csum;
+    unsigned long long wtotal = (unsigned long long)(u32)__shfl((int)pre,
+                                                                63, 64);
     long long basell = 0;
     if (lane == 63 && wtotal)
       basell = (long long)atomicAdd(ctr, wtotal);
     u64 base = (u64)__shfl((int64_t)basell, 63, 64) + excl;
-    if (!active || c == 0 || base + c > cap) continue;
-    // second walk emits from L2-hot lines
-    i64 d1 = ddiffs[i];
-    const DevBatch &b = bl.b[bi];
+    if (csum == 0 || base + csum > cap) continue;
+    // Phase C: emit from L2-hot lines, fields written straight to the
+    // reserved global slots (closure filters run before any field write)
     u64 o = base;
-    for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
-      const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-      const u8 *v1 = swap ? lv : dv;
-      const u8 *v2 = swap ? dv : lv;
-      u64 okey[MAX_KW];
-      u8 oval[MAX_VB];
-      if (!d_closure_apply(&cl, key, v1, v2, okey, oval)) continue;
-      for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
-        u64 tout;
-        if (bl.allpass[bi]) {
-          tout = t;
-        } else if (mode == PM_JOIN) {
-          u64 t2 = b.times[u];
-          tout = t2 > t ? t2 : t;
-        } else {
-          u64 t2 = b.times[u];
-          if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
-          tout = t;
+#pragma unroll
+    for (int tt = 0; tt < PROBE_TILE; tt++) {
+      if (cc[tt] == 0) continue;
+      u64 idx = start + (it0 + tt) * stride;
+      u64 i = idx % n;
+      int bi = (int)(idx / n);
+      const DevBatch &b = bl.b[bi];
+      const u64 *key = dkeys + i * kw;
+      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+      u64 t = dtimes[i];
+      i64 d1 = ddiffs[i];
+      for (u32 j = (u32)kvr[tt]; j < (u32)(kvr[tt] >> 32); j++) {
+        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+        const u8 *v1 = swap ? lv : dv;
+        const u8 *v2 = swap ? dv : lv;
+        for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
+          u64 tout;
+          if (bl.allpass[bi]) {
+            tout = t;
+          } else if (mode == PM_JOIN) {
+            u64 t2 = b.times[u];
+            tout = t2 > t ? t2 : t;
+          } else {
+            u64 t2 = b.times[u];
+            if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
+            tout = t;
+          }
+          if (!d_closure_apply(&cl, key, v1, v2, okeys + o * okw,
+                               ovals + o * ovb))
+            continue;
+          otimes[o] = tout;
+          odiffs[o] = wmul(d1, b.diffs[u]);
+          o++;
         }
-        for (u32 w = 0; w < okw; w++) okeys[o * okw + w] = okey[w];
-        for (u32 cc = 0; cc < ovb; cc++) ovals[o * ovb + cc] = oval[cc];
-        otimes[o] = tout;
-        odiffs[o] = wmul(d1, b.diffs[u]);
-        o++;
       }
     }
   }
@@ -603,6 +632,17 @@ struct RowLess {
     return times[a] < times[b];
   }
 };
+
+// Bucket of each delta row in a batch's hash-table slot order: the top 8
+// bits of its slot index. Probing in this order walks the table
+// ~sequentially (ascending slot windows), so a table far larger than one
+// XCD's L2 streams instead of thrashing (hash-order probing, see
+// probe_impl).
+__global__ void k_probe_bucket(const u64 *keys, u32 kw, u64 n, u64 slots,
+                               u32 shift, u32 *bucket) {
+  GRID_STRIDE(i, n)
+  bucket[i] = (u32)((route_hash(keys + i * kw, kw) & (slots - 1)) >> shift);
+}
 
 // expansion: flatten a DevBatch back to per-update (key,val,time,diff)
 __global__ void k_expand_batch(DevBatch b, u32 kw, u32 vb, u64 frontier,
@@ -1975,10 +2015,10 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   u64 slots = 16;
   while (slots < 2 * cap) slots <<= 1;
   b.hash_slots = slots;
-  b.hash = dnew<u64>(c, slots * (kw + 2));
+  b.hash = dnew<u64>(c, slots * (kw + 1));
   // full-line 0xFF fill: lookups check the idx-word sentinel before key
   // compares, so poisoned key words are never read
-  fill_u64(c, b.hash, slots * (kw + 2), ~0ull);
+  fill_u64(c, b.hash, slots * (kw + 1), ~0ull);
   hipLaunchKernelGGL(k_hash_build, dim3(ngrid(cap)), dim3(BLK), 0,
                      c->stream, b.hash, slots, b.keys, kw, kid, b.kv_off,
                      cap, dcounts);
@@ -2598,7 +2638,7 @@ int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
   for (auto &b : a->batches) {
     n += b.n_upds;
     by += b.n_keys * kw * 8 + b.n_vals * vb + b.n_upds * 16 +
-          b.hash_slots * (kw + 2) * 8 + (b.n_keys + b.n_vals) * 4 +
+          b.hash_slots * (kw + 1) * 8 + (b.n_keys + b.n_vals) * 4 +
           b.n_upds * 4 + b.n_vals * 4;
   }
   *n_updates = n;
@@ -2709,6 +2749,65 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1), dnew<u64>(ctx, 1),
                     dnew<i64>(ctx, 1), 0, okw, ovb);
     return 0;
+  }
+  // Hash-order probing: when the largest table in the list exceeds the
+  // per-XCD L2 (32 MB), reorder the delta by that table's slot order so
+  // concurrent waves probe one ascending table window at a time —
+  // line-sequential table reads with an L2-resident working set instead
+  // of uniform random 128 B lines (the measured random ceiling was
+  // ~1.1 TB/s; ordered windows stream).
+  static const u64 ORDER_MIN = [] {
+    const char *e = getenv("MZ_PROBE_ORDER_MIN_MB");
+    return (u64)(e ? atoll(e) : 24) << 20;
+  }();
+  static const bool ORDER_EN = [] {
+    const char *e = getenv("MZ_PROBE_ORDER");
+    return !(e && e[0] == '0');
+  }();
+  u64 big_bytes = 0;
+  int big = -1;
+  for (int b2 = 0; b2 < bl.n; b2++) {
+    u64 tb = bl.b[b2].hash_slots * (kw + 1) * 8;
+    if (tb > big_bytes) {
+      big_bytes = tb;
+      big = b2;
+    }
+  }
+  if (ORDER_EN && big >= 0 && big_bytes > ORDER_MIN && n >= 32768) {
+    u64 slots = bl.b[big].hash_slots;
+    u32 shift = 0;
+    while ((slots >> shift) > 256) shift++;
+    u32 *bucket = (u32 *)S.get(n * 4);
+    u32 *bucket_o = (u32 *)S.get(n * 4);
+    u32 *io = (u32 *)S.get(n * 4);
+    u32 *ord = (u32 *)S.get(n * 4);
+    hipLaunchKernelGGL(k_probe_bucket, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.keys, kw, n, slots, shift, bucket);
+    hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
+                       io, n);
+    size_t need = 0;
+    (void)rocprim::radix_sort_pairs(nullptr, need, bucket, bucket_o, io,
+                                    ord, (unsigned)n, 0, 8, ctx->stream);
+    void *tmp = S.get(need);
+    (void)rocprim::radix_sort_pairs(tmp, need, bucket, bucket_o, io, ord,
+                                    (unsigned)n, 0, 8, ctx->stream);
+    u64 *gk = (u64 *)S.get(n * kw * 8);
+    u8 *gv = stream_vb ? (u8 *)S.get(n * stream_vb) : nullptr;
+    u64 *gt = (u64 *)S.get(n * 8);
+    i64 *gd = (i64 *)S.get(n * 8);
+    hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.keys, kw, ord, gk, n);
+    if (stream_vb)
+      hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
+                         ctx->stream, d.vals, stream_vb, ord, gv, n);
+    hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.times, ord, gt, n);
+    hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0,
+                       ctx->stream, d.diffs, ord, gd, n);
+    d.keys = gk;
+    d.vals = gv;
+    d.times = gt;
+    d.diffs = gd;
   }
   u64 nb2 = n * (u64)bl.n;
   // Single-walk probe: allocate the output queue from the arrangement's
