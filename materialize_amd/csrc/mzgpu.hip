@@ -633,17 +633,6 @@ struct RowLess {
   }
 };
 
-// Bucket of each delta row in a batch's hash-table slot order: the top 8
-// bits of its slot index. Probing in this order walks the table
-// ~sequentially (ascending slot windows), so a table far larger than one
-// XCD's L2 streams instead of thrashing (hash-order probing, see
-// probe_impl).
-__global__ void k_probe_bucket(const u64 *keys, u32 kw, u64 n, u64 slots,
-                               u32 shift, u32 *bucket) {
-  GRID_STRIDE(i, n)
-  bucket[i] = (u32)((route_hash(keys + i * kw, kw) & (slots - 1)) >> shift);
-}
-
 // expansion: flatten a DevBatch back to per-update (key,val,time,diff)
 __global__ void k_expand_batch(DevBatch b, u32 kw, u32 vb, u64 frontier,
                                u64 *okeys, u8 *ovals, u64 *otimes,
@@ -2749,65 +2738,6 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1), dnew<u64>(ctx, 1),
                     dnew<i64>(ctx, 1), 0, okw, ovb);
     return 0;
-  }
-  // Hash-order probing: when the largest table in the list exceeds the
-  // per-XCD L2 (32 MB), reorder the delta by that table's slot order so
-  // concurrent waves probe one ascending table window at a time —
-  // line-sequential table reads with an L2-resident working set instead
-  // of uniform random 128 B lines (the measured random ceiling was
-  // ~1.1 TB/s; ordered windows stream).
-  static const u64 ORDER_MIN = [] {
-    const char *e = getenv("MZ_PROBE_ORDER_MIN_MB");
-    return (u64)(e ? atoll(e) : 24) << 20;
-  }();
-  static const bool ORDER_EN = [] {
-    const char *e = getenv("MZ_PROBE_ORDER");
-    return !(e && e[0] == '0');
-  }();
-  u64 big_bytes = 0;
-  int big = -1;
-  for (int b2 = 0; b2 < bl.n; b2++) {
-    u64 tb = bl.b[b2].hash_slots * (kw + 1) * 8;
-    if (tb > big_bytes) {
-      big_bytes = tb;
-      big = b2;
-    }
-  }
-  if (ORDER_EN && big >= 0 && big_bytes > ORDER_MIN && n >= 32768) {
-    u64 slots = bl.b[big].hash_slots;
-    u32 shift = 0;
-    while ((slots >> shift) > 256) shift++;
-    u32 *bucket = (u32 *)S.get(n * 4);
-    u32 *bucket_o = (u32 *)S.get(n * 4);
-    u32 *io = (u32 *)S.get(n * 4);
-    u32 *ord = (u32 *)S.get(n * 4);
-    hipLaunchKernelGGL(k_probe_bucket, dim3(ngrid(n)), dim3(BLK), 0,
-                       ctx->stream, d.keys, kw, n, slots, shift, bucket);
-    hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, ctx->stream,
-                       io, n);
-    size_t need = 0;
-    (void)rocprim::radix_sort_pairs(nullptr, need, bucket, bucket_o, io,
-                                    ord, (unsigned)n, 0, 8, ctx->stream);
-    void *tmp = S.get(need);
-    (void)rocprim::radix_sort_pairs(tmp, need, bucket, bucket_o, io, ord,
-                                    (unsigned)n, 0, 8, ctx->stream);
-    u64 *gk = (u64 *)S.get(n * kw * 8);
-    u8 *gv = stream_vb ? (u8 *)S.get(n * stream_vb) : nullptr;
-    u64 *gt = (u64 *)S.get(n * 8);
-    i64 *gd = (i64 *)S.get(n * 8);
-    hipLaunchKernelGGL(k_gather_keyrows, dim3(ngrid(n)), dim3(BLK), 0,
-                       ctx->stream, d.keys, kw, ord, gk, n);
-    if (stream_vb)
-      hipLaunchKernelGGL(k_gather_valrows, dim3(ngrid(n)), dim3(BLK), 0,
-                         ctx->stream, d.vals, stream_vb, ord, gv, n);
-    hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0,
-                       ctx->stream, d.times, ord, gt, n);
-    hipLaunchKernelGGL(k_gather_i64, dim3(ngrid(n)), dim3(BLK), 0,
-                       ctx->stream, d.diffs, ord, gd, n);
-    d.keys = gk;
-    d.vals = gv;
-    d.times = gt;
-    d.diffs = gd;
   }
   u64 nb2 = n * (u64)bl.n;
   // Single-walk probe: allocate the output queue from the arrangement's
