@@ -54,6 +54,12 @@ typedef struct {
   uint64_t n;
   uint64_t lower, upper;  /* batch time bounds [lower, upper)              */
   int32_t  on_device;
+  /* 1 when rows are already sorted ascending by (key, val, time) in the
+   * canonical order (the form mz_gpu_consolidate produces). Purely an
+   * optimization hint: sorted delta streams probe large arrangements by
+   * merge scan (streaming both sorted sides) instead of per-row hash
+   * lookups. 0 is always safe. */
+  int32_t  sorted;
 } mz_gpu_updates;
 
 /* ----------------------------------------------------------- closures

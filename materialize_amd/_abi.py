@@ -42,6 +42,10 @@ class Updates(C.Structure):
         ("lower", C.c_uint64),
         ("upper", C.c_uint64),
         ("on_device", C.c_int32),
+        # sorted=1: rows already in canonical (key, val, time) order (the
+        # consolidate output form) -> large-table probes take the
+        # streaming merge path instead of hash lookups. 0 always safe.
+        ("sorted", C.c_int32),
     ]
 
 
@@ -147,7 +151,8 @@ def _as_u64(a):
     return np.ascontiguousarray(a, dtype=np.uint64)
 
 
-def make_updates(keys, vals, times, diffs, lower, upper, on_device=0):
+def make_updates(keys, vals, times, diffs, lower, upper, on_device=0,
+                 sorted=0):
     """Build an Updates descriptor over numpy arrays (host memory).
 
     keys: int64/uint64 array of n*key_words; vals: uint8 array of
@@ -172,6 +177,7 @@ def make_updates(keys, vals, times, diffs, lower, upper, on_device=0):
     u.lower = lower
     u.upper = upper
     u.on_device = on_device
+    u.sorted = sorted
     u._refs = (keys, vals, times, diffs)
     return u
 

@@ -304,9 +304,9 @@ class GpuCtx:
 
     # --- device-resident variants (outputs stay on the GPU; used by the
     # render layer to chain stages without host round trips) ---
-    def _dev_out(self, outp):
+    def _dev_out(self, outp, sorted=False):
         from .render import DevOut
-        return DevOut(self, outp)
+        return DevOut(self, outp, sorted=sorted)
 
     def _take_copy(self, outp):
         """Copy an OutBatch to host WITHOUT releasing it."""
@@ -331,7 +331,7 @@ class GpuCtx:
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_consolidate(self.ctx, C.byref(sch),
                                                 C.byref(upd), C.byref(outp)))
-        return self._dev_out(outp)
+        return self._dev_out(outp, sorted=True)
 
     def halfjoin_dev(self, lookup, upd, stream_vb, le, cl):
         # raw (unconsolidated) output: the render layer's consumers —
@@ -362,26 +362,26 @@ class GpuCtx:
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_join_push(self.ctx, op, side,
                                               C.byref(upd), C.byref(outp)))
-        return self._dev_out(outp)
+        return self._dev_out(outp, sorted=True)  # consolidated output
 
     def threshold_push_dev(self, op, upd):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_threshold_push(self.ctx, op,
                                                    C.byref(upd),
                                                    C.byref(outp)))
-        return self._dev_out(outp)
+        return self._dev_out(outp, sorted=True)  # consolidated output
 
     def reduce_push_dev(self, op, upd):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_reduce_push(self.ctx, op, C.byref(upd),
                                                 C.byref(outp)))
-        return self._dev_out(outp)
+        return self._dev_out(outp, sorted=True)  # consolidated output
 
     def reduce_push2_dev(self, op, u1, u2):
         outp = C.POINTER(OutBatch)()
         self._check(self.lib.mz_gpu_reduce_push2(
             self.ctx, op, C.byref(u1), C.byref(u2), C.byref(outp)))
-        return self._dev_out(outp)
+        return self._dev_out(outp, sorted=True)  # consolidated output
 
     def set_kernel_timing(self, on):
         self.lib.mz_gpu_set_kernel_timing(self.ctx, 1 if on else 0)

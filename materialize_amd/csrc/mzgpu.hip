@@ -478,6 +478,98 @@ struct BatchList {
 
 enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 
+// Count the output rows of one (delta row, batch) pair: walk the key's
+// val range applying the closure filter and the time mode.
+__device__ __forceinline__ u32 d_count_pair(const DevBatch &b, int allpass,
+                                            const u64 *key, const u8 *dv,
+                                            u64 t, u64 kvr, int mode,
+                                            int swap,
+                                            const mz_gpu_closure &cl,
+                                            u32 lvb) {
+  u32 c = 0;
+  for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
+    const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+    const u8 *v1 = swap ? lv : dv;
+    const u8 *v2 = swap ? dv : lv;
+    if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
+    u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+    if (mode == PM_JOIN || allpass) {
+      c += hi - lo;
+    } else {
+      for (u32 u = lo; u < hi; u++) {
+        u64 t2 = b.times[u];
+        c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+      }
+    }
+  }
+  return c;
+}
+
+// Emit the counted rows at queue offset o (fields written straight to
+// global — closure filters run before any field write). Returns new o.
+__device__ __forceinline__ u64 d_emit_pair(const DevBatch &b, int allpass,
+                                           const u64 *key, const u8 *dv,
+                                           u64 t, i64 d1, u64 kvr, int mode,
+                                           int swap,
+                                           const mz_gpu_closure &cl,
+                                           u32 lvb, u32 okw, u32 ovb, u64 o,
+                                           u64 *okeys, u8 *ovals,
+                                           u64 *otimes, i64 *odiffs) {
+  for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
+    const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
+    const u8 *v1 = swap ? lv : dv;
+    const u8 *v2 = swap ? dv : lv;
+    for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
+      u64 tout;
+      if (allpass) {
+        tout = t;
+      } else if (mode == PM_JOIN) {
+        u64 t2 = b.times[u];
+        tout = t2 > t ? t2 : t;
+      } else {
+        u64 t2 = b.times[u];
+        if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
+        tout = t;
+      }
+      if (!d_closure_apply(&cl, key, v1, v2, okeys + o * okw,
+                           ovals + o * ovb))
+        continue;
+      otimes[o] = tout;
+      odiffs[o] = wmul(d1, b.diffs[u]);
+      o++;
+    }
+  }
+  return o;
+}
+
+// Wave-aggregated output-queue reservation: exclusive prefix of c across
+// the wavefront, one atomicAdd per wave. ALL 64 lanes must participate.
+__device__ __forceinline__ u64 wave_reserve(unsigned long long *ctr, u32 c,
+                                            u32 lane) {
+  u32 pre = c;
+  for (int d = 1; d < 64; d <<= 1) {
+    u32 up = __shfl_up(pre, d, 64);
+    if ((int)lane >= d) pre += up;
+  }
+  u32 excl = pre - c;
+  unsigned long long wtotal = (unsigned long long)(u32)__shfl((int)pre, 63,
+                                                              64);
+  long long basell = 0;
+  if (lane == 63 && wtotal)
+    basell = (long long)atomicAdd(ctr, wtotal);
+  return (u64)__shfl((int64_t)basell, 63, 64) + excl;
+}
+
+// canonical key order: i64-tuple ascending
+__device__ __forceinline__ int d_key_cmp(const u64 *a, const u64 *b,
+                                         u32 kw) {
+  for (u32 w = 0; w < kw; w++) {
+    i64 x = (i64)a[w], y = (i64)b[w];
+    if (x != y) return x < y ? -1 : 1;
+  }
+  return 0;
+}
+
 // Single-walk probe (DESIGN §9 candidate (b), the half_join2 probe loop
 // replacement, delta_join.rs:500,544): one kernel probes AND emits —
 // each (row, batch) pair counts its matches from the just-read range
@@ -532,43 +624,12 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
       u64 idx = start + (it0 + tt) * stride;
       u64 i = idx % n;
       int bi = (int)(idx / n);
-      const DevBatch &b = bl.b[bi];
-      const u64 *key = dkeys + i * kw;
-      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-      u64 t = dtimes[i];
-      u32 c = 0;
-      for (u32 j = (u32)kvr[tt]; j < (u32)(kvr[tt] >> 32); j++) {
-        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-        const u8 *v1 = swap ? lv : dv;
-        const u8 *v2 = swap ? dv : lv;
-        if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
-        u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
-        if (mode == PM_JOIN || bl.allpass[bi]) {
-          c += hi - lo;
-        } else {
-          for (u32 u = lo; u < hi; u++) {
-            u64 t2 = b.times[u];
-            c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
-          }
-        }
-      }
-      cc[tt] = c;
-      csum += c;
+      cc[tt] = d_count_pair(bl.b[bi], bl.allpass[bi], dkeys + i * kw,
+                            dvals ? dvals + i * dvb : nullptr, dtimes[i],
+                            kvr[tt], mode, swap, cl, lvb);
+      csum += cc[tt];
     }
-    // wave-aggregated queue reservation: exclusive prefix of csum across
-    // the wavefront, one atomicAdd per wave (all 64 lanes participate)
-    u32 pre = csum;
-    for (int d = 1; d < 64; d <<= 1) {
-      u32 up = __shfl_up(pre, d, 64);
-      if ((int)lane >= d) pre += up;
-    }
-    u32 excl = pre - csum;
-    unsigned long long wtotal = (unsigned long long)(u32)__shfl((int)pre,
-                                                                63, 64);
-    long long basell = 0;
-    if (lane == 63 && wtotal)
-      basell = (long long)atomicAdd(ctr, wtotal);
-    u64 base = (u64)__shfl((int64_t)basell, 63, 64) + excl;
+    u64 base = wave_reserve(ctr, csum, lane);
     if (csum == 0 || base + csum > cap) continue;
     // Phase C: emit from L2-hot lines, fields written straight to the
     // reserved global slots (closure filters run before any field write)
@@ -579,36 +640,111 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
       u64 idx = start + (it0 + tt) * stride;
       u64 i = idx % n;
       int bi = (int)(idx / n);
-      const DevBatch &b = bl.b[bi];
+      o = d_emit_pair(bl.b[bi], bl.allpass[bi], dkeys + i * kw,
+                      dvals ? dvals + i * dvb : nullptr, dtimes[i],
+                      ddiffs[i], kvr[tt], mode, swap, cl, lvb, okw, ovb, o,
+                      okeys, ovals, otimes, odiffs);
+    }
+  }
+}
+
+// Merge probe for SORTED delta streams against one large batch: both
+// sides are ascending in the canonical key order, so each block narrows
+// the batch's key array to its delta rows' range with two binary
+// searches, stages that window in LDS, and every thread resolves its
+// rows' val ranges by an LDS search — the batch's keys/kv_off/vals/upds
+// are then read in ASCENDING order across the grid (streaming) and the
+// hash table is never touched. This is the sort-merge restatement of
+// half_join2's per-key cursor seek (delta_join.rs:500,544;
+// mz_join_core.rs:647-662) for the MI355X memory system: the measured
+// uniform-random 128 B line ceiling is ~1.1 TB/s while sequential
+// windows stream at HBM rates.
+#define MERGE_DROWS 1024
+#define MERGE_LDSW 8192
+__global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                              const u64 *dtimes, const i64 *ddiffs, u64 n,
+                              u32 kw, u32 lvb, DevBatch b, int allpass,
+                              int mode, int swap, const mz_gpu_closure cl,
+                              u64 cap, unsigned long long *ctr, u64 *okeys,
+                              u8 *ovals, u64 *otimes, i64 *odiffs) {
+  __shared__ u64 lk[MERGE_LDSW];
+  __shared__ u64 sklo, skhi;
+  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
+  u64 r0 = (u64)blockIdx.x * MERGE_DROWS;
+  if (r0 >= n) return;
+  u64 r1 = r0 + MERGE_DROWS < n ? r0 + MERGE_DROWS : n;
+  if (threadIdx.x == 0) {
+    const u64 *k0 = dkeys + r0 * kw;
+    const u64 *k1 = dkeys + (r1 - 1) * kw;
+    u64 lo = 0, hi = b.n_keys;
+    while (lo < hi) {
+      u64 mid = (lo + hi) / 2;
+      if (d_key_cmp(b.keys + mid * kw, k0, kw) < 0)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    sklo = lo;
+    u64 lo2 = lo;
+    hi = b.n_keys;
+    while (lo2 < hi) {
+      u64 mid = (lo2 + hi) / 2;
+      if (d_key_cmp(b.keys + mid * kw, k1, kw) <= 0)
+        lo2 = mid + 1;
+      else
+        hi = mid;
+    }
+    skhi = lo2;
+  }
+  __syncthreads();
+  u64 klo = sklo, khi = skhi;
+  u64 nk = khi - klo;
+  bool use_lds = nk * kw <= MERGE_LDSW;
+  if (use_lds) {
+    for (u64 w = threadIdx.x; w < nk * kw; w += blockDim.x)
+      lk[w] = b.keys[klo * kw + w];
+    __syncthreads();
+  }
+  u32 lane = threadIdx.x & 63;
+  const int RPT = MERGE_DROWS / BLK;  // rows per thread
+  u64 kvr[RPT];
+  u32 cc[RPT];
+  u32 csum = 0;
+  for (int q = 0; q < RPT; q++) {
+    u64 i = r0 + threadIdx.x + (u64)q * blockDim.x;  // coalesced
+    kvr[q] = ~0ull;
+    cc[q] = 0;
+    if (i < r1) {
       const u64 *key = dkeys + i * kw;
-      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-      u64 t = dtimes[i];
-      i64 d1 = ddiffs[i];
-      for (u32 j = (u32)kvr[tt]; j < (u32)(kvr[tt] >> 32); j++) {
-        const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
-        const u8 *v1 = swap ? lv : dv;
-        const u8 *v2 = swap ? dv : lv;
-        for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
-          u64 tout;
-          if (bl.allpass[bi]) {
-            tout = t;
-          } else if (mode == PM_JOIN) {
-            u64 t2 = b.times[u];
-            tout = t2 > t ? t2 : t;
-          } else {
-            u64 t2 = b.times[u];
-            if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
-            tout = t;
-          }
-          if (!d_closure_apply(&cl, key, v1, v2, okeys + o * okw,
-                               ovals + o * ovb))
-            continue;
-          otimes[o] = tout;
-          odiffs[o] = wmul(d1, b.diffs[u]);
-          o++;
-        }
+      const u64 *base = use_lds ? lk : b.keys + klo * kw;
+      u64 lo = 0, hi = nk;
+      while (lo < hi) {
+        u64 mid = (lo + hi) / 2;
+        if (d_key_cmp(base + mid * kw, key, kw) < 0)
+          lo = mid + 1;
+        else
+          hi = mid;
+      }
+      if (lo < nk && d_key_cmp(base + lo * kw, key, kw) == 0) {
+        u64 kidx = klo + lo;
+        kvr[q] = (u64)b.kv_off[kidx] | ((u64)b.kv_off[kidx + 1] << 32);
+        cc[q] = d_count_pair(b, allpass, key,
+                             dvals ? dvals + i * dvb : nullptr, dtimes[i],
+                             kvr[q], mode, swap, cl, lvb);
+        csum += cc[q];
       }
     }
+  }
+  u64 base_o = wave_reserve(ctr, csum, lane);
+  if (csum == 0 || base_o + csum > cap) return;
+  u64 o = base_o;
+  for (int q = 0; q < RPT; q++) {
+    if (!cc[q]) continue;
+    u64 i = r0 + threadIdx.x + (u64)q * blockDim.x;
+    o = d_emit_pair(b, allpass, dkeys + i * kw,
+                    dvals ? dvals + i * dvb : nullptr, dtimes[i], ddiffs[i],
+                    kvr[q], mode, swap, cl, lvb, okw, ovb, o, okeys, ovals,
+                    otimes, odiffs);
   }
 }
 
@@ -1807,11 +1943,13 @@ struct DevUpdates {
   const u64 *times;
   const i64 *diffs;
   u64 n;
+  int sorted = 0;  // canonical (key,val,time) ascending (see mz_gpu.h)
 };
 
 DevUpdates stage_updates(Ctx *c, const mz_gpu_updates *u, u32 kw, u32 vb) {
   DevUpdates d;
   d.n = u->n;
+  d.sorted = u->sorted;
   if (u->on_device) {
     d.keys = u->keys;
     d.vals = u->vals;
@@ -2496,7 +2634,15 @@ static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
   u64 *ot = dnew<u64>(ctx, capn);
   i64 *od = dnew<i64>(ctx, capn);
   u64 *dcounts = (u64 *)S.get(3 * 8);
-  consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  if (d.sorted && d.n) {
+    // already in canonical order: skip the radix passes, identity perm
+    u32 *perm = (u32 *)S.get(d.n * 4);
+    hipLaunchKernelGGL(k_iota, dim3(ngrid(d.n)), dim3(BLK), 0, ctx->stream,
+                       perm, d.n);
+    consolidate_with_perm(ctx, kw, vb, d, perm, ok, ov, ot, od, dcounts);
+  } else {
+    consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  }
   DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, lower,
                                 upper, dcounts);
   u64 cnt[3] = {0, 0, 0};
@@ -2531,7 +2677,14 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
   u64 *ot = dnew<u64>(ctx, capn);
   i64 *od = dnew<i64>(ctx, capn);
   u64 *dcounts = (u64 *)S.get(3 * 8);
-  consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  if (d.sorted && d.n) {
+    u32 *perm = (u32 *)S.get(d.n * 4);
+    hipLaunchKernelGGL(k_iota, dim3(ngrid(d.n)), dim3(BLK), 0, ctx->stream,
+                       perm, d.n);
+    consolidate_with_perm(ctx, kw, vb, d, perm, ok, ov, ot, od, dcounts);
+  } else {
+    consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+  }
   DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, u->lower,
                                 u->upper, dcounts);
   a->pending.active = 1;
@@ -2739,7 +2892,32 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                     dnew<i64>(ctx, 1), 0, okw, ovb);
     return 0;
   }
-  u64 nb2 = n * (u64)bl.n;
+  // SORTED deltas probe large batches by merge scan (k_probe_merge:
+  // streaming both sorted sides); everything else takes the hash walk.
+  // read per call (not latched): tests force the merge path with tiny
+  // thresholds mid-process
+  const char *emn = getenv("MZ_PROBE_MERGE_MIN_MB");
+  const u64 MERGE_MIN = (u64)(emn ? atoll(emn) : 24) << 20;
+  const char *eme = getenv("MZ_PROBE_MERGE");
+  const bool MERGE_EN = !(eme && eme[0] == '0');
+  BatchList blw;
+  blw.n = 0;
+  struct MergeTarget {
+    DevBatch b;
+    u8 allpass;
+  };
+  std::vector<MergeTarget> mts;
+  for (int b2 = 0; b2 < bl.n; b2++) {
+    u64 tb = bl.b[b2].hash_slots * (kw + 1) * 8;
+    if (MERGE_EN && d.sorted && tb > MERGE_MIN && bl.b[b2].n_keys >= 4096)
+      mts.push_back({bl.b[b2], bl.allpass[b2]});
+    else {
+      blw.b[blw.n] = bl.b[b2];
+      blw.allpass[blw.n++] = bl.allpass[b2];
+    }
+  }
+  u64 nb2 = n * (u64)blw.n;
+  u64 mgrid = (n + MERGE_DROWS - 1) / MERGE_DROWS;
   // Single-walk probe: allocate the output queue from the arrangement's
   // emit-ratio hint, relaunch once with the exact count on overflow.
   u64 cap = (lookup->probe_cap_hint ? lookup->probe_cap_hint + 1 : 2) * n +
@@ -2750,11 +2928,20 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
   u64 *pt = dnew<u64>(ctx, cap);
   i64 *pd = dnew<i64>(ctx, cap);
+  auto launch_probes = [&]() {
+    for (auto &mt : mts)
+      hipLaunchKernelGGL(k_probe_merge, dim3((u32)mgrid), dim3(BLK), 0,
+                         ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                         d.diffs, n, kw, lvb, mt.b, (int)mt.allpass, mode,
+                         swap, *cl, cap, ctr, pk, pv, pt, pd);
+    if (blw.n)
+      hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
+                         ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                         d.diffs, n, kw, lvb, blw, mode, swap, *cl, cap,
+                         ctr, pk, pv, pt, pd);
+  };
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
-  hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
-                     ctx->stream, d.keys, d.vals, stream_vb, d.times,
-                     d.diffs, n, kw, lvb, bl, mode, swap, *cl, cap, ctr, pk,
-                     pv, pt, pd);
+  launch_probes();
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
   unsigned long long M = 0;
   HIP_CHECK(hipMemcpyAsync(&M, ctr, 8, hipMemcpyDeviceToHost, ctx->stream));
@@ -2770,10 +2957,7 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     pd = dnew<i64>(ctx, cap);
     fill_u64(ctx, (u64 *)ctr, 1, 0);
     if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
-    hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
-                       ctx->stream, d.keys, d.vals, stream_vb, d.times,
-                       d.diffs, n, kw, lvb, bl, mode, swap, *cl, cap, ctr,
-                       pk, pv, pt, pd);
+    launch_probes();
     if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
     launches = 2;
   }

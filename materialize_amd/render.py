@@ -23,10 +23,12 @@ class DevOut:
     """An engine out-batch kept where the engine produced it (device for
     GpuCtx, host for OracleCtx), usable as the next stage's input."""
 
-    def __init__(self, ctx, outp, host_cols=None):
+    def __init__(self, ctx, outp, host_cols=None, sorted=False):
         self.ctx = ctx
         self.outp = outp  # POINTER(OutBatch) or None (host)
         self.host_cols = host_cols
+        # canonical (key,val,time)-sorted content (consolidated producers)
+        self.sorted = sorted
 
     @property
     def n(self):
@@ -53,6 +55,7 @@ class DevOut:
             u.lower = lower
             u.upper = upper
             u.on_device = 1
+            u.sorted = 1 if self.sorted else 0
             u._ref = self
             return u
         k, v, t, d = self.host_cols

@@ -217,19 +217,33 @@ class Q3Dataflow:
         Updates descriptors with times == t. Returns the corrections
         DevOut (or None)."""
         ctx = self.ctx
-        # async: each arrangement's consolidate+build runs on its own HIP
-        # stream; the first probe of each flushes it (overlap of the three
-        # independent maintenance pipelines)
+        # Consolidate each relation's churn ONCE (the only radix sorts of
+        # the step), then hand the canonically-SORTED batch to both
+        # consumers: the arrangement insert skips its sort (identity-perm
+        # consolidation on the lane) and the delta-path probes take the
+        # streaming merge path against large arrangements
+        # (sorted-delta/sorted-batch merge scan — the MergeBatcher-feeds-
+        # both pattern of mz_arrange_core, extensions/arrange.rs:69).
+        cons = {}
         for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
-            ctx.arr_insert_async(self.arrs[name], upd[name])
+            kw, vb = self.SCHEMAS[name]
+            cons[name] = ctx.consolidate_dev(abi.schema(kw, vb), upd[name])
+        for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
+            ctx.arr_insert_async(self.arrs[name],
+                                 cons[name].updates(t, t + 1))
         outs = []
         for rel, src in (("orders", "orders_by_custkey"),
                          ("lineitem", "lineitem")):
             o = self.join.push_path_updates(
-                self.paths[rel], upd[src], t,
+                self.paths[rel], cons[src].updates(t, t + 1), t,
                 final_exchange=self._final_exchange)
             if o is not None:
                 outs.append(o)
+        # async-insert lifetime: orders_by_custkey is not probed this step,
+        # so flush it before releasing its consolidated source buffers
+        ctx.arr_flush(self.arrs["orders_by_custkey"])
+        for c in cons.values():
+            c.release()
         if not outs:
             return None
         if len(outs) == 1:
