@@ -213,6 +213,15 @@ typedef struct {
   mz_gpu_schema schema;
 } mz_gpu_out;
 
+/* Flush like mz_gpu_arr_flush, and additionally hand back the pending
+ * insert's CONSOLIDATED flat rows as a sorted out-batch (*out = NULL when
+ * no insert was pending or it was empty). This is the arrangement's
+ * update stream: the same sealed rows the reference's mz_arrange_core
+ * publishes to downstream operators (extensions/arrange.rs:69-114) —
+ * delta-path probes consume it with `sorted = 1`. Caller releases. */
+int mz_gpu_arr_flush_take(mz_gpu_ctx *ctx, mz_gpu_arr *arr,
+                          mz_gpu_out **out);
+
 void mz_gpu_out_release(mz_gpu_ctx *ctx, mz_gpu_out *out);
 /* Copy an out-batch's columns to caller host buffers (sized n). */
 int  mz_gpu_out_to_host(mz_gpu_ctx *ctx, const mz_gpu_out *out,

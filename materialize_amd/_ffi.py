@@ -92,6 +92,8 @@ def load():
     lib.mz_gpu_arr_insert_async.argtypes = [C.c_void_p, C.c_void_p,
                                             C.POINTER(Updates)]
     lib.mz_gpu_arr_flush.argtypes = [C.c_void_p, C.c_void_p]
+    lib.mz_gpu_arr_flush_take.argtypes = [C.c_void_p, C.c_void_p,
+                                          C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_prof_dump.argtypes = [C.c_void_p]
     lib.mz_gpu_topk_create.restype = C.c_void_p
     lib.mz_gpu_topk_create.argtypes = [C.c_void_p, C.POINTER(TopKSpec)]
@@ -238,6 +240,18 @@ class GpuCtx:
 
     def arr_flush(self, arr):
         self._check(self.lib.mz_gpu_arr_flush(self.ctx, arr))
+
+    def arr_flush_take(self, arr):
+        """Flush, returning the pending insert's consolidated rows as a
+        sorted DevOut (None when nothing was pending/empty) — the
+        arrangement's update stream (mz_arrange_core's published sealed
+        rows, extensions/arrange.rs:69-114)."""
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_arr_flush_take(self.ctx, arr,
+                                                   C.byref(outp)))
+        if not outp:
+            return None
+        return self._dev_out(outp, sorted=True)
 
     def prof_dump(self):
         self.lib.mz_gpu_prof_dump(self.ctx)

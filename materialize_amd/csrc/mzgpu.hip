@@ -661,43 +661,53 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
 // windows stream at HBM rates.
 #define MERGE_DROWS 1024
 #define MERGE_LDSW 8192
-__global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
-                              const u64 *dtimes, const i64 *ddiffs, u64 n,
-                              u32 kw, u32 lvb, DevBatch b, int allpass,
-                              int mode, int swap, const mz_gpu_closure cl,
-                              u64 cap, unsigned long long *ctr, u64 *okeys,
-                              u8 *ovals, u64 *otimes, i64 *odiffs) {
-  __shared__ u64 lk[MERGE_LDSW];
-  __shared__ u64 sklo, skhi;
-  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
-  u64 r0 = (u64)blockIdx.x * MERGE_DROWS;
-  if (r0 >= n) return;
-  u64 r1 = r0 + MERGE_DROWS < n ? r0 + MERGE_DROWS : n;
-  if (threadIdx.x == 0) {
+// Per-block key-window bounds for k_probe_merge, one THREAD per block
+// (the in-block thread-0 search serialized ~21 dependent global loads
+// while 255 lanes idled — measured 243 us/launch; precomputing all
+// windows in one parallel kernel removes that).
+__global__ void k_merge_bounds(const u64 *dkeys, u64 n, u32 kw,
+                               const u64 *bkeys, u64 n_keys, u64 nblocks,
+                               u64 *bounds) {
+  GRID_STRIDE(blk, nblocks) {
+    u64 r0 = blk * MERGE_DROWS;
+    u64 r1 = r0 + MERGE_DROWS < n ? r0 + MERGE_DROWS : n;
     const u64 *k0 = dkeys + r0 * kw;
     const u64 *k1 = dkeys + (r1 - 1) * kw;
-    u64 lo = 0, hi = b.n_keys;
+    u64 lo = 0, hi = n_keys;
     while (lo < hi) {
       u64 mid = (lo + hi) / 2;
-      if (d_key_cmp(b.keys + mid * kw, k0, kw) < 0)
+      if (d_key_cmp(bkeys + mid * kw, k0, kw) < 0)
         lo = mid + 1;
       else
         hi = mid;
     }
-    sklo = lo;
+    bounds[2 * blk] = lo;
     u64 lo2 = lo;
-    hi = b.n_keys;
+    hi = n_keys;
     while (lo2 < hi) {
       u64 mid = (lo2 + hi) / 2;
-      if (d_key_cmp(b.keys + mid * kw, k1, kw) <= 0)
+      if (d_key_cmp(bkeys + mid * kw, k1, kw) <= 0)
         lo2 = mid + 1;
       else
         hi = mid;
     }
-    skhi = lo2;
+    bounds[2 * blk + 1] = lo2;
   }
-  __syncthreads();
-  u64 klo = sklo, khi = skhi;
+}
+
+__global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                              const u64 *dtimes, const i64 *ddiffs, u64 n,
+                              u32 kw, u32 lvb, DevBatch b, int allpass,
+                              int mode, int swap, const mz_gpu_closure cl,
+                              const u64 *bounds, u64 cap,
+                              unsigned long long *ctr, u64 *okeys,
+                              u8 *ovals, u64 *otimes, i64 *odiffs) {
+  __shared__ u64 lk[MERGE_LDSW];
+  u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
+  u64 r0 = (u64)blockIdx.x * MERGE_DROWS;
+  if (r0 >= n) return;
+  u64 r1 = r0 + MERGE_DROWS < n ? r0 + MERGE_DROWS : n;
+  u64 klo = bounds[2 * blockIdx.x], khi = bounds[2 * blockIdx.x + 1];
   u64 nk = khi - klo;
   bool use_lds = nk * kw <= MERGE_LDSW;
   if (use_lds) {
@@ -1441,6 +1451,11 @@ struct mz_gpu_arr {
     u64 cnt[3] = {0, 0, 0};
     DevBatch batch;
     u64 upper = 0;
+    // the insert's consolidated FLAT key/val rows (same order as
+    // batch.times/diffs), kept so mz_gpu_arr_flush_take can hand the
+    // sorted rows to the delta-path probes without re-sorting
+    u64 *flat_keys = nullptr;
+    u8 *flat_vals = nullptr;
   } pending;
   // deferred spine merge: the merged batch is computed on the lane while
   // probes keep using the pre-merge batch list (identical logical
@@ -2100,7 +2115,7 @@ void free_batch(Ctx *c, DevBatch &b) {
 // batch's host-side counts from dcounts.
 DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
                           u64 *times, i64 *diffs, u64 cap, u64 lower,
-                          u64 upper, u64 *dcounts) {
+                          u64 upper, u64 *dcounts, int keep_flat = 0) {
   MZ_PROF(c, "build_batch");
   auto &S = (*c->scr);
   DevBatch b;
@@ -2150,8 +2165,11 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
                      c->stream, b.hash, slots, b.keys, kw, kid, b.kv_off,
                      cap, dcounts);
   // flat key/val arrays were re-packed; stream-ordered free is safe
-  dfree(c, keys);
-  dfree(c, vals);
+  // (unless the caller keeps them for a flush_take hand-off)
+  if (!keep_flat) {
+    dfree(c, keys);
+    dfree(c, vals);
+  }
   return b;
 }
 
@@ -2686,15 +2704,19 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
     consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
   }
   DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, u->lower,
-                                u->upper, dcounts);
+                                u->upper, dcounts, /*keep_flat=*/1);
   a->pending.active = 1;
   a->pending.batch = b;
   a->pending.upper = u->upper;
+  a->pending.flat_keys = ok;
+  a->pending.flat_vals = ov;
   HIP_CHECK(hipMemcpyAsync(a->pending.cnt, dcounts, 3 * 8,
                            hipMemcpyDeviceToHost, ctx->stream));
 }
 
-static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
+static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
+                                mz_gpu_out **take) {
+  if (take) *take = nullptr;
   if (!a->pending.active && !a->pending_merge.active) return;
   MZ_PROF(ctx, "arr_flush");
   LaneGuard lane(ctx, a);
@@ -2708,6 +2730,30 @@ static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
     a->batches.push_back(b);
     a->upper = std::max(a->upper, a->pending.upper);
     a->pending.active = 0;
+    u32 kw = a->schema.kw, vb = a->schema.vb;
+    if (take && b.n_upds) {
+      // hand the consolidated flat rows over as a sorted out-batch:
+      // keys/vals transfer ownership; times/diffs are copied (the batch
+      // owns its columns and a spine merge may free them)
+      u64 *tc = dnew<u64>(ctx, b.n_upds);
+      i64 *dc = dnew<i64>(ctx, b.n_upds);
+      HIP_CHECK(hipMemcpyAsync(tc, b.times, b.n_upds * 8,
+                               hipMemcpyDeviceToDevice, ctx->stream));
+      HIP_CHECK(hipMemcpyAsync(dc, b.diffs, b.n_upds * 8,
+                               hipMemcpyDeviceToDevice, ctx->stream));
+      *take = make_out(a->pending.flat_keys, a->pending.flat_vals, tc, dc,
+                       b.n_upds, kw, vb);
+      a->pending.flat_keys = nullptr;
+      a->pending.flat_vals = nullptr;
+      // the out's consumers run on other streams; its columns must be
+      // complete before the flush returns
+      HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    } else {
+      dfree(ctx, a->pending.flat_keys);
+      dfree(ctx, a->pending.flat_vals);
+      a->pending.flat_keys = nullptr;
+      a->pending.flat_vals = nullptr;
+    }
   }
   // Measured on the 1M churn config: deferring merges off the probe path
   // LOSES ~10% — probes pay for the deeper pre-merge spine and the
@@ -2727,6 +2773,10 @@ static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
   }
 }
 
+static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a) {
+  arr_flush_take_impl(ctx, a, nullptr);
+}
+
 int mz_gpu_arr_insert_async(mz_gpu_ctx *c, mz_gpu_arr *a,
                             const mz_gpu_updates *u) {
   arr_insert_async_impl(&c->impl, a, u);
@@ -2735,6 +2785,11 @@ int mz_gpu_arr_insert_async(mz_gpu_ctx *c, mz_gpu_arr *a,
 
 int mz_gpu_arr_flush(mz_gpu_ctx *c, mz_gpu_arr *a) {
   arr_flush_impl(&c->impl, a);
+  return 0;
+}
+
+int mz_gpu_arr_flush_take(mz_gpu_ctx *c, mz_gpu_arr *a, mz_gpu_out **out) {
+  arr_flush_take_impl(&c->impl, a, out);
   return 0;
 }
 
@@ -2898,8 +2953,12 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   // thresholds mid-process
   const char *emn = getenv("MZ_PROBE_MERGE_MIN_MB");
   const u64 MERGE_MIN = (u64)(emn ? atoll(emn) : 24) << 20;
+  // Default OFF: measured 1.4x slower than the compressed-slot hash walk
+  // at the Q3 1M config (LDS staging + lockstep searches beat by 16 B
+  // random slots over half-L2-resident tables); kept as a parity-tested
+  // option for larger-than-L2 regimes. MZ_PROBE_MERGE=1 enables.
   const char *eme = getenv("MZ_PROBE_MERGE");
-  const bool MERGE_EN = !(eme && eme[0] == '0');
+  const bool MERGE_EN = eme && eme[0] == '1';
   BatchList blw;
   blw.n = 0;
   struct MergeTarget {
@@ -2928,12 +2987,23 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
   u64 *pt = dnew<u64>(ctx, cap);
   i64 *pd = dnew<i64>(ctx, cap);
+  u64 *mbounds = nullptr;
+  if (!mts.empty()) {
+    mbounds = (u64 *)S.get(mts.size() * mgrid * 2 * 8);
+    for (size_t mi = 0; mi < mts.size(); mi++)
+      hipLaunchKernelGGL(k_merge_bounds, dim3(ngrid(mgrid)), dim3(BLK), 0,
+                         ctx->stream, d.keys, n, kw, mts[mi].b.keys,
+                         mts[mi].b.n_keys, mgrid,
+                         mbounds + mi * mgrid * 2);
+  }
   auto launch_probes = [&]() {
-    for (auto &mt : mts)
+    for (size_t mi = 0; mi < mts.size(); mi++)
       hipLaunchKernelGGL(k_probe_merge, dim3((u32)mgrid), dim3(BLK), 0,
                          ctx->stream, d.keys, d.vals, stream_vb, d.times,
-                         d.diffs, n, kw, lvb, mt.b, (int)mt.allpass, mode,
-                         swap, *cl, cap, ctr, pk, pv, pt, pd);
+                         d.diffs, n, kw, lvb, mts[mi].b,
+                         (int)mts[mi].allpass, mode, swap, *cl,
+                         mbounds + mi * mgrid * 2, cap, ctr, pk, pv, pt,
+                         pd);
     if (blw.n)
       hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
                          ctx->stream, d.keys, d.vals, stream_vb, d.times,

@@ -217,33 +217,34 @@ class Q3Dataflow:
         Updates descriptors with times == t. Returns the corrections
         DevOut (or None)."""
         ctx = self.ctx
-        # Consolidate each relation's churn ONCE (the only radix sorts of
-        # the step), then hand the canonically-SORTED batch to both
-        # consumers: the arrangement insert skips its sort (identity-perm
-        # consolidation on the lane) and the delta-path probes take the
-        # streaming merge path against large arrangements
-        # (sorted-delta/sorted-batch merge scan — the MergeBatcher-feeds-
-        # both pattern of mz_arrange_core, extensions/arrange.rs:69).
-        cons = {}
+        # Each relation's churn is consolidated ONCE, on its arrangement's
+        # own lane (the three sort pipelines overlap); the path probes
+        # then consume the arrangements' published update streams — the
+        # flush_take hand-off of the sealed sorted rows, exactly
+        # mz_arrange_core feeding both the trace and downstream operators
+        # (extensions/arrange.rs:69-114). No per-path re-sort, and the
+        # sorted flag lets large-table probes take the merge path.
         for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
-            kw, vb = self.SCHEMAS[name]
-            cons[name] = ctx.consolidate_dev(abi.schema(kw, vb), upd[name])
-        for name in ("lineitem", "orders_by_orderkey", "orders_by_custkey"):
-            ctx.arr_insert_async(self.arrs[name],
-                                 cons[name].updates(t, t + 1))
+            ctx.arr_insert_async(self.arrs[name], upd[name])
+        take = {}
+        for name in ("orders_by_custkey", "lineitem"):
+            take[name] = (ctx.arr_flush_take(self.arrs[name])
+                          if hasattr(ctx, "arr_flush_take") else None)
         outs = []
         for rel, src in (("orders", "orders_by_custkey"),
                          ("lineitem", "lineitem")):
+            if take.get(src) is not None:
+                u = take[src].updates(t, t + 1)
+            else:  # oracle / fallback: the raw staged batch
+                u = upd[src]
             o = self.join.push_path_updates(
-                self.paths[rel], cons[src].updates(t, t + 1), t,
+                self.paths[rel], u, t,
                 final_exchange=self._final_exchange)
             if o is not None:
                 outs.append(o)
-        # async-insert lifetime: orders_by_custkey is not probed this step,
-        # so flush it before releasing its consolidated source buffers
-        ctx.arr_flush(self.arrs["orders_by_custkey"])
-        for c in cons.values():
-            c.release()
+        for tk in take.values():
+            if tk is not None:
+                tk.release()
         if not outs:
             return None
         if len(outs) == 1:
