@@ -1,0 +1,127 @@
+#!/usr/bin/env python3
+"""Generate and verify the SF0.01 exact-TPCH golden fixture.
+
+Runs materialize_amd.tpch_exact (the draw-exact restatement of the
+reference's TPCH load generator, tpch.rs) at SCALE FACTOR .01 / seed 0 —
+the configuration of /root/reference/test/testdrive/tpch.td — and
+verifies this restatement against the reference's own pinned MD5s
+BEFORE writing anything:
+
+  Q3  (127 rows) 637be0ff3f50cd612b004a69958bfccb   (tpch.td:193-215)
+  Q6  (1 value)  d9c979f1eed5940788ff3653321acac4   (tpch.td:268-278)
+  Q12 (2 rows)   3c31b94c99bd77e96003c2059416ed7a   (tpch.td:462-491)
+
+Then writes tpch_sf001.npz: the snapshot in the engine's column formats,
+8 churn batches (the reference's retract/regenerate protocol), and the
+expected Q3 result rows after the snapshot and after each churn step.
+Run from the repo root: python tests/golden/make_tpch_sf001.py
+"""
+import hashlib
+import json
+import os
+import sys
+from datetime import date
+
+import numpy as np
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+from materialize_amd.tpch_exact import (CUTOFF, ExactEngineData, TpchExact,
+                                        q3_md5, q3_result,
+                                        render_revenue_1e2)
+
+Q3_MD5 = "637be0ff3f50cd612b004a69958bfccb"
+Q6_MD5 = "d9c979f1eed5940788ff3653321acac4"
+Q12_MD5 = "3c31b94c99bd77e96003c2059416ed7a"
+
+PRIORITIES = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED"]
+MODES = ["REG AIR", "AIR", "RAIL", "SHIP", "TRUCK", "MAIL", "FOB"]
+
+
+def verify_q6(lineitems):
+    lo, hi = date(1994, 1, 1), date(1995, 1, 1)
+    tot = 0
+    for (okey, pk, q, ep, d, sd, cd, rd, md, rf) in lineitems:
+        if q < 24 and lo <= sd < hi and 5 <= d <= 7:
+            tot += ep * d
+    h = hashlib.md5()
+    h.update(render_revenue_1e2(tot).encode())
+    return h.hexdigest()
+
+
+def verify_q12(orders, lineitems):
+    oprio = {t[0]: t[3] for t in orders}
+    lo, hi = date(1994, 1, 1), date(1995, 1, 1)
+    agg = {}
+    for (okey, pk, q, ep, d, sd, cd, rd, md, rf) in lineitems:
+        mode = MODES[md]
+        if mode not in ("MAIL", "SHIP"):
+            continue
+        if not (cd < rd and sd < cd and lo <= rd < hi):
+            continue
+        p = PRIORITIES[oprio[okey]]
+        high = 1 if p in ("1-URGENT", "2-HIGH") else 0
+        a = agg.setdefault(mode, [0, 0])
+        a[0] += high
+        a[1] += 1 - high
+    h = hashlib.md5()
+    for mode in sorted(agg):
+        h.update(mode.encode())
+        h.update(str(agg[mode][0]).encode())
+        h.update(str(agg[mode][1]).encode())
+    return h.hexdigest()
+
+
+def main():
+    gen = TpchExact(sf=0.01, seed=0)
+    customers, orders, lineitems = gen.snapshot()
+
+    rows = q3_result(customers, orders, lineitems)
+    assert len(rows) == 127, len(rows)
+    got = q3_md5(rows)
+    assert got == Q3_MD5, f"Q3 {got}"
+    assert verify_q6(lineitems) == Q6_MD5, "Q6"
+    assert verify_q12(orders, lineitems) == Q12_MD5, "Q12"
+    print("reference golden verification: Q3(127 rows) / Q6 / Q12 all ok")
+
+    data = ExactEngineData(gen, customers, orders, lineitems)
+
+    # churn batches + expected Q3 rows after each (maintained exact state)
+    state_orders = {t[0]: t for t in orders}
+    state_lines = {}
+    for t in lineitems:
+        state_lines.setdefault(t[0], []).append(t)
+    churn_npz = {}
+    expected = [rows]
+    n_churn = 8
+    for b in range(n_churn):
+        batch = gen.churn_batch()
+        (okey, ck_o, od_o, p_o), old_lines, new_o, new_lines = batch
+        state_orders[okey] = (okey,) + new_o[1:]
+        state_lines[okey] = new_lines
+        eng = ExactEngineData.churn_to_engine(batch)
+        for rel, (k, v, d) in eng.items():
+            churn_npz[f"b{b}_{rel}_keys"] = k
+            churn_npz[f"b{b}_{rel}_vals"] = v
+            churn_npz[f"b{b}_{rel}_diffs"] = d
+        cur_lines = [t for ls in state_lines.values() for t in ls]
+        expected.append(q3_result(customers, list(state_orders.values()),
+                                  cur_lines))
+
+    out = os.path.join(os.path.dirname(__file__), "tpch_sf001.npz")
+    np.savez_compressed(
+        out,
+        c_custkey=data.c_custkey, c_mktsegment=data.c_mktsegment,
+        o_orderkey=data.o_orderkey, o_custkey=data.o_custkey,
+        o_orderdate=data.o_orderdate,
+        l_orderkey=data.l_orderkey, l_extendedprice=data.l_extendedprice,
+        l_discount=data.l_discount, l_shipdate=data.l_shipdate,
+        n_churn=np.array([n_churn]),
+        expected_json=np.frombuffer(
+            json.dumps(expected).encode(), dtype=np.uint8),
+        **churn_npz)
+    print(f"wrote {out} "
+          f"({os.path.getsize(out) / 1e6:.2f} MB, {n_churn} churn batches)")
+
+
+if __name__ == "__main__":
+    main()
