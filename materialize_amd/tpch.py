@@ -212,3 +212,20 @@ class TpchGen:
             "orders": (o_keys, o_vals, o_diffs),
             "orders_by_cust": (oc_keys, oc_vals, o_diffs.copy()),
         }
+
+    def churn_customers(self, k):
+        """Retract k customers and re-insert them with fresh segments
+        (exercises the customer -> orders -> lineitem delta path under
+        retractions; the reference's loadgen keeps customers static, so
+        this is engine-coverage churn, not a reference-protocol shape).
+        Mutates c_mktsegment. Returns (keys, vals, diffs)."""
+        rng = self.rng
+        idx = rng.choice(self.n_customer, k, replace=False)
+        keys = np.concatenate([self.c_custkey[idx], self.c_custkey[idx]])
+        old_vals = self.c_mktsegment[idx].copy()
+        self.c_mktsegment[idx] = rng.integers(0, SEGMENTS, k)
+        vals = np.concatenate([old_vals,
+                               self.c_mktsegment[idx]]).reshape(-1, 1)
+        diffs = np.concatenate([-np.ones(k, np.int64),
+                                np.ones(k, np.int64)])
+        return keys, vals, diffs

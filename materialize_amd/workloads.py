@@ -174,14 +174,28 @@ class Q3Dataflow:
         l_keys, l_vals, l_diffs = churn["lineitem"]
         o_keys, o_vals, o_diffs = churn["orders"]
         oc_keys, oc_vals, oc_diffs = churn["orders_by_cust"]
-        rows = len(l_keys) + len(o_keys)
+        c_churn = churn.get("customer")  # optional customer churn
+        rows = len(l_keys) + len(o_keys) + \
+            (len(c_churn[0]) if c_churn is not None else 0)
         # 1. arrangements first (the paths' le/lt tie-breaks then count
         #    concurrent cross-terms exactly once — DESIGN.md §5)
         self._seal_push("lineitem", l_keys, l_vals, l_diffs, t)
         self._seal_push("orders_by_orderkey", o_keys, o_vals, o_diffs, t)
         self._seal_push("orders_by_custkey", oc_keys, oc_vals, oc_diffs, t)
-        # 2. delta paths (customer static in the churn workload)
+        if c_churn is not None:
+            self._seal_push("customer", c_churn[0],
+                            np.ascontiguousarray(c_churn[1],
+                                                 np.int64).view(np.uint8),
+                            c_churn[2], t)
+        # 2. delta paths, one per updated source relation
         outs = []
+        if c_churn is not None:
+            o = self.join.push_path(
+                self.paths["customer"], c_churn[0],
+                np.ascontiguousarray(c_churn[1], np.int64).view(np.uint8)
+                .reshape(-1, 8), c_churn[2], t)
+            if o is not None:
+                outs.append(o)
         o = self.join.push_path(self.paths["orders"], oc_keys, oc_vals,
                                 oc_diffs, t)
         if o is not None:

@@ -153,3 +153,36 @@ def test_compaction_bounds_arrangements():
     # ~200 lineitem rows churned per step x 24 steps would add ~5k rows
     # uncompacted; compaction keeps it within ~25% of base
     assert n_after < base_n * 1.25, (base_n, n_after)
+
+
+def test_q3_customer_churn_oracle():
+    """All three delta paths under retractions: customers change segment
+    (retract+insert) alongside the order/lineitem churn; the maintained
+    result must track the naive recompute."""
+    gen = TpchGen(sf=0.002, seed=11)
+    ctx = OracleCtx()
+    df = Q3Dataflow(ctx)
+    state = {}
+    holder = []
+    orig = df.reduce.push
+
+    def capture(u):
+        o = orig(u)
+        holder.append(o.to_host())
+        return o
+
+    df.reduce.push = capture
+    df.load(gen)
+    for c in holder:
+        apply_corrections(state, c)
+    for t in range(1, 6):
+        holder.clear()
+        churn = gen.churn(300)
+        churn["customer"] = gen.churn_customers(20)
+        _, corr = df.step(churn, t)
+        if corr is not None:
+            corr.release()
+        for c in holder:
+            apply_corrections(state, c)
+        assert state == naive_q3(gen), f"mismatch after customer churn {t}"
+    ctx.close()
