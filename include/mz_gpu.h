@@ -87,7 +87,12 @@ enum {
    * a SUM_I64 aggregate slot's i128 value, with the COUNT slot's i64 at
    * arg1+24 (the 24-byte aggregate slot layout). NULL count (count==0 in
    * the reference's CASE, tpch_create_index.slt:1470) fails the filter. */
-  MZ_COMPUTE_Q17_QTYLT = 2
+  MZ_COMPUTE_Q17_QTYLT = 2,
+  /* i64 division as an OUT FIELD: arg0 / arg1 (both i64 at the given
+   * offsets). arg1 == 0 raises MZ_ERR_DIVISION_BY_ZERO: the row is
+   * diverted to the out-batch's error stream — the could_error ok/err
+   * split of linear_join.rs:495-541. */
+  MZ_COMPUTE_DIV_I64 = 3
 };
 
 typedef struct {
@@ -211,7 +216,24 @@ typedef struct {
   uint64_t n;
   int32_t on_device;      /* 1: device pointers (default)              */
   mz_gpu_schema schema;
+  /* Error-row stream: the ok/err split of the reference's join closure
+   * (JoinClosure::could_error, linear_join.rs:495-541). A closure field
+   * whose evaluation errors (e.g. MZ_COMPUTE_DIV_I64 by zero) diverts
+   * the would-be output row here as (error code, time, diff),
+   * consolidated like any update stream. err_* are device arrays owned
+   * by the out-batch (NULL/0 when the closure cannot error). */
+  uint64_t err_n;
+  uint64_t *err_codes;    /* MZ_ERR_* per row */
+  uint64_t *err_times;
+  int64_t  *err_diffs;
 } mz_gpu_out;
+
+enum { MZ_ERR_DIVISION_BY_ZERO = 1 };
+
+/* Copy an out-batch's error rows to caller host buffers (sized err_n). */
+int mz_gpu_out_err_to_host(mz_gpu_ctx *ctx, const mz_gpu_out *out,
+                           uint64_t *codes, uint64_t *times,
+                           int64_t *diffs);
 
 /* Flush like mz_gpu_arr_flush, and additionally hand back the pending
  * insert's CONSOLIDATED flat rows as a sorted out-batch (*out = NULL when

@@ -18,6 +18,7 @@ MZ_CMP_LT, MZ_CMP_LE, MZ_CMP_GT, MZ_CMP_GE, MZ_CMP_EQ, MZ_CMP_NE = range(6)
 MZ_COMPUTE_REVENUE = 0
 MZ_COMPUTE_CONST0 = 1
 MZ_COMPUTE_Q17_QTYLT = 2
+MZ_COMPUTE_DIV_I64 = 3
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 
 MZ_GPU_MAX_FILTERS = 4
@@ -127,6 +128,9 @@ class TopKSpec(C.Structure):
     ]
 
 
+MZ_ERR_DIVISION_BY_ZERO = 1
+
+
 class OutBatch(C.Structure):
     _fields_ = [
         ("keys", C.POINTER(C.c_uint64)),
@@ -136,6 +140,12 @@ class OutBatch(C.Structure):
         ("n", C.c_uint64),
         ("on_device", C.c_int32),
         ("schema", Schema),
+        # error-row stream (the could_error ok/err split,
+        # linear_join.rs:495-541): (code, time, diff) rows
+        ("err_n", C.c_uint64),
+        ("err_codes", C.POINTER(C.c_uint64)),
+        ("err_times", C.POINTER(C.c_uint64)),
+        ("err_diffs", C.POINTER(C.c_int64)),
     ]
 
 

@@ -66,6 +66,9 @@ def load():
     lib.mz_gpu_out_to_host.argtypes = [
         C.c_void_p, C.POINTER(OutBatch), C.POINTER(C.c_uint64),
         C.POINTER(C.c_uint8), C.POINTER(C.c_uint64), C.POINTER(C.c_int64)]
+    lib.mz_gpu_out_err_to_host.argtypes = [
+        C.c_void_p, C.POINTER(OutBatch), C.POINTER(C.c_uint64),
+        C.POINTER(C.c_uint64), C.POINTER(C.c_int64)]
     lib.mz_gpu_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
                                        C.POINTER(Updates),
                                        C.POINTER(C.POINTER(OutBatch))]
@@ -165,7 +168,9 @@ class GpuCtx:
             raise MzGpuError(self.lib.mz_gpu_last_error(self.ctx).decode())
 
     def _take(self, outp):
-        """Copy an out-batch to host numpy arrays and release it."""
+        """Copy an out-batch to host numpy arrays and release it. Any
+        error rows (ok/err split) land in self.last_errs as
+        (codes, times, diffs)."""
         import numpy as np
         ob = outp.contents
         n = ob.n
@@ -181,6 +186,17 @@ class GpuCtx:
                 vals.ctypes.data_as(C.POINTER(C.c_uint8)),
                 times.ctypes.data_as(C.POINTER(C.c_uint64)),
                 diffs.ctypes.data_as(C.POINTER(C.c_int64))))
+        en = ob.err_n
+        ecodes = np.empty(en, np.uint64)
+        etimes = np.empty(en, np.uint64)
+        ediffs = np.empty(en, np.int64)
+        if en:
+            self._check(self.lib.mz_gpu_out_err_to_host(
+                self.ctx, outp,
+                ecodes.ctypes.data_as(C.POINTER(C.c_uint64)),
+                etimes.ctypes.data_as(C.POINTER(C.c_uint64)),
+                ediffs.ctypes.data_as(C.POINTER(C.c_int64))))
+        self.last_errs = (ecodes, etimes, ediffs)
         self.lib.mz_gpu_out_release(self.ctx, outp)
         return keys.view("int64"), vals, times, diffs
 

@@ -408,10 +408,12 @@ __device__ __forceinline__ const u8 *d_cl_src(const u64 *key, const u8 *v1,
   }
 }
 
-// mirrors oracle closure_apply; v1 = input-1/stream val, v2 = input-2/lookup
-__device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
-                                const u8 *v1, const u8 *v2, u64 *okey,
-                                u8 *oval) {
+// mirrors oracle closure_apply; v1 = input-1/stream val, v2 = input-2/lookup.
+// Returns 0 = filtered out, 1 = ok, 2 = evaluation ERROR (division by
+// zero) — the err branch of the could_error split, linear_join.rs:495.
+__device__ int d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
+                               const u8 *v1, const u8 *v2, u64 *okey,
+                               u8 *oval) {
   for (u32 i = 0; i < cl->n_filters; i++) {
     const mz_gpu_filter f = cl->filters[i];
     if (f.src == MZ_SRC_COMPUTE) {
@@ -436,9 +438,25 @@ __device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
       case MZ_CMP_EQ: ok = x == f.imm; break;
       default: ok = x != f.imm; break;
     }
-    if (!ok) return false;
+    if (!ok) return 0;
   }
-  if (!okey) return true;
+  if (!okey) {
+    // count mode: fields are not materialized, but erroring computes
+    // must still be CHECKED so ok/err counts match the emit phase
+    for (int which = 0; which < 2; which++) {
+      u32 nf = which ? cl->n_val_fields : cl->n_key_fields;
+      const mz_gpu_field *fs = which ? cl->val_fields : cl->key_fields;
+      for (u32 i = 0; i < nf; i++) {
+        const mz_gpu_field f = fs[i];
+        if (f.src == MZ_SRC_COMPUTE && f.off == MZ_COMPUTE_DIV_I64) {
+          i64 b = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1,
+                             8);
+          if (b == 0) return 2;
+        }
+      }
+    }
+    return 1;
+  }
   for (int which = 0; which < 2; which++) {
     u32 nf = which ? cl->n_val_fields : cl->n_key_fields;
     const mz_gpu_field *fs = which ? cl->val_fields : cl->key_fields;
@@ -451,6 +469,11 @@ __device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
           i64 ep = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
           i64 disc = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
           v = ep * (10000 - disc);
+        } else if (f.off == MZ_COMPUTE_DIV_I64) {
+          i64 a = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 b = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          if (b == 0) return 2;  // -> error stream
+          v = a / b;
         }
         memcpy(dst, &v, 8);
         dst += 8;
@@ -461,7 +484,7 @@ __device__ bool d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
       }
     }
   }
-  return true;
+  return 1;
 }
 
 // --------------------------------------------------------------- probe
@@ -479,46 +502,60 @@ struct BatchList {
 enum ProbeMode { PM_JOIN = 0, PM_HALF_LE = 1, PM_HALF_LT = 2 };
 
 // Count the output rows of one (delta row, batch) pair: walk the key's
-// val range applying the closure filter and the time mode.
-__device__ __forceinline__ u32 d_count_pair(const DevBatch &b, int allpass,
+// val range applying the closure filter and the time mode. Returns
+// ok_count | err_count << 32 (err = closure evaluation errors, the
+// could_error split of linear_join.rs:495-541).
+__device__ __forceinline__ u64 d_count_pair(const DevBatch &b, int allpass,
                                             const u64 *key, const u8 *dv,
                                             u64 t, u64 kvr, int mode,
                                             int swap,
                                             const mz_gpu_closure &cl,
                                             u32 lvb) {
-  u32 c = 0;
+  u32 c = 0, e = 0;
   for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
     const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
     const u8 *v1 = swap ? lv : dv;
     const u8 *v2 = swap ? dv : lv;
-    if (!d_closure_apply(&cl, key, v1, v2, nullptr, nullptr)) continue;
+    int cls = d_closure_apply(&cl, key, v1, v2, nullptr, nullptr);
+    if (cls == 0) continue;
     u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+    u32 m;
     if (mode == PM_JOIN || allpass) {
-      c += hi - lo;
+      m = hi - lo;
     } else {
+      m = 0;
       for (u32 u = lo; u < hi; u++) {
         u64 t2 = b.times[u];
-        c += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+        m += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
       }
     }
+    if (cls == 1)
+      c += m;
+    else
+      e += m;
   }
-  return c;
+  return (u64)c | ((u64)e << 32);
 }
 
-// Emit the counted rows at queue offset o (fields written straight to
-// global — closure filters run before any field write). Returns new o.
-__device__ __forceinline__ u64 d_emit_pair(const DevBatch &b, int allpass,
-                                           const u64 *key, const u8 *dv,
-                                           u64 t, i64 d1, u64 kvr, int mode,
-                                           int swap,
-                                           const mz_gpu_closure &cl,
-                                           u32 lvb, u32 okw, u32 ovb, u64 o,
-                                           u64 *okeys, u8 *ovals,
-                                           u64 *otimes, i64 *odiffs) {
+// Emit the counted rows at queue offsets *o (ok) / *eo (err) — ok fields
+// written straight to global (closure filters run before any field
+// write); error rows carry (code, time, d1*d2).
+__device__ __forceinline__ void d_emit_pair(const DevBatch &b, int allpass,
+                                            const u64 *key, const u8 *dv,
+                                            u64 t, i64 d1, u64 kvr,
+                                            int mode, int swap,
+                                            const mz_gpu_closure &cl,
+                                            u32 lvb, u32 okw, u32 ovb,
+                                            u64 *o, u64 *okeys, u8 *ovals,
+                                            u64 *otimes, i64 *odiffs,
+                                            u64 *eo, u64 *ecodes,
+                                            u64 *etimes, i64 *ediffs) {
   for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
     const u8 *lv = b.vals ? b.vals + (u64)j * lvb : nullptr;
     const u8 *v1 = swap ? lv : dv;
     const u8 *v2 = swap ? dv : lv;
+    int cls = d_closure_apply(&cl, key, v1, v2, nullptr, nullptr);
+    if (cls == 0) continue;
     for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
       u64 tout;
       if (allpass) {
@@ -531,15 +568,20 @@ __device__ __forceinline__ u64 d_emit_pair(const DevBatch &b, int allpass,
         if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
         tout = t;
       }
-      if (!d_closure_apply(&cl, key, v1, v2, okeys + o * okw,
-                           ovals + o * ovb))
+      if (cls == 2) {
+        ecodes[*eo] = MZ_ERR_DIVISION_BY_ZERO;
+        etimes[*eo] = tout;
+        ediffs[*eo] = wmul(d1, b.diffs[u]);
+        (*eo)++;
         continue;
-      otimes[o] = tout;
-      odiffs[o] = wmul(d1, b.diffs[u]);
-      o++;
+      }
+      (void)d_closure_apply(&cl, key, v1, v2, okeys + *o * okw,
+                            ovals + *o * ovb);
+      otimes[*o] = tout;
+      odiffs[*o] = wmul(d1, b.diffs[u]);
+      (*o)++;
     }
   }
-  return o;
 }
 
 // Wave-aggregated output-queue reservation: exclusive prefix of c across
@@ -588,8 +630,10 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
                              const u64 *dtimes, const i64 *ddiffs, u64 n,
                              u32 kw, u32 lvb, BatchList bl, int mode,
                              int swap, const mz_gpu_closure cl, u64 cap,
-                             unsigned long long *ctr, u64 *okeys, u8 *ovals,
-                             u64 *otimes, i64 *odiffs) {
+                             u64 ecap, unsigned long long *ctr,
+                             u64 *okeys, u8 *ovals, u64 *otimes,
+                             i64 *odiffs, u64 *ecodes, u64 *etimes,
+                             i64 *ediffs) {
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   u64 total = n * (u64)bl.n;
   u64 stride = (u64)gridDim.x * blockDim.x;
@@ -602,7 +646,7 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
   // lockstep convergence off the per-pair critical path.
   for (u64 it0 = 0; it0 < iters; it0 += PROBE_TILE) {
     u64 kvr[PROBE_TILE];
-    u32 cc[PROBE_TILE];
+    u64 cc[PROBE_TILE];
     // Phase A: hash lookups (independent random lines in flight)
 #pragma unroll
     for (int tt = 0; tt < PROBE_TILE; tt++) {
@@ -616,8 +660,9 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
                                     dkeys + (idx % n) * kw, kw);
       }
     }
-    // Phase B: count matches per slot (filters applied)
-    u32 csum = 0;
+    // Phase B: count matches per slot (filters applied); low 32 bits ok
+    // rows, high 32 error rows
+    u32 csum = 0, esum = 0;
 #pragma unroll
     for (int tt = 0; tt < PROBE_TILE; tt++) {
       if (kvr[tt] == ~0ull) continue;
@@ -627,23 +672,27 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
       cc[tt] = d_count_pair(bl.b[bi], bl.allpass[bi], dkeys + i * kw,
                             dvals ? dvals + i * dvb : nullptr, dtimes[i],
                             kvr[tt], mode, swap, cl, lvb);
-      csum += cc[tt];
+      csum += (u32)cc[tt];
+      esum += (u32)(cc[tt] >> 32);
     }
     u64 base = wave_reserve(ctr, csum, lane);
-    if (csum == 0 || base + csum > cap) continue;
+    u64 ebase = wave_reserve(ctr + 1, esum, lane);
+    if ((csum == 0 && esum == 0) || base + csum > cap ||
+        ebase + esum > ecap)
+      continue;
     // Phase C: emit from L2-hot lines, fields written straight to the
     // reserved global slots (closure filters run before any field write)
-    u64 o = base;
+    u64 o = base, eo = ebase;
 #pragma unroll
     for (int tt = 0; tt < PROBE_TILE; tt++) {
       if (cc[tt] == 0) continue;
       u64 idx = start + (it0 + tt) * stride;
       u64 i = idx % n;
       int bi = (int)(idx / n);
-      o = d_emit_pair(bl.b[bi], bl.allpass[bi], dkeys + i * kw,
-                      dvals ? dvals + i * dvb : nullptr, dtimes[i],
-                      ddiffs[i], kvr[tt], mode, swap, cl, lvb, okw, ovb, o,
-                      okeys, ovals, otimes, odiffs);
+      d_emit_pair(bl.b[bi], bl.allpass[bi], dkeys + i * kw,
+                  dvals ? dvals + i * dvb : nullptr, dtimes[i], ddiffs[i],
+                  kvr[tt], mode, swap, cl, lvb, okw, ovb, &o, okeys, ovals,
+                  otimes, odiffs, &eo, ecodes, etimes, ediffs);
     }
   }
 }
@@ -699,9 +748,10 @@ __global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
                               const u64 *dtimes, const i64 *ddiffs, u64 n,
                               u32 kw, u32 lvb, DevBatch b, int allpass,
                               int mode, int swap, const mz_gpu_closure cl,
-                              const u64 *bounds, u64 cap,
+                              const u64 *bounds, u64 cap, u64 ecap,
                               unsigned long long *ctr, u64 *okeys,
-                              u8 *ovals, u64 *otimes, i64 *odiffs) {
+                              u8 *ovals, u64 *otimes, i64 *odiffs,
+                              u64 *ecodes, u64 *etimes, i64 *ediffs) {
   __shared__ u64 lk[MERGE_LDSW];
   u32 okw = cl.out.key_words, ovb = cl.out.val_bytes;
   u64 r0 = (u64)blockIdx.x * MERGE_DROWS;
@@ -718,8 +768,8 @@ __global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
   u32 lane = threadIdx.x & 63;
   const int RPT = MERGE_DROWS / BLK;  // rows per thread
   u64 kvr[RPT];
-  u32 cc[RPT];
-  u32 csum = 0;
+  u64 cc[RPT];
+  u32 csum = 0, esum = 0;
   for (int q = 0; q < RPT; q++) {
     u64 i = r0 + threadIdx.x + (u64)q * blockDim.x;  // coalesced
     kvr[q] = ~0ull;
@@ -741,20 +791,24 @@ __global__ void k_probe_merge(const u64 *dkeys, const u8 *dvals, u32 dvb,
         cc[q] = d_count_pair(b, allpass, key,
                              dvals ? dvals + i * dvb : nullptr, dtimes[i],
                              kvr[q], mode, swap, cl, lvb);
-        csum += cc[q];
+        csum += (u32)cc[q];
+        esum += (u32)(cc[q] >> 32);
       }
     }
   }
   u64 base_o = wave_reserve(ctr, csum, lane);
-  if (csum == 0 || base_o + csum > cap) return;
-  u64 o = base_o;
+  u64 base_e = wave_reserve(ctr + 1, esum, lane);
+  if ((csum == 0 && esum == 0) || base_o + csum > cap ||
+      base_e + esum > ecap)
+    return;
+  u64 o = base_o, eo = base_e;
   for (int q = 0; q < RPT; q++) {
     if (!cc[q]) continue;
     u64 i = r0 + threadIdx.x + (u64)q * blockDim.x;
-    o = d_emit_pair(b, allpass, dkeys + i * kw,
-                    dvals ? dvals + i * dvb : nullptr, dtimes[i], ddiffs[i],
-                    kvr[q], mode, swap, cl, lvb, okw, ovb, o, okeys, ovals,
-                    otimes, odiffs);
+    d_emit_pair(b, allpass, dkeys + i * kw,
+                dvals ? dvals + i * dvb : nullptr, dtimes[i], ddiffs[i],
+                kvr[q], mode, swap, cl, lvb, okw, ovb, &o, okeys, ovals,
+                otimes, odiffs, &eo, ecodes, etimes, ediffs);
   }
 }
 
@@ -2008,6 +2062,7 @@ mz_gpu_out *make_out(u64 *k, u8 *v, u64 *t, i64 *d, u64 n, u32 kw, u32 vb) {
   o->vals = v;
   o->times = t;
   o->diffs = d;
+  o->pub_ = mz_gpu_out{};  // err fields start empty
   o->pub_.keys = k;
   o->pub_.vals = v;
   o->pub_.times = t;
@@ -2017,6 +2072,15 @@ mz_gpu_out *make_out(u64 *k, u8 *v, u64 *t, i64 *d, u64 n, u32 kw, u32 vb) {
   o->pub_.schema.key_words = kw;
   o->pub_.schema.val_bytes = vb;
   return &o->pub_;
+}
+
+// Attach a consolidated error stream to an out-batch (ownership moves).
+void out_attach_errs(mz_gpu_out *out, u64 *codes, u64 *times, i64 *diffs,
+                     u64 n) {
+  out->err_n = n;
+  out->err_codes = codes;
+  out->err_times = times;
+  out->err_diffs = diffs;
 }
 
 // Core consolidation: sort + group + sum + compact. Returns owned device
@@ -2846,9 +2910,26 @@ int mz_gpu_arr_stats(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t *n_batches,
 void mz_gpu_out_release(mz_gpu_ctx *c, mz_gpu_out *o) {
   OutOwned *oo = reinterpret_cast<OutOwned *>(o);
   for (void *p : {(void *)oo->keys, (void *)oo->vals, (void *)oo->times,
-                  (void *)oo->diffs})
+                  (void *)oo->diffs, (void *)o->err_codes,
+                  (void *)o->err_times, (void *)o->err_diffs})
     dfree(&c->impl, p);
   delete oo;
+}
+
+int mz_gpu_out_err_to_host(mz_gpu_ctx *c, const mz_gpu_out *o,
+                           uint64_t *codes, uint64_t *times,
+                           int64_t *diffs) {
+  Ctx *ctx = &c->impl;
+  u64 n = o->err_n;
+  if (n == 0) return 0;
+  HIP_CHECK(hipMemcpyAsync(codes, o->err_codes, n * 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(times, o->err_times, n * 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(diffs, o->err_diffs, n * 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return 0;
 }
 
 int mz_gpu_out_to_host(mz_gpu_ctx *c, const mz_gpu_out *o, uint64_t *keys,
@@ -2981,12 +3062,16 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   // emit-ratio hint, relaunch once with the exact count on overflow.
   u64 cap = (lookup->probe_cap_hint ? lookup->probe_cap_hint + 1 : 2) * n +
             1024;
-  unsigned long long *ctr = (unsigned long long *)S.get(8);
-  fill_u64(ctx, (u64 *)ctr, 1, 0);
+  u64 ecap = n / 4 + 1024;  // error rows are exceptional; exact relaunch
+  unsigned long long *ctr = (unsigned long long *)S.get(16);
+  fill_u64(ctx, (u64 *)ctr, 2, 0);  // [0] ok rows, [1] err rows
   u64 *pk = dnew<u64>(ctx, cap * okw);
   u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
   u64 *pt = dnew<u64>(ctx, cap);
   i64 *pd = dnew<i64>(ctx, cap);
+  u64 *ek = dnew<u64>(ctx, ecap);
+  u64 *et = dnew<u64>(ctx, ecap);
+  i64 *ed = dnew<i64>(ctx, ecap);
   u64 *mbounds = nullptr;
   if (!mts.empty()) {
     mbounds = (u64 *)S.get(mts.size() * mgrid * 2 * 8);
@@ -3002,30 +3087,37 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                          ctx->stream, d.keys, d.vals, stream_vb, d.times,
                          d.diffs, n, kw, lvb, mts[mi].b,
                          (int)mts[mi].allpass, mode, swap, *cl,
-                         mbounds + mi * mgrid * 2, cap, ctr, pk, pv, pt,
-                         pd);
+                         mbounds + mi * mgrid * 2, cap, ecap, ctr, pk, pv,
+                         pt, pd, ek, et, ed);
     if (blw.n)
       hipLaunchKernelGGL(k_probe_walk, dim3(ngrid(nb2)), dim3(BLK), 0,
                          ctx->stream, d.keys, d.vals, stream_vb, d.times,
                          d.diffs, n, kw, lvb, blw, mode, swap, *cl, cap,
-                         ctr, pk, pv, pt, pd);
+                         ecap, ctr, pk, pv, pt, pd, ek, et, ed);
   };
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   launch_probes();
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
-  unsigned long long M = 0;
-  HIP_CHECK(hipMemcpyAsync(&M, ctr, 8, hipMemcpyDeviceToHost, ctx->stream));
+  unsigned long long MB[2] = {0, 0};
+  HIP_CHECK(hipMemcpyAsync(MB, ctr, 16, hipMemcpyDeviceToHost,
+                           ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  u64 M = MB[0], E = MB[1];
   u64 launches = 1;
-  if (M > cap) {  // rare: queue overflow — exact-size relaunch
-    for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
+  if (M > cap || E > ecap) {  // rare: queue overflow — exact relaunch
+    for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd,
+                    (void *)ek, (void *)et, (void *)ed})
       dfree(ctx, p);
-    cap = M;
+    cap = std::max<u64>(M, 1);
+    ecap = std::max<u64>(E, 1);
     pk = dnew<u64>(ctx, cap * okw);
     pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
     pt = dnew<u64>(ctx, cap);
     pd = dnew<i64>(ctx, cap);
-    fill_u64(ctx, (u64 *)ctr, 1, 0);
+    ek = dnew<u64>(ctx, ecap);
+    et = dnew<u64>(ctx, ecap);
+    ed = dnew<i64>(ctx, ecap);
+    fill_u64(ctx, (u64 *)ctr, 2, 0);
     if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
     launch_probes();
     if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
@@ -3048,9 +3140,23 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
         n * (8ull * kw + stream_vb + 16) + n * 128ull * (u64)bl.n +
         M * (lvb + 16ull) + M * (8ull * okw + ovb + 16);
   }
+  // error stream: consolidate by (code, time) — the err collection is a
+  // first-class update stream in the reference (linear_join.rs:516-541)
+  u64 *cek = nullptr, *cet = nullptr;
+  i64 *ced = nullptr;
+  u64 Ec = 0;
+  if (E) {
+    DevUpdates epin{ek, nullptr, et, ed, E};
+    u8 *unused_v;
+    consolidate_dev(ctx, 1, 0, epin, &cek, &unused_v, &cet, &ced, &Ec);
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    dfree(ctx, unused_v);
+  }
+  for (void *p : {(void *)ek, (void *)et, (void *)ed}) dfree(ctx, p);
   if (!consolidate_out) {
-    // raw emitted pairs (deterministic order; consumer consolidates)
+    // raw emitted pairs (deterministic content; consumer consolidates)
     *out = make_out(pk, pv, pt, pd, M, okw, ovb);
+    out_attach_errs(*out, cek, cet, ced, Ec);
     return 0;
   }
   // consolidate the emitted pairs
@@ -3065,6 +3171,7 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd})
     dfree(ctx, (p));
   *out = make_out(ok, ov, ot, od, Mc, okw, ovb);
+  out_attach_errs(*out, cek, cet, ced, Ec);
   return 0;
 }
 

@@ -248,8 +248,10 @@ inline const u8 *cl_src(const u64 *key, const u8 *v1, const u8 *v2, u8 src) {
   }
 }
 
-bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
-                   const u8 *v2, u64 *out_key, u8 *out_val) {
+// 0 = filtered out, 1 = ok, 2 = evaluation ERROR (division by zero) —
+// the could_error ok/err split of linear_join.rs:495-541.
+int closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
+                  const u8 *v2, u64 *out_key, u8 *out_val) {
   for (u32 i = 0; i < cl->n_filters; i++) {
     const auto &f = cl->filters[i];
     if (f.src == MZ_SRC_COMPUTE) {
@@ -262,7 +264,7 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
         i128 S;
         std::memcpy(&S, slot, 16);
         i64 C = read_int(slot + 24, 8);
-        if (!(C > 0 && (i128)5 * q * C < S)) return false;
+        if (!(C > 0 && (i128)5 * q * C < S)) return 0;
       }
       continue;
     }
@@ -276,7 +278,7 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
       case MZ_CMP_EQ: ok = x == f.imm; break;
       default: ok = x != f.imm; break;
     }
-    if (!ok) return false;
+    if (!ok) return 0;
   }
   u8 *outs[2] = {(u8 *)out_key, out_val};
   for (int which = 0; which < 2; which++) {
@@ -293,6 +295,11 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
           i64 ep = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
           i64 disc = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
           v = ep * (10000 - disc);
+        } else if (f.off == MZ_COMPUTE_DIV_I64) {
+          i64 a = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 b = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          if (b == 0) return 2;  // -> error stream
+          v = a / b;
         }  // MZ_COMPUTE_CONST0 leaves v = 0
         std::memcpy(dst, &v, 8);
         dst += 8;
@@ -302,7 +309,7 @@ bool closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
       }
     }
   }
-  return true;
+  return 1;
 }
 
 // ------------------------------------------------------ linear join (oracle)
@@ -442,12 +449,16 @@ struct ValueHistory {
 // the delta came from input 2 (history1 is then the trace side) — result_fn
 // argument order is always (key, val-of-input1, val-of-input2).
 void join_key(const mz_gpu_closure *cl, const u64 *key, EditList &h1,
-              EditList &h2, Cols &out, const Schema &os) {
+              EditList &h2, Cols &out, const Schema &os,
+              std::vector<std::tuple<u64, u64, i64>> *errs = nullptr) {
   std::vector<u64> okey(os.kw);
   std::vector<u8> oval(os.vb ? os.vb : 1);
   auto emit = [&](const u8 *v1, const u8 *v2, u64 t, i64 r) {
-    if (closure_apply(cl, key, v1, v2, okey.data(), oval.data()))
+    int cls = closure_apply(cl, key, v1, v2, okey.data(), oval.data());
+    if (cls == 1)
       out.push(okey.data(), os.kw, oval.data(), os.vb, t, r);
+    else if (cls == 2 && errs)
+      errs->push_back({MZ_ERR_DIVISION_BY_ZERO, t, r});
   };
   if (h1.len() < 10 || h2.len() < 10) {
     // simple strategy: full cross product (:755-767)
@@ -751,6 +762,8 @@ struct Out {
   mz_gpu_out pub_;
   Cols cols;
   Schema schema;
+  std::vector<u64> err_codes, err_times;
+  std::vector<i64> err_diffs;
 };
 
 }  // namespace
@@ -865,10 +878,49 @@ JoinOp *orc_join_create(orc_ctx *c, Arr *a1, Arr *a2,
   return p;
 }
 
+using ErrRows = std::vector<std::tuple<u64, u64, i64>>;  // code,time,diff
+
+static void consolidate_errs(ErrRows &e) {
+  std::sort(e.begin(), e.end(),
+            [](const auto &a, const auto &b) {
+              if (std::get<0>(a) != std::get<0>(b))
+                return std::get<0>(a) < std::get<0>(b);
+              return std::get<1>(a) < std::get<1>(b);
+            });
+  ErrRows out;
+  for (auto &r : e) {
+    if (!out.empty() && std::get<0>(out.back()) == std::get<0>(r) &&
+        std::get<1>(out.back()) == std::get<1>(r))
+      std::get<2>(out.back()) =
+          (i64)((u64)std::get<2>(out.back()) + (u64)std::get<2>(r));
+    else
+      out.push_back(r);
+  }
+  out.erase(std::remove_if(out.begin(), out.end(),
+                           [](const auto &r) { return std::get<2>(r) == 0; }),
+            out.end());
+  e = std::move(out);
+}
+
+static Out *attach_errs(Out *o, ErrRows &&e) {
+  consolidate_errs(e);
+  for (auto &[code, t, d] : e) {
+    o->err_codes.push_back(code);
+    o->err_times.push_back(t);
+    o->err_diffs.push_back(d);
+  }
+  o->pub_.err_n = o->err_codes.size();
+  o->pub_.err_codes = o->err_codes.data();
+  o->pub_.err_times = o->err_times.data();
+  o->pub_.err_diffs = o->err_diffs.data();
+  return o;
+}
+
 static Out *make_out(Cols &&c, const Schema &s) {
   Out *o = new Out();
   o->cols = std::move(c);
   o->schema = s;
+  o->pub_ = mz_gpu_out{};
   o->pub_.keys = o->cols.keys.data();
   o->pub_.vals = o->cols.vals.data();
   o->pub_.times = o->cols.times.data();
@@ -896,6 +948,7 @@ int orc_join_push(orc_ctx *c, JoinOp *op, int side, const mz_gpu_updates *u,
   u64 meet = u->lower;
   Schema os = {op->cl.out.key_words, op->cl.out.val_bytes};
   Cols result;
+  ErrRows errs;
   EditList el_delta, el_trace;
   // merge scan over delta's distinct keys (start_work, :644-663): the delta
   // cursor drives; trace cursors seek.
@@ -918,12 +971,13 @@ int orc_join_push(orc_ctx *c, JoinOp *op, int side, const mz_gpu_updates *u,
     editlist_load(ts, el_trace, tr, meet);
     if (el_delta.len() == 0 || el_trace.len() == 0) continue;
     if (side == 1)
-      join_key(&op->cl, key, el_delta, el_trace, result, os);
+      join_key(&op->cl, key, el_delta, el_trace, result, os, &errs);
     else
-      join_key(&op->cl, key, el_trace, el_delta, result, os);
+      join_key(&op->cl, key, el_trace, el_delta, result, os, &errs);
   }
   consolidate(os, result);
-  *out = &make_out(std::move(result), os)->pub_;
+  *out = &attach_errs(make_out(std::move(result), os),
+                      std::move(errs))->pub_;
   return 0;
 }
 
@@ -938,6 +992,7 @@ int orc_halfjoin(orc_ctx *c, Arr *lookup, const mz_gpu_updates *u,
   const Schema &ls = lookup->schema;
   Schema os = {cl->out.key_words, cl->out.val_bytes};
   Cols result;
+  ErrRows errs;
   std::vector<u64> okey(os.kw);
   std::vector<u8> oval(os.vb ? os.vb : 1);
   for (u64 i = 0; i < u->n; i++) {
@@ -951,16 +1006,20 @@ int orc_halfjoin(orc_ctx *c, Arr *lookup, const mz_gpu_updates *u,
         u64 t2 = b->cols.times[j];
         if (le ? (t2 <= t) : (t2 < t)) {
           const u8 *lval = ls.vb ? b->cols.vals.data() + j * ls.vb : nullptr;
-          if (closure_apply(cl, key, sval, lval, okey.data(), oval.data())) {
-            i64 d = (i64)((u64)d1 * (u64)b->cols.diffs[j]);
+          int cls = closure_apply(cl, key, sval, lval, okey.data(),
+                                  oval.data());
+          i64 d = (i64)((u64)d1 * (u64)b->cols.diffs[j]);
+          if (cls == 1)
             result.push(okey.data(), os.kw, oval.data(), os.vb, t, d);
-          }
+          else if (cls == 2)
+            errs.push_back({MZ_ERR_DIVISION_BY_ZERO, t, d});
         }
       }
     }
   }
   consolidate(os, result);
-  *out = &make_out(std::move(result), os)->pub_;
+  *out = &attach_errs(make_out(std::move(result), os),
+                      std::move(errs))->pub_;
   return 0;
 }
 
@@ -975,7 +1034,8 @@ int orc_map(orc_ctx *c, const mz_gpu_schema *in, const mz_gpu_updates *u,
   for (u64 i = 0; i < u->n; i++) {
     const u64 *key = u->keys + i * s.kw;
     const u8 *val = s.vb ? u->vals + i * s.vb : nullptr;
-    if (closure_apply(cl, key, val, nullptr, okey.data(), oval.data()))
+    if (closure_apply(cl, key, val, nullptr, okey.data(),
+                      oval.data()) == 1)
       result.push(okey.data(), os.kw, oval.data(), os.vb, u->times[i],
                   u->diffs[i]);
   }

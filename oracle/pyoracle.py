@@ -101,7 +101,18 @@ class OracleCtx:
         self.close()
 
     def _take(self, outp):
-        res = out_to_numpy(outp.contents)
+        import numpy as np
+        ob = outp.contents
+        res = out_to_numpy(ob)
+        en = ob.err_n
+        ecodes = np.empty(en, np.uint64)
+        etimes = np.empty(en, np.uint64)
+        ediffs = np.empty(en, np.int64)
+        for i in range(en):  # host pointers: direct reads
+            ecodes[i] = ob.err_codes[i]
+            etimes[i] = ob.err_times[i]
+            ediffs[i] = ob.err_diffs[i]
+        self.last_errs = (ecodes, etimes, ediffs)
         self.lib.orc_out_release(self.ctx, outp)
         return res
 
