@@ -221,6 +221,117 @@ class HostOut:
 
 
 @dataclass
+class LinearStagePlan:
+    """One stage of a linear join (compute-types/src/plan/join/
+    linear_join.rs:50-65): probe `lookup_relation` with the accumulated
+    stream; `stream_key` (when set) is the key-prep closure — the
+    stream_key exprs + stream_thinning re-keying of linear_join.rs:349-385,
+    evaluated via mz_gpu_map; `closure` is the stage's JoinClosure over
+    (key, stream val, lookup val)."""
+    lookup_relation: str
+    closure: abi.Closure
+    stream_key: Optional[abi.Closure] = None
+    stream_key_words: int = 1
+    stream_val_bytes: int = 0
+
+
+@dataclass
+class LinearJoinPlan:
+    """compute-types/src/plan/join/linear_join.rs:27-42: a source
+    relation, an optional initial closure, a sequence of stages, an
+    optional final closure. `source_relation` names the arrangement that
+    plays arranged1 of the FIRST stage's mz_join_core (the reference's
+    source_key arrangement); interior stages arrange the accumulated
+    stream in op-owned "JoinStage" arrangements (linear_join.rs:461)."""
+    source_relation: str
+    stage_plans: list
+    initial_closure: Optional[abi.Closure] = None
+    final_closure: Optional[abi.Closure] = None
+
+
+class LinearJoinOp:
+    """render_join / differential_join (linear_join.rs:204-543): executes
+    a LinearJoinPlan over the engine. Each stage is the two-sided
+    mz_join_core between the stream's arrangement and the lookup
+    arrangement; `step` drives one timestamp's deltas through the
+    drain-input-1-first discipline (mz_join_core.rs:237-368):
+    the stream delta probes the lookup BEFORE the lookup's delta is
+    installed, then the lookup delta probes the updated stream side —
+    each concurrent pair counted exactly once."""
+
+    def __init__(self, ctx, arrangements, plan: LinearJoinPlan):
+        self.ctx = ctx
+        self.arrangements = arrangements
+        self.plan = plan
+        self.joins = []
+        self.stage_arrs = []  # op-owned interior stream arrangements
+        stream_arr = arrangements[plan.source_relation]
+        for i, st in enumerate(plan.stage_plans):
+            lookup = arrangements[st.lookup_relation]
+            j = ctx.join_create(stream_arr, lookup, st.closure)
+            self.joins.append(j)
+            if i + 1 < len(plan.stage_plans):
+                nxt = plan.stage_plans[i + 1]
+                stream_arr = ctx.arr_create(
+                    abi.schema(nxt.stream_key_words, nxt.stream_val_bytes))
+                self.stage_arrs.append(stream_arr)
+
+    def _map(self, cl, cols, t, in_kw, in_vb):
+        k, v, tm, d = cols
+        u = abi.make_updates(k, v, tm, d, t, t + 1)
+        return self.ctx.map(abi.schema(in_kw, in_vb), u, cl)
+
+    def step(self, t, source_cols, lookup_deltas=None):
+        """One timestamp: `source_cols` = (keys, vals, times, diffs) of
+        the source relation's delta (the caller has ALREADY installed it
+        in the source arrangement — it is shared state, like the
+        reference's CollectionBundle arrangements); `lookup_deltas` maps
+        lookup relation name -> (updates_desc, already_installed: bool).
+        The op installs not-yet-installed lookup deltas at the correct
+        drain point. Returns the final output columns (host)."""
+        ctx = self.ctx
+        lookup_deltas = lookup_deltas or {}
+        plan = self.plan
+        cur = source_cols
+        if plan.initial_closure is not None:
+            st0 = plan.stage_plans[0]
+            cur = self._map(plan.initial_closure, cur, t,
+                            st0.stream_key_words, st0.stream_val_bytes)
+        for i, st in enumerate(plan.stage_plans):
+            if st.stream_key is not None:
+                cur = self._map(st.stream_key, cur, t, st.stream_key_words,
+                                st.stream_val_bytes)
+            u = abi.make_updates(cur[0], cur[1], cur[2], cur[3], t, t + 1)
+            if i > 0:
+                # interior stage: arrange the accumulated stream
+                ctx.arr_insert(self.stage_arrs[i - 1], u)
+            out1 = ctx.join_push(self.joins[i], 1, u)  # probes lookup OLD
+            ld = lookup_deltas.get(st.lookup_relation)
+            if ld is not None:
+                lu, installed = ld
+                if not installed:
+                    ctx.arr_insert(self.arrangements[st.lookup_relation],
+                                   lu)
+                out2 = ctx.join_push(self.joins[i], 2, lu)  # probes NEW
+                cur = tuple(np.concatenate([a, b])
+                            for a, b in zip(out1, out2))
+            else:
+                cur = out1
+        if plan.final_closure is not None:
+            lst = plan.stage_plans[-1]
+            cur = self._map(plan.final_closure, cur, t,
+                            lst.closure.out.key_words,
+                            lst.closure.out.val_bytes)
+        return cur
+
+
+def render_join(ctx, arrangements, plan: LinearJoinPlan) -> LinearJoinOp:
+    """Context::render_join dispatch for Plan::Join{Linear}
+    (render.rs:1321-1350)."""
+    return LinearJoinOp(ctx, arrangements, plan)
+
+
+@dataclass
 class ReducePlan:
     """render_reduce / AccumulablePlan surface (reduce.rs:71,
     plan/reduce.rs:233)."""
