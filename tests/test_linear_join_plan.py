@@ -14,13 +14,15 @@ F = abi.field
 
 
 def _plan():
-    # A(k1) val=[k2 i64, a i64]; B(k2) val=[k3 i64]; C(k3) val=[c i64]
-    # stage1: probe B with key k2; out key := B.val(k3), val := [a]
+    # A arranged by k2, val=[a i64] (the source_key arrangement matches
+    # stage 1's stream key, as the reference requires); B(k2) val=[k3];
+    # C(k3) val=[c]
+    # stage1: probe B on k2; out key := B.val (k3), val := [a]
     cl1 = abi.closure(
         [], [F(abi.MZ_SRC_VAL_LOOKUP, 0, 8)],
-        [F(abi.MZ_SRC_VAL_STREAM, 8, 8)],
+        [F(abi.MZ_SRC_VAL_STREAM, 0, 8)],
         abi.schema(1, 8))
-    # stage2: probe C with key k3; out key := k3, val := [a, c]
+    # stage2: probe C on k3; out key := k3, val := [a, c]
     cl2 = abi.closure(
         [], [F(abi.MZ_SRC_KEY, 0, 8)],
         [F(abi.MZ_SRC_VAL_STREAM, 0, 8), F(abi.MZ_SRC_VAL_LOOKUP, 0, 8)],
@@ -29,7 +31,7 @@ def _plan():
         source_relation="A",
         stage_plans=[
             LinearStagePlan("B", cl1, stream_key_words=1,
-                            stream_val_bytes=16),
+                            stream_val_bytes=8),
             LinearStagePlan("C", cl2, stream_key_words=1,
                             stream_val_bytes=8),
         ])
@@ -51,8 +53,8 @@ def _naive(A, B, C):
     for k3, c in zip(*C):
         cmap.setdefault(int(k3), []).append(int(c))
     out = Counter()
-    ak, av2, aa = A
-    for k1, k2, a in zip(ak, av2, aa):
+    ak2, aa = A
+    for k2, a in zip(ak2, aa):
         for k3 in bmap.get(int(k2), []):
             for c in cmap.get(int(k3), []):
                 out[(int(k3), int(a), int(c))] += 1
@@ -61,7 +63,7 @@ def _naive(A, B, C):
 
 def _run(ctx, seed=5):
     rng = np.random.default_rng(seed)
-    arrs = {"A": ctx.arr_create(abi.schema(1, 16)),
+    arrs = {"A": ctx.arr_create(abi.schema(1, 8)),
             "B": ctx.arr_create(abi.schema(1, 8)),
             "C": ctx.arr_create(abi.schema(1, 8))}
     op = render_join(ctx, arrs, _plan())
@@ -72,20 +74,17 @@ def _run(ctx, seed=5):
                              np.zeros(len(k), np.uint64),
                              np.ones(len(k), np.int64), 0, 1)
         ctx.arr_insert(arrs[name], u)
-    # source delta at t=1
+    # source delta at t=1 (A keyed by k2)
     n = 400
-    ak = rng.integers(0, 40, n).astype(np.int64)
     ak2 = rng.integers(0, 50, n).astype(np.int64)
     aa = rng.integers(0, 1000, n).astype(np.int64)
-    av = np.zeros((n, 16), np.uint8)
-    av[:, :8] = ak2.reshape(-1, 1).view(np.uint8).reshape(n, 8)
-    av[:, 8:] = aa.reshape(-1, 1).view(np.uint8).reshape(n, 8)
+    av = aa.reshape(-1, 1).view(np.uint8).reshape(n, 8)
     times = np.full(n, 1, np.uint64)
     diffs = np.ones(n, np.int64)
-    au = abi.make_updates(ak, av, times, diffs, 1, 2)
+    au = abi.make_updates(ak2, av, times, diffs, 1, 2)
     ctx.arr_insert(arrs["A"], au)
-    cols = op.step(1, (ak, av.reshape(-1), times, diffs))
-    return cols, (ak, ak2, bk, bv, ck, cv, aa)
+    cols = op.step(1, (ak2, av.reshape(-1), times, diffs))
+    return cols, (ak2, bk, bv, ck, cv, aa)
 
 
 def test_two_stage_plan_oracle_vs_naive():
@@ -93,7 +92,7 @@ def test_two_stage_plan_oracle_vs_naive():
 
     from pyoracle import OracleCtx
     ctx = OracleCtx()
-    cols, (ak, ak2, bk, bv, ck, cv, aa) = _run(ctx)
+    cols, (ak2, bk, bv, ck, cv, aa) = _run(ctx)
     k, v, t, d = cols
     v = np.asarray(v).reshape(-1, 16)
     got = Counter()
@@ -101,7 +100,7 @@ def test_two_stage_plan_oracle_vs_naive():
         a = int(v[i][:8].view(np.int64)[0])
         c = int(v[i][8:].view(np.int64)[0])
         got[(int(k[i]), a, c)] += int(d[i])
-    want = _naive((ak, ak2, aa), (bk, bv), (ck, cv))
+    want = _naive((ak2, aa), (bk, bv), (ck, cv))
     ctx.close()
     assert +got == +want
 
