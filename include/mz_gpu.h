@@ -36,8 +36,14 @@ typedef struct mz_gpu_red mz_gpu_red;   /* an accumulable-reduce operator */
  * Blueprint: RowRowSpine columnar layout, src/row-spine/src/lib.rs:56-135. */
 typedef struct {
   uint32_t key_words;  /* 1 or 2; keys compared as i64-tuple ascending */
-  uint32_t val_bytes;  /* 0..64; vals compared as lexicographic bytes   */
+  uint32_t val_bytes;  /* 0..64 fixed width, or MZ_GPU_VARLEN: vals are
+                          variable-length byte strings compared
+                          lexicographically (shorter-prefix-first), the
+                          reference's byte-arena row layout
+                          (row-spine/src/lib.rs:110-135)              */
 } mz_gpu_schema;
+
+#define MZ_GPU_VARLEN 0xFFFFFFFFu
 
 /* A set of updates ((key, val), time, diff), SoA columns.
  * `on_device` = 1 when the pointers are HIP device pointers on the ctx's
@@ -60,6 +66,9 @@ typedef struct {
    * merge scan (streaming both sorted sides) instead of per-row hash
    * lookups. 0 is always safe. */
   int32_t  sorted;
+  /* VARLEN schemas only: n+1 offsets into `vals` (the byte arena); row
+   * i's val is vals[val_offs[i] .. val_offs[i+1]). NULL otherwise. */
+  const uint32_t *val_offs;
 } mz_gpu_updates;
 
 /* ----------------------------------------------------------- closures
@@ -226,9 +235,18 @@ typedef struct {
   uint64_t *err_codes;    /* MZ_ERR_* per row */
   uint64_t *err_times;
   int64_t  *err_diffs;
+  /* VARLEN schemas only (schema.val_bytes == MZ_GPU_VARLEN): n+1
+   * offsets into `vals`; val_arena_bytes = val_offs[n]. */
+  uint32_t *val_offs;
+  uint64_t val_arena_bytes;
 } mz_gpu_out;
 
 enum { MZ_ERR_DIVISION_BY_ZERO = 1 };
+
+/* VARLEN out-batches: copy the [n+1] val offsets to the host; `vals`
+ * passed to mz_gpu_out_to_host must then be sized val_arena_bytes. */
+int mz_gpu_out_voffs_to_host(mz_gpu_ctx *ctx, const mz_gpu_out *out,
+                             uint32_t *offs);
 
 /* Copy an out-batch's error rows to caller host buffers (sized err_n). */
 int mz_gpu_out_err_to_host(mz_gpu_ctx *ctx, const mz_gpu_out *out,

@@ -19,6 +19,7 @@ MZ_COMPUTE_REVENUE = 0
 MZ_COMPUTE_CONST0 = 1
 MZ_COMPUTE_Q17_QTYLT = 2
 MZ_COMPUTE_DIV_I64 = 3
+MZ_GPU_VARLEN = 0xFFFFFFFF
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 
 MZ_GPU_MAX_FILTERS = 4
@@ -47,6 +48,8 @@ class Updates(C.Structure):
         # consolidate output form) -> large-table probes take the
         # streaming merge path instead of hash lookups. 0 always safe.
         ("sorted", C.c_int32),
+        # VARLEN schemas: [n+1] offsets into the vals byte arena
+        ("val_offs", C.POINTER(C.c_uint32)),
     ]
 
 
@@ -146,6 +149,9 @@ class OutBatch(C.Structure):
         ("err_codes", C.POINTER(C.c_uint64)),
         ("err_times", C.POINTER(C.c_uint64)),
         ("err_diffs", C.POINTER(C.c_int64)),
+        # VARLEN: [n+1] offsets into vals (the byte arena)
+        ("val_offs", C.POINTER(C.c_uint32)),
+        ("val_arena_bytes", C.c_uint64),
     ]
 
 
@@ -162,7 +168,7 @@ def _as_u64(a):
 
 
 def make_updates(keys, vals, times, diffs, lower, upper, on_device=0,
-                 sorted=0):
+                 sorted=0, val_offs=None):
     """Build an Updates descriptor over numpy arrays (host memory).
 
     keys: int64/uint64 array of n*key_words; vals: uint8 array of
@@ -188,7 +194,10 @@ def make_updates(keys, vals, times, diffs, lower, upper, on_device=0,
     u.upper = upper
     u.on_device = on_device
     u.sorted = sorted
-    u._refs = (keys, vals, times, diffs)
+    if val_offs is not None:
+        val_offs = np.ascontiguousarray(val_offs, np.uint32).ravel()
+        u.val_offs = val_offs.ctypes.data_as(C.POINTER(C.c_uint32))
+    u._refs = (keys, vals, times, diffs, val_offs)
     return u
 
 

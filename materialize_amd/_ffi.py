@@ -69,6 +69,8 @@ def load():
     lib.mz_gpu_out_err_to_host.argtypes = [
         C.c_void_p, C.POINTER(OutBatch), C.POINTER(C.c_uint64),
         C.POINTER(C.c_uint64), C.POINTER(C.c_int64)]
+    lib.mz_gpu_out_voffs_to_host.argtypes = [
+        C.c_void_p, C.POINTER(OutBatch), C.POINTER(C.c_uint32)]
     lib.mz_gpu_consolidate.argtypes = [C.c_void_p, C.POINTER(Schema),
                                        C.POINTER(Updates),
                                        C.POINTER(C.POINTER(OutBatch))]
@@ -172,11 +174,22 @@ class GpuCtx:
         error rows (ok/err split) land in self.last_errs as
         (codes, times, diffs)."""
         import numpy as np
+        from ._abi import MZ_GPU_VARLEN
         ob = outp.contents
         n = ob.n
         kw, vb = ob.schema.key_words, ob.schema.val_bytes
         keys = np.empty(n * kw, np.uint64)
-        vals = np.empty(n * vb, np.uint8)
+        if vb == MZ_GPU_VARLEN:
+            vals = np.empty(int(ob.val_arena_bytes), np.uint8)
+            voffs = np.zeros(n + 1, np.uint32)
+            if n:
+                self._check(self.lib.mz_gpu_out_voffs_to_host(
+                    self.ctx, outp,
+                    voffs.ctypes.data_as(C.POINTER(C.c_uint32))))
+            self.last_voffs = voffs
+        else:
+            vals = np.empty(n * vb, np.uint8)
+            self.last_voffs = None
         times = np.empty(n, np.uint64)
         diffs = np.empty(n, np.int64)
         if n:

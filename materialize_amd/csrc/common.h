@@ -38,6 +38,7 @@ struct DevBatch {
   u32 *kv_off = nullptr;  // [n_keys + 1]
   u8 *vals = nullptr;     // [vb * n_vals], sorted (byte-lex) within key
   u32 *vu_off = nullptr;  // [n_vals + 1]
+  u32 *v_offs = nullptr;  // [n_vals + 1] VARLEN val offsets into vals
   u32 *val_key = nullptr; // [n_vals] -> owning key index
   u64 *times = nullptr;   // [n_upds]
   i64 *diffs = nullptr;   // [n_upds]

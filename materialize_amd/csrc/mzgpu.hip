@@ -2013,7 +2013,10 @@ struct DevUpdates {
   const i64 *diffs;
   u64 n;
   int sorted = 0;  // canonical (key,val,time) ascending (see mz_gpu.h)
+  const u32 *val_offs = nullptr;  // VARLEN: [n+1] offsets into vals
 };
+
+static inline bool is_varlen(u32 vb) { return vb == 0xFFFFFFFFu; }
 
 DevUpdates stage_updates(Ctx *c, const mz_gpu_updates *u, u32 kw, u32 vb) {
   DevUpdates d;
@@ -2024,24 +2027,38 @@ DevUpdates stage_updates(Ctx *c, const mz_gpu_updates *u, u32 kw, u32 vb) {
     d.vals = u->vals;
     d.times = u->times;
     d.diffs = u->diffs;
+    d.val_offs = u->val_offs;
     return d;
   }
   auto &S = (*c->scr);
   u64 *k = (u64 *)S.get(u->n * kw * 8);
-  u8 *v = vb ? (u8 *)S.get(u->n * vb) : nullptr;
+  u8 *v = nullptr;
+  u32 *vo = nullptr;
+  if (is_varlen(vb)) {
+    u64 bytes = u->n ? u->val_offs[u->n] : 0;
+    vo = (u32 *)S.get((u->n + 1) * 4);
+    v = (u8 *)S.get(std::max<u64>(bytes, 1));
+    HIP_CHECK(hipMemcpyAsync(vo, u->val_offs, (u->n + 1) * 4,
+                             hipMemcpyHostToDevice, c->stream));
+    if (bytes)
+      HIP_CHECK(hipMemcpyAsync(v, u->vals, bytes, hipMemcpyHostToDevice,
+                               c->stream));
+  } else if (vb) {
+    v = (u8 *)S.get(u->n * vb);
+    HIP_CHECK(hipMemcpyAsync(v, u->vals, u->n * vb, hipMemcpyHostToDevice,
+                             c->stream));
+  }
   u64 *t = (u64 *)S.get(u->n * 8);
   i64 *df = (i64 *)S.get(u->n * 8);
   HIP_CHECK(hipMemcpyAsync(k, u->keys, u->n * kw * 8, hipMemcpyHostToDevice,
                            c->stream));
-  if (vb)
-    HIP_CHECK(hipMemcpyAsync(v, u->vals, u->n * vb, hipMemcpyHostToDevice,
-                             c->stream));
   HIP_CHECK(hipMemcpyAsync(t, u->times, u->n * 8, hipMemcpyHostToDevice,
                            c->stream));
   HIP_CHECK(hipMemcpyAsync(df, u->diffs, u->n * 8, hipMemcpyHostToDevice,
                            c->stream));
   d.keys = k;
   d.vals = v;
+  d.val_offs = vo;
   d.times = t;
   d.diffs = df;
   return d;
@@ -2167,8 +2184,9 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
 
 void free_batch(Ctx *c, DevBatch &b) {
   for (void *p : {(void *)b.keys, (void *)b.kv_off, (void *)b.vals,
-                  (void *)b.vu_off, (void *)b.val_key, (void *)b.times,
-                  (void *)b.diffs, (void *)b.upd_val, (void *)b.hash})
+                  (void *)b.vu_off, (void *)b.v_offs, (void *)b.val_key,
+                  (void *)b.times, (void *)b.diffs, (void *)b.upd_val,
+                  (void *)b.hash})
     dfree(c, p);
   b = DevBatch();
 }
@@ -2262,9 +2280,17 @@ DevBatch build_batch(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals, u64 *times,
 // a->pending_merge — the inputs stay in the batch list (probes keep
 // using them; the merged batch holds the same logical updates) until
 // merge_install replaces them at the next flush.
+// varlen merges are defined after this namespace (they reuse the varlen
+// consolidation machinery); dispatched through this hook
+void (*merge_range_vl_fn)(Ctx *, mz_gpu_arr *, size_t, size_t) = nullptr;
+
 void merge_range(Ctx *c, mz_gpu_arr *a, size_t from, size_t to,
                  int deferred = 0) {
   if (to - from <= 1) return;
+  if (is_varlen(a->schema.vb)) {
+    merge_range_vl_fn(c, a, from, to);
+    return;
+  }
   MZ_PROF(c, "merge_range");
   auto &S = (*c->scr);
   S.reset();
@@ -2435,6 +2461,614 @@ void merge_install(Ctx *c, mz_gpu_arr *a) {
 }  // namespace
 
 // ================================================================ C ABI
+
+
+// ================================================= VARLEN arrangements
+// Variable-length vals (schema.val_bytes == MZ_GPU_VARLEN): the
+// reference's byte-arena row layout (row-spine/src/lib.rs:110-135).
+// Vals are arbitrary byte strings (embedded NULs legal) compared
+// lexicographically with shorter-prefix-first. Consolidation sorts row
+// INDICES with a comparator merge sort (rocprim::merge_sort) instead of
+// the radix pipeline — varlen is off the benchmark hot path, and the
+// comparator path is exact for any bytes. Batches carry a per-distinct-
+// val offset array (v_offs) beside the arena; the key hash index and
+// spine policy are unchanged.
+
+__device__ __forceinline__ int d_bytes_cmp(const u8 *a, u32 la,
+                                           const u8 *b, u32 lb) {
+  u32 m = la < lb ? la : lb;
+  for (u32 i = 0; i < m; i++)
+    if (a[i] != b[i]) return a[i] < b[i] ? -1 : 1;
+  if (la != lb) return la < lb ? -1 : 1;
+  return 0;
+}
+
+// (key, val, time) canonical order over varlen rows addressed by
+// (vstarts, vends) into the arena (for monotone offsets pass offs and
+// offs+1).
+struct VlRowLess {
+  const u64 *keys;
+  const u32 *vstarts, *vends;
+  const u8 *arena;
+  const u64 *times;
+  u32 kw;
+  __device__ bool operator()(u32 a, u32 b) const {
+    for (u32 w = 0; w < kw; w++) {
+      i64 x = (i64)keys[(u64)a * kw + w], y = (i64)keys[(u64)b * kw + w];
+      if (x != y) return x < y;
+    }
+    int vc = d_bytes_cmp(arena + vstarts[a], vends[a] - vstarts[a],
+                         arena + vstarts[b], vends[b] - vstarts[b]);
+    if (vc) return vc < 0;
+    return times[a] < times[b];
+  }
+};
+
+__global__ void k_vl_head_flags(const u64 *keys, u32 kw, const u32 *vstarts,
+                                const u32 *vends, const u8 *arena,
+                                const u64 *times, const u32 *perm,
+                                u32 *flags, u64 n, int with_time) {
+  GRID_STRIDE(i, n) {
+    if (i == 0) {
+      flags[0] = 1;
+      continue;
+    }
+    u32 a = perm[i], b = perm[i - 1];
+    bool neq = false;
+    for (u32 w = 0; w < kw; w++)
+      neq |= keys[(u64)a * kw + w] != keys[(u64)b * kw + w];
+    if (!neq)
+      neq = d_bytes_cmp(arena + vstarts[a], vends[a] - vstarts[a],
+                        arena + vstarts[b], vends[b] - vstarts[b]) != 0;
+    if (with_time && !neq) neq = times[a] != times[b];
+    flags[i] = neq ? 1u : 0u;
+  }
+}
+
+// val length of each SURVIVING group's representative row, at its
+// compacted output position (zeros elsewhere so the scan is exact)
+__global__ void k_vl_survivor_lens(const u32 *starts, const u32 *gid,
+                                   const u32 *nz, const u32 *nzpos, u64 n,
+                                   const u32 *perm, const u32 *vstarts,
+                                   const u32 *vends, u32 *lens) {
+  u64 G = n ? gid[n - 1] : 0;
+  GRID_STRIDE(g, n) {
+    if (g >= G) {
+      lens[g] = 0;
+      continue;
+    }
+    if (!nz[g]) continue;  // positions covered by surviving groups only
+    u32 r = perm[starts[g]];
+    lens[nzpos[g]] = vends[r] - vstarts[r];
+  }
+}
+
+__global__ void k_vl_emit_consolidated(
+    const u64 *keys, u32 kw, const u32 *vstarts, const u32 *vends,
+    const u8 *arena, const u64 *times, const u32 *perm, const u32 *starts,
+    const i64 *gsum, const u32 *nz, const u32 *nzpos, const u32 *gid,
+    u64 n, const u32 *ovoffs /*exclusive [M+1]*/, u64 *okeys, u8 *oarena,
+    u64 *otimes, i64 *odiffs) {
+  u64 G = n ? gid[n - 1] : 0;
+  GRID_STRIDE(g, G) {
+    if (!nz[g]) continue;
+    u32 o = nzpos[g];
+    u32 r = perm[starts[g]];
+    for (u32 w = 0; w < kw; w++) okeys[(u64)o * kw + w] = keys[(u64)r * kw + w];
+    otimes[o] = times[r];
+    odiffs[o] = gsum[g];
+    u32 lo = ovoffs[o], len = vends[r] - vstarts[r];
+    for (u32 b = 0; b < len; b++) oarena[lo + b] = arena[vstarts[r] + b];
+  }
+}
+
+__global__ void k_vl_change_flags(const u64 *keys, u32 kw, const u32 *voffs,
+                                  const u8 *arena, u32 *kc, u32 *vc, u64 n) {
+  GRID_STRIDE(i, n) {
+    if (i == 0) {
+      kc[0] = 1;
+      vc[0] = 1;
+      continue;
+    }
+    bool kneq = false;
+    for (u32 w = 0; w < kw; w++)
+      kneq |= keys[i * kw + w] != keys[(i - 1) * kw + w];
+    bool vneq = kneq ||
+        d_bytes_cmp(arena + voffs[i], voffs[i + 1] - voffs[i],
+                    arena + voffs[i - 1], voffs[i] - voffs[i - 1]) != 0;
+    kc[i] = kneq ? 1u : 0u;
+    vc[i] = vneq ? 1u : 0u;
+  }
+}
+
+__global__ void k_vl_distinct_lens(const u32 *vc, const u32 *voffs, u64 n,
+                                   u32 *dl) {
+  GRID_STRIDE(i, n) dl[i] = vc[i] ? voffs[i + 1] - voffs[i] : 0;
+}
+
+__global__ void k_vl_scatter(const u64 *keys, u32 kw, const u32 *voffs,
+                             const u8 *arena, const u32 *kc, const u32 *vc,
+                             const u32 *kid, const u32 *vid,
+                             const u32 *apos /*inclusive scan of dl*/,
+                             u64 n, u64 n_keys, u64 n_vals, u64 abytes,
+                             u64 *bkeys, u32 *kv_off, u32 *b_voffs,
+                             u8 *barena, u32 *vu_off, u32 *val_key,
+                             u32 *upd_val) {
+  GRID_STRIDE(i, n) {
+    u32 k = kid[i] - 1, v = vid[i] - 1;
+    upd_val[i] = v;
+    if (kc[i]) {
+      for (u32 w = 0; w < kw; w++) bkeys[(u64)k * kw + w] = keys[i * kw + w];
+      kv_off[k] = v;
+      if (k == 0) kv_off[n_keys] = (u32)n_vals;
+    }
+    if (vc[i]) {
+      u32 len = voffs[i + 1] - voffs[i];
+      u32 dst = apos[i] - len;  // exclusive position
+      for (u32 b = 0; b < len; b++) barena[dst + b] = arena[voffs[i] + b];
+      b_voffs[v] = dst;
+      if (v == 0) b_voffs[n_vals] = (u32)abytes;
+      vu_off[v] = (u32)i;
+      val_key[v] = k;
+      if (v == 0) vu_off[n_vals] = (u32)n;
+    }
+  }
+}
+
+// expansion of a varlen batch back to flat rows: lengths pass + copy pass
+__global__ void k_vl_expand_lens(DevBatch b, u32 *lens, u64 base) {
+  GRID_STRIDE(i, b.n_upds) {
+    u32 v = b.upd_val[i];
+    lens[base + i] = b.v_offs[v + 1] - b.v_offs[v];
+  }
+}
+__global__ void k_vl_expand_copy(DevBatch b, u32 kw, u64 frontier,
+                                 const u32 *ovoffs /*exclusive*/,
+                                 u64 *okeys, u8 *oarena, u64 *otimes,
+                                 i64 *odiffs, u64 base) {
+  GRID_STRIDE(i, b.n_upds) {
+    u32 v = b.upd_val[i];
+    u32 k = b.val_key[v];
+    u64 o = base + i;
+    for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = b.keys[(u64)k * kw + w];
+    u64 t = b.times[i];
+    otimes[o] = t < frontier ? frontier : t;
+    odiffs[o] = b.diffs[i];
+    u32 lo = ovoffs[o], src = b.v_offs[v], len = b.v_offs[v + 1] - src;
+    for (u32 c = 0; c < len; c++) oarena[lo + c] = b.vals[src + c];
+  }
+}
+
+// consolidated varlen columns (owned device arrays, exact sizes)
+struct VlCols {
+  u64 *keys = nullptr;
+  u32 *voffs = nullptr;  // [n+1] monotone
+  u8 *arena = nullptr;
+  u64 *times = nullptr;
+  i64 *diffs = nullptr;
+  u64 n = 0, bytes = 0;
+};
+
+// Sort + consolidate varlen rows addressed by (vstarts, vends) into
+// canonical (key, val, time) order. Synchronizes.
+VlCols consolidate_vl(Ctx *c, u32 kw, const u64 *keys, const u32 *vstarts,
+                      const u32 *vends, const u8 *arena, const u64 *times,
+                      const i64 *diffs, u64 n) {
+  auto &S = (*c->scr);
+  VlCols out;
+  if (n == 0) {
+    out.keys = dnew<u64>(c, 1);
+    out.voffs = dnew<u32>(c, 1);
+    out.arena = (u8 *)dmalloc(c, 1);
+    out.times = dnew<u64>(c, 1);
+    out.diffs = dnew<i64>(c, 1);
+    fill_u32(c, out.voffs, 1, 0);
+    return out;
+  }
+  u32 *perm = (u32 *)S.get(n * 4);
+  u32 *perm_out = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_iota, dim3(ngrid(n)), dim3(BLK), 0, c->stream, perm,
+                     n);
+  VlRowLess cmp{keys, vstarts, vends, arena, times, kw};
+  size_t need = 0;
+  (void)rocprim::merge_sort(nullptr, need, perm, perm_out, n, cmp,
+                            c->stream);
+  void *tmp = S.get(need);
+  (void)rocprim::merge_sort(tmp, need, perm, perm_out, n, cmp, c->stream);
+  std::swap(perm, perm_out);
+  u32 *flags = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_vl_head_flags, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, keys, kw, vstarts, vends, arena, times,
+                     perm, flags, n, 1);
+  u32 *gid = (u32 *)S.get(n * 4);
+  inclusive_scan_u32(c, flags, gid, n);
+  u32 *starts = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_group_starts, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, flags, gid, starts, n);
+  u64 *pdiff = (u64 *)S.get(n * 8);
+  hipLaunchKernelGGL(k_gather_u64, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     (const u64 *)diffs, perm, pdiff, n);
+  u64 *pref = (u64 *)S.get(n * 8);
+  inclusive_scan_u64(c, pdiff, pref, n);
+  i64 *gsum = (i64 *)S.get(n * 8);
+  u32 *nz = (u32 *)S.get((n + 1) * 4);
+  hipLaunchKernelGGL(k_group_sums, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     starts, gid, n, pref, gsum, nz);
+  u32 *nzpos = (u32 *)S.get((n + 1) * 4);
+  u64 M = exclusive_scan_u32(c, nz, nzpos, n);  // syncs
+  u32 *lens = (u32 *)S.get((n + 1) * 4);
+  hipLaunchKernelGGL(k_vl_survivor_lens, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, starts, gid, nz, nzpos, n, perm, vstarts,
+                     vends, lens);
+  u32 *ovoffs = dnew<u32>(c, M + 1);
+  u64 bytes = exclusive_scan_u32(c, lens, ovoffs, M);  // syncs; [M+1]
+  out.n = M;
+  out.bytes = bytes;
+  out.keys = dnew<u64>(c, std::max<u64>(M, 1) * kw);
+  out.arena = (u8 *)dmalloc(c, std::max<u64>(bytes, 1));
+  out.times = dnew<u64>(c, std::max<u64>(M, 1));
+  out.diffs = dnew<i64>(c, std::max<u64>(M, 1));
+  out.voffs = ovoffs;
+  if (M)
+    hipLaunchKernelGGL(k_vl_emit_consolidated, dim3(ngrid(n)), dim3(BLK),
+                       0, c->stream, keys, kw, vstarts, vends, arena,
+                       times, perm, starts, gsum, nz, nzpos, gid, n,
+                       ovoffs, out.keys, out.arena, out.times, out.diffs);
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return out;
+}
+
+// Build a sealed varlen batch from consolidated columns (takes ownership
+// of times/diffs; copies keys/arena into dedup form and frees them).
+DevBatch build_batch_vl(Ctx *c, u32 kw, VlCols &&in, u64 lower, u64 upper) {
+  auto &S = (*c->scr);
+  DevBatch b;
+  b.lower = lower;
+  b.upper = upper;
+  u64 n = in.n;
+  b.n_upds = n;
+  if (n == 0) {
+    b.keys = dnew<u64>(c, 1);
+    b.kv_off = dnew<u32>(c, 1);
+    b.vals = (u8 *)dmalloc(c, 1);
+    b.vu_off = dnew<u32>(c, 1);
+    b.v_offs = dnew<u32>(c, 1);
+    b.val_key = dnew<u32>(c, 1);
+    b.upd_val = dnew<u32>(c, 1);
+    b.times = in.times;
+    b.diffs = in.diffs;
+    fill_u32(c, b.kv_off, 1, 0);
+    fill_u32(c, b.vu_off, 1, 0);
+    fill_u32(c, b.v_offs, 1, 0);
+    for (void *p : {(void *)in.keys, (void *)in.voffs, (void *)in.arena})
+      dfree(c, p);
+    return b;
+  }
+  u32 *kc = (u32 *)S.get(n * 4);
+  u32 *vc = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_vl_change_flags, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, in.keys, kw, in.voffs, in.arena, kc, vc, n);
+  u32 *kid = (u32 *)S.get(n * 4);
+  u32 *vid = (u32 *)S.get(n * 4);
+  inclusive_scan_u32(c, kc, kid, n);
+  inclusive_scan_u32(c, vc, vid, n);
+  u32 *dl = (u32 *)S.get(n * 4);
+  hipLaunchKernelGGL(k_vl_distinct_lens, dim3(ngrid(n)), dim3(BLK), 0,
+                     c->stream, vc, in.voffs, n, dl);
+  u32 *apos = (u32 *)S.get(n * 4);
+  inclusive_scan_u32(c, dl, apos, n);
+  // host-read the three totals in one go
+  u32 tails[3];
+  HIP_CHECK(hipMemcpyAsync(&tails[0], kid + (n - 1), 4,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipMemcpyAsync(&tails[1], vid + (n - 1), 4,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipMemcpyAsync(&tails[2], apos + (n - 1), 4,
+                           hipMemcpyDeviceToHost, c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  u64 n_keys = tails[0], n_vals = tails[1], abytes = tails[2];
+  b.n_keys = n_keys;
+  b.n_vals = n_vals;
+  b.keys = dnew<u64>(c, n_keys * kw);
+  b.kv_off = dnew<u32>(c, n_keys + 1);
+  b.vals = (u8 *)dmalloc(c, std::max<u64>(abytes, 1));
+  b.v_offs = dnew<u32>(c, n_vals + 1);
+  b.vu_off = dnew<u32>(c, n_vals + 1);
+  b.val_key = dnew<u32>(c, n_vals);
+  b.upd_val = dnew<u32>(c, n);
+  hipLaunchKernelGGL(k_vl_scatter, dim3(ngrid(n)), dim3(BLK), 0, c->stream,
+                     in.keys, kw, in.voffs, in.arena, kc, vc, kid, vid,
+                     apos, n, n_keys, n_vals, abytes, b.keys, b.kv_off,
+                     b.v_offs, b.vals, b.vu_off, b.val_key, b.upd_val);
+  b.times = in.times;
+  b.diffs = in.diffs;
+  u64 slots = 16;
+  while (slots < 2 * n_keys) slots <<= 1;
+  b.hash_slots = slots;
+  b.hash = dnew<u64>(c, slots * (kw + 1));
+  fill_u64(c, b.hash, slots * (kw + 1), ~0ull);
+  // reuse the fixed-width hash build: it reads keys/kid/kv_off only.
+  // kid for distinct keys = identity over n_keys; pass a counting setup:
+  {
+    u32 *kid2 = (u32 *)S.get(std::max<u64>(n_keys, 1) * 4);
+    hipLaunchKernelGGL(k_iota_off, dim3(ngrid(std::max<u64>(n_keys, 1))),
+                       dim3(BLK), 0, c->stream, kid2,
+                       std::max<u64>(n_keys, 1), 1);
+    u64 *dc = (u64 *)S.get(8);
+    hipLaunchKernelGGL(k_write_u64, dim3(1), dim3(1), 0, c->stream, dc,
+                       n_keys);
+    hipLaunchKernelGGL(k_hash_build, dim3(ngrid(std::max<u64>(n_keys, 1))),
+                       dim3(BLK), 0, c->stream, b.hash, slots, b.keys, kw,
+                       kid2, b.kv_off, n_keys, dc);
+  }
+  for (void *p : {(void *)in.keys, (void *)in.voffs, (void *)in.arena})
+    dfree(c, p);
+  return b;
+}
+
+
+// Varlen spine merge: expand the batches to flat rows (logical
+// compaction advanced), re-consolidate, re-seal — semantics identical to
+// the fixed-width merge (concat + advance + consolidate).
+void merge_range_vl(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
+  auto &S = (*c->scr);
+  S.reset();
+  u32 kw = a->schema.kw;
+  u64 total = 0, lo = UINT64_MAX, hi = 0;
+  for (size_t i = from; i < to; i++) {
+    total += a->batches[i].n_upds;
+    lo = std::min(lo, a->batches[i].lower);
+    hi = std::max(hi, a->batches[i].upper);
+  }
+  u64 capn = std::max<u64>(total, 1);
+  u64 *keys = (u64 *)S.get(capn * kw * 8);
+  u32 *lens = (u32 *)S.get((capn + 1) * 4);
+  u64 *times = (u64 *)S.get(capn * 8);
+  i64 *diffs = (i64 *)S.get(capn * 8);
+  u64 base = 0;
+  for (size_t i = from; i < to; i++) {
+    DevBatch &b = a->batches[i];
+    if (b.n_upds)
+      hipLaunchKernelGGL(k_vl_expand_lens, dim3(ngrid(b.n_upds)),
+                         dim3(BLK), 0, c->stream, b, lens, base);
+    base += b.n_upds;
+  }
+  u32 *voffs = (u32 *)S.get((capn + 1) * 4);
+  u64 bytes = total ? exclusive_scan_u32(c, lens, voffs, total) : 0;
+  u8 *arena = (u8 *)S.get(std::max<u64>(bytes, 1));
+  base = 0;
+  for (size_t i = from; i < to; i++) {
+    DevBatch &b = a->batches[i];
+    if (b.n_upds)
+      hipLaunchKernelGGL(k_vl_expand_copy, dim3(ngrid(b.n_upds)),
+                         dim3(BLK), 0, c->stream, b, kw,
+                         a->logical_compaction, voffs + base, keys, arena,
+                         times, diffs, base);
+    base += b.n_upds;
+  }
+  VlCols cc = consolidate_vl(c, kw, keys, voffs, voffs + 1, arena, times,
+                             diffs, total);
+  DevBatch merged = build_batch_vl(c, kw, std::move(cc),
+                                   lo == UINT64_MAX ? 0 : lo, hi);
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  for (size_t i = from; i < to; i++) free_batch(c, a->batches[i]);
+  a->batches.erase(a->batches.begin() + from, a->batches.begin() + to);
+  a->batches.insert(a->batches.begin() + from, merged);
+}
+
+static const bool _mrvl_registered = [] {
+  merge_range_vl_fn = merge_range_vl;
+  return true;
+}();
+
+// Varlen probe: one thread per (delta row, batch); counts rows AND arena
+// bytes, reserves both queues per wave, emits key fields + passthrough
+// varlen val bytes. The closure must not reference the varlen (lookup)
+// side except as the single whole-val passthrough (validated host-side).
+__global__ void k_probe_vl(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                           const u64 *dtimes, const i64 *ddiffs, u64 n,
+                           u32 kw, BatchList bl, int mode, int swap,
+                           const mz_gpu_closure cl, u64 cap, u64 bcap,
+                           unsigned long long *ctr, u64 *okeys,
+                           u32 *ovstarts, u32 *ovends, u8 *oarena,
+                           u64 *otimes, i64 *odiffs) {
+  u32 okw = cl.out.key_words;
+  u64 total = n * (u64)bl.n;
+  u64 stride = (u64)gridDim.x * blockDim.x;
+  u64 start = blockIdx.x * (u64)blockDim.x + threadIdx.x;
+  u64 iters = (total + stride - 1) / stride;
+  u32 lane = threadIdx.x & 63;
+  for (u64 it = 0; it < iters; it++) {
+    u64 idx = start + it * stride;
+    bool active = idx < total;
+    u64 kvr = ~0ull;
+    u32 c = 0, bytes = 0;
+    u64 i = 0;
+    int bi = 0;
+    int cls = 0;
+    if (active) {
+      i = idx % n;
+      bi = (int)(idx / n);
+      const DevBatch &b = bl.b[bi];
+      const u64 *key = dkeys + i * kw;
+      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+      cls = d_closure_apply(&cl, key, swap ? nullptr : dv,
+                            swap ? dv : nullptr, nullptr, nullptr);
+      if (cls == 1) {
+        kvr = hash_lookup_range(b.hash, b.hash_slots, key, kw);
+        if (kvr != ~0ull) {
+          u64 t = dtimes[i];
+          for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
+            u32 len = b.v_offs[j + 1] - b.v_offs[j];
+            u32 lo = b.vu_off[j], hi = b.vu_off[j + 1];
+            u32 m;
+            if (mode == PM_JOIN || bl.allpass[bi]) {
+              m = hi - lo;
+            } else {
+              m = 0;
+              for (u32 u = lo; u < hi; u++) {
+                u64 t2 = b.times[u];
+                m += (mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t);
+              }
+            }
+            c += m;
+            bytes += m * len;
+          }
+        }
+      }
+    }
+    u64 base = wave_reserve(ctr, c, lane);
+    u64 bbase = wave_reserve(ctr + 1, bytes, lane);
+    if (c == 0 || base + c > cap || bbase + bytes > bcap) continue;
+    const DevBatch &b = bl.b[bi];
+    const u64 *key = dkeys + i * kw;
+    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+    u64 t = dtimes[i];
+    i64 d1 = ddiffs[i];
+    u64 o = base, bo = bbase;
+    for (u32 j = (u32)kvr; j < (u32)(kvr >> 32); j++) {
+      u32 vsrc = b.v_offs[j], len = b.v_offs[j + 1] - vsrc;
+      for (u32 u = b.vu_off[j]; u < b.vu_off[j + 1]; u++) {
+        u64 tout;
+        if (bl.allpass[bi]) {
+          tout = t;
+        } else if (mode == PM_JOIN) {
+          u64 t2 = b.times[u];
+          tout = t2 > t ? t2 : t;
+        } else {
+          u64 t2 = b.times[u];
+          if (!((mode == PM_HALF_LE) ? (t2 <= t) : (t2 < t))) continue;
+          tout = t;
+        }
+        (void)d_closure_apply(&cl, key, swap ? nullptr : dv,
+                              swap ? dv : nullptr, okeys + o * okw,
+                              (u8 *)(okeys + o * okw));
+        otimes[o] = tout;
+        odiffs[o] = wmul(d1, b.diffs[u]);
+        ovstarts[o] = (u32)bo;
+        ovends[o] = (u32)(bo + len);
+        for (u32 cc = 0; cc < len; cc++) oarena[bo + cc] = b.vals[vsrc + cc];
+        o++;
+        bo += len;
+      }
+    }
+  }
+}
+
+mz_gpu_out *make_out_vl(u64 *k, u8 *arena, u32 *voffs, u64 *t, i64 *d,
+                        u64 n, u64 bytes, u32 kw) {
+  mz_gpu_out *o = make_out(k, arena, t, d, n, kw, 0xFFFFFFFFu);
+  o->val_offs = voffs;
+  o->val_arena_bytes = bytes;
+  return o;
+}
+
+static void arr_flush_impl(Ctx *ctx, mz_gpu_arr *a);
+static void arr_insert_vl(Ctx *ctx, mz_gpu_arr *a, const mz_gpu_updates *u);
+
+// varlen probe host path (halfjoin / linear join over a varlen lookup):
+// validates the closure shape, runs k_probe_vl with row+byte queues,
+// consolidates (canonical varlen order) and returns a varlen out-batch.
+static int probe_vl_impl(Ctx *ctx, mz_gpu_arr *lookup,
+                         const mz_gpu_updates *u, u32 stream_vb, int mode,
+                         int swap, const mz_gpu_closure *cl,
+                         mz_gpu_out **out) {
+  arr_flush_impl(ctx, lookup);
+  if (lookup->stream)
+    (void)hipStreamWaitEvent(ctx->stream, lookup->ev_ready, 0);
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
+  u32 kw = lookup->schema.kw;
+  u32 okw = cl->out.key_words;
+  // closure validation: no reference to the varlen side except the
+  // single whole-val passthrough
+  u8 banned = swap ? MZ_SRC_VAL_STREAM : MZ_SRC_VAL_LOOKUP;
+  for (u32 i = 0; i < cl->n_filters; i++)
+    if (cl->filters[i].src == banned ||
+        cl->filters[i].src == MZ_SRC_COMPUTE) {
+      ctx->err = "varlen probe: filters may not touch the varlen side";
+      return -1;
+    }
+  for (u32 i = 0; i < cl->n_key_fields; i++)
+    if (cl->key_fields[i].src == banned) {
+      ctx->err = "varlen probe: key fields may not touch the varlen side";
+      return -1;
+    }
+  if (!(cl->n_val_fields == 1 && cl->val_fields[0].src == banned &&
+        cl->val_fields[0].width == 0) ||
+      cl->out.val_bytes != 0xFFFFFFFFu) {
+    ctx->err = "varlen probe: out val must be the whole varlen val "
+               "(one field, src = lookup side, width 0)";
+    return -1;
+  }
+  mz_gpu_closure cl2 = *cl;
+  cl2.n_val_fields = 0;  // key fields evaluated in-kernel; val is copied
+  DevUpdates d = stage_updates(ctx, u, kw, stream_vb);
+  BatchList bl;
+  bl.n = 0;
+  for (auto &b : lookup->batches) {
+    if (b.n_upds == 0) continue;
+    if (bl.n >= 12) {
+      merge_range(ctx, lookup, 0, lookup->batches.size());
+      return probe_vl_impl(ctx, lookup, u, stream_vb, mode, swap, cl, out);
+    }
+    u64 tmax_excl = std::max(b.upper, lookup->logical_compaction + 1);
+    bl.allpass[bl.n] = tmax_excl <= u->lower ? 1 : 0;
+    bl.b[bl.n++] = b;
+  }
+  u64 n = d.n;
+  if (n == 0 || bl.n == 0) {
+    *out = make_out_vl(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                       dnew<u32>(ctx, 1), dnew<u64>(ctx, 1),
+                       dnew<i64>(ctx, 1), 0, 0, okw);
+    fill_u32(ctx, (*out)->val_offs, 1, 0);
+    return 0;
+  }
+  u64 nb2 = n * (u64)bl.n;
+  u64 cap = 2 * n + 1024;
+  u64 bcap = 64 * cap;
+  unsigned long long *ctr = (unsigned long long *)S.get(16);
+  u64 *pk = dnew<u64>(ctx, cap * okw);
+  u32 *pvs = dnew<u32>(ctx, cap);
+  u32 *pve = dnew<u32>(ctx, cap);
+  u8 *pa = (u8 *)dmalloc(ctx, bcap);
+  u64 *pt = dnew<u64>(ctx, cap);
+  i64 *pd = dnew<i64>(ctx, cap);
+  for (int attempt = 0; attempt < 2; attempt++) {
+    fill_u64(ctx, (u64 *)ctr, 2, 0);
+    hipLaunchKernelGGL(k_probe_vl, dim3(ngrid(nb2)), dim3(BLK), 0,
+                       ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                       d.diffs, n, kw, bl, mode, swap, cl2, cap, bcap, ctr,
+                       pk, pvs, pve, pa, pt, pd);
+    unsigned long long MB[2];
+    HIP_CHECK(hipMemcpyAsync(MB, ctr, 16, hipMemcpyDeviceToHost,
+                             ctx->stream));
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    if (MB[0] <= cap && MB[1] <= bcap) {
+      VlCols cc = consolidate_vl(ctx, okw, pk, pvs, pve, pa, pt, pd,
+                                 MB[0]);
+      for (void *p : {(void *)pk, (void *)pvs, (void *)pve, (void *)pa,
+                      (void *)pt, (void *)pd})
+        dfree(ctx, p);
+      *out = make_out_vl(cc.keys, cc.arena, cc.voffs, cc.times, cc.diffs,
+                         cc.n, cc.bytes, okw);
+      return 0;
+    }
+    for (void *p : {(void *)pk, (void *)pvs, (void *)pve, (void *)pa,
+                    (void *)pt, (void *)pd})
+      dfree(ctx, p);
+    cap = std::max<u64>(MB[0], 1);
+    bcap = std::max<u64>(MB[1], 1);
+    pk = dnew<u64>(ctx, cap * okw);
+    pvs = dnew<u32>(ctx, cap);
+    pve = dnew<u32>(ctx, cap);
+    pa = (u8 *)dmalloc(ctx, bcap);
+    pt = dnew<u64>(ctx, cap);
+    pd = dnew<i64>(ctx, cap);
+  }
+  ctx->err = "varlen probe: queue overflow after exact relaunch";
+  return -1;
+}
 
 struct mz_gpu_ctx {
   Ctx impl;
@@ -2670,6 +3304,10 @@ static void spine_policy(Ctx *ctx, mz_gpu_arr *a) {
 int mz_gpu_arr_push_batch(mz_gpu_ctx *c, mz_gpu_arr *a,
                           const mz_gpu_updates *u) {
   Ctx *ctx = &c->impl;
+  if (is_varlen(a->schema.vb)) {
+    arr_insert_vl(ctx, a, u);  // sealed input re-consolidates (idempotent)
+    return 0;
+  }
   arr_flush_impl(ctx, a);
   LaneGuard lane(ctx, a);
   (*ctx->scr).reset();
@@ -2744,9 +3382,31 @@ static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
 // arrangement's own lane; the counts readback is issued asynchronously
 // and the sealed batch joins the spine at mz_gpu_arr_flush. Independent
 // arrangements' inserts overlap this way (one per GPU stream).
+// varlen insert: synchronous consolidate + build + push (no lane
+// pipeline — varlen is off the benchmark hot path)
+static void arr_insert_vl(Ctx *ctx, mz_gpu_arr *a,
+                          const mz_gpu_updates *u) {
+  (*ctx->scr).reset();
+  u32 kw = a->schema.kw;
+  DevUpdates d = stage_updates(ctx, u, kw, a->schema.vb);
+  // copy-consolidate then seal: consolidate_vl owns fresh arrays
+  VlCols cc = consolidate_vl(ctx, kw, d.keys,
+                             d.val_offs, d.val_offs + 1, d.vals, d.times,
+                             d.diffs, d.n);
+  DevBatch b = build_batch_vl(ctx, kw, std::move(cc), u->lower, u->upper);
+  a->batches.push_back(b);
+  a->upper = std::max(a->upper, u->upper);
+  spine_policy(ctx, a);
+  if (a->stream) (void)hipEventRecord(a->ev_ready, a->stream);
+}
+
 static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
                                   const mz_gpu_updates *u) {
   MZ_PROF(ctx, "arr_insert");
+  if (is_varlen(a->schema.vb)) {
+    arr_insert_vl(ctx, a, u);
+    return;
+  }
   if (a->pending.active) arr_flush_impl(ctx, a);
   LaneGuard lane(ctx, a);
   auto &S = (*ctx->scr);
@@ -2911,9 +3571,20 @@ void mz_gpu_out_release(mz_gpu_ctx *c, mz_gpu_out *o) {
   OutOwned *oo = reinterpret_cast<OutOwned *>(o);
   for (void *p : {(void *)oo->keys, (void *)oo->vals, (void *)oo->times,
                   (void *)oo->diffs, (void *)o->err_codes,
-                  (void *)o->err_times, (void *)o->err_diffs})
+                  (void *)o->err_times, (void *)o->err_diffs,
+                  (void *)o->val_offs})
     dfree(&c->impl, p);
   delete oo;
+}
+
+/* VARLEN out-batches: copy the [n+1] val offsets to the host. */
+int mz_gpu_out_voffs_to_host(mz_gpu_ctx *c, const mz_gpu_out *o,
+                             uint32_t *offs) {
+  Ctx *ctx = &c->impl;
+  HIP_CHECK(hipMemcpyAsync(offs, o->val_offs, (o->n + 1) * 4,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  return 0;
 }
 
 int mz_gpu_out_err_to_host(mz_gpu_ctx *c, const mz_gpu_out *o,
@@ -2940,7 +3611,11 @@ int mz_gpu_out_to_host(mz_gpu_ctx *c, const mz_gpu_out *o, uint64_t *keys,
   u32 kw = o->schema.key_words, vb = o->schema.val_bytes;
   HIP_CHECK(hipMemcpyAsync(keys, o->keys, n * kw * 8, hipMemcpyDeviceToHost,
                            ctx->stream));
-  if (vb)
+  if (is_varlen(vb)) {
+    if (o->val_arena_bytes)
+      HIP_CHECK(hipMemcpyAsync(vals, o->vals, o->val_arena_bytes,
+                               hipMemcpyDeviceToHost, ctx->stream));
+  } else if (vb)
     HIP_CHECK(hipMemcpyAsync(vals, o->vals, n * vb, hipMemcpyDeviceToHost,
                              ctx->stream));
   HIP_CHECK(hipMemcpyAsync(times, o->times, n * 8, hipMemcpyDeviceToHost,
@@ -2957,6 +3632,14 @@ int mz_gpu_consolidate(mz_gpu_ctx *c, const mz_gpu_schema *s,
   (*ctx->scr).reset();
   u32 kw = s->key_words, vb = s->val_bytes;
   DevUpdates d = stage_updates(ctx, u, kw, vb);
+  if (is_varlen(vb)) {
+    VlCols cc = consolidate_vl(ctx, kw, d.keys, d.val_offs,
+                               d.val_offs + 1, d.vals, d.times, d.diffs,
+                               d.n);
+    *out = make_out_vl(cc.keys, cc.arena, cc.voffs, cc.times, cc.diffs,
+                       cc.n, cc.bytes, kw);
+    return 0;
+  }
   u64 *ok;
   u8 *ov;
   u64 *ot;
@@ -2991,6 +3674,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
                       u32 stream_vb, int mode, int swap,
                       const mz_gpu_closure *cl, int consolidate_out,
                       mz_gpu_out **out) {
+  if (is_varlen(lookup->schema.vb))
+    return probe_vl_impl(ctx, lookup, u, stream_vb, mode, swap, cl, out);
   // a probe after insert_async must see the batch (probes of the OLD
   // state precede the insert call entirely), and must run after the
   // lookup lane's enqueued maintenance
@@ -3234,6 +3919,10 @@ int mz_gpu_halfjoin_raw(mz_gpu_ctx *c, mz_gpu_arr *lookup,
 mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,
                                  const mz_gpu_reduce_spec *spec) {
   Ctx *ctx = &c->impl;
+  if (is_varlen(spec->in.val_bytes) || is_varlen(spec->out.val_bytes)) {
+    ctx->err = "varlen vals unsupported in reduce";
+    return nullptr;
+  }
   mz_gpu_red *r = new mz_gpu_red();
   r->spec = *spec;
   u32 kw = spec->in.key_words;
@@ -3553,6 +4242,10 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
 
 mz_gpu_thr *mz_gpu_threshold_create(mz_gpu_ctx *c, const mz_gpu_schema *s) {
   Ctx *ctx = &c->impl;
+  if (is_varlen(s->val_bytes)) {
+    ctx->err = "varlen vals unsupported in threshold";
+    return nullptr;
+  }
   (*ctx->scr).reset();
   mz_gpu_thr *r = new mz_gpu_thr();
   r->s = *s;
