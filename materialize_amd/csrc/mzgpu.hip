@@ -1452,6 +1452,16 @@ struct Scratch {
   void *base = nullptr;
   size_t cap = 0, used = 0;
   void reset() { used = 0; }
+  // fini teardown: the arena is plain hipMalloc memory; without this
+  // every context leaked its (pre-grown, multi-GB) arena — a long test
+  // session with many contexts ran the device out of memory
+  void destroy() {
+    if (base) (void)hipFree(base);
+    for (void *p : retired) (void)hipFree(p);
+    base = nullptr;
+    retired.clear();
+    cap = used = 0;
+  }
   void *get(size_t bytes) {
     bytes = (bytes + 255) & ~size_t(255);
     if (used + bytes > cap) {
@@ -2842,7 +2852,7 @@ void merge_range_vl(Ctx *c, mz_gpu_arr *a, size_t from, size_t to) {
     if (b.n_upds)
       hipLaunchKernelGGL(k_vl_expand_copy, dim3(ngrid(b.n_upds)),
                          dim3(BLK), 0, c->stream, b, kw,
-                         a->logical_compaction, voffs + base, keys, arena,
+                         a->logical_compaction, voffs, keys, arena,
                          times, diffs, base);
     base += b.n_upds;
   }
@@ -3156,18 +3166,41 @@ void mz_gpu_prof_dump(mz_gpu_ctx *c) {
 
 void mz_gpu_fini(mz_gpu_ctx *c) {
   if (!c) return;
-  for (mz_gpu_arr *a : c->impl.arrs) {
+  Ctx *ctx = &c->impl;
+  // release everything the library owns: lane resources, arrangement
+  // batches, operator state tables, scratch arenas (leaking these across
+  // many short-lived contexts exhausted HBM)
+  for (mz_gpu_arr *a : ctx->arrs) {
     if (a->stream) {
       (void)hipStreamSynchronize(a->stream);
       (void)hipStreamDestroy(a->stream);
       (void)hipEventDestroy(a->ev_done);
       (void)hipEventDestroy(a->ev_gate);
       (void)hipEventDestroy(a->ev_ready);
+      if (a->lane_scr) a->lane_scr->destroy();
       delete a->lane_scr;
       a->stream = nullptr;
     }
   }
-  (void)hipStreamSynchronize(c->impl.stream);
+  (void)hipStreamSynchronize(ctx->stream);
+  for (mz_gpu_arr *a : ctx->arrs) {
+    for (auto &b : a->batches) free_batch(ctx, b);
+    dfree(ctx, a->pending.flat_keys);
+    dfree(ctx, a->pending.flat_vals);
+    delete a;
+  }
+  ctx->arrs.clear();
+  for (mz_gpu_red *r : ctx->reds) {
+    for (void *p : {(void *)r->st.hash, (void *)r->st.rows,
+                    (void *)r->d_nrows, (void *)r->d_err})
+      dfree(ctx, p);
+    delete r;
+  }
+  ctx->reds.clear();
+  for (mz_gpu_join *j : ctx->joins) delete j;
+  ctx->joins.clear();
+  (void)hipStreamSynchronize(ctx->stream);
+  ctx->scratch.destroy();
   delete c;
 }
 
@@ -3206,6 +3239,7 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
     (void)hipEventDestroy(a->ev_done);
     (void)hipEventDestroy(a->ev_gate);
     (void)hipEventDestroy(a->ev_ready);
+    if (a->lane_scr) a->lane_scr->destroy();
     delete a->lane_scr;
     a->stream = nullptr;
     a->lane_scr = nullptr;
