@@ -165,6 +165,16 @@ class GpuCtx:
             self.lib.mz_gpu_fini(self.ctx)
             self.ctx = None
 
+    def __del__(self):
+        # contexts own multi-GB device state; dropping one without
+        # close() must not leak it for the rest of the process
+        if os.environ.get("MZ_NO_GCCLOSE"):
+            return
+        try:
+            self.close()
+        except Exception:
+            pass
+
     def _check(self, rc):
         if rc != 0:
             raise MzGpuError(self.lib.mz_gpu_last_error(self.ctx).decode())

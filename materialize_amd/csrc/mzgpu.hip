@@ -1683,7 +1683,17 @@ void *dmalloc(Ctx *c, size_t bytes) {
 }
 
 void dfree(Ctx *c, void *p) {
-  if (p) HIP_CHECK(hipFreeAsync(p, c->stream));
+  if (!p) return;
+  hipError_t rc = hipFreeAsync(p, c->stream);
+  if (rc != hipSuccess) {
+    hipPointerAttribute_t at{};
+    hipError_t arc = hipPointerGetAttributes(&at, p);
+    fprintf(stderr,
+            "dfree FAIL %s ptr=%p attr_rc=%d type=%d device=%d\n",
+            hipGetErrorString(rc), p, (int)arc, (int)at.type,
+            (int)at.device);
+    if (!getenv("MZ_DFREE_SOFT")) abort();
+  }
 }
 
 template <typename T>
@@ -2193,11 +2203,17 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
 }
 
 void free_batch(Ctx *c, DevBatch &b) {
-  for (void *p : {(void *)b.keys, (void *)b.kv_off, (void *)b.vals,
+  const char *names[10] = {"keys", "kv_off", "vals", "vu_off", "v_offs",
+                           "val_key", "times", "diffs", "upd_val", "hash"};
+  void *ps[10] = {(void *)b.keys, (void *)b.kv_off, (void *)b.vals,
                   (void *)b.vu_off, (void *)b.v_offs, (void *)b.val_key,
                   (void *)b.times, (void *)b.diffs, (void *)b.upd_val,
-                  (void *)b.hash})
-    dfree(c, p);
+                  (void *)b.hash};
+  for (int i = 0; i < 10; i++) {
+    if (getenv("MZ_DBG_FINI") && ps[i])
+      fprintf(stderr, "[free_batch] %s %p\n", names[i], ps[i]);
+    dfree(c, ps[i]);
+  }
   b = DevBatch();
 }
 
@@ -2214,8 +2230,12 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
   b.lower = lower;
   b.upper = upper;
   if (cap == 0) {
-    b.keys = keys;
-    b.vals = vals;
+    // placeholders: NEVER adopt the flat arrays here — with keep_flat
+    // the caller retains them (flush_take hand-off), and adopting gave
+    // the batch a second owner of the same pointers (double free; the
+    // recycled VAs then corrupted unrelated batches)
+    b.keys = dnew<u64>(c, 1);
+    b.vals = (u8 *)dmalloc(c, 1);
     b.times = times;
     b.diffs = diffs;
     b.kv_off = dnew<u32>(c, 1);
@@ -2224,6 +2244,10 @@ DevBatch build_batch_core(Ctx *c, u32 kw, u32 vb, u64 *keys, u8 *vals,
     b.upd_val = dnew<u32>(c, 1);
     fill_u32(c, b.kv_off, 1, 0);
     fill_u32(c, b.vu_off, 1, 0);
+    if (!keep_flat) {
+      dfree(c, keys);
+      dfree(c, vals);
+    }
     return b;
   }
   u32 *kc = (u32 *)S.get(cap * 4);
@@ -3183,7 +3207,10 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
     }
   }
   (void)hipStreamSynchronize(ctx->stream);
+  const char *dbg = getenv("MZ_DBG_FINI");
   for (mz_gpu_arr *a : ctx->arrs) {
+    if (dbg) fprintf(stderr, "[fini] arr %p batches=%zu\n", (void *)a,
+                     a->batches.size());
     for (auto &b : a->batches) free_batch(ctx, b);
     dfree(ctx, a->pending.flat_keys);
     dfree(ctx, a->pending.flat_vals);
@@ -3191,6 +3218,7 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
   }
   ctx->arrs.clear();
   for (mz_gpu_red *r : ctx->reds) {
+    if (dbg) fprintf(stderr, "[fini] red %p\n", (void *)r);
     for (void *p : {(void *)r->st.hash, (void *)r->st.rows,
                     (void *)r->d_nrows, (void *)r->d_err})
       dfree(ctx, p);
@@ -3201,6 +3229,17 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
   ctx->joins.clear();
   (void)hipStreamSynchronize(ctx->stream);
   ctx->scratch.destroy();
+  // hand the pool's now-unused reservations back to the OS: the infinite
+  // release threshold (set in init for steady-state speed) otherwise
+  // accumulates reserved-but-free memory across short-lived contexts
+  // until plain hipMalloc (scratch arenas) cannot back a new one
+  if (!getenv("MZ_NO_TRIM")) {
+    int dev = 0;
+    (void)hipGetDevice(&dev);
+    hipMemPool_t pool = nullptr;
+    if (hipDeviceGetDefaultMemPool(&pool, dev) == hipSuccess && pool)
+      (void)hipMemPoolTrimTo(pool, 0);
+  }
   delete c;
 }
 
