@@ -487,8 +487,8 @@ def main():
     # the one profiled
     try:
         import glob as _glob
-        for tf in _glob.glob(os.path.join(REPO, "profiles",
-                                          "traffic_*.json")):
+        for tf in sorted(_glob.glob(os.path.join(REPO, "profiles",
+                                                 "traffic_*.json"))):
             td = json.load(open(tf))
             if td.get("workload") == (f"tpch_q3_sf{sf:g}_churn{batch_rows}"
                                       if N == 1 else None):
