@@ -3891,11 +3891,13 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     ctx->probe_launches += launches;
     ctx->probe_pairs += M;
     ctx->probe_batches += (u64)bl.n;
-    // Algorithmic bytes of the probe (SURVEY §8d model): delta tuple +
-    // one hash line per probed batch + matched val+upd read + output
-    // write — each touched once cold in the single walk.
+    // Algorithmic bytes of the probe (SURVEY §8d model, updated for the
+    // compressed 16 B hash slots: one 64 B line per probed batch instead
+    // of the r1 layout's 128 B): delta tuple + one hash line per batch +
+    // matched val+upd read + output write — each touched once cold in
+    // the single walk.
     ctx->probe_alg_bytes +=
-        n * (8ull * kw + stream_vb + 16) + n * 128ull * (u64)bl.n +
+        n * (8ull * kw + stream_vb + 16) + n * 64ull * (u64)bl.n +
         M * (lvb + 16ull) + M * (8ull * okw + ovb + 16);
   }
   // error stream: consolidate by (code, time) — the err collection is a
