@@ -1593,6 +1593,9 @@ struct Ctx {
   std::vector<mz_gpu_arr *> arrs;
   std::vector<mz_gpu_join *> joins;
   std::vector<mz_gpu_red *> reds;
+  // pinned staging for small device->host readbacks (pageable-staged
+  // async copies cost tens of microseconds each; the step does ~a dozen)
+  void *pin = nullptr;
   // probe-kernel timing (for bench roofline): accumulated ns and bytes
   double probe_ms = 0;
   u64 probe_rows = 0, probe_launches = 0;
@@ -1701,6 +1704,15 @@ T *dnew(Ctx *c, u64 n) {
   return (T *)dmalloc(c, n * sizeof(T));
 }
 
+// Copy small device data into the ctx's pinned staging page and sync;
+// the returned pointer is valid until the next d2h_pinned on this ctx.
+void *d2h_pinned(Ctx *c, const void *dev, size_t bytes) {
+  HIP_CHECK(hipMemcpyAsync(c->pin, dev, bytes, hipMemcpyDeviceToHost,
+                           c->stream));
+  HIP_CHECK(hipStreamSynchronize(c->stream));
+  return c->pin;
+}
+
 // Kernel-based fills for SEMANTIC device state (hash sentinels, counters,
 // zero-padding that later passes read as content). hipMemsetAsync fills
 // into a freshly-carved hipMallocAsync block were observed to be silently
@@ -1792,12 +1804,9 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                        dim3(ngrid(n), 1 + vwords + kw), dim3(BLK), 0,
                        c->stream, keys, kw, vals, vb, times, n, vwords, 1,
                        dmin, dmax);
-  u64 hmin[MAX_PASSES], hmax[MAX_PASSES];
-  HIP_CHECK(hipMemcpyAsync(hmin, dmin, MAX_PASSES * 8,
-                           hipMemcpyDeviceToHost, c->stream));
-  HIP_CHECK(hipMemcpyAsync(hmax, dmax, MAX_PASSES * 8,
-                           hipMemcpyDeviceToHost, c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
+  // one pinned staged copy for both halves (dmin/dmax are contiguous)
+  u64 *mm = (u64 *)d2h_pinned(c, dminmax, 2 * MAX_PASSES * 8);
+  u64 *hmin = mm, *hmax = mm + MAX_PASSES;
   // pass slot layout from k_pass_minmax: [time][val words][key words]
   struct Pass {
     int kind;  // 0 = time, 1 = val word, 2 = key word
@@ -1973,10 +1982,7 @@ u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   void *tmp = S.get(need);
   (void)rocprim::exclusive_scan(tmp, need, pad, out, 0u, n + 1,
                           rocprim::plus<u32>(), c->stream);
-  u32 total;
-  HIP_CHECK(hipMemcpyAsync(&total, out + n, 4, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
+  u32 total = *(u32 *)d2h_pinned(c, out + n, 4);
   return total;
 }
 
@@ -2196,9 +2202,7 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
   consolidate_core(c, kw, vb, in, *okeys, *ovals, *otimes, *odiffs,
                    dcounts);
   u64 M = 0;
-  HIP_CHECK(hipMemcpyAsync(&M, dcounts, 8, hipMemcpyDeviceToHost,
-                           c->stream));
-  HIP_CHECK(hipStreamSynchronize(c->stream));
+  M = *(u64 *)d2h_pinned(c, dcounts, 8);
   *out_n = M;
 }
 
@@ -3123,6 +3127,7 @@ mz_gpu_ctx *mz_gpu_init(const mz_gpu_cfg *cfg) {
   c->impl.scr = &c->impl.scratch;
   HIP_CHECK(hipEventCreate(&c->impl.ev_a));
   HIP_CHECK(hipEventCreate(&c->impl.ev_b));
+  HIP_CHECK(hipHostMalloc(&c->impl.pin, 4096));
   // Keep freed stream-ordered allocations in the pool forever (288 GB of
   // HBM — never hand memory back to the OS mid-run; pool misses showed up
   // as multi-ms host stalls before large allocations).
@@ -3228,6 +3233,7 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
   for (mz_gpu_join *j : ctx->joins) delete j;
   ctx->joins.clear();
   (void)hipStreamSynchronize(ctx->stream);
+  if (ctx->pin) (void)hipHostFree(ctx->pin);
   ctx->scratch.destroy();
   // hand the pool's now-unused reservations back to the OS: the infinite
   // release threshold (set in init for steady-state speed) otherwise
@@ -3438,10 +3444,7 @@ static DevBatch *arr_insert_dev(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
   }
   DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, lower,
                                 upper, dcounts);
-  u64 cnt[3] = {0, 0, 0};
-  HIP_CHECK(hipMemcpyAsync(cnt, dcounts, 3 * 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  u64 *cnt = (u64 *)d2h_pinned(ctx, dcounts, 3 * 8);
   b.n_upds = cnt[0];
   b.n_keys = cnt[1];
   b.n_vals = cnt[2];
@@ -3856,10 +3859,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
   launch_probes();
   if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
-  unsigned long long MB[2] = {0, 0};
-  HIP_CHECK(hipMemcpyAsync(MB, ctr, 16, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  unsigned long long *MB =
+      (unsigned long long *)d2h_pinned(ctx, ctr, 16);
   u64 M = MB[0], E = MB[1];
   u64 launches = 1;
   if (M > cap || E > ecap) {  // rare: queue overflow — exact relaunch
@@ -4289,15 +4290,16 @@ static int reduce_push_dev_impl(Ctx *ctx, mz_gpu_red *op, DevUpdates d,
   }
   // consolidate the actual emitted corrections (one count readback —
   // sorting the 2n-capacity zero-padded buffer dominated this path)
-  unsigned long long emitted = 0;
-  u64 errflag = 0;
-  HIP_CHECK(hipMemcpyAsync(&emitted, ocount, 8, hipMemcpyDeviceToHost,
+  u64 *stage = (u64 *)ctx->pin;
+  HIP_CHECK(hipMemcpyAsync(stage, ocount, 8, hipMemcpyDeviceToHost,
                            ctx->stream));
-  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipMemcpyAsync(&op->n_rows, op->d_nrows, 8,
+  HIP_CHECK(hipMemcpyAsync(stage + 1, op->d_err, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(stage + 2, op->d_nrows, 8,
                            hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  u64 emitted = stage[0], errflag = stage[1];
+  op->n_rows = stage[2];
   DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
   u8 *ov;
@@ -4435,15 +4437,16 @@ int mz_gpu_threshold_push(mz_gpu_ctx *c, mz_gpu_thr *op,
     hipLaunchKernelGGL(k_bump_ctr, dim3(1), dim3(1), 0, ctx->stream,
                        op->d_nrows, misspos, gid, m);
   }
-  unsigned long long emitted = 0;
-  u64 errflag = 0;
-  HIP_CHECK(hipMemcpyAsync(&emitted, ocount, 8, hipMemcpyDeviceToHost,
+  u64 *stage = (u64 *)ctx->pin;
+  HIP_CHECK(hipMemcpyAsync(stage, ocount, 8, hipMemcpyDeviceToHost,
                            ctx->stream));
-  HIP_CHECK(hipMemcpyAsync(&errflag, op->d_err, 8, hipMemcpyDeviceToHost,
-                           ctx->stream));
-  HIP_CHECK(hipMemcpyAsync(&op->n_rows, op->d_nrows, 8,
+  HIP_CHECK(hipMemcpyAsync(stage + 1, op->d_err, 8,
+                           hipMemcpyDeviceToHost, ctx->stream));
+  HIP_CHECK(hipMemcpyAsync(stage + 2, op->d_nrows, 8,
                            hipMemcpyDeviceToHost, ctx->stream));
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  u64 emitted = stage[0], errflag = stage[1];
+  op->n_rows = stage[2];
   DevUpdates pin{pk, pv, pt, pd, (u64)emitted};
   u64 *ok;
   u8 *ov;
