@@ -62,9 +62,10 @@ class TpchGen:
         return extprice, discount_bp, shipdate, partkey, quantity
 
     def _gen_part(self):
-        """part table (Q17): brand 0..24 (Brand#23 = 23), container 0..39
-        (MED BOX = 17) — 1/1000 selectivity like the reference's
-        predicates."""
+        """part table (Q17): brand code 10*m+n (Brand#23 = 23),
+        container code 10*c1+c2 (MED BOX = 10) — the exact generator's
+        encodings; synthetic draws cover the same ranges with ~1/1000
+        joint selectivity like the reference's predicates."""
         rng = self.rng
         self.p_partkey = np.arange(1, self.n_part + 1, dtype=np.int64)
         self.p_brand = rng.integers(0, 25, self.n_part).astype(np.int64)

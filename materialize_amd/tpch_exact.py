@@ -296,8 +296,8 @@ class TpchExact:
     def _part_row(self):
         r = self.rng
         r.floyd_5_of(PARTNAMES_LEN)       # name
-        r.r32_incl(1, 5)                  # m
-        r.r32_incl(1, 5)                  # n
+        m = r.r32_incl(1, 5)
+        n = r.r32_incl(1, 5)
         for _ in range(4):                # partsupp rows
             r.r64_incl(0, 3)              # suppkey term
             r.r32_incl(1, 9_999)          # availqty
@@ -306,9 +306,12 @@ class TpchExact:
         for ln in TYPES_LENS:             # type syllables
             r.choose_idx(ln)
         r.r32_incl(1, 50)                 # size
-        for ln in CONTAINERS_LENS:        # container syllables
-            r.choose_idx(ln)
+        c1 = r.choose_idx(CONTAINERS_LENS[0])
+        c2 = r.choose_idx(CONTAINERS_LENS[1])
         r.d_text(49, 198, self.TEXT_LEN)
+        # engine encodings: brand code = 10*m+n ('Brand#23' = 23);
+        # container code = 10*c1+c2 ('MED BOX' = 10)
+        return m * 10 + n, c1 * 10 + c2
 
     def _customer_row(self, key):
         r = self.rng
@@ -368,8 +371,10 @@ class TpchExact:
         shipdate, partkey, quantity)]."""
         for _ in range(self.count_supplier):
             self._supplier_row()
-        for _ in range(1, self.count_part + 1):
-            self._part_row()
+        self.parts = []  # (partkey, brand_code, container_code)
+        for key in range(1, self.count_part + 1):
+            b, c = self._part_row()
+            self.parts.append((key, b, c))
         customers = []
         for key in range(1, self.count_customer + 1):
             customers.append(self._customer_row(key))
@@ -477,8 +482,10 @@ class ExactEngineData:
     extendedprice int64 cents, discount int64 basis points, BUILDING=0).
     Duck-types the TpchGen surface Q3Dataflow.load/churn consume."""
 
-    def __init__(self, gen, customers, orders, lineitems):
+    def __init__(self, gen, customers, orders, lineitems, parts=None):
         self._gen = gen
+        self.parts = parts if parts is not None else getattr(gen, "parts",
+                                                             [])
         self.n_customer = len(customers)
         self.c_custkey = np.array([c[0] for c in customers], np.int64)
         self.c_mktsegment = np.array([_seg_code(c[1]) for c in customers],
@@ -489,15 +496,6 @@ class ExactEngineData:
         self.o_orderdate = np.array([_days(t[2]) for t in orders], np.int32)
         self.o_shippriority = np.zeros(self.n_orders, np.int32)
         self._pack_lineitems(lineitems)
-
-    def _pack_lineitems(self, lineitems):
-        self.l_orderkey = np.array([t[0] for t in lineitems], np.int64)
-        self.l_extendedprice = np.array([t[3] * 100 for t in lineitems],
-                                        np.int64)  # cents
-        self.l_discount = np.array([t[4] * 100 for t in lineitems],
-                                   np.int64)       # basis points
-        self.l_shipdate = np.array([_days(t[5]) for t in lineitems],
-                                   np.int32)
 
     # ---- TpchGen surface used by Q3Dataflow.load ----
     def customer_updates(self):
@@ -532,6 +530,40 @@ class ExactEngineData:
         return self.l_orderkey, self.lineitem_vals(
             self.l_extendedprice, self.l_discount, self.l_shipdate)
 
+    # ---- Q17 surfaces (lineitem by partkey, part) ----
+    def _pack_lineitems(self, lineitems):
+        self.l_orderkey = np.array([t[0] for t in lineitems], np.int64)
+        self.l_partkey = np.array([t[1] for t in lineitems], np.int64)
+        self.l_quantity = np.array([t[2] for t in lineitems], np.int64)
+        self.l_extendedprice = np.array([t[3] * 100 for t in lineitems],
+                                        np.int64)  # cents
+        self.l_discount = np.array([t[4] * 100 for t in lineitems],
+                                   np.int64)       # basis points
+        self.l_shipdate = np.array([_days(t[5]) for t in lineitems],
+                                   np.int32)
+
+    @staticmethod
+    def lineitem_bypart_vals(quantity, extprice):
+        n = len(quantity)
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = quantity.view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = extprice.view(np.uint8).reshape(n, 8)
+        return v
+
+    def lineitem_bypart_updates(self):
+        return self.l_partkey, self.lineitem_bypart_vals(
+            self.l_quantity, self.l_extendedprice)
+
+    def part_updates(self):
+        n = len(self.parts)
+        keys = np.array([p[0] for p in self.parts], np.int64)
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = np.array([p[1] for p in self.parts], np.int64) \
+            .view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = np.array([p[2] for p in self.parts], np.int64) \
+            .view(np.uint8).reshape(n, 8)
+        return keys, v
+
     @staticmethod
     def churn_to_engine(batch):
         """Map one exact churn batch (old_order, old_lines, new_order,
@@ -548,11 +580,21 @@ class ExactEngineData:
             sd = np.array([_days(t[5]) for t in lines], np.int32)
             return keys, ExactEngineData.lineitem_vals(ep, d, sd),                 np.full(n, diff, np.int64)
 
+        def pack_bypart(lines, diff):
+            n = len(lines)
+            pk = np.array([t[1] for t in lines], np.int64)
+            q = np.array([t[2] for t in lines], np.int64)
+            ep = np.array([t[3] * 100 for t in lines], np.int64)
+            return (pk, ExactEngineData.lineitem_bypart_vals(q, ep),
+                    np.full(n, diff, np.int64))
+
         lk_o, lv_o, ld_o = pack_lines(old_lines, -1)
         lk_n, lv_n, ld_n = pack_lines(new_lines, 1)
         l_keys = np.concatenate([lk_o, lk_n])
         l_vals = np.concatenate([lv_o, lv_n])
         l_diffs = np.concatenate([ld_o, ld_n])
+        bp_o = pack_bypart(old_lines, -1)
+        bp_n = pack_bypart(new_lines, 1)
 
         def pack_order(ck, od, diff, by_cust):
             v = np.zeros((1, 16), np.uint8)
@@ -568,6 +610,9 @@ class ExactEngineData:
         ck_kn, cv_n, cd_n = pack_order(ck_n, od_n, 1, True)
         return {
             "lineitem": (l_keys, l_vals, l_diffs),
+            "lineitem_by_part": (np.concatenate([bp_o[0], bp_n[0]]),
+                                 np.concatenate([bp_o[1], bp_n[1]]),
+                                 np.concatenate([bp_o[2], bp_n[2]])),
             "orders": (np.concatenate([ok_o, ok_n]),
                        np.concatenate([ov_o, ov_n]),
                        np.concatenate([od_do, od_dn])),
@@ -575,3 +620,37 @@ class ExactEngineData:
                                np.concatenate([cv_o, cv_n]),
                                np.concatenate([cd_o, cd_n])),
         }
+
+
+def q17_avg_yearly(parts, lineitems):
+    """Reference Q17 over the current state, exactly: per selected part
+    (Brand#23, MED BOX), lineitems with quantity < 0.2 * avg(quantity)
+    contribute extendedprice; result = sum / 7.0 rendered in decNumber
+    39-digit standard notation (cx_datum semantics emulated with Python
+    decimal, the same General Decimal Arithmetic spec). Returns the
+    rendered string, or None for an empty result (SQL NULL)."""
+    from collections import defaultdict
+    from decimal import ROUND_HALF_EVEN, Decimal, localcontext
+    sel = {pk for (pk, b, cc) in parts if b == 23 and cc == 10}
+    with localcontext() as ctx:
+        ctx.prec = 39
+        ctx.rounding = ROUND_HALF_EVEN
+        bypk = defaultdict(list)
+        for t in lineitems:
+            if t[1] in bypk or t[1] in sel:
+                bypk[t[1]].append(t)
+        total = Decimal(0)
+        nrows = 0
+        for pk in sel:
+            lines = bypk.get(pk, [])
+            if not lines:
+                continue
+            avg = Decimal(sum(t[2] for t in lines)) / Decimal(len(lines))
+            thr = Decimal("0.2") * avg
+            for t in lines:
+                if Decimal(t[2]) < thr:
+                    total += Decimal(t[3])
+                    nrows += 1
+        if nrows == 0:
+            return None
+        return format(total / Decimal("7.0"), "f")

@@ -386,12 +386,12 @@ def q17_closures():
     l1 = lineitem ⋈ part (brand/container filters); per-partkey
     sum(quantity)/count over Distinct(l1.partkey) ⋈ lineitem; final
     l1 ⋈ avg with quantity < 0.2*avg (exact integer form),
-    then a global SUM(extendedprice). Brand#23 = code 23, MED BOX = 17."""
+    then a global SUM(extendedprice). Brand#23 = code 23, MED BOX = 10."""
     # join1: lineitem(by partkey) ⋈ part — input1 = lineitem
     #   lineitem val: [quantity i64][extprice i64]; part val: [brand][cont]
     cl_j1 = abi.closure(
         [FL(VL, 0, 8, abi.MZ_CMP_EQ, 23),        # p_brand = Brand#23
-         FL(VL, 8, 8, abi.MZ_CMP_EQ, 17)],       # p_container = MED BOX
+         FL(VL, 8, 8, abi.MZ_CMP_EQ, 10)],       # p_container = MED BOX
         [F(KEY, 0, 8)],                          # key := partkey
         [F(VS, 0, 16)],                          # val := (qty, extprice)
         abi.schema(1, 16))

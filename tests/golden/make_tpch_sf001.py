@@ -26,12 +26,13 @@ import numpy as np
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 from materialize_amd.tpch_exact import (CUTOFF, ExactEngineData, TpchExact,
-                                        q3_md5, q3_result,
+                                        q3_md5, q3_result, q17_avg_yearly,
                                         render_revenue_1e2)
 
 Q3_MD5 = "637be0ff3f50cd612b004a69958bfccb"
 Q6_MD5 = "d9c979f1eed5940788ff3653321acac4"
 Q12_MD5 = "3c31b94c99bd77e96003c2059416ed7a"
+Q17_MD5 = "6ea48615d6dd1ff31045cd67a15ef60a"
 
 PRIORITIES = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED"]
 MODES = ["REG AIR", "AIR", "RAIL", "SHIP", "TRUCK", "MAIL", "FOB"]
@@ -81,7 +82,12 @@ def main():
     assert got == Q3_MD5, f"Q3 {got}"
     assert verify_q6(lineitems) == Q6_MD5, "Q6"
     assert verify_q12(orders, lineitems) == Q12_MD5, "Q12"
-    print("reference golden verification: Q3(127 rows) / Q6 / Q12 all ok")
+    q17_str = q17_avg_yearly(gen.parts, lineitems)
+    h = hashlib.md5()
+    h.update((q17_str if q17_str is not None else "<null>").encode())
+    assert h.hexdigest() == Q17_MD5, f"Q17 {q17_str!r}"
+    print("reference golden verification: Q3(127 rows) / Q6 / Q12 / Q17 "
+          "all ok")
 
     data = ExactEngineData(gen, customers, orders, lineitems)
 
@@ -92,6 +98,7 @@ def main():
         state_lines.setdefault(t[0], []).append(t)
     churn_npz = {}
     expected = [rows]
+    expected_q17 = [q17_str]
     n_churn = 8
     for b in range(n_churn):
         batch = gen.churn_batch()
@@ -106,6 +113,7 @@ def main():
         cur_lines = [t for ls in state_lines.values() for t in ls]
         expected.append(q3_result(customers, list(state_orders.values()),
                                   cur_lines))
+        expected_q17.append(q17_avg_yearly(gen.parts, cur_lines))
 
     out = os.path.join(os.path.dirname(__file__), "tpch_sf001.npz")
     np.savez_compressed(
@@ -116,8 +124,14 @@ def main():
         l_orderkey=data.l_orderkey, l_extendedprice=data.l_extendedprice,
         l_discount=data.l_discount, l_shipdate=data.l_shipdate,
         n_churn=np.array([n_churn]),
+        p_partkey=np.array([p[0] for p in gen.parts], np.int64),
+        p_brand=np.array([p[1] for p in gen.parts], np.int64),
+        p_container=np.array([p[2] for p in gen.parts], np.int64),
+        l_partkey=data.l_partkey, l_quantity=data.l_quantity,
         expected_json=np.frombuffer(
             json.dumps(expected).encode(), dtype=np.uint8),
+        expected_q17_json=np.frombuffer(
+            json.dumps(expected_q17).encode(), dtype=np.uint8),
         **churn_npz)
     print(f"wrote {out} "
           f"({os.path.getsize(out) / 1e6:.2f} MB, {n_churn} churn batches)")

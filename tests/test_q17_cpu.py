@@ -11,7 +11,7 @@ from materialize_amd.tpch import TpchGen
 from materialize_amd.workloads import Q17Dataflow
 from pyoracle import OracleCtx
 
-BRAND, CONTAINER = 23, 17
+BRAND, CONTAINER = 23, 10
 
 
 def naive_q17(gen):

@@ -20,6 +20,7 @@ from materialize_amd.tpch_exact import q3_md5, render_revenue_1e2
 from materialize_amd.workloads import Q3Dataflow
 
 Q3_MD5 = "637be0ff3f50cd612b004a69958bfccb"
+Q17_MD5 = "6ea48615d6dd1ff31045cd67a15ef60a"
 START = date(1992, 1, 1)
 FIXTURE = os.path.join(os.path.dirname(__file__), "golden",
                        "tpch_sf001.npz")
@@ -40,6 +41,11 @@ class FixtureData:
         self.l_extendedprice = z["l_extendedprice"]
         self.l_discount = z["l_discount"]
         self.l_shipdate = z["l_shipdate"]
+        self.l_partkey = z["l_partkey"]
+        self.l_quantity = z["l_quantity"]
+        self.p_partkey = z["p_partkey"]
+        self.p_brand = z["p_brand"]
+        self.p_container = z["p_container"]
 
     def customer_updates(self):
         return self.c_custkey, self.c_mktsegment.reshape(-1, 1)
@@ -65,6 +71,20 @@ class FixtureData:
         v[:, 8:16] = self.l_discount.view(np.uint8).reshape(n, 8)
         v[:, 16:20] = self.l_shipdate.view(np.uint8).reshape(n, 4)
         return self.l_orderkey, v
+
+    def lineitem_bypart_updates(self):
+        n = len(self.l_partkey)
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = self.l_quantity.view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = self.l_extendedprice.view(np.uint8).reshape(n, 8)
+        return self.l_partkey, v
+
+    def part_updates(self):
+        n = len(self.p_partkey)
+        v = np.zeros((n, 16), np.uint8)
+        v[:, 0:8] = self.p_brand.view(np.uint8).reshape(n, 8)
+        v[:, 8:16] = self.p_container.view(np.uint8).reshape(n, 8)
+        return self.p_partkey, v
 
 
 def _apply_corrections(state, cols):
@@ -146,3 +166,54 @@ def test_oracle_matches_reference_golden():
 def test_gpu_matches_reference_golden():
     from materialize_amd._ffi import GpuCtx
     _run(GpuCtx())
+
+
+def _q17_render(cents):
+    """avg_yearly = sum(extendedprice)/7.0 in the reference's decNumber
+    39-digit pipeline (engine keeps cents; cents/100 is exact, so the
+    single rounding matches sum_dollars/7.0)."""
+    from decimal import ROUND_HALF_EVEN, Decimal, localcontext
+    with localcontext() as c:
+        c.prec = 39
+        c.rounding = ROUND_HALF_EVEN
+        return format((Decimal(cents) / Decimal(100)) / Decimal("7.0"),
+                      "f")
+
+
+def _run_q17(ctx):
+    import hashlib
+
+    from materialize_amd.workloads import Q17Dataflow
+    z = np.load(FIXTURE)
+    gen = FixtureData(z)
+    expected = json.loads(bytes(z["expected_q17_json"]).decode())
+    df = Q17Dataflow(ctx)
+    df.load(gen)
+
+    def got():
+        s = df.result.get(0)
+        return None if s is None else _q17_render(int(s))
+
+    g0 = got()
+    assert g0 == expected[0]
+    h = hashlib.md5()
+    h.update((g0 if g0 is not None else "<null>").encode())
+    assert h.hexdigest() == Q17_MD5, "snapshot Q17 != reference golden"
+    for b in range(int(z["n_churn"][0])):
+        churn = {"lineitem_by_part": (z[f"b{b}_lineitem_by_part_keys"],
+                                      z[f"b{b}_lineitem_by_part_vals"],
+                                      z[f"b{b}_lineitem_by_part_diffs"])}
+        df.step(churn, b + 1)
+        assert got() == expected[b + 1], f"churn {b}"
+    ctx.close()
+
+
+def test_q17_oracle_matches_reference_golden():
+    from pyoracle import OracleCtx
+    _run_q17(OracleCtx())
+
+
+@pytest.mark.gpu
+def test_q17_gpu_matches_reference_golden():
+    from materialize_amd._ffi import GpuCtx
+    _run_q17(GpuCtx())
