@@ -48,6 +48,8 @@ def parse_args():
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--call-profile", action="store_true",
                    help="print per-engine-call wall time breakdown")
+    p.add_argument("--no-pipeline", action="store_true",
+                   help="disable the 1-deep insert pipeline (A/B control)")
     p.add_argument("--workload", default="q3", choices=["q3", "q17"],
                    help="q3 = BASELINE config 2/3 (default, the metric's "
                         "workload); q17 = config 5 shape at N=1")
@@ -405,22 +407,35 @@ def main():
     df.maintain()
     ctx.lib.mz_gpu_sync(ctx.ctx)
 
-    # pre-generate + pre-filter + pre-stage all churn batches (untimed)
+    # pre-generate + pre-filter + pre-stage all churn batches (untimed).
+    # One extra batch beyond the window: the 1-deep insert pipeline means
+    # step i also enqueues batch i+1's lane consolidations, so the timed
+    # region executes exactly K ingest pipelines (batches W+1..W+K; batch
+    # W's ran during warmup, priming the pipeline) — the closing
+    # mz_gpu_sync waits for the lanes, keeping the window honest.
     K, W = args.steps, args.warmup
+    pipeline = not args.no_pipeline
     staged = []
     rows_per_step = []
-    for i in range(W + K):
+    for i in range(W + K + (1 if pipeline else 0)):
         churn = gen.churn(batch_rows)
-        rows_per_step.append(len(churn["lineitem"][0]) +
-                             len(churn["orders"][0]))
+        if i < W + K:
+            rows_per_step.append(len(churn["lineitem"][0]) +
+                                 len(churn["orders"][0]))
         churn = filter_shard(churn, world, rank)
         staged.append(stage_churn(churn, i + 1, device))
 
+    def nxt(i):
+        return staged[i + 1] if pipeline and i + 1 < len(staged) else None
+
     for i in range(W):
-        corr = df.step_dev(staged[i], i + 1)
+        corr = df.step_dev(staged[i], i + 1, next_upd=nxt(i))
         if corr is not None:
             corr.release()
-    ctx.lib.mz_gpu_sync(ctx.ctx)
+    # device-wide sync WITHOUT mz_gpu_sync: that call force-flushes
+    # pending inserts, which would install batch W before the window and
+    # break the primed 1-deep pipeline (the closing sync below still
+    # installs/drains everything inside the timed region)
     torch.cuda.synchronize()
     if dist:
         dist.barrier()
@@ -433,11 +448,11 @@ def main():
     step_times = [] if os.environ.get("MZ_BENCH_STEP_TIMES") else None
     t0 = time.perf_counter()
     for i in range(W, W + K):
-        corr = df.step_dev(staged[i], i + 1)
+        corr = df.step_dev(staged[i], i + 1, next_upd=nxt(i))
         if corr is not None:
             corr.release()
         if step_times is not None:
-            ctx.lib.mz_gpu_sync(ctx.ctx)
+            torch.cuda.synchronize()  # not mz_gpu_sync: keep pendings
             step_times.append(time.perf_counter() - t0 -
                               sum(step_times))
     ctx.lib.mz_gpu_sync(ctx.ctx)

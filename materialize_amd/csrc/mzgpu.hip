@@ -1514,6 +1514,9 @@ struct mz_gpu_arr {
     int active = 0;
     u64 cnt[3] = {0, 0, 0};
     DevBatch batch;
+    u64 lower = 0;  // batch frontier: a time-filtered probe whose delta
+                    // upper <= lower cannot see this batch, so it need
+                    // not force the flush (1-deep insert pipelining)
     u64 upper = 0;
     // the insert's consolidated FLAT key/val rows (same order as
     // batch.times/diffs), kept so mz_gpu_arr_flush_take can hand the
@@ -1639,7 +1642,12 @@ struct LaneGuard {
   mz_gpu_arr *a;
   hipStream_t ps;
   Scratch *pscr;
-  LaneGuard(Ctx *ctx, mz_gpu_arr *arr) : c(ctx), a(arr) {
+  // gate=false: lane work does NOT wait for the main stream's prior
+  // enqueues. Only legal for pipelines that touch exclusively fresh
+  // memory (the async-insert consolidation: stage + sort + emit into
+  // newly-allocated arrays) — install/merge/free paths MUST gate, since
+  // they retire batches that in-flight main-stream probes still read.
+  LaneGuard(Ctx *ctx, mz_gpu_arr *arr, bool gate = true) : c(ctx), a(arr) {
     ps = c->stream;
     pscr = c->scr;
     if (!a->stream) {
@@ -1656,7 +1664,7 @@ struct LaneGuard {
       HIP_CHECK(hipEventCreate(&a->ev_ready));
       (void)hipEventRecord(a->ev_ready, a->stream);
     }
-    if (c->stream != a->stream) {
+    if (gate && c->stream != a->stream) {
       (void)hipEventRecord(a->ev_gate, c->stream);
       (void)hipStreamWaitEvent(a->stream, a->ev_gate, 0);
     }
@@ -3484,7 +3492,10 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
     return;
   }
   if (a->pending.active) arr_flush_impl(ctx, a);
-  LaneGuard lane(ctx, a);
+  // gate=false: this pipeline writes only freshly-allocated memory, so
+  // it may run concurrently with the main stream's in-flight probes of
+  // the arrangement's CURRENT batches (the 1-deep insert pipeline).
+  LaneGuard lane(ctx, a, /*gate=*/false);
   auto &S = (*ctx->scr);
   S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
@@ -3507,6 +3518,7 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
                                 u->upper, dcounts, /*keep_flat=*/1);
   a->pending.active = 1;
   a->pending.batch = b;
+  a->pending.lower = u->lower;
   a->pending.upper = u->upper;
   a->pending.flat_keys = ok;
   a->pending.flat_vals = ov;
@@ -3754,8 +3766,17 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
     return probe_vl_impl(ctx, lookup, u, stream_vb, mode, swap, cl, out);
   // a probe after insert_async must see the batch (probes of the OLD
   // state precede the insert call entirely), and must run after the
-  // lookup lane's enqueued maintenance
-  arr_flush_impl(ctx, lookup);
+  // lookup lane's enqueued maintenance. Exception (1-deep insert
+  // pipeline): a time-filtered probe (le/lt half-join) whose delta times
+  // all precede a pending batch's lower frontier cannot match any of its
+  // rows (t2 >= lower >= delta.upper > t1), so the pending insert may
+  // keep consolidating on its lane while this probe runs — skipping the
+  // flush (and its lane sync) entirely.
+  bool future_pending =
+      (mode == PM_HALF_LE || mode == PM_HALF_LT) &&
+      lookup->pending.active && !lookup->pending_merge.active &&
+      lookup->pending.lower >= u->upper;
+  if (!future_pending) arr_flush_impl(ctx, lookup);
   if (lookup->stream)
     (void)hipStreamWaitEvent(ctx->stream, lookup->ev_ready, 0);
   (*ctx->scr).reset();
