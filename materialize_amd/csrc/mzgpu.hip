@@ -2215,17 +2215,23 @@ void consolidate_dev(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 **okeys,
 }
 
 void free_batch(Ctx *c, DevBatch &b) {
+  // Batches are probed on the MAIN stream; freeing there (in-order after
+  // every probe enqueued so far) is always safe, and lets lane/flush
+  // paths retire batches without gating the lane on the main stream.
   const char *names[10] = {"keys", "kv_off", "vals", "vu_off", "v_offs",
                            "val_key", "times", "diffs", "upd_val", "hash"};
   void *ps[10] = {(void *)b.keys, (void *)b.kv_off, (void *)b.vals,
                   (void *)b.vu_off, (void *)b.v_offs, (void *)b.val_key,
                   (void *)b.times, (void *)b.diffs, (void *)b.upd_val,
                   (void *)b.hash};
+  hipStream_t ps_stream = c->stream;
+  if (c->main_stream) c->stream = c->main_stream;
   for (int i = 0; i < 10; i++) {
     if (getenv("MZ_DBG_FINI") && ps[i])
       fprintf(stderr, "[free_batch] %s %p\n", names[i], ps[i]);
     dfree(c, ps[i]);
   }
+  c->stream = ps_stream;
   b = DevBatch();
 }
 
@@ -2488,16 +2494,9 @@ void merge_install(Ctx *c, mz_gpu_arr *a) {
   merged.n_upds = pm.cnt[0];
   merged.n_keys = pm.cnt[1];
   merged.n_vals = pm.cnt[2];
-  // gate: lane frees must run after main-stream probes of the inputs
-  (void)hipEventRecord(a->ev_gate, c->main_stream);
-  (void)hipStreamWaitEvent(a->stream ? a->stream : c->stream, a->ev_gate,
-                           0);
-  {
-    hipStream_t ps = c->stream;
-    if (a->stream) c->stream = a->stream;
-    for (size_t i = pm.from; i < pm.to; i++) free_batch(c, a->batches[i]);
-    c->stream = ps;
-  }
+  // frees run on the MAIN stream (free_batch), in-order after every
+  // probe of the inputs enqueued so far — no lane gating needed
+  for (size_t i = pm.from; i < pm.to; i++) free_batch(c, a->batches[i]);
   a->batches.erase(a->batches.begin() + pm.from,
                    a->batches.begin() + pm.to);
   a->batches.insert(a->batches.begin() + pm.from, merged);
@@ -3531,7 +3530,12 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
   if (take) *take = nullptr;
   if (!a->pending.active && !a->pending_merge.active) return;
   MZ_PROF(ctx, "arr_flush");
-  LaneGuard lane(ctx, a);
+  // gate=false: the flush waits only for the LANE's own pipeline (the
+  // pending consolidation/merge), not for the main stream's in-flight
+  // probes of the previous step — installs mutate only the host batch
+  // list, and retired batches are freed on the main stream (free_batch),
+  // in-order after every probe that reads them.
+  LaneGuard lane(ctx, a, /*gate=*/false);
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
   merge_install(ctx, a);
   if (a->pending.active) {
