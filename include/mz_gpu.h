@@ -301,6 +301,21 @@ int  mz_gpu_halfjoin(mz_gpu_ctx *ctx, mz_gpu_arr *lookup,
                      const mz_gpu_updates *delta, uint32_t stream_val_bytes,
                      int le, const mz_gpu_closure *cl, mz_gpu_out **out);
 
+/* Fused two-stage delta path (one path's two lookup stages,
+ * delta_join.rs:338-472, in a single kernel): delta -> lookup1 (le1,
+ * closure1 produces the intermediate key/val in registers) -> lookup2
+ * (le2, closure2 produces the output). Equivalent to two halfjoin calls
+ * with the intermediate stream never materialized; output is RAW
+ * (unconsolidated — the consumer consolidates) with the err stream
+ * attached. Constraints: fixed-width lookups; closure1's output key
+ * must be lookup2's key schema, its output <= 2 key words / 48 val
+ * bytes. */
+int  mz_gpu_halfjoin2(mz_gpu_ctx *ctx, mz_gpu_arr *lookup1, int le1,
+                      const mz_gpu_closure *cl1, mz_gpu_arr *lookup2,
+                      int le2, const mz_gpu_closure *cl2,
+                      const mz_gpu_updates *delta,
+                      uint32_t stream_val_bytes, mz_gpu_out **out);
+
 /* --------------------------------------------------------------- reduce
  * Replaces build_accumulable + mz_reduce_abelian (reduce.rs:1357-1581,
  * extensions/reduce.rs:131). The operator owns the resident accumulator

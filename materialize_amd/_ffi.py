@@ -85,6 +85,11 @@ def load():
                                     C.POINTER(Closure),
                                     C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_halfjoin_raw.argtypes = lib.mz_gpu_halfjoin.argtypes
+    lib.mz_gpu_halfjoin2.argtypes = [C.c_void_p, C.c_void_p, C.c_int,
+                                     C.POINTER(Closure), C.c_void_p,
+                                     C.c_int, C.POINTER(Closure),
+                                     C.POINTER(Updates), C.c_uint32,
+                                     C.POINTER(C.POINTER(OutBatch))]
     lib.mz_gpu_reduce_create.restype = C.c_void_p
     lib.mz_gpu_reduce_create.argtypes = [C.c_void_p, C.POINTER(ReduceSpec)]
     lib.mz_gpu_reduce_push.argtypes = [C.c_void_p, C.c_void_p,
@@ -393,6 +398,15 @@ class GpuCtx:
         self._check(self.lib.mz_gpu_halfjoin_raw(
             self.ctx, lookup, C.byref(upd), stream_vb, 1 if le else 0,
             C.byref(cl), C.byref(outp)))
+        return self._dev_out(outp)
+
+    def halfjoin2_dev(self, lk1, le1, cl1, lk2, le2, cl2, upd, stream_vb):
+        # fused two-stage delta path (k_probe_path2); raw output
+        outp = C.POINTER(OutBatch)()
+        self._check(self.lib.mz_gpu_halfjoin2(
+            self.ctx, lk1, 1 if le1 else 0, C.byref(cl1), lk2,
+            1 if le2 else 0, C.byref(cl2), C.byref(upd), stream_vb,
+            C.byref(outp)))
         return self._dev_out(outp)
 
     def partition_dev(self, sch, upd, nshards, out_tensors):

@@ -697,6 +697,135 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
   }
 }
 
+// Fused two-stage delta-path probe (one delta path's two lookup stages,
+// delta_join.rs:338-472, in a single kernel): a stage-1 match produces
+// its intermediate (key2, val2) in REGISTERS and immediately probes the
+// stage-2 arrangement — the intermediate stream is never materialized in
+// HBM, and the per-stage launch + counter-readback round trip
+// disappears. Half-join modes only (le/lt per stage, no swap): times are
+// total-ordered u64s and each stage's time filter keeps t2 (<=|<) t, so
+// the promoted output time stays the delta time t and the stage-2 filter
+// compares against the same t (delta_join.rs:157-160,362,372). Counting
+// and emission walk the same nested ranges (emission L2-hot); *ctr[0/1]
+// always accumulate exact ok/err totals (host relaunches on overflow),
+// ctr[2] counts intermediate rows (stats/roofline only).
+#define P2_MAX_KW 2
+#define P2_MAX_VB 48
+__global__ void k_probe_path2(const u64 *dkeys, const u8 *dvals, u32 dvb,
+                              const u64 *dtimes, const i64 *ddiffs, u64 n,
+                              u32 kw, u32 lvb1, BatchList bl1, int mode1,
+                              const mz_gpu_closure cl1, u32 lvb2,
+                              BatchList bl2, int mode2,
+                              const mz_gpu_closure cl2, u64 cap, u64 ecap,
+                              unsigned long long *ctr, u64 *okeys,
+                              u8 *ovals, u64 *otimes, i64 *odiffs,
+                              u64 *ecodes, u64 *etimes, i64 *ediffs) {
+  u32 okw = cl2.out.key_words, ovb = cl2.out.val_bytes;
+  u32 k2w = cl1.out.key_words;
+  u64 total = n * (u64)bl1.n;
+  u64 stride = (u64)gridDim.x * blockDim.x;
+  u64 start = blockIdx.x * (u64)blockDim.x + threadIdx.x;
+  u64 iters = (total + stride - 1) / stride;  // uniform across the wave
+  u32 lane = threadIdx.x & 63;
+  for (u64 it0 = 0; it0 < iters; it0++) {
+    u64 idx = start + it0 * stride;
+    u64 kvr1 = ~0ull;
+    u64 i = 0;
+    int bi = 0;
+    if (idx < total) {
+      i = idx % n;
+      bi = (int)(idx / n);
+      kvr1 = hash_lookup_range(bl1.b[bi].hash, bl1.b[bi].hash_slots,
+                               dkeys + i * kw, kw);
+    }
+    // count: final emissions + stage-1 error rows for this pair
+    u32 c = 0, e = 0;
+    u64 inter = 0;
+    if (kvr1 != ~0ull) {
+      const DevBatch &b1 = bl1.b[bi];
+      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+      u64 t = dtimes[i];
+      for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
+        const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
+        int cls = d_closure_apply(&cl1, dkeys + i * kw, dv, lv, nullptr,
+                                  nullptr);
+        if (cls == 0) continue;
+        u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
+        u32 m1 = 0;
+        if (bl1.allpass[bi]) {
+          m1 = hi - lo;
+        } else {
+          for (u32 u = lo; u < hi; u++)
+            m1 += (mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
+                                        : (b1.times[u] < t);
+        }
+        if (m1 == 0) continue;
+        if (cls == 2) {
+          e += m1;
+          continue;
+        }
+        inter += m1;
+        u64 k2[P2_MAX_KW];
+        u8 v2buf[P2_MAX_VB];
+        (void)d_closure_apply(&cl1, dkeys + i * kw, dv, lv, k2, v2buf);
+        for (int b2i = 0; b2i < bl2.n; b2i++) {
+          const DevBatch &b2 = bl2.b[b2i];
+          u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
+          if (kvr2 == ~0ull) continue;
+          u64 ce2 = d_count_pair(b2, bl2.allpass[b2i], k2, v2buf, t, kvr2,
+                                 mode2, 0, cl2, lvb2);
+          c += m1 * (u32)ce2;
+          e += m1 * (u32)(ce2 >> 32);
+        }
+      }
+    }
+    u64 base = wave_reserve(ctr, c, lane);
+    u64 ebase = wave_reserve(ctr + 1, e, lane);
+    if (inter) atomicAdd(ctr + 2, (unsigned long long)inter);
+    if ((c == 0 && e == 0) || base + c > cap || ebase + e > ecap)
+      continue;
+    // emit (second walk over L2-hot ranges)
+    u64 o = base, eo = ebase;
+    const DevBatch &b1 = bl1.b[bi];
+    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
+    u64 t = dtimes[i];
+    i64 d0 = ddiffs[i];
+    for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
+      const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
+      int cls = d_closure_apply(&cl1, dkeys + i * kw, dv, lv, nullptr,
+                                nullptr);
+      if (cls == 0) continue;
+      u64 k2[P2_MAX_KW];
+      u8 v2buf[P2_MAX_VB];
+      if (cls == 1)
+        (void)d_closure_apply(&cl1, dkeys + i * kw, dv, lv, k2, v2buf);
+      u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
+      for (u32 u = lo; u < hi; u++) {
+        if (!bl1.allpass[bi] &&
+            !((mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
+                                    : (b1.times[u] < t)))
+          continue;
+        i64 d1 = wmul(d0, b1.diffs[u]);
+        if (cls == 2) {
+          ecodes[eo] = MZ_ERR_DIVISION_BY_ZERO;
+          etimes[eo] = t;
+          ediffs[eo] = d1;
+          eo++;
+          continue;
+        }
+        for (int b2i = 0; b2i < bl2.n; b2i++) {
+          const DevBatch &b2 = bl2.b[b2i];
+          u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
+          if (kvr2 == ~0ull) continue;
+          d_emit_pair(b2, bl2.allpass[b2i], k2, v2buf, t, d1, kvr2, mode2,
+                      0, cl2, lvb2, okw, ovb, &o, okeys, ovals, otimes,
+                      odiffs, &eo, ecodes, etimes, ediffs);
+        }
+      }
+    }
+  }
+}
+
 // Merge probe for SORTED delta streams against one large batch: both
 // sides are ascending in the canonical key order, so each block narrows
 // the batch's key array to its delta rows' range with two binary
@@ -1497,6 +1626,8 @@ struct mz_gpu_arr {
   u64 upper = 0;
   u64 probe_cap_hint = 0;  // last probe's ceil(matches/row): sizes the
                            // single-walk output queue (k_probe_walk)
+  u64 path_cap_hint = 0;   // same, for the fused two-stage path probe
+                           // starting at this arrangement (k_probe_path2)
   Ctx *ctx = nullptr;
   // Per-arrangement lane: inserts/merges run on this stream with this
   // scratch arena so independent arrangements' maintenance overlaps;
@@ -4015,6 +4146,155 @@ int mz_gpu_halfjoin_raw(mz_gpu_ctx *c, mz_gpu_arr *lookup,
                         const mz_gpu_closure *cl, mz_gpu_out **out) {
   return probe_impl(&c->impl, lookup, delta, stream_val_bytes,
                     le ? PM_HALF_LE : PM_HALF_LT, 0, cl, 0, out);
+}
+
+// Probe-visible BatchList (with the >12-batch synchronous catch-up of
+// probe_impl); returns false after a catch-up merge — retry.
+static bool build_probe_batchlist(Ctx *ctx, mz_gpu_arr *a, u64 delta_lower,
+                                  BatchList &bl) {
+  bl.n = 0;
+  for (auto &b : a->batches) {
+    if (b.n_upds == 0) continue;
+    if (bl.n >= 12) {
+      if (a->pending_merge.active) {
+        LaneGuard l2(ctx, a);
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        merge_install(ctx, a);
+      }
+      merge_range(ctx, a, 0, a->batches.size());
+      return false;
+    }
+    u64 tmax_excl = std::max(b.upper, a->logical_compaction + 1);
+    bl.allpass[bl.n] = tmax_excl <= delta_lower ? 1 : 0;
+    bl.b[bl.n++] = b;
+  }
+  return true;
+}
+
+// Fused two-stage delta path (k_probe_path2): delta -> lookup1 (le1) ->
+// lookup2 (le2), raw (unconsolidated) output with the err stream
+// attached. The render layer uses this when a DeltaPathPlan has exactly
+// two local stages; semantics identical to two mz_gpu_halfjoin_raw
+// calls with the intermediate consolidation deferred (every consumer
+// consolidates).
+int mz_gpu_halfjoin2(mz_gpu_ctx *c, mz_gpu_arr *lk1, int le1,
+                     const mz_gpu_closure *cl1, mz_gpu_arr *lk2, int le2,
+                     const mz_gpu_closure *cl2, const mz_gpu_updates *u,
+                     uint32_t stream_vb, mz_gpu_out **out) {
+  Ctx *ctx = &c->impl;
+  if (is_varlen(lk1->schema.vb) || is_varlen(lk2->schema.vb)) {
+    ctx->err = "halfjoin2: varlen lookups unsupported (use halfjoin)";
+    return 1;
+  }
+  if (cl1->out.key_words == 0 || cl1->out.key_words > P2_MAX_KW ||
+      cl1->out.val_bytes > P2_MAX_VB ||
+      cl1->out.key_words != lk2->schema.kw) {
+    ctx->err = "halfjoin2: stage-1 closure output shape unsupported";
+    return 1;
+  }
+  for (mz_gpu_arr *lk : {lk1, lk2}) {
+    bool future_pending = lk->pending.active &&
+                          !lk->pending_merge.active &&
+                          lk->pending.lower >= u->upper;
+    if (!future_pending) arr_flush_impl(ctx, lk);
+    if (lk->stream)
+      (void)hipStreamWaitEvent(ctx->stream, lk->ev_ready, 0);
+  }
+  BatchList bl1, bl2;
+  while (!build_probe_batchlist(ctx, lk1, u->lower, bl1)) {}
+  while (!build_probe_batchlist(ctx, lk2, u->lower, bl2)) {}
+  (*ctx->scr).reset();
+  auto &S = (*ctx->scr);
+  u32 kw = lk1->schema.kw, lvb1 = lk1->schema.vb, lvb2 = lk2->schema.vb;
+  u32 okw = cl2->out.key_words, ovb = cl2->out.val_bytes;
+  DevUpdates d = stage_updates(ctx, u, kw, stream_vb);
+  u64 n = d.n;
+  if (n == 0 || bl1.n == 0 || bl2.n == 0) {
+    *out = make_out(dnew<u64>(ctx, 1), (u8 *)dmalloc(ctx, 1),
+                    dnew<u64>(ctx, 1), dnew<i64>(ctx, 1), 0, okw, ovb);
+    return 0;
+  }
+  int mode1 = le1 ? PM_HALF_LE : PM_HALF_LT;
+  int mode2 = le2 ? PM_HALF_LE : PM_HALF_LT;
+  u64 cap = (lk1->path_cap_hint ? lk1->path_cap_hint + 1 : 2) * n + 1024;
+  u64 ecap = n / 4 + 1024;
+  unsigned long long *ctr = (unsigned long long *)S.get(24);
+  fill_u64(ctx, (u64 *)ctr, 3, 0);  // [0] ok, [1] err, [2] intermediates
+  u64 *pk = dnew<u64>(ctx, cap * okw);
+  u8 *pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
+  u64 *pt = dnew<u64>(ctx, cap);
+  i64 *pd = dnew<i64>(ctx, cap);
+  u64 *ek = dnew<u64>(ctx, ecap);
+  u64 *et = dnew<u64>(ctx, ecap);
+  i64 *ed = dnew<i64>(ctx, ecap);
+  u64 total = n * (u64)bl1.n;
+  auto launch = [&]() {
+    hipLaunchKernelGGL(k_probe_path2, dim3(ngrid(total)), dim3(BLK), 0,
+                       ctx->stream, d.keys, d.vals, stream_vb, d.times,
+                       d.diffs, n, kw, lvb1, bl1, mode1, *cl1, lvb2, bl2,
+                       mode2, *cl2, cap, ecap, ctr, pk, pv, pt, pd, ek,
+                       et, ed);
+  };
+  if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+  launch();
+  if (ctx->time_kernels) HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+  unsigned long long *MB = (unsigned long long *)d2h_pinned(ctx, ctr, 24);
+  u64 M = MB[0], E = MB[1], I = MB[2];
+  u64 launches = 1;
+  if (M > cap || E > ecap) {  // rare: queue overflow — exact relaunch
+    for (void *p : {(void *)pk, (void *)pv, (void *)pt, (void *)pd,
+                    (void *)ek, (void *)et, (void *)ed})
+      dfree(ctx, p);
+    cap = std::max<u64>(M, 1);
+    ecap = std::max<u64>(E, 1);
+    pk = dnew<u64>(ctx, cap * okw);
+    pv = (u8 *)dmalloc(ctx, std::max<u64>(cap * ovb, 1));
+    pt = dnew<u64>(ctx, cap);
+    pd = dnew<i64>(ctx, cap);
+    ek = dnew<u64>(ctx, ecap);
+    et = dnew<u64>(ctx, ecap);
+    ed = dnew<i64>(ctx, ecap);
+    fill_u64(ctx, (u64 *)ctr, 3, 0);
+    if (ctx->time_kernels)
+      HIP_CHECK(hipEventRecord(ctx->ev_a, ctx->stream));
+    launch();
+    if (ctx->time_kernels)
+      HIP_CHECK(hipEventRecord(ctx->ev_b, ctx->stream));
+    launches = 2;
+  }
+  lk1->path_cap_hint = (M + n - 1) / n;
+  if (ctx->time_kernels) {
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_a, ctx->ev_b));
+    ctx->probe_ms += ms;
+    ctx->probe_rows += n;
+    ctx->probe_launches += launches;
+    ctx->probe_pairs += M;
+    ctx->probe_batches += (u64)bl1.n + (u64)bl2.n;
+    // algorithmic bytes: delta tuple + one hash line per stage-1 batch
+    // + matched stage-1 val/upds + one hash line per stage-2 batch per
+    // intermediate + matched stage-2 val/upds + output write. The
+    // intermediate tuple itself stays in registers (never HBM).
+    ctx->probe_alg_bytes +=
+        n * (8ull * kw + stream_vb + 16) + n * 64ull * (u64)bl1.n +
+        I * (lvb1 + 16ull) + I * 64ull * (u64)bl2.n +
+        M * (lvb2 + 16ull) + M * (8ull * okw + ovb + 16);
+  }
+  u64 *cek = nullptr, *cet = nullptr;
+  i64 *ced = nullptr;
+  u64 Ec = 0;
+  if (E) {
+    DevUpdates epin{ek, nullptr, et, ed, E};
+    u8 *unused_v;
+    consolidate_dev(ctx, 1, 0, epin, &cek, &unused_v, &cet, &ced, &Ec);
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    dfree(ctx, unused_v);
+  }
+  for (void *p : {(void *)ek, (void *)et, (void *)ed}) dfree(ctx, p);
+  *out = make_out(pk, pv, pt, pd, M, okw, ovb);
+  out_attach_errs(*out, cek, cet, ced, Ec);
+  return 0;
 }
 
 mz_gpu_red *mz_gpu_reduce_create(mz_gpu_ctx *c,

@@ -161,6 +161,24 @@ class DeltaJoinOp:
         into HBM and passes device columns)."""
         ctx = self.ctx
         collective = self.exchange is not None
+        # Fused two-stage path (k_probe_path2): one kernel runs both
+        # lookups with the intermediate in registers. Only when no
+        # inter-stage exchange is needed and the intermediate shape fits
+        # the kernel's register buffers (C side validates too).
+        if (not collective and len(path.stages) == 2
+                and hasattr(ctx, "halfjoin2_dev")):
+            s1, s2 = path.stages
+            o1 = s1.closure.out
+            arr2 = self.arrangements[s2.lookup_relation]
+            if 0 < o1.key_words <= 2 and o1.val_bytes <= 48:
+                cur = ctx.halfjoin2_dev(
+                    self.arrangements[s1.lookup_relation], s1.le,
+                    s1.closure, arr2, s2.le, s2.closure, u,
+                    s1.stream_val_bytes)
+                if cur.n == 0:
+                    cur.release()
+                    return None
+                return cur
         cur = None
         for i, st in enumerate(path.stages):
             arr = self.arrangements[st.lookup_relation]
