@@ -1640,6 +1640,12 @@ struct mz_gpu_arr {
   // final for the step (after install+push, BEFORE a deferred merge is
   // enqueued — the merge must not block the probes it runs under)
   hipEvent_t ev_ready = nullptr;
+  // Deferred spine merges run on their OWN stream with their own scratch
+  // so they never queue ahead of the lane's insert consolidations;
+  // ev_mdone (recorded after the merge's count copy) gates the install.
+  hipStream_t mstream = nullptr;
+  Scratch *merge_scr = nullptr;
+  hipEvent_t ev_mdone = nullptr;
   // deferred insert (arr_insert_async): counts land here asynchronously
   struct Pending {
     int active = 0;
@@ -1804,6 +1810,34 @@ struct LaneGuard {
   }
   ~LaneGuard() {
     (void)hipEventRecord(a->ev_done, a->stream);
+    c->stream = ps;
+    c->scr = pscr;
+  }
+};
+
+// Swap the ctx onto an arrangement's MERGE stream (own scratch): a
+// deferred merge reads only installed, fully-computed batches, so it
+// needs no ordering edge at all — it runs concurrently with the lane's
+// consolidations AND the main stream's probes. Its install is gated by
+// ev_mdone (recorded here at scope exit, after the count copy).
+struct MergeGuard {
+  Ctx *c;
+  mz_gpu_arr *a;
+  hipStream_t ps;
+  Scratch *pscr;
+  MergeGuard(Ctx *ctx, mz_gpu_arr *arr) : c(ctx), a(arr) {
+    ps = c->stream;
+    pscr = c->scr;
+    if (!a->mstream) {
+      HIP_CHECK(hipStreamCreate(&a->mstream));
+      a->merge_scr = new Scratch();
+      HIP_CHECK(hipEventCreate(&a->ev_mdone));
+    }
+    c->stream = a->mstream;
+    c->scr = a->merge_scr;
+  }
+  ~MergeGuard() {
+    (void)hipEventRecord(a->ev_mdone, a->mstream);
     c->stream = ps;
     c->scr = pscr;
   }
@@ -3348,6 +3382,14 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
       delete a->lane_scr;
       a->stream = nullptr;
     }
+    if (a->mstream) {
+      (void)hipStreamSynchronize(a->mstream);
+      (void)hipStreamDestroy(a->mstream);
+      (void)hipEventDestroy(a->ev_mdone);
+      if (a->merge_scr) a->merge_scr->destroy();
+      delete a->merge_scr;
+      a->mstream = nullptr;
+    }
   }
   (void)hipStreamSynchronize(ctx->stream);
   const char *dbg = getenv("MZ_DBG_FINI");
@@ -3357,6 +3399,8 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
     for (auto &b : a->batches) free_batch(ctx, b);
     dfree(ctx, a->pending.flat_keys);
     dfree(ctx, a->pending.flat_vals);
+    if (a->pending.active) free_batch(ctx, a->pending.batch);
+    if (a->pending_merge.active) free_batch(ctx, a->pending_merge.merged);
     delete a;
   }
   ctx->arrs.clear();
@@ -3392,7 +3436,12 @@ const char *mz_gpu_last_error(mz_gpu_ctx *c) { return c->impl.err.c_str(); }
 int mz_gpu_sync(mz_gpu_ctx *c) {
   for (mz_gpu_arr *a : c->impl.arrs) {
     arr_flush_impl(&c->impl, a);
+    if (a->pending_merge.active) {
+      if (a->mstream) HIP_CHECK(hipStreamSynchronize(a->mstream));
+      merge_install(&c->impl, a);
+    }
     if (a->stream) HIP_CHECK(hipStreamSynchronize(a->stream));
+    if (a->mstream) HIP_CHECK(hipStreamSynchronize(a->mstream));
   }
   HIP_CHECK(hipStreamSynchronize(c->impl.stream));
   return 0;
@@ -3409,8 +3458,7 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
   Ctx *ctx = &c->impl;
   arr_flush_impl(ctx, a);
   if (a->pending_merge.active) {  // in-flight deferred merge reads batches
-    LaneGuard lane(ctx, a);
-    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    if (a->mstream) HIP_CHECK(hipStreamSynchronize(a->mstream));
     merge_install(ctx, a);
   }
   if (a->stream) (void)hipStreamSynchronize(a->stream);
@@ -3426,6 +3474,14 @@ void mz_gpu_arr_drop(mz_gpu_ctx *c, mz_gpu_arr *a) {
     delete a->lane_scr;
     a->stream = nullptr;
     a->lane_scr = nullptr;
+  }
+  if (a->mstream) {
+    (void)hipStreamSynchronize(a->mstream);
+    (void)hipStreamDestroy(a->mstream);
+    (void)hipEventDestroy(a->ev_mdone);
+    if (a->merge_scr) a->merge_scr->destroy();
+    delete a->merge_scr;
+    a->mstream = nullptr;
   }
   auto &v = ctx->arrs;
   v.erase(std::remove(v.begin(), v.end(), a), v.end());
@@ -3459,12 +3515,14 @@ static void spine_policy_deferred(Ctx *ctx, mz_gpu_arr *a) {
       (double)a->batches[nb - 2].n_upds <=
           GEO * (double)a->batches[nb - 1].n_upds &&
       a->batches[nb - 2].n_upds + a->batches[nb - 1].n_upds >= SMALL) {
+    MergeGuard mg(ctx, a);
     merge_range(ctx, a, nb - 2, nb, 1);
     return;
   }
   size_t i = nb;
   while (i > 0 && a->batches[i - 1].n_upds < SMALL) i--;
   if ((long)(nb - i) > POOL) {
+    MergeGuard mg(ctx, a);
     merge_range(ctx, a, i, nb, 1);
     return;
   }
@@ -3668,7 +3726,11 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
   // in-order after every probe that reads them.
   LaneGuard lane(ctx, a, /*gate=*/false);
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
-  merge_install(ctx, a);
+  // install an in-flight deferred merge only when its stream is done —
+  // otherwise leave it running (the pre-merge batch list stays valid)
+  if (a->pending_merge.active &&
+      (!a->ev_mdone || hipEventQuery(a->ev_mdone) == hipSuccess))
+    merge_install(ctx, a);
   if (a->pending.active) {
     DevBatch b = a->pending.batch;
     b.n_upds = a->pending.cnt[0];
@@ -3702,16 +3764,21 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
       a->pending.flat_vals = nullptr;
     }
   }
-  // Measured on the 1M churn config: deferring merges off the probe path
-  // LOSES ~10% — probes pay for the deeper pre-merge spine and the
-  // merges are bandwidth-bound anyway (no free overlap). Kept as an
-  // option for latency-sensitive shapes. ev_ready (what probes wait on)
-  // is recorded BEFORE a deferred merge but AFTER synchronous ones.
-  static const bool DEFER = [] {
-    const char *e = getenv("MZ_GPU_DEFER_MERGE");
+  // Deferred merges (default): the merge runs on the arrangement's OWN
+  // merge stream (MergeGuard) — overlapping the lane's consolidations
+  // and the main stream's probes — and installs at a later flush once
+  // ev_mdone reports completion. Probes pay for a slightly deeper spine
+  // for a step or two; the flush never blocks on a merge. (Round 1's
+  // lane-resident deferral lost ~10% at 1M because merges queued AHEAD
+  // of the next insert's consolidation on the same stream; the separate
+  // stream removes that. MZ_GPU_SYNC_MERGE=1 restores synchronous
+  // merges for A/B.) ev_ready (what probes wait on) is recorded BEFORE
+  // a deferred merge but AFTER synchronous ones.
+  static const bool SYNC_MERGE = [] {
+    const char *e = getenv("MZ_GPU_SYNC_MERGE");
     return e && e[0] && e[0] != '0';
   }();
-  if (DEFER) {
+  if (!SYNC_MERGE) {
     (void)hipEventRecord(a->ev_ready, ctx->stream);
     spine_policy_deferred(ctx, a);
   } else {
@@ -3766,7 +3833,7 @@ int mz_gpu_arr_maintain(mz_gpu_ctx *c, mz_gpu_arr *a, uint64_t fuel) {
   arr_flush_impl(ctx, a);  // pending insert joins; pending merge installs
   LaneGuard lane(ctx, a);
   if (a->pending_merge.active) {  // flush's policy may have enqueued one
-    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    if (a->mstream) HIP_CHECK(hipStreamSynchronize(a->mstream));
     merge_install(ctx, a);
   }
   merge_range(ctx, a, 0, a->batches.size());
@@ -3927,8 +3994,8 @@ static int probe_impl(Ctx *ctx, mz_gpu_arr *lookup, const mz_gpu_updates *u,
       // too many batches: catch up synchronously (install any deferred
       // merge first — its inputs are in the list being merged)
       if (lookup->pending_merge.active) {
-        LaneGuard l2(ctx, lookup);
-        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        if (lookup->mstream)
+          HIP_CHECK(hipStreamSynchronize(lookup->mstream));
         merge_install(ctx, lookup);
       }
       merge_range(ctx, lookup, 0, lookup->batches.size());
@@ -4157,8 +4224,7 @@ static bool build_probe_batchlist(Ctx *ctx, mz_gpu_arr *a, u64 delta_lower,
     if (b.n_upds == 0) continue;
     if (bl.n >= 12) {
       if (a->pending_merge.active) {
-        LaneGuard l2(ctx, a);
-        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        if (a->mstream) HIP_CHECK(hipStreamSynchronize(a->mstream));
         merge_install(ctx, a);
       }
       merge_range(ctx, a, 0, a->batches.size());
