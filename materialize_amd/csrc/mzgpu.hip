@@ -3764,21 +3764,21 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
       a->pending.flat_vals = nullptr;
     }
   }
-  // Deferred merges (default): the merge runs on the arrangement's OWN
-  // merge stream (MergeGuard) — overlapping the lane's consolidations
-  // and the main stream's probes — and installs at a later flush once
-  // ev_mdone reports completion. Probes pay for a slightly deeper spine
-  // for a step or two; the flush never blocks on a merge. (Round 1's
-  // lane-resident deferral lost ~10% at 1M because merges queued AHEAD
-  // of the next insert's consolidation on the same stream; the separate
-  // stream removes that. MZ_GPU_SYNC_MERGE=1 restores synchronous
-  // merges for A/B.) ev_ready (what probes wait on) is recorded BEFORE
-  // a deferred merge but AFTER synchronous ones.
-  static const bool SYNC_MERGE = [] {
-    const char *e = getenv("MZ_GPU_SYNC_MERGE");
+  // Synchronous merges by default: deferral was re-measured (round 2)
+  // with the merge on its OWN stream + own scratch + event-gated
+  // install, and still loses on BOTH configs (100k: 2.50 vs 2.26
+  // ms/step; 1M: 9.3 vs 5.9) — probes pay ~64 B/row for every extra
+  // batch in the pre-merge list (the fused two-stage probe doubly so),
+  // and at 1M the merges are bandwidth-bound, so the overlap has no
+  // spare bandwidth to use. MZ_GPU_DEFER_MERGE=1 enables the
+  // merge-stream deferral for latency-sensitive shapes. ev_ready (what
+  // probes wait on) is recorded BEFORE a deferred merge but AFTER
+  // synchronous ones.
+  static const bool DEFER = [] {
+    const char *e = getenv("MZ_GPU_DEFER_MERGE");
     return e && e[0] && e[0] != '0';
   }();
-  if (!SYNC_MERGE) {
+  if (DEFER) {
     (void)hipEventRecord(a->ev_ready, ctx->stream);
     spine_policy_deferred(ctx, a);
   } else {
