@@ -711,6 +711,92 @@ __global__ void k_probe_walk(const u64 *dkeys, const u8 *dvals, u32 dvb,
 // ctr[2] counts intermediate rows (stats/roofline only).
 #define P2_MAX_KW 2
 #define P2_MAX_VB 48
+#define P2_TILE 4
+
+// Count one (delta row, stage-1 batch) pair's final emissions through
+// both stages; returns ok | err<<32 and adds this pair's intermediate
+// row count to *inter (stats).
+__device__ __forceinline__ u64 d_path2_count(
+    const u64 *key, const u8 *dv, u64 t, const DevBatch &b1, int allpass1,
+    u64 kvr1, int mode1, const mz_gpu_closure &cl1, u32 lvb1, u32 k2w,
+    const BatchList &bl2, int mode2, const mz_gpu_closure &cl2, u32 lvb2,
+    u64 *inter) {
+  u32 c = 0, e = 0;
+  for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
+    const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
+    int cls = d_closure_apply(&cl1, key, dv, lv, nullptr, nullptr);
+    if (cls == 0) continue;
+    u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
+    u32 m1 = 0;
+    if (allpass1) {
+      m1 = hi - lo;
+    } else {
+      for (u32 u = lo; u < hi; u++)
+        m1 += (mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
+                                    : (b1.times[u] < t);
+    }
+    if (m1 == 0) continue;
+    if (cls == 2) {
+      e += m1;
+      continue;
+    }
+    *inter += m1;
+    u64 k2[P2_MAX_KW];
+    u8 v2buf[P2_MAX_VB];
+    (void)d_closure_apply(&cl1, key, dv, lv, k2, v2buf);
+    for (int b2i = 0; b2i < bl2.n; b2i++) {
+      const DevBatch &b2 = bl2.b[b2i];
+      u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
+      if (kvr2 == ~0ull) continue;
+      u64 ce2 = d_count_pair(b2, bl2.allpass[b2i], k2, v2buf, t, kvr2,
+                             mode2, 0, cl2, lvb2);
+      c += m1 * (u32)ce2;
+      e += m1 * (u32)(ce2 >> 32);
+    }
+  }
+  return (u64)c | ((u64)e << 32);
+}
+
+// Emit one pair's rows at queue offsets *o / *eo (same nested walk as
+// the count — ranges are L2-hot on this second pass).
+__device__ __forceinline__ void d_path2_emit(
+    const u64 *key, const u8 *dv, u64 t, i64 d0, const DevBatch &b1,
+    int allpass1, u64 kvr1, int mode1, const mz_gpu_closure &cl1, u32 lvb1,
+    u32 k2w, const BatchList &bl2, int mode2, const mz_gpu_closure &cl2,
+    u32 lvb2, u32 okw, u32 ovb, u64 *o, u64 *okeys, u8 *ovals, u64 *otimes,
+    i64 *odiffs, u64 *eo, u64 *ecodes, u64 *etimes, i64 *ediffs) {
+  for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
+    const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
+    int cls = d_closure_apply(&cl1, key, dv, lv, nullptr, nullptr);
+    if (cls == 0) continue;
+    u64 k2[P2_MAX_KW];
+    u8 v2buf[P2_MAX_VB];
+    if (cls == 1) (void)d_closure_apply(&cl1, key, dv, lv, k2, v2buf);
+    u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
+    for (u32 u = lo; u < hi; u++) {
+      if (!allpass1 && !((mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
+                                               : (b1.times[u] < t)))
+        continue;
+      i64 d1 = wmul(d0, b1.diffs[u]);
+      if (cls == 2) {
+        ecodes[*eo] = MZ_ERR_DIVISION_BY_ZERO;
+        etimes[*eo] = t;
+        ediffs[*eo] = d1;
+        (*eo)++;
+        continue;
+      }
+      for (int b2i = 0; b2i < bl2.n; b2i++) {
+        const DevBatch &b2 = bl2.b[b2i];
+        u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
+        if (kvr2 == ~0ull) continue;
+        d_emit_pair(b2, bl2.allpass[b2i], k2, v2buf, t, d1, kvr2, mode2,
+                    0, cl2, lvb2, okw, ovb, o, okeys, ovals, otimes,
+                    odiffs, eo, ecodes, etimes, ediffs);
+      }
+    }
+  }
+}
+
 __global__ void k_probe_path2(const u64 *dkeys, const u8 *dvals, u32 dvb,
                               const u64 *dtimes, const i64 *ddiffs, u64 n,
                               u32 kw, u32 lvb1, BatchList bl1, int mode1,
@@ -727,101 +813,59 @@ __global__ void k_probe_path2(const u64 *dkeys, const u8 *dvals, u32 dvb,
   u64 start = blockIdx.x * (u64)blockDim.x + threadIdx.x;
   u64 iters = (total + stride - 1) / stride;  // uniform across the wave
   u32 lane = threadIdx.x & 63;
-  for (u64 it0 = 0; it0 < iters; it0++) {
-    u64 idx = start + it0 * stride;
-    u64 kvr1 = ~0ull;
-    u64 i = 0;
-    int bi = 0;
-    if (idx < total) {
-      i = idx % n;
-      bi = (int)(idx / n);
-      kvr1 = hash_lookup_range(bl1.b[bi].hash, bl1.b[bi].hash_slots,
-                               dkeys + i * kw, kw);
+  // P2_TILE pairs per macro-iteration (same shape as k_probe_walk):
+  // phase A issues the independent stage-1 hash lines, phase B walks the
+  // nested counts, ONE wave-aggregated reservation per tile, phase C
+  // emits from L2-hot ranges.
+  for (u64 it0 = 0; it0 < iters; it0 += P2_TILE) {
+    u64 kvr1[P2_TILE];
+    u64 cc[P2_TILE];
+#pragma unroll
+    for (int tt = 0; tt < P2_TILE; tt++) {
+      u64 idx = start + (it0 + tt) * stride;
+      kvr1[tt] = ~0ull;
+      cc[tt] = 0;
+      if (it0 + tt < iters && idx < total) {
+        int bi = (int)(idx / n);
+        const DevBatch &b = bl1.b[bi];
+        kvr1[tt] = hash_lookup_range(b.hash, b.hash_slots,
+                                     dkeys + (idx % n) * kw, kw);
+      }
     }
-    // count: final emissions + stage-1 error rows for this pair
-    u32 c = 0, e = 0;
+    u32 csum = 0, esum = 0;
     u64 inter = 0;
-    if (kvr1 != ~0ull) {
-      const DevBatch &b1 = bl1.b[bi];
-      const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-      u64 t = dtimes[i];
-      for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
-        const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
-        int cls = d_closure_apply(&cl1, dkeys + i * kw, dv, lv, nullptr,
-                                  nullptr);
-        if (cls == 0) continue;
-        u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
-        u32 m1 = 0;
-        if (bl1.allpass[bi]) {
-          m1 = hi - lo;
-        } else {
-          for (u32 u = lo; u < hi; u++)
-            m1 += (mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
-                                        : (b1.times[u] < t);
-        }
-        if (m1 == 0) continue;
-        if (cls == 2) {
-          e += m1;
-          continue;
-        }
-        inter += m1;
-        u64 k2[P2_MAX_KW];
-        u8 v2buf[P2_MAX_VB];
-        (void)d_closure_apply(&cl1, dkeys + i * kw, dv, lv, k2, v2buf);
-        for (int b2i = 0; b2i < bl2.n; b2i++) {
-          const DevBatch &b2 = bl2.b[b2i];
-          u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
-          if (kvr2 == ~0ull) continue;
-          u64 ce2 = d_count_pair(b2, bl2.allpass[b2i], k2, v2buf, t, kvr2,
-                                 mode2, 0, cl2, lvb2);
-          c += m1 * (u32)ce2;
-          e += m1 * (u32)(ce2 >> 32);
-        }
-      }
+#pragma unroll
+    for (int tt = 0; tt < P2_TILE; tt++) {
+      if (kvr1[tt] == ~0ull) continue;
+      u64 idx = start + (it0 + tt) * stride;
+      u64 i = idx % n;
+      int bi = (int)(idx / n);
+      cc[tt] = d_path2_count(dkeys + i * kw,
+                             dvals ? dvals + i * dvb : nullptr, dtimes[i],
+                             bl1.b[bi], bl1.allpass[bi], kvr1[tt], mode1,
+                             cl1, lvb1, k2w, bl2, mode2, cl2, lvb2,
+                             &inter);
+      csum += (u32)cc[tt];
+      esum += (u32)(cc[tt] >> 32);
     }
-    u64 base = wave_reserve(ctr, c, lane);
-    u64 ebase = wave_reserve(ctr + 1, e, lane);
+    u64 base = wave_reserve(ctr, csum, lane);
+    u64 ebase = wave_reserve(ctr + 1, esum, lane);
     if (inter) atomicAdd(ctr + 2, (unsigned long long)inter);
-    if ((c == 0 && e == 0) || base + c > cap || ebase + e > ecap)
+    if ((csum == 0 && esum == 0) || base + csum > cap ||
+        ebase + esum > ecap)
       continue;
-    // emit (second walk over L2-hot ranges)
     u64 o = base, eo = ebase;
-    const DevBatch &b1 = bl1.b[bi];
-    const u8 *dv = dvals ? dvals + i * dvb : nullptr;
-    u64 t = dtimes[i];
-    i64 d0 = ddiffs[i];
-    for (u32 j = (u32)kvr1; j < (u32)(kvr1 >> 32); j++) {
-      const u8 *lv = b1.vals ? b1.vals + (u64)j * lvb1 : nullptr;
-      int cls = d_closure_apply(&cl1, dkeys + i * kw, dv, lv, nullptr,
-                                nullptr);
-      if (cls == 0) continue;
-      u64 k2[P2_MAX_KW];
-      u8 v2buf[P2_MAX_VB];
-      if (cls == 1)
-        (void)d_closure_apply(&cl1, dkeys + i * kw, dv, lv, k2, v2buf);
-      u32 lo = b1.vu_off[j], hi = b1.vu_off[j + 1];
-      for (u32 u = lo; u < hi; u++) {
-        if (!bl1.allpass[bi] &&
-            !((mode1 == PM_HALF_LE) ? (b1.times[u] <= t)
-                                    : (b1.times[u] < t)))
-          continue;
-        i64 d1 = wmul(d0, b1.diffs[u]);
-        if (cls == 2) {
-          ecodes[eo] = MZ_ERR_DIVISION_BY_ZERO;
-          etimes[eo] = t;
-          ediffs[eo] = d1;
-          eo++;
-          continue;
-        }
-        for (int b2i = 0; b2i < bl2.n; b2i++) {
-          const DevBatch &b2 = bl2.b[b2i];
-          u64 kvr2 = hash_lookup_range(b2.hash, b2.hash_slots, k2, k2w);
-          if (kvr2 == ~0ull) continue;
-          d_emit_pair(b2, bl2.allpass[b2i], k2, v2buf, t, d1, kvr2, mode2,
-                      0, cl2, lvb2, okw, ovb, &o, okeys, ovals, otimes,
-                      odiffs, &eo, ecodes, etimes, ediffs);
-        }
-      }
+#pragma unroll
+    for (int tt = 0; tt < P2_TILE; tt++) {
+      if (cc[tt] == 0) continue;
+      u64 idx = start + (it0 + tt) * stride;
+      u64 i = idx % n;
+      int bi = (int)(idx / n);
+      d_path2_emit(dkeys + i * kw, dvals ? dvals + i * dvb : nullptr,
+                   dtimes[i], ddiffs[i], bl1.b[bi], bl1.allpass[bi],
+                   kvr1[tt], mode1, cl1, lvb1, k2w, bl2, mode2, cl2, lvb2,
+                   okw, ovb, &o, okeys, ovals, otimes, odiffs, &eo,
+                   ecodes, etimes, ediffs);
     }
   }
 }

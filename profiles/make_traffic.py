@@ -41,7 +41,8 @@ def main(fdb, wdb, workload, outpath):
     w = per_kernel(wdb, 1.0)
     probe = {}
     total = 0.0
-    for k in ("k_probe_walk", "k_probe_merge", "k_probe_vl"):
+    for k in ("k_probe_path2", "k_probe_walk", "k_probe_merge",
+              "k_probe_vl"):
         if k in f or k in w:
             fe = f.get(k, {})
             we = w.get(k, {})
@@ -61,7 +62,8 @@ def main(fdb, wdb, workload, outpath):
                    "separate passes; FETCH_SIZE x2 gfx950 "
                    "wide-coalesced-read correction "
                    "(MI355X_MICROARCH.md §HBM); per-launch averages; "
-                   "single-walk probe kernel (round 2)"),
+                   "single-walk / fused two-stage probe kernels "
+                   "(round 2)"),
     }
     json.dump(doc, open(outpath, "w"), indent=1)
     print(json.dumps(doc, indent=1))
