@@ -66,6 +66,24 @@ class DevOut:
             return self.ctx._take_copy(self.outp)
         return self.host_cols
 
+    def errs_to_host(self):
+        """Copy the out-batch's consolidated (code, time, diff) error
+        stream to host without releasing the batch."""
+        import ctypes as C
+
+        import numpy as np
+        en = int(self.outp.contents.err_n) if self.outp is not None else 0
+        ecodes = np.empty(en, np.uint64)
+        etimes = np.empty(en, np.uint64)
+        ediffs = np.empty(en, np.int64)
+        if en:
+            self.ctx._check(self.ctx.lib.mz_gpu_out_err_to_host(
+                self.ctx.ctx, self.outp,
+                ecodes.ctypes.data_as(C.POINTER(C.c_uint64)),
+                etimes.ctypes.data_as(C.POINTER(C.c_uint64)),
+                ediffs.ctypes.data_as(C.POINTER(C.c_int64))))
+        return ecodes, etimes, ediffs
+
     def release(self):
         if self.outp is not None:
             self.ctx.lib.mz_gpu_out_release(self.ctx.ctx, self.outp)

@@ -76,3 +76,94 @@ def test_gpu_err_stream_matches_oracle():
     assert len(eg[0]) > 0
     g.close()
     o.close()
+
+
+def _consolidate_host(keys, vals, times, diffs, kw=1, vb=8):
+    """Canonical (key, val, time) consolidation of a raw update stream."""
+    n = len(times)
+    if n == 0:
+        return [], [], [], []
+    k = np.asarray(keys).reshape(n, kw)
+    v = np.asarray(vals).reshape(n, vb) if vb else np.zeros((n, 0),
+                                                            np.uint8)
+    t = np.asarray(times)
+    d = np.asarray(diffs)
+    rows = {}
+    for i in range(n):
+        key = (tuple(int(x) for x in k[i]), bytes(v[i].tobytes()),
+               int(t[i]))
+        rows[key] = rows.get(key, 0) + int(d[i])
+    out = sorted((key, s) for key, s in rows.items() if s != 0)
+    return out
+
+
+@pytest.mark.gpu
+def test_gpu_halfjoin2_fused_matches_two_stage_with_errs():
+    """The fused two-stage path probe (k_probe_path2) must equal the
+    two-call halfjoin sequence — ok rows AND stage-2 error rows — on a
+    path whose stage-2 closure divides by a sometimes-zero lookup val."""
+    from materialize_amd._ffi import GpuCtx
+    from pyoracle import OracleCtx
+    g, o = GpuCtx(), OracleCtx()
+    rng = np.random.default_rng(77)
+    # stage 1: lookup val = (k2, carry) ; stage 2: carry / divisor(k2)
+    cl1 = abi.closure(
+        [],
+        [abi.field(abi.MZ_SRC_VAL_LOOKUP, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_LOOKUP, 8, 8)],
+        abi.schema(1, 8))
+    cl2 = abi.closure(
+        [],
+        [abi.field(abi.MZ_SRC_KEY, 0, 8)],
+        [abi.field(abi.MZ_SRC_COMPUTE, abi.MZ_COMPUTE_DIV_I64, 8,
+                   arg0=0, arg1=0, arg0_src=abi.MZ_SRC_VAL_STREAM,
+                   arg1_src=abi.MZ_SRC_VAL_LOOKUP)],
+        abi.schema(1, 8))
+    n1, n2, m = 400, 60, 300
+    k1 = rng.integers(0, 50, n1).astype(np.int64)
+    v1 = np.zeros((n1, 16), np.uint8)
+    v1[:, 0:8] = rng.integers(0, 60, n1).astype(np.int64) \
+        .reshape(-1, 1).view(np.uint8).reshape(n1, 8)
+    v1[:, 8:16] = rng.integers(1, 1000, n1).astype(np.int64) \
+        .reshape(-1, 1).view(np.uint8).reshape(n1, 8)
+    k2 = np.arange(60).astype(np.int64)
+    v2 = rng.integers(0, 4, n2).astype(np.int64)  # zeros common
+    pk = rng.integers(0, 50, m).astype(np.int64)
+    pd = rng.choice([-1, 1, 1], m).astype(np.int64)
+
+    def build(ctx):
+        a1, a2 = (ctx.arr_create(abi.schema(1, 16)),
+                  ctx.arr_create(abi.schema(1, 8)))
+        ctx.arr_insert(a1, abi.make_updates(
+            k1, v1, np.zeros(n1, np.uint64), np.ones(n1, np.int64), 0, 1))
+        ctx.arr_insert(a2, abi.make_updates(
+            k2, v2.reshape(-1, 1).view(np.uint8), np.zeros(n2, np.uint64),
+            np.ones(n2, np.int64), 0, 1))
+        return a1, a2
+
+    ga1, ga2 = build(g)
+    pu = abi.make_updates(pk, None, np.full(m, 1, np.uint64), pd, 1, 2)
+    fused = g.halfjoin2_dev(ga1, True, cl1, ga2, True, cl2, pu, 0)
+    fk, fv, ft, fd = fused.to_host()
+    g_errs = fused.errs_to_host()
+    fused.release()
+    # reference: two-call sequence on the oracle
+    oa1, oa2 = build(o)
+    s1 = o.halfjoin(oa1, abi.make_updates(
+        pk, None, np.full(m, 1, np.uint64), pd, 1, 2), 0, True, cl1)
+    s1k, s1v, s1t, s1d = s1
+    u2 = abi.make_updates(np.asarray(s1k, np.int64).reshape(-1),
+                          np.asarray(s1v, np.uint8),
+                          np.asarray(s1t, np.uint64),
+                          np.asarray(s1d, np.int64), 1, 2)
+    s2 = o.halfjoin(oa2, u2, 8, True, cl2)
+    o_errs = o.last_errs
+    assert _consolidate_host(fk, fv, ft, fd) == \
+        _consolidate_host(s2[0], s2[1], s2[2], s2[3])
+    # err streams are consolidated on both sides already
+    for x, y, what in zip(g_errs, o_errs, ("codes", "times", "diffs")):
+        np.testing.assert_array_equal(np.asarray(x), np.asarray(y),
+                                      err_msg=f"err {what}")
+    assert len(g_errs[0]) > 0, "zero divisors must produce error rows"
+    g.close()
+    o.close()
