@@ -101,7 +101,11 @@ enum {
    * offsets). arg1 == 0 raises MZ_ERR_DIVISION_BY_ZERO: the row is
    * diverted to the out-batch's error stream — the could_error ok/err
    * split of linear_join.rs:495-541. */
-  MZ_COMPUTE_DIV_I64 = 3
+  MZ_COMPUTE_DIV_I64 = 3,
+  /* i64 wrapping multiply as an OUT FIELD: arg0 * arg1 (both i64 at the
+   * given offsets; Diff-style wrapping per overflowing.rs:24-31). Q6's
+   * `l_extendedprice * l_discount` term. */
+  MZ_COMPUTE_MUL_I64 = 4
 };
 
 typedef struct {
@@ -123,7 +127,7 @@ typedef struct {
   uint8_t  arg0_src, arg1_src;
 } mz_gpu_field;
 
-#define MZ_GPU_MAX_FILTERS 4
+#define MZ_GPU_MAX_FILTERS 6
 #define MZ_GPU_MAX_FIELDS  8
 
 typedef struct {

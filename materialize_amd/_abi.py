@@ -19,10 +19,11 @@ MZ_COMPUTE_REVENUE = 0
 MZ_COMPUTE_CONST0 = 1
 MZ_COMPUTE_Q17_QTYLT = 2
 MZ_COMPUTE_DIV_I64 = 3
+MZ_COMPUTE_MUL_I64 = 4
 MZ_GPU_VARLEN = 0xFFFFFFFF
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 
-MZ_GPU_MAX_FILTERS = 4
+MZ_GPU_MAX_FILTERS = 6
 MZ_GPU_MAX_FIELDS = 8
 MZ_GPU_MAX_AGGS = 4
 

@@ -474,6 +474,10 @@ __device__ int d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
           i64 b = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
           if (b == 0) return 2;  // -> error stream
           v = a / b;
+        } else if (f.off == MZ_COMPUTE_MUL_I64) {
+          i64 a = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 b = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          v = wmul(a, b);
         }
         memcpy(dst, &v, 8);
         dst += 8;

@@ -300,6 +300,10 @@ int closure_apply(const mz_gpu_closure *cl, const u64 *key, const u8 *v1,
           i64 b = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
           if (b == 0) return 2;  // -> error stream
           v = a / b;
+        } else if (f.off == MZ_COMPUTE_MUL_I64) {
+          i64 a = read_int(cl_src(key, v1, v2, f.arg0_src) + f.arg0, 8);
+          i64 b = read_int(cl_src(key, v1, v2, f.arg1_src) + f.arg1, 8);
+          v = (i64)((u64)a * (u64)b);  // wrapping, overflowing.rs:24-31
         }  // MZ_COMPUTE_CONST0 leaves v = 0
         std::memcpy(dst, &v, 8);
         dst += 8;

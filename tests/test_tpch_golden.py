@@ -217,3 +217,134 @@ def test_q17_oracle_matches_reference_golden():
 def test_q17_gpu_matches_reference_golden():
     from materialize_amd._ffi import GpuCtx
     _run_q17(GpuCtx())
+
+
+# ---- Q6: filtered global SUM(ep*disc), pinned to tpch.td:268-278 ----
+Q6_MD5 = "d9c979f1eed5940788ff3653321acac4"
+D94, D95 = 731, 1096  # days since 1992-01-01
+
+
+def _q6_closure():
+    """revenue = SUM(l_extendedprice * l_discount) WHERE quantity < 24
+    AND discount BETWEEN 0.05 AND 0.07 AND shipdate IN [1994, 1995) —
+    engine units: cents, basis points, int32 days. Input val layout:
+    q(8) ep(8) d(8) sd(4)+pad4."""
+    V = abi.MZ_SRC_VAL_STREAM
+    return abi.closure(
+        [abi.filt(V, 0, 8, abi.MZ_CMP_LT, 24),
+         abi.filt(V, 16, 8, abi.MZ_CMP_GE, 500),
+         abi.filt(V, 16, 8, abi.MZ_CMP_LE, 700),
+         abi.filt(V, 24, 4, abi.MZ_CMP_GE, D94),
+         abi.filt(V, 24, 4, abi.MZ_CMP_LT, D95)],
+        [abi.field(abi.MZ_SRC_COMPUTE, abi.MZ_COMPUTE_CONST0, 8)],
+        [abi.field(abi.MZ_SRC_COMPUTE, abi.MZ_COMPUTE_MUL_I64, 8,
+                   arg0=8, arg1=16, arg0_src=abi.MZ_SRC_VAL_STREAM,
+                   arg1_src=abi.MZ_SRC_VAL_STREAM)],
+        abi.schema(1, 8))
+
+
+def _q6_host_total(q, ep, d, sd, diffs):
+    m = (q < 24) & (d >= 500) & (d <= 700) & (sd >= D94) & (sd < D95)
+    return int((ep[m].astype(object) * d[m] * diffs[m]).sum())
+
+
+def _q6_render(total):
+    from materialize_amd.tpch_exact import render_revenue_1e2
+    assert total % 10**4 == 0  # cents*bp = 1e4 x (dollars x 1e-2)
+    return render_revenue_1e2(total // 10**4)
+
+
+def _q6_vals(q, ep, d, sd):
+    n = len(q)
+    v = np.zeros((n, 32), np.uint8)
+    v[:, 0:8] = np.ascontiguousarray(q, np.int64).view(np.uint8) \
+        .reshape(n, 8)
+    v[:, 8:16] = np.ascontiguousarray(ep, np.int64).view(np.uint8) \
+        .reshape(n, 8)
+    v[:, 16:24] = np.ascontiguousarray(d, np.int64).view(np.uint8) \
+        .reshape(n, 8)
+    v[:, 24:28] = np.ascontiguousarray(sd, np.int32).view(np.uint8) \
+        .reshape(n, 4)
+    return v
+
+
+def _run_q6(ctx):
+    import hashlib
+    z = np.load(FIXTURE)
+    in_sch = abi.schema(1, 32)
+    cl = _q6_closure()
+    red = ctx.reduce_create(abi.reduce_spec(
+        [abi.Aggregate(func=abi.MZ_AGG_SUM_I64, off=0, width=8,
+                       is_float=0, nullable=0)], abi.schema(1, 8)))
+    state = {}
+
+    def push(keys, vals, diffs, t):
+        n = len(keys)
+        u = abi.make_updates(np.ascontiguousarray(keys, np.int64), vals,
+                             np.full(n, t, np.uint64),
+                             np.ascontiguousarray(diffs, np.int64),
+                             t, t + 1)
+        mk, mv, mt, md = ctx.map(in_sch, u, cl)
+        if len(mt) == 0:
+            return
+        ck, cv, ct, cd = ctx.reduce_push(red, abi.make_updates(
+            np.asarray(mk, np.int64), np.asarray(mv, np.uint8),
+            np.asarray(mt, np.uint64), np.asarray(md, np.int64), t, t + 1))
+        n2 = len(ct)
+        cv = np.asarray(cv).reshape(n2, 24) if n2 else cv
+        rows = []
+        for i in range(n2):
+            k = int(np.asarray(ck).reshape(-1)[i])
+            slot = cv[i]
+            assert slot[0] == 0
+            v = (int(slot[16:24].view(np.int64)[0]) * 2**64 +
+                 int(slot[8:16].view(np.uint64)[0]))
+            rows.append((k, v, int(cd[i])))
+        # consolidated order is by (key, val) — apply retracts first
+        for k, v, dd in [r for r in rows if r[2] == -1]:
+            assert state.get(k) == v
+            del state[k]
+        for k, v, dd in [r for r in rows if r[2] == 1]:
+            state[k] = v
+
+    # snapshot
+    q, ep = z["l_quantity"], z["l_extendedprice"]
+    d, sd = z["l_discount"], z["l_shipdate"]
+    push(z["l_orderkey"], _q6_vals(q, ep, d, sd),
+         np.ones(len(q), np.int64), 0)
+    expected = _q6_host_total(q, ep, d, sd, np.ones(len(q), np.int64))
+    assert state.get(0, 0) == expected
+    h = hashlib.md5()
+    h.update(_q6_render(expected).encode())
+    assert h.hexdigest() == Q6_MD5, "snapshot Q6 != reference golden"
+    # churn: q comes from the by-part columns (same row order as the
+    # lineitem columns — both pack old_lines then new_lines)
+    for b in range(int(z["n_churn"][0])):
+        lv = z[f"b{b}_lineitem_vals"].reshape(-1, 24)
+        bv = z[f"b{b}_lineitem_by_part_vals"].reshape(-1, 16)
+        diffs = z[f"b{b}_lineitem_diffs"]
+        np.testing.assert_array_equal(diffs,
+                                      z[f"b{b}_lineitem_by_part_diffs"])
+        qq = bv[:, 0:8].copy().view(np.int64).reshape(-1)
+        eep = lv[:, 0:8].copy().view(np.int64).reshape(-1)
+        dd = lv[:, 8:16].copy().view(np.int64).reshape(-1)
+        ssd = lv[:, 16:20].copy().view(np.int32).reshape(-1)
+        np.testing.assert_array_equal(
+            eep, bv[:, 8:16].copy().view(np.int64).reshape(-1))
+        push(z[f"b{b}_lineitem_keys"], _q6_vals(qq, eep, dd, ssd), diffs,
+             b + 1)
+        expected += _q6_host_total(qq, eep, dd, ssd, diffs)
+        assert state.get(0, 0) == expected, f"churn {b}"
+        _q6_render(expected)  # stays renderable (units invariant)
+    ctx.close()
+
+
+def test_q6_oracle_matches_reference_golden():
+    from pyoracle import OracleCtx
+    _run_q6(OracleCtx())
+
+
+@pytest.mark.gpu
+def test_q6_gpu_matches_reference_golden():
+    from materialize_amd._ffi import GpuCtx
+    _run_q6(GpuCtx())
