@@ -105,7 +105,13 @@ enum {
   /* i64 wrapping multiply as an OUT FIELD: arg0 * arg1 (both i64 at the
    * given offsets; Diff-style wrapping per overflowing.rs:24-31). Q6's
    * `l_extendedprice * l_discount` term. */
-  MZ_COMPUTE_MUL_I64 = 4
+  MZ_COMPUTE_MUL_I64 = 4,
+  /* Field-vs-field compare as a FILTER compute: passes iff
+   * cmp(arg0_field, arg1_field) holds, both read at the filter's `width`
+   * (4 or 8, signed LE) from (arg0_src, arg0) / (arg1_src, arg1) with
+   * the filter's `cmp`. Q12's `l_commit_date < l_receipt_date` /
+   * `l_ship_date < l_commit_date` predicates. */
+  MZ_COMPUTE_CMP_FIELDS = 5
 };
 
 typedef struct {

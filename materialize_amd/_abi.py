@@ -20,6 +20,7 @@ MZ_COMPUTE_CONST0 = 1
 MZ_COMPUTE_Q17_QTYLT = 2
 MZ_COMPUTE_DIV_I64 = 3
 MZ_COMPUTE_MUL_I64 = 4
+MZ_COMPUTE_CMP_FIELDS = 5
 MZ_GPU_VARLEN = 0xFFFFFFFF
 MZ_AGG_COUNT, MZ_AGG_SUM_I64, MZ_AGG_SUM_F64 = range(3)
 

@@ -426,6 +426,22 @@ __device__ int d_closure_apply(const mz_gpu_closure *cl, const u64 *key,
         i64 C = d_read_int(slot + 24, 8);
         if (!(C > 0 && (i128)5 * q * C < S)) return false;
       }
+      if (f.off == MZ_COMPUTE_CMP_FIELDS) {
+        i64 x = d_read_int(d_cl_src(key, v1, v2, f.arg0_src) + f.arg0,
+                           f.width);
+        i64 y = d_read_int(d_cl_src(key, v1, v2, f.arg1_src) + f.arg1,
+                           f.width);
+        bool ok;
+        switch (f.cmp) {
+          case MZ_CMP_LT: ok = x < y; break;
+          case MZ_CMP_LE: ok = x <= y; break;
+          case MZ_CMP_GT: ok = x > y; break;
+          case MZ_CMP_GE: ok = x >= y; break;
+          case MZ_CMP_EQ: ok = x == y; break;
+          default: ok = x != y; break;
+        }
+        if (!ok) return 0;
+      }
       continue;
     }
     i64 x = d_read_int(d_cl_src(key, v1, v2, f.src) + f.off, f.width);

@@ -26,8 +26,8 @@ import numpy as np
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
 from materialize_amd.tpch_exact import (CUTOFF, ExactEngineData, TpchExact,
-                                        q3_md5, q3_result, q17_avg_yearly,
-                                        render_revenue_1e2)
+                                        _days, q3_md5, q3_result,
+                                        q17_avg_yearly, render_revenue_1e2)
 
 Q3_MD5 = "637be0ff3f50cd612b004a69958bfccb"
 Q6_MD5 = "d9c979f1eed5940788ff3653321acac4"
@@ -49,7 +49,8 @@ def verify_q6(lineitems):
     return h.hexdigest()
 
 
-def verify_q12(orders, lineitems):
+def q12_result(orders, lineitems):
+    """{mode_name: [high_line_count, low_line_count]} per tpch.td Q12."""
     oprio = {t[0]: t[3] for t in orders}
     lo, hi = date(1994, 1, 1), date(1995, 1, 1)
     agg = {}
@@ -64,6 +65,11 @@ def verify_q12(orders, lineitems):
         a = agg.setdefault(mode, [0, 0])
         a[0] += high
         a[1] += 1 - high
+    return agg
+
+
+def verify_q12(orders, lineitems):
+    agg = q12_result(orders, lineitems)
     h = hashlib.md5()
     for mode in sorted(agg):
         h.update(mode.encode())
@@ -99,6 +105,7 @@ def main():
     churn_npz = {}
     expected = [rows]
     expected_q17 = [q17_str]
+    expected_q12 = [q12_result(orders, lineitems)]
     n_churn = 8
     for b in range(n_churn):
         batch = gen.churn_batch()
@@ -110,10 +117,22 @@ def main():
             churn_npz[f"b{b}_{rel}_keys"] = k
             churn_npz[f"b{b}_{rel}_vals"] = v
             churn_npz[f"b{b}_{rel}_diffs"] = d
+        # Q12 churn columns (aligned with b{b}_lineitem_* row order:
+        # old_lines then new_lines)
+        alll = old_lines + new_lines
+        churn_npz[f"b{b}_l_cd"] = np.array([_days(t[6]) for t in alll],
+                                           np.int32)
+        churn_npz[f"b{b}_l_rd"] = np.array([_days(t[7]) for t in alll],
+                                           np.int32)
+        churn_npz[f"b{b}_l_md"] = np.array([t[8] for t in alll], np.int64)
+        churn_npz[f"b{b}_o_prio"] = np.array(
+            [batch[0][3], batch[2][3]], np.int64)  # old, new (diff -1/+1)
         cur_lines = [t for ls in state_lines.values() for t in ls]
         expected.append(q3_result(customers, list(state_orders.values()),
                                   cur_lines))
         expected_q17.append(q17_avg_yearly(gen.parts, cur_lines))
+        expected_q12.append(q12_result(list(state_orders.values()),
+                                       cur_lines))
 
     out = os.path.join(os.path.dirname(__file__), "tpch_sf001.npz")
     np.savez_compressed(
@@ -128,10 +147,16 @@ def main():
         p_brand=np.array([p[1] for p in gen.parts], np.int64),
         p_container=np.array([p[2] for p in gen.parts], np.int64),
         l_partkey=data.l_partkey, l_quantity=data.l_quantity,
+        o_orderpriority=np.array([t[3] for t in orders], np.int64),
+        l_commitdate=np.array([_days(t[6]) for t in lineitems], np.int32),
+        l_receiptdate=np.array([_days(t[7]) for t in lineitems], np.int32),
+        l_shipmode=np.array([t[8] for t in lineitems], np.int64),
         expected_json=np.frombuffer(
             json.dumps(expected).encode(), dtype=np.uint8),
         expected_q17_json=np.frombuffer(
             json.dumps(expected_q17).encode(), dtype=np.uint8),
+        expected_q12_json=np.frombuffer(
+            json.dumps(expected_q12).encode(), dtype=np.uint8),
         **churn_npz)
     print(f"wrote {out} "
           f"({os.path.getsize(out) / 1e6:.2f} MB, {n_churn} churn batches)")

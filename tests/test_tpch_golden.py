@@ -348,3 +348,201 @@ def test_q6_oracle_matches_reference_golden():
 def test_q6_gpu_matches_reference_golden():
     from materialize_amd._ffi import GpuCtx
     _run_q6(GpuCtx())
+
+
+# ---- Q12: join + conditional counts per shipmode (tpch.td:462-491) ----
+Q12_MD5 = "3c31b94c99bd77e96003c2059416ed7a"
+MODE_SHIP, MODE_MAIL = 3, 5
+MODE_NAMES = {3: "SHIP", 5: "MAIL"}
+
+
+def _q12_ingest_closure(mode_code):
+    """Pushed-down lineitem filter for one shipmode: commit < receipt,
+    ship < commit, receipt in [1994, 1995). In val layout: md(8) sd(4)
+    cd(4) rd(4) +pad. Out: key = l_orderkey, val = mode."""
+    V = abi.MZ_SRC_VAL_STREAM
+    C = abi.MZ_SRC_COMPUTE
+    return abi.closure(
+        [abi.filt(V, 0, 8, abi.MZ_CMP_EQ, mode_code),
+         abi.filt(C, abi.MZ_COMPUTE_CMP_FIELDS, 4, abi.MZ_CMP_LT, 0,
+                  arg0=12, arg1=16, arg0_src=V, arg1_src=V),  # cd < rd
+         abi.filt(C, abi.MZ_COMPUTE_CMP_FIELDS, 4, abi.MZ_CMP_LT, 0,
+                  arg0=8, arg1=12, arg0_src=V, arg1_src=V),   # sd < cd
+         abi.filt(V, 16, 4, abi.MZ_CMP_GE, D94),
+         abi.filt(V, 16, 4, abi.MZ_CMP_LT, D95)],
+        [abi.field(abi.MZ_SRC_KEY, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 8)],
+        abi.schema(1, 8))
+
+
+def _q12_vals(md, sd, cd, rd):
+    n = len(md)
+    v = np.zeros((n, 24), np.uint8)
+    v[:, 0:8] = np.ascontiguousarray(md, np.int64).view(np.uint8) \
+        .reshape(n, 8)
+    v[:, 8:12] = np.ascontiguousarray(sd, np.int32).view(np.uint8) \
+        .reshape(n, 4)
+    v[:, 12:16] = np.ascontiguousarray(cd, np.int32).view(np.uint8) \
+        .reshape(n, 4)
+    v[:, 16:20] = np.ascontiguousarray(rd, np.int32).view(np.uint8) \
+        .reshape(n, 4)
+    return v
+
+
+def _run_q12(ctx):
+    import hashlib
+    z = np.load(FIXTURE)
+    expected = json.loads(bytes(z["expected_q12_json"]).decode())
+    in_l = abi.schema(1, 24)
+    cls_ingest = [_q12_ingest_closure(m) for m in (MODE_SHIP, MODE_MAIL)]
+    # join closures: stream side carries (okey -> mode|prio), lookup the
+    # other relation; out key = mode, val = prio
+    cl_L = abi.closure(  # lineitem delta probes orders_prio
+        [], [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_LOOKUP, 0, 8)], abi.schema(1, 8))
+    cl_O = abi.closure(  # orders_prio delta probes lineitem_q12
+        [], [abi.field(abi.MZ_SRC_VAL_LOOKUP, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 8)], abi.schema(1, 8))
+    cl_high = abi.closure(
+        [abi.filt(abi.MZ_SRC_VAL_STREAM, 0, 8, abi.MZ_CMP_LE, 1)],
+        [abi.field(abi.MZ_SRC_KEY, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 8)], abi.schema(1, 8))
+    cl_low = abi.closure(
+        [abi.filt(abi.MZ_SRC_VAL_STREAM, 0, 8, abi.MZ_CMP_GE, 2)],
+        [abi.field(abi.MZ_SRC_KEY, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 8)], abi.schema(1, 8))
+    arr_l = ctx.arr_create(abi.schema(1, 8))
+    arr_o = ctx.arr_create(abi.schema(1, 8))
+    reds = {}
+    states = {}
+    for name, c in (("high", cl_high), ("low", cl_low)):
+        reds[name] = ctx.reduce_create(abi.reduce_spec(
+            [abi.Aggregate(func=abi.MZ_AGG_COUNT, off=0, width=8,
+                           is_float=0, nullable=0)], abi.schema(1, 8)))
+        states[name] = {}
+
+    def apply(name, ck, cv, ct, cd_):
+        n2 = len(ct)
+        cv = np.asarray(cv).reshape(n2, 24) if n2 else cv
+        rows = []
+        for i in range(n2):
+            k = int(np.asarray(ck).reshape(-1)[i])
+            slot = cv[i]
+            v = int(slot[8:16].view(np.uint64)[0]) + \
+                int(slot[16:24].view(np.int64)[0]) * 2**64
+            rows.append((k, v, int(cd_[i])))
+        st = states[name]
+        for k, v, dd in [r for r in rows if r[2] == -1]:
+            assert st.get(k) == v
+            del st[k]
+        for k, v, dd in [r for r in rows if r[2] == 1]:
+            st[k] = v
+
+    def reduce_push(joined_cols, t):
+        jk, jv, jt, jd = joined_cols
+        if len(jt) == 0:
+            return
+        u = abi.make_updates(np.asarray(jk, np.int64),
+                             np.asarray(jv, np.uint8),
+                             np.asarray(jt, np.uint64),
+                             np.asarray(jd, np.int64), t, t + 1)
+        for name in ("high", "low"):
+            mk, mv, mt, md_ = ctx.map(abi.schema(1, 8), u,
+                                      cl_high if name == "high"
+                                      else cl_low)
+            if len(mt) == 0:
+                continue
+            cols = ctx.reduce_push(reds[name], abi.make_updates(
+                np.asarray(mk, np.int64), np.asarray(mv, np.uint8),
+                np.asarray(mt, np.uint64), np.asarray(md_, np.int64),
+                t, t + 1))
+            apply(name, *cols)
+
+    def step(lkeys, lvals, ldiffs, okeys, oprio, odiffs, t, snapshot):
+        # pushed-down lineitem filter (both modes), arrangement inserts
+        lq_cols = []
+        lu = abi.make_updates(np.ascontiguousarray(lkeys, np.int64),
+                              lvals, np.full(len(lkeys), t, np.uint64),
+                              np.ascontiguousarray(ldiffs, np.int64),
+                              t, t + 1)
+        for c in cls_ingest:
+            lq_cols.append(ctx.map(in_l, lu, c))
+        fk = np.concatenate([np.asarray(c[0], np.int64).reshape(-1)
+                             for c in lq_cols])
+        fv = np.concatenate([np.asarray(c[1], np.uint8).reshape(-1)
+                             for c in lq_cols])
+        fd = np.concatenate([np.asarray(c[3], np.int64) for c in lq_cols])
+        n = len(fd)
+        flu = abi.make_updates(fk, fv.reshape(n, 8) if n else fv,
+                               np.full(n, t, np.uint64), fd, t, t + 1)
+        ou = abi.make_updates(np.ascontiguousarray(okeys, np.int64),
+                              np.ascontiguousarray(oprio, np.int64)
+                              .reshape(-1, 1).view(np.uint8),
+                              np.full(len(okeys), t, np.uint64),
+                              np.ascontiguousarray(odiffs, np.int64),
+                              t, t + 1)
+        ctx.arr_insert(arr_l, flu)
+        ctx.arr_insert(arr_o, ou)
+        # drain: path L (src 0, le) always; path O (src 1, lt) only has
+        # deltas after the snapshot (at the as-of only path 0 emits,
+        # delta_join.rs:752-798)
+        flu2 = abi.make_updates(fk, fv.reshape(n, 8) if n else fv,
+                                np.full(n, t, np.uint64), fd, t, t + 1)
+        reduce_push(ctx.halfjoin(arr_o, flu2, 8, True, cl_L), t)
+        if not snapshot:
+            ou2 = abi.make_updates(np.ascontiguousarray(okeys, np.int64),
+                                   np.ascontiguousarray(oprio, np.int64)
+                                   .reshape(-1, 1).view(np.uint8),
+                                   np.full(len(okeys), t, np.uint64),
+                                   np.ascontiguousarray(odiffs, np.int64),
+                                   t, t + 1)
+            reduce_push(ctx.halfjoin(arr_l, ou2, 8, False, cl_O), t)
+
+    def got():
+        out = {}
+        for mcode, mname in MODE_NAMES.items():
+            h = states["high"].get(mcode, 0)
+            lo = states["low"].get(mcode, 0)
+            if h or lo:
+                out[mname] = [h, lo]
+        return out
+
+    # snapshot at t=0
+    nl = len(z["l_orderkey"])
+    step(z["l_orderkey"],
+         _q12_vals(z["l_shipmode"], z["l_shipdate"], z["l_commitdate"],
+                   z["l_receiptdate"]),
+         np.ones(nl, np.int64),
+         z["o_orderkey"], z["o_orderpriority"],
+         np.ones(len(z["o_orderkey"]), np.int64), 0, snapshot=True)
+    want0 = {k: list(v) for k, v in expected[0].items()}
+    assert got() == want0
+    h = hashlib.md5()
+    for mname in sorted(want0):
+        h.update(mname.encode())
+        h.update(str(want0[mname][0]).encode())
+        h.update(str(want0[mname][1]).encode())
+    assert h.hexdigest() == Q12_MD5, "snapshot Q12 != reference golden"
+    for b in range(int(z["n_churn"][0])):
+        step(z[f"b{b}_lineitem_keys"],
+             _q12_vals(z[f"b{b}_l_md"],
+                       z[f"b{b}_lineitem_vals"].reshape(-1, 24)[:, 16:20]
+                       .copy().view(np.int32).reshape(-1),
+                       z[f"b{b}_l_cd"], z[f"b{b}_l_rd"]),
+             z[f"b{b}_lineitem_diffs"],
+             z[f"b{b}_orders_keys"], z[f"b{b}_o_prio"],
+             z[f"b{b}_orders_diffs"], b + 1, snapshot=False)
+        assert got() == {k: list(v) for k, v in expected[b + 1].items()}, \
+            f"churn {b}"
+    ctx.close()
+
+
+def test_q12_oracle_matches_reference_golden():
+    from pyoracle import OracleCtx
+    _run_q12(OracleCtx())
+
+
+@pytest.mark.gpu
+def test_q12_gpu_matches_reference_golden():
+    from materialize_amd._ffi import GpuCtx
+    _run_q12(GpuCtx())
