@@ -1682,6 +1682,16 @@ struct Scratch {
 
 }  // namespace
 
+struct DevUpdates {
+  const u64 *keys;
+  const u8 *vals;
+  const u64 *times;
+  const i64 *diffs;
+  u64 n;
+  int sorted = 0;  // canonical (key,val,time) ascending (see mz_gpu.h)
+  const u32 *val_offs = nullptr;  // VARLEN: [n+1] offsets into vals
+};
+
 struct mz_gpu_arr {
   DevSchema schema;
   std::vector<DevBatch> batches;
@@ -1711,9 +1721,24 @@ struct mz_gpu_arr {
   Scratch *merge_scr = nullptr;
   hipEvent_t ev_mdone = nullptr;
   // deferred insert (arr_insert_async): counts land here asynchronously
+  // Cached radix sort plan for the insert lane (churn batches have
+  // constant per-batch times and keys from a fixed keyspace, so the
+  // column min/max plan repeats every step): sort_updates reuses the
+  // cached plan WITHOUT the minmax readback sync and enqueues a
+  // device-side validity check instead; a mismatch (flag) triggers a
+  // synchronous rebuild at the flush before anything is installed.
+  struct SortPlan {
+    int valid = 0;
+    u32 kw = 0, vb = 0;
+    u64 hmin[MAX_PASSES], hmax[MAX_PASSES];
+    u64 *dev = nullptr;      // device copy [2 * MAX_PASSES]
+    u32 *d_flag = nullptr;   // 1 = this batch's minmax not covered
+  } sort_plan;
   struct Pending {
     int active = 0;
     u64 cnt[3] = {0, 0, 0};
+    u64 flag[1] = {0};       // host copy of sort_plan.d_flag
+    DevUpdates staged;       // inputs retained for the redo path
     DevBatch batch;
     u64 lower = 0;  // batch frontier: a time-filtered probe whose delta
                     // upper <= lower cannot see this batch, so it need
@@ -2014,11 +2039,28 @@ static void red_compact_grow(Ctx *c, RedState &st, u64 *d_nrows,
   fill_u64(c, d_nrows, 1, live);
 }
 
+// Cached-plan validity: slot s covered iff [min,max] within the cached
+// bounds, or both plans see a constant column (cached bits = 0 and the
+// new column is constant — the pass is skipped either way, so the
+// constant's VALUE is irrelevant; this is what lets single-timestamp
+// batches reuse the plan as t advances).
+__global__ void k_plan_check(const u64 *dminmax, const u64 *cached,
+                             u32 nslots, u32 max_passes, u32 *flag) {
+  u32 s = blockIdx.x * blockDim.x + threadIdx.x;
+  if (s >= nslots) return;
+  u64 nmin = dminmax[s], nmax = dminmax[s + max_passes];
+  u64 cmin = cached[s], cmax = cached[s + max_passes];
+  bool ok = (nmin >= cmin && nmax <= cmax) ||
+            (cmin == cmax && nmin == nmax);
+  if (!ok) atomicOr(flag, 1u);
+}
+
 // composite stable sort: returns perm ordering rows by (key, val, time) —
 // or (time, key) when for_reduce (primary time).
 // Sort passes use rocprim radix_sort_pairs (AMD-native primitive).
 void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
-                  const u64 *times, u64 n, u32 *perm, bool time_major) {
+                  const u64 *times, u64 n, u32 *perm, bool time_major,
+                  mz_gpu_arr::SortPlan *plan = nullptr) {
   MZ_PROF(c, "sort_updates");
   auto &S = (*c->scr);
   u64 *skey = (u64 *)S.get(n * 8);
@@ -2041,9 +2083,38 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
                        dim3(ngrid(n), 1 + vwords + kw), dim3(BLK), 0,
                        c->stream, keys, kw, vals, vb, times, n, vwords, 1,
                        dmin, dmax);
-  // one pinned staged copy for both halves (dmin/dmax are contiguous)
-  u64 *mm = (u64 *)d2h_pinned(c, dminmax, 2 * MAX_PASSES * 8);
-  u64 *hmin = mm, *hmax = mm + MAX_PASSES;
+  u64 *hmin, *hmax;
+  if (plan && plan->valid && plan->kw == kw && plan->vb == vb &&
+      !time_major) {
+    // cached plan: enqueue the device validity check, no host sync
+    hipLaunchKernelGGL(k_plan_check, dim3(1), dim3(64), 0, c->stream,
+                       dminmax, plan->dev, 1 + vwords + kw, MAX_PASSES,
+                       plan->d_flag);
+    hmin = plan->hmin;
+    hmax = plan->hmax;
+  } else {
+    // one pinned staged copy for both halves (dmin/dmax are contiguous)
+    u64 *mm = (u64 *)d2h_pinned(c, dminmax, 2 * MAX_PASSES * 8);
+    hmin = mm;
+    hmax = mm + MAX_PASSES;
+    if (plan && !time_major) {
+      // (re)prime the cache from this batch's exact plan
+      if (!plan->dev) {
+        plan->dev = dnew<u64>(c, 2 * MAX_PASSES);
+        plan->d_flag = (u32 *)dmalloc(c, 4);
+        fill_u32(c, plan->d_flag, 1, 0);
+      }
+      memcpy(plan->hmin, hmin, MAX_PASSES * 8);
+      memcpy(plan->hmax, hmax, MAX_PASSES * 8);
+      HIP_CHECK(hipMemcpyAsync(plan->dev, dminmax, 2 * MAX_PASSES * 8,
+                               hipMemcpyDeviceToDevice, c->stream));
+      hmin = plan->hmin;  // pinned page gets reused by later readbacks
+      hmax = plan->hmax;
+      plan->kw = kw;
+      plan->vb = vb;
+      plan->valid = 1;
+    }
+  }
   // pass slot layout from k_pass_minmax: [time][val words][key words]
   struct Pass {
     int kind;  // 0 = time, 1 = val word, 2 = key word
@@ -2269,15 +2340,6 @@ void inclusive_scan_u64(Ctx *c, const u64 *in, u64 *out, u64 n) {
 }
 
 // Stage updates onto the device (if host) and return device pointers.
-struct DevUpdates {
-  const u64 *keys;
-  const u8 *vals;
-  const u64 *times;
-  const i64 *diffs;
-  u64 n;
-  int sorted = 0;  // canonical (key,val,time) ascending (see mz_gpu.h)
-  const u32 *val_offs = nullptr;  // VARLEN: [n+1] offsets into vals
-};
 
 static inline bool is_varlen(u32 vb) { return vb == 0xFFFFFFFFu; }
 
@@ -2374,7 +2436,8 @@ void consolidate_with_perm(Ctx *c, u32 kw, u32 vb, DevUpdates in,
                            u64 *otimes, i64 *odiffs, u64 *dcounts);
 
 void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
-                      u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts) {
+                      u8 *ovals, u64 *otimes, i64 *odiffs, u64 *dcounts,
+                      mz_gpu_arr::SortPlan *plan = nullptr) {
   MZ_PROF(c, "consolidate_core");
   auto &S = (*c->scr);
   u64 n = in.n;
@@ -2383,7 +2446,8 @@ void consolidate_core(Ctx *c, u32 kw, u32 vb, DevUpdates in, u64 *okeys,
     return;
   }
   u32 *perm = (u32 *)S.get(n * 4);
-  sort_updates(c, in.keys, kw, in.vals, vb, in.times, n, perm, false);
+  sort_updates(c, in.keys, kw, in.vals, vb, in.times, n, perm, false,
+               plan);
   consolidate_with_perm(c, kw, vb, in, perm, okeys, ovals, otimes, odiffs,
                         dcounts);
 }
@@ -3463,6 +3527,8 @@ void mz_gpu_fini(mz_gpu_ctx *c) {
     for (auto &b : a->batches) free_batch(ctx, b);
     dfree(ctx, a->pending.flat_keys);
     dfree(ctx, a->pending.flat_vals);
+    dfree(ctx, a->sort_plan.dev);
+    dfree(ctx, a->sort_plan.d_flag);
     if (a->pending.active) free_batch(ctx, a->pending.batch);
     if (a->pending_merge.active) free_batch(ctx, a->pending_merge.merged);
     delete a;
@@ -3736,22 +3802,18 @@ static void arr_insert_vl(Ctx *ctx, mz_gpu_arr *a,
   if (a->stream) (void)hipEventRecord(a->ev_ready, a->stream);
 }
 
-static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
-                                  const mz_gpu_updates *u) {
-  MZ_PROF(ctx, "arr_insert");
-  if (is_varlen(a->schema.vb)) {
-    arr_insert_vl(ctx, a, u);
-    return;
-  }
-  if (a->pending.active) arr_flush_impl(ctx, a);
-  // gate=false: this pipeline writes only freshly-allocated memory, so
-  // it may run concurrently with the main stream's in-flight probes of
-  // the arrangement's CURRENT batches (the 1-deep insert pipeline).
-  LaneGuard lane(ctx, a, /*gate=*/false);
+// The enqueue-only insert body. `plan` non-null (device inputs only)
+// uses the arrangement's cached sort plan without the minmax sync; the
+// device validity flag is copied back with the counts and checked at
+// the flush, which rebuilds synchronously on a mismatch (rare: the
+// batch's column ranges escaped the cached bounds).
+static void insert_pipeline(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
+                            u64 lower, u64 upper,
+                            mz_gpu_arr::SortPlan *plan) {
   auto &S = (*ctx->scr);
   S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
-  DevUpdates d = stage_updates(ctx, u, kw, vb);
+  if (plan && plan->d_flag) fill_u32(ctx, plan->d_flag, 1, 0);
   u64 capn = std::max<u64>(d.n, 1);
   u64 *ok = dnew<u64>(ctx, capn * kw);
   u8 *ov = (u8 *)dmalloc(ctx, std::max<u64>(capn * vb, 1));
@@ -3764,18 +3826,44 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
                        perm, d.n);
     consolidate_with_perm(ctx, kw, vb, d, perm, ok, ov, ot, od, dcounts);
   } else {
-    consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts);
+    consolidate_core(ctx, kw, vb, d, ok, ov, ot, od, dcounts, plan);
   }
-  DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, u->lower,
-                                u->upper, dcounts, /*keep_flat=*/1);
+  DevBatch b = build_batch_core(ctx, kw, vb, ok, ov, ot, od, d.n, lower,
+                                upper, dcounts, /*keep_flat=*/1);
   a->pending.active = 1;
   a->pending.batch = b;
-  a->pending.lower = u->lower;
-  a->pending.upper = u->upper;
+  a->pending.staged = d;
+  a->pending.lower = lower;
+  a->pending.upper = upper;
   a->pending.flat_keys = ok;
   a->pending.flat_vals = ov;
+  a->pending.flag[0] = 0;
   HIP_CHECK(hipMemcpyAsync(a->pending.cnt, dcounts, 3 * 8,
                            hipMemcpyDeviceToHost, ctx->stream));
+  if (plan && plan->d_flag)
+    HIP_CHECK(hipMemcpyAsync(a->pending.flag, plan->d_flag, 4,
+                             hipMemcpyDeviceToHost, ctx->stream));
+}
+
+static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
+                                  const mz_gpu_updates *u) {
+  MZ_PROF(ctx, "arr_insert");
+  if (is_varlen(a->schema.vb)) {
+    arr_insert_vl(ctx, a, u);
+    return;
+  }
+  if (a->pending.active) arr_flush_impl(ctx, a);
+  // gate=false: this pipeline writes only freshly-allocated memory, so
+  // it may run concurrently with the main stream's in-flight probes of
+  // the arrangement's CURRENT batches (the 1-deep insert pipeline).
+  LaneGuard lane(ctx, a, /*gate=*/false);
+  (*ctx->scr).reset();
+  DevUpdates d = stage_updates(ctx, u, a->schema.kw, a->schema.vb);
+  // the cached-plan path needs the inputs retained for the redo; only
+  // device inputs outlive the call (the async-insert lifetime contract)
+  mz_gpu_arr::SortPlan *plan =
+      (u->on_device && !d.sorted) ? &a->sort_plan : nullptr;
+  insert_pipeline(ctx, a, d, u->lower, u->upper, plan);
 }
 
 static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
@@ -3790,6 +3878,17 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
   // in-order after every probe that reads them.
   LaneGuard lane(ctx, a, /*gate=*/false);
   HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  if (a->pending.active && a->pending.flag[0]) {
+    // the cached sort plan did not cover this batch: discard the
+    // mis-sorted build and redo synchronously (re-priming the cache)
+    free_batch(ctx, a->pending.batch);
+    dfree(ctx, a->pending.flat_keys);
+    dfree(ctx, a->pending.flat_vals);
+    a->sort_plan.valid = 0;
+    insert_pipeline(ctx, a, a->pending.staged, a->pending.lower,
+                    a->pending.upper, &a->sort_plan);
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+  }
   // install an in-flight deferred merge only when its stream is done —
   // otherwise leave it running (the pre-merge batch list stays valid)
   if (a->pending_merge.active &&
