@@ -3810,8 +3810,9 @@ static void arr_insert_vl(Ctx *ctx, mz_gpu_arr *a,
 static void insert_pipeline(Ctx *ctx, mz_gpu_arr *a, DevUpdates d,
                             u64 lower, u64 upper,
                             mz_gpu_arr::SortPlan *plan) {
+  // NOTE: does NOT reset the lane scratch — host-input callers staged
+  // `d` into it (arr_insert_async_impl resets before staging)
   auto &S = (*ctx->scr);
-  S.reset();
   u32 kw = a->schema.kw, vb = a->schema.vb;
   if (plan && plan->d_flag) fill_u32(ctx, plan->d_flag, 1, 0);
   u64 capn = std::max<u64>(d.n, 1);
@@ -3885,6 +3886,7 @@ static void arr_flush_take_impl(Ctx *ctx, mz_gpu_arr *a,
     dfree(ctx, a->pending.flat_keys);
     dfree(ctx, a->pending.flat_vals);
     a->sort_plan.valid = 0;
+    (*ctx->scr).reset();  // redo inputs are device-resident, not staged
     insert_pipeline(ctx, a, a->pending.staged, a->pending.lower,
                     a->pending.upper, &a->sort_plan);
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
