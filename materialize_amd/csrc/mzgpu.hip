@@ -2098,16 +2098,33 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
     hmin = mm;
     hmax = mm + MAX_PASSES;
     if (plan && !time_major) {
-      // (re)prime the cache from this batch's exact plan
+      // (re)prime the cache from this batch's plan, PADDED by one span
+      // on each side of every non-constant column: batches are samples
+      // of a fixed population, so the sampled min/max jitters — exact
+      // bounds would flag (and synchronously redo) nearly every later
+      // batch. Constant columns stay exact (their validity rule is
+      // value-independent constancy). One extra covered span costs at
+      // most ~2 radix bits per column.
       if (!plan->dev) {
         plan->dev = dnew<u64>(c, 2 * MAX_PASSES);
         plan->d_flag = (u32 *)dmalloc(c, 4);
         fill_u32(c, plan->d_flag, 1, 0);
       }
-      memcpy(plan->hmin, hmin, MAX_PASSES * 8);
-      memcpy(plan->hmax, hmax, MAX_PASSES * 8);
-      HIP_CHECK(hipMemcpyAsync(plan->dev, dminmax, 2 * MAX_PASSES * 8,
-                               hipMemcpyDeviceToDevice, c->stream));
+      for (u32 s = 0; s < MAX_PASSES; s++) {
+        u64 lo = hmin[s], hi2 = hmax[s];
+        if (lo < hi2) {
+          u64 span = hi2 - lo;
+          lo -= std::min(span, lo);
+          hi2 = (hi2 + span < hi2) ? ~0ull : hi2 + span;
+        }
+        plan->hmin[s] = lo;
+        plan->hmax[s] = hi2;
+      }
+      HIP_CHECK(hipMemcpyAsync(plan->dev, plan->hmin, MAX_PASSES * 8,
+                               hipMemcpyHostToDevice, c->stream));
+      HIP_CHECK(hipMemcpyAsync(plan->dev + MAX_PASSES, plan->hmax,
+                               MAX_PASSES * 8, hipMemcpyHostToDevice,
+                               c->stream));
       hmin = plan->hmin;  // pinned page gets reused by later readbacks
       hmax = plan->hmax;
       plan->kw = kw;
@@ -3862,8 +3879,13 @@ static void arr_insert_async_impl(Ctx *ctx, mz_gpu_arr *a,
   DevUpdates d = stage_updates(ctx, u, a->schema.kw, a->schema.vb);
   // the cached-plan path needs the inputs retained for the redo; only
   // device inputs outlive the call (the async-insert lifetime contract)
+  static const bool NO_PLAN_CACHE = [] {
+    const char *e = getenv("MZ_NO_SORT_PLAN_CACHE");
+    return e && e[0] && e[0] != '0';
+  }();
   mz_gpu_arr::SortPlan *plan =
-      (u->on_device && !d.sorted) ? &a->sort_plan : nullptr;
+      (!NO_PLAN_CACHE && u->on_device && !d.sorted) ? &a->sort_plan
+                                                    : nullptr;
   insert_pipeline(ctx, a, d, u->lower, u->upper, plan);
 }
 
