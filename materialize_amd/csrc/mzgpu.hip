@@ -1257,67 +1257,83 @@ __global__ void k_reduce_apply(const u64 *keys, const u8 *vals, u32 kw,
   u32 ovb = spec.out.val_bytes;
   if (gidn) G = hi_row ? gidn[hi_row - 1] : 0;
   if (d_nrows) base = *d_nrows;
-  GRID_STRIDE(g, G) {
-    u64 lo = gstart[g];
-    u64 end = (g + 1 < G) ? gstart[g + 1] : hi_row;
-    const u64 *key = keys + lo * kw;
-    u64 idx = miss[g] ? base + misspos[g] : found[g];
-    if (idx >= st.capacity) continue;  // overflow flagged by insert
-    u64 *row = st.rows + idx * st.stride_w;
-    i64 *total_p = (i64 *)(row + kw);
-    Acc5 *accs = (Acc5 *)(row + kw + 1);
-    // snapshot old
-    i64 old_total = *total_p;
-    Acc5 old_a[MZ_GPU_MAX_AGGS];
-    for (u32 a = 0; a < na; a++) old_a[a] = accs[a];
-    // apply updates (explode_one * diff + semigroup merge; wrapping)
-    for (u64 r = lo; r < end; r++) {
-      i64 d = diffs[r];
-      const u8 *v = vals + r * vb;
-      for (u32 a = 0; a < na; a++) {
-        Acc5 c = d_datum_to_acc(spec.aggs[a], v);
-        accs[a].accum += c.accum * (u128)(i128)d;
-        accs[a].nn += (u64)c.nn * (u64)d;
-        accs[a].pi += (u64)c.pi * (u64)d;
-        accs[a].ni += (u64)c.ni * (u64)d;
-        accs[a].nan += (u64)c.nan * (u64)d;
-      }
-      *total_p = wadd(*total_p, d);
-    }
-    i64 new_total = *total_p;
-    // exists = any nonzero component (reduce_abelian: keys with nonempty
-    // input accumulation produce one output row)
-    auto exists = [&](const Acc5 *as, i64 tot) {
-      if (tot != 0) return true;
-      for (u32 a = 0; a < na; a++)
-        if (as[a].accum != 0 || as[a].nn || as[a].pi || as[a].ni || as[a].nan)
-          return true;
-      return false;
-    };
-    bool oe = exists(old_a, old_total), ne = exists(accs, new_total);
+  // Uniform-iteration loop with ONE wave-aggregated output reservation
+  // per iteration (the per-row atomicAdds on the shared counter
+  // serialized within each wave — same cure as the probe queues).
+  u64 stride = (u64)gridDim.x * blockDim.x;
+  u64 start0 = blockIdx.x * (u64)blockDim.x + threadIdx.x;
+  u64 iters = (G + stride - 1) / stride;
+  u32 lane = threadIdx.x & 63;
+  for (u64 it = 0; it < iters; it++) {
+    u64 g = start0 + it * stride;
+    bool oe = false, ne = false;
+    const u64 *key = nullptr;
     u8 oldrow[MZ_GPU_MAX_AGGS * 24], newrow[MZ_GPU_MAX_AGGS * 24];
-    if (oe)
-      for (u32 a = 0; a < na; a++)
-        d_finalize(spec.aggs[a], old_a[a], old_total, oldrow + 24 * a);
-    if (ne)
-      for (u32 a = 0; a < na; a++)
-        d_finalize(spec.aggs[a], accs[a], new_total, newrow + 24 * a);
-    if (oe && ne) {
-      bool same = true;
-      for (u32 c = 0; c < ovb; c++) same &= oldrow[c] == newrow[c];
-      if (same) continue;
+    if (g < G) {
+      u64 lo = gstart[g];
+      u64 end = (g + 1 < G) ? gstart[g + 1] : hi_row;
+      key = keys + lo * kw;
+      u64 idx = miss[g] ? base + misspos[g] : found[g];
+      if (idx < st.capacity) {  // overflow flagged by insert
+        u64 *row = st.rows + idx * st.stride_w;
+        i64 *total_p = (i64 *)(row + kw);
+        Acc5 *accs = (Acc5 *)(row + kw + 1);
+        // snapshot old
+        i64 old_total = *total_p;
+        Acc5 old_a[MZ_GPU_MAX_AGGS];
+        for (u32 a = 0; a < na; a++) old_a[a] = accs[a];
+        // apply updates (explode_one * diff + semigroup merge; wrapping)
+        for (u64 r = lo; r < end; r++) {
+          i64 d = diffs[r];
+          const u8 *v = vals + r * vb;
+          for (u32 a = 0; a < na; a++) {
+            Acc5 c = d_datum_to_acc(spec.aggs[a], v);
+            accs[a].accum += c.accum * (u128)(i128)d;
+            accs[a].nn += (u64)c.nn * (u64)d;
+            accs[a].pi += (u64)c.pi * (u64)d;
+            accs[a].ni += (u64)c.ni * (u64)d;
+            accs[a].nan += (u64)c.nan * (u64)d;
+          }
+          *total_p = wadd(*total_p, d);
+        }
+        i64 new_total = *total_p;
+        // exists = any nonzero component (reduce_abelian: keys with
+        // nonempty input accumulation produce one output row)
+        auto exists = [&](const Acc5 *as, i64 tot) {
+          if (tot != 0) return true;
+          for (u32 a = 0; a < na; a++)
+            if (as[a].accum != 0 || as[a].nn || as[a].pi || as[a].ni ||
+                as[a].nan)
+              return true;
+          return false;
+        };
+        oe = exists(old_a, old_total);
+        ne = exists(accs, new_total);
+        if (oe)
+          for (u32 a = 0; a < na; a++)
+            d_finalize(spec.aggs[a], old_a[a], old_total, oldrow + 24 * a);
+        if (ne)
+          for (u32 a = 0; a < na; a++)
+            d_finalize(spec.aggs[a], accs[a], new_total, newrow + 24 * a);
+        if (oe && ne) {
+          bool same = true;
+          for (u32 c = 0; c < ovb; c++) same &= oldrow[c] == newrow[c];
+          if (same) oe = ne = false;
+        }
+      }
     }
+    u32 c = (oe ? 1u : 0u) + (ne ? 1u : 0u);
+    u64 o = wave_reserve(ocount, c, lane);
     if (oe) {
-      u64 o = atomicAdd(ocount, 1ull);
       for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = key[w];
-      for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = oldrow[c];
+      for (u32 cb = 0; cb < ovb; cb++) ovals[o * ovb + cb] = oldrow[cb];
       otimes[o] = t;
       odiffs[o] = -1;
+      o++;
     }
     if (ne) {
-      u64 o = atomicAdd(ocount, 1ull);
       for (u32 w = 0; w < kw; w++) okeys[o * kw + w] = key[w];
-      for (u32 c = 0; c < ovb; c++) ovals[o * ovb + c] = newrow[c];
+      for (u32 cb = 0; cb < ovb; cb++) ovals[o * ovb + cb] = newrow[cb];
       otimes[o] = t;
       odiffs[o] = 1;
     }
