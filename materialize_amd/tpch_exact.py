@@ -342,7 +342,7 @@ class TpchExact:
             r.r64_incl(0, 3)               # suppkey term
             quantity = r.r32_incl(1, 50)
             discount = r.d_decimal(0, 8)   # /100
-            r.d_decimal(0, 10)             # tax
+            tax = r.d_decimal(0, 10)       # /100
             shipdate = r.d_date(orderdate, 1, 121)
             receiptdate = r.d_date(shipdate, 1, 30)
             # packer draw order: returnflag (only when receipted),
@@ -356,7 +356,7 @@ class TpchExact:
             r.d_text(10, 43, self.TEXT_LEN)
             ep = quantity * partkey_retailprice(partkey)
             lines.append((partkey, quantity, ep, discount, shipdate,
-                          commitdate, receiptdate, mode, retflag))
+                          commitdate, receiptdate, mode, retflag, tax))
         prio = r.choose_idx(PRIORITIES_LEN)  # orderpriority
         r.r64_incl(1, self.count_clerk)      # clerk
         r.d_text(19, 78, self.TEXT_LEN)
@@ -428,7 +428,7 @@ def q3_result(customers, orders, lineitems):
         if ck in building and od < CUTOFF:
             odate[okey] = od
     groups = {}  # okey -> sum in 1e-2 dollars
-    for (okey, _pk, _q, ep, d, sd, _cd, _rd, _md, _rf) in lineitems:
+    for (okey, _pk, _q, ep, d, sd, *_rest) in lineitems:
         if sd <= CUTOFF:
             continue
         od = odate.get(okey)
@@ -441,6 +441,18 @@ def q3_result(customers, orders, lineitems):
                      "0"])
     rows.sort()
     return rows
+
+
+def render_dec_reduced(units, scale):
+    """decNumber cx.reduce rendering of an exact value `units` * 10^-scale
+    (non-negative): trailing fractional zeros stripped, integer when the
+    fraction vanishes (the Numeric sum rendering, reduce.rs:2192)."""
+    s = 10 ** scale
+    whole, frac = divmod(units, s)
+    if frac == 0:
+        return str(whole)
+    fs = str(frac).rjust(scale, "0").rstrip("0")
+    return f"{whole}.{fs}"
 
 
 def render_revenue_1e2(r2):
