@@ -2311,19 +2311,27 @@ void sort_updates(Ctx *c, const u64 *keys, u32 kw, const u8 *vals, u32 vb,
   }
 }
 
+// Bounds-checked scan input: in[i] for i < n, 0 for the padding slot —
+// replaces the per-scan device copy + fill (two extra launches and a
+// full re-read of the input).
+struct ScanPadIn {
+  const u32 *in;
+  u64 n;
+  __host__ __device__ u32 operator()(u64 i) const {
+    return i < n ? in[i] : 0u;
+  }
+};
+
 u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
   // returns total; out = exclusive prefix (out has n+1 slots).
-  // The n+1-sized scan would read in[n]; pad a copy to stay defined.
   auto &S = (*c->scr);
-  u32 *pad = (u32 *)S.get((n + 1) * 4);
-  HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
-                           c->stream));
-  fill_u32(c, pad + n, 1, 0);
+  auto it = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<u64>(0), ScanPadIn{in, n});
   size_t need = 0;
-  (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
+  (void)rocprim::exclusive_scan(nullptr, need, it, out, 0u, n + 1,
                           rocprim::plus<u32>(), c->stream);
   void *tmp = S.get(need);
-  (void)rocprim::exclusive_scan(tmp, need, pad, out, 0u, n + 1,
+  (void)rocprim::exclusive_scan(tmp, need, it, out, 0u, n + 1,
                           rocprim::plus<u32>(), c->stream);
   u32 total = *(u32 *)d2h_pinned(c, out + n, 4);
   return total;
@@ -2332,15 +2340,13 @@ u64 exclusive_scan_u32(Ctx *c, const u32 *in, u32 *out, u64 n) {
 // enqueue-only exclusive scan (out has n+1 slots; no readback)
 void exclusive_scan_u32_ns(Ctx *c, const u32 *in, u32 *out, u64 n) {
   auto &S = (*c->scr);
-  u32 *pad = (u32 *)S.get((n + 1) * 4);
-  HIP_CHECK(hipMemcpyAsync(pad, in, n * 4, hipMemcpyDeviceToDevice,
-                           c->stream));
-  fill_u32(c, pad + n, 1, 0);
+  auto it = rocprim::make_transform_iterator(
+      rocprim::make_counting_iterator<u64>(0), ScanPadIn{in, n});
   size_t need = 0;
-  (void)rocprim::exclusive_scan(nullptr, need, pad, out, 0u, n + 1,
+  (void)rocprim::exclusive_scan(nullptr, need, it, out, 0u, n + 1,
                                 rocprim::plus<u32>(), c->stream);
   void *tmp = S.get(need);
-  (void)rocprim::exclusive_scan(tmp, need, pad, out, 0u, n + 1,
+  (void)rocprim::exclusive_scan(tmp, need, it, out, 0u, n + 1,
                                 rocprim::plus<u32>(), c->stream);
 }
 
