@@ -1898,8 +1898,17 @@ struct LaneGuard {
       // pre-grow AND touch: lane arenas otherwise carve + fault their
       // pages inside the first big merges (KFD zeroes fresh VRAM —
       // measured as a 350ms step on a cold box's first process)
-      void *w = a->lane_scr->get(2ull << 30);
-      (void)hipMemsetAsync(w, 0, 2ull << 30, a->stream);
+      // 6 GB default: the periodic giant merge expands ~30M+ rows into
+      // lane scratch (~3-5 GB at the SF1/1M config); outgrowing the
+      // arena mid-run costs a synchronous hipMalloc of KFD-zeroed pages
+      // (measured as one 459 ms step at 1M-step ~50). Carving happens
+      // here, at arrangement creation (untimed).
+      static const u64 LANE_ARENA = [] {
+        const char *e = getenv("MZ_GPU_LANE_ARENA_GB");
+        return (u64)((e ? atof(e) : 6.0) * (double)(1ull << 30));
+      }();
+      void *w = a->lane_scr->get(LANE_ARENA);
+      (void)hipMemsetAsync(w, 0, LANE_ARENA, a->stream);
       a->lane_scr->reset();
       HIP_CHECK(hipEventCreate(&a->ev_done));
       HIP_CHECK(hipEventCreate(&a->ev_gate));
