@@ -167,3 +167,46 @@ def test_gpu_halfjoin2_fused_matches_two_stage_with_errs():
     assert len(g_errs[0]) > 0, "zero divisors must produce error rows"
     g.close()
     o.close()
+
+
+def test_cmp_fields_signed_widths():
+    """MZ_COMPUTE_CMP_FIELDS reads SIGNED ints at the filter's width —
+    negative i32 dates must order correctly (oracle semantics)."""
+    from pyoracle import OracleCtx
+    o = OracleCtx()
+    sch_in = abi.schema(1, 16)
+    # val = (a i32 @0, b i32 @4, pad)
+    cl = abi.closure(
+        [abi.filt(abi.MZ_SRC_COMPUTE, abi.MZ_COMPUTE_CMP_FIELDS, 4,
+                  abi.MZ_CMP_LT, 0, arg0=0, arg1=4,
+                  arg0_src=abi.MZ_SRC_VAL_STREAM,
+                  arg1_src=abi.MZ_SRC_VAL_STREAM)],
+        [abi.field(abi.MZ_SRC_KEY, 0, 8)],
+        [abi.field(abi.MZ_SRC_VAL_STREAM, 0, 4)],
+        abi.schema(1, 4))
+    pairs = [(-5, -3), (-3, -5), (-1, 2), (2, -1), (7, 7), (0, 1)]
+    n = len(pairs)
+    v = np.zeros((n, 16), np.uint8)
+    v[:, 0:4] = np.array([p[0] for p in pairs], np.int32) \
+        .reshape(-1, 1).view(np.uint8).reshape(n, 4)
+    v[:, 4:8] = np.array([p[1] for p in pairs], np.int32) \
+        .reshape(-1, 1).view(np.uint8).reshape(n, 4)
+    u = abi.make_updates(np.arange(n, dtype=np.int64), v,
+                         np.zeros(n, np.uint64), np.ones(n, np.int64),
+                         0, 1)
+    k, _, _, _ = o.map(sch_in, u, cl)
+    kept = sorted(int(x) for x in np.asarray(k).reshape(-1))
+    want = sorted(i for i, (a, b) in enumerate(pairs) if a < b)
+    assert kept == want
+    o.close()
+
+
+def test_render_dec_reduced():
+    from materialize_amd.tpch_exact import render_dec_reduced
+    assert render_dec_reduced(123400, 4) == "12.34"
+    assert render_dec_reduced(123000, 4) == "12.3"
+    assert render_dec_reduced(120000, 4) == "12"
+    assert render_dec_reduced(5, 4) == "0.0005"
+    assert render_dec_reduced(0, 2) == "0"
+    assert render_dec_reduced(100, 2) == "1"
+    assert render_dec_reduced(101, 2) == "1.01"
